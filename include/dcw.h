@@ -74,6 +74,12 @@ typedef struct dcw_grandparent {
   uint64_t file_size;
 } dcw_grandparent;
 
+/* User-key-range list of one level's files (sorted, non-overlapping). */
+typedef struct dcw_level_files {
+  const dcw_grandparent* files; /* reuses the range+size triple */
+  uint32_t num_files;
+} dcw_level_files;
+
 /* Serialized subset of CompactionParams actually consumed by the worker.
  * Field citations: struct CompactionParams (compaction_executor.h:33-118)
  * and its fill at compaction_job.cc:944-963. */
@@ -111,11 +117,15 @@ typedef struct dcw_job_desc {
   const dcw_grandparent* grandparents; /* CompactionParams::grandparents */
   uint32_t num_grandparents;
 
-  /* 1 iff no file in any level below output_level overlaps the job's key
-   * range (Compaction::KeyNotExistsBeyondOutputLevel for every key).
-   * The DB-side plugin computes this from the version like the reference
-   * does per key; the conservative value is 0. */
-  int32_t key_not_exists_beyond_output_level;
+  /* Per-key KeyNotExistsBeyondOutputLevel (compaction.cc:548-586) needs the
+   * user-key ranges of the files in every level BELOW output_level.  The
+   * DB-side plugin ships them; levels_below_valid=0 reproduces the reference
+   * worker's conservative `is_compaction_woker_` branch (always false when
+   * not bottommost, compaction.cc:555-556).  At bottommost the check is
+   * always true regardless. */
+  int32_t levels_below_valid;
+  const struct dcw_level_files* levels_below; /* one per level > output_level */
+  uint32_t num_levels_below;
 
   /* BlockBasedTableOptions consumed by the build path (include/rocksdb/table.h):
    * defaults block_size=4096 (:276), block_restart_interval=16 (:289),

@@ -32,6 +32,10 @@ class Grandparent(C.Structure):
     ]
 
 
+class LevelFiles(C.Structure):
+    _fields_ = [("files", C.POINTER(Grandparent)), ("num_files", C.c_uint32)]
+
+
 class JobDesc(C.Structure):
     _fields_ = [
         ("struct_size", C.c_uint32),
@@ -57,7 +61,9 @@ class JobDesc(C.Structure):
         ("oldest_ancester_time", C.c_uint64),
         ("grandparents", C.POINTER(Grandparent)),
         ("num_grandparents", C.c_uint32),
-        ("key_not_exists_beyond_output_level", C.c_int32),
+        ("levels_below_valid", C.c_int32),
+        ("levels_below", C.POINTER(LevelFiles)),
+        ("num_levels_below", C.c_uint32),
         ("block_size", C.c_uint32),
         ("block_restart_interval", C.c_uint32),
         ("format_version", C.c_uint32),
@@ -318,8 +324,35 @@ def make_job(runs, output_dir, **kw) -> JobDesc:
         d._keep.append(garr)
         d.grandparents = garr
         d.num_grandparents = len(gps)
-    d.key_not_exists_beyond_output_level = kw.pop(
-        "key_not_exists_beyond_output_level", 1)
+    def _ranges(triples):
+        arr = (Grandparent * len(triples))()
+        for i, (sm, lg, fsz) in enumerate(triples):
+            smb = (C.c_uint8 * len(sm)).from_buffer_copy(sm)
+            lgb = (C.c_uint8 * len(lg)).from_buffer_copy(lg)
+            d._keep += [smb, lgb]
+            arr[i].smallest_ukey = smb
+            arr[i].smallest_len = len(sm)
+            arr[i].largest_ukey = lgb
+            arr[i].largest_len = len(lg)
+            arr[i].file_size = fsz
+        d._keep.append(arr)
+        return arr
+
+    # levels_below: list (one per level below output) of [(sm, lg, size)]
+    lvls = kw.pop("levels_below", None)
+    d.levels_below_valid = kw.pop("levels_below_valid",
+                                  1 if lvls is not None else 1)
+    if lvls is None:
+        lvls = []
+    if lvls or d.levels_below_valid:
+        larr = (LevelFiles * max(len(lvls), 1))()
+        for i, files in enumerate(lvls):
+            rarr = _ranges(files)
+            larr[i].files = rarr
+            larr[i].num_files = len(files)
+        d._keep.append(larr)
+        d.levels_below = larr
+        d.num_levels_below = len(lvls)
     d.block_size = kw.pop("block_size", 4096)
     d.block_restart_interval = kw.pop("block_restart_interval", 16)
     d.format_version = 5

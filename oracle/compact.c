@@ -264,9 +264,37 @@ typedef struct citer {
   int at_next;
   /* stats */
   uint64_t num_input_records, num_output_records;
+  /* KeyNotExistsBeyondOutputLevel monotonic level pointers (compaction.cc:564) */
+  size_t level_ptrs[16];
   char err[256];
   int failed;
 } citer;
+
+static int ukey_cmp2(const uint8_t* a, size_t alen, const uint8_t* b, size_t blen) {
+  size_t n = alen < blen ? alen : blen;
+  int c = memcmp(a, b, n);
+  if (c) return c;
+  return alen < blen ? -1 : (alen > blen ? 1 : 0);
+}
+
+/* Compaction::KeyNotExistsBeyondOutputLevel (compaction.cc:548-586) */
+static int key_not_exists_beyond(citer* c, const uint8_t* ukey, size_t ulen) {
+  const dcw_job_desc* d = c->d;
+  if (d->bottommost_level) return 1;
+  if (!d->levels_below_valid) return 0; /* reference worker branch (:555-556) */
+  for (uint32_t lvl = 0; lvl < d->num_levels_below && lvl < 16; lvl++) {
+    const dcw_level_files* lf = &d->levels_below[lvl];
+    for (; c->level_ptrs[lvl] < lf->num_files; c->level_ptrs[lvl]++) {
+      const dcw_grandparent* f = &lf->files[c->level_ptrs[lvl]];
+      if (ukey_cmp2(ukey, ulen, f->largest_ukey, f->largest_len) <= 0) {
+        if (ukey_cmp2(ukey, ulen, f->smallest_ukey, f->smallest_len) >= 0)
+          return 0;
+        break;
+      }
+    }
+  }
+  return 1;
+}
 
 /* findEarliestVisibleSnapshot (compaction_iterator.cc:1343-1369, no checker):
  * first snapshot >= seq; prev = the one below (0 if none). */
@@ -397,7 +425,7 @@ static void ci_next_from_input(citer* c) {
         /* end of input or different key (compaction_iterator.cc:856-890) */
         c->has_current_user_key = 0;
         if (seq <= c->earliest_snapshot &&
-            c->d->key_not_exists_beyond_output_level) {
+            key_not_exists_beyond(c, c->cur_ukey, c->cur_ukey_len)) {
           /* drop SD */
         } else if (c->last_key_seq_zeroed) {
           /* drop */
@@ -411,7 +439,7 @@ static void ci_next_from_input(citer* c) {
       /* rule (A): hidden by newer entry for the same user key (:894-916) */
       if (km_next(in) != 0) goto in_err;
     } else if (type == DCW_TYPE_DELETION && seq <= c->earliest_snapshot &&
-               c->d->key_not_exists_beyond_output_level) {
+               key_not_exists_beyond(c, c->cur_ukey, c->cur_ukey_len)) {
       /* obsolete deletion marker (:917-952) */
       if (km_next(in) != 0) goto in_err;
     } else if (type == DCW_TYPE_DELETION && c->d->bottommost_level) {
