@@ -288,7 +288,7 @@ uint32_t orc_block_checksum(uint32_t type, const void* data, size_t n, uint8_t l
 /* ---------------- snappy-format codec ----------------
  * Decoder: full public snappy format (any compliant producer).
  * Encoder: DCW-DETERMINISTIC spec: greedy matcher,
- *   hash table 1<<14 over 4-byte windows, hash = (load32(p)*0x1e35a7bd)>>18,
+ *   hash table 1<<12 over 4-byte windows, hash = (load32(p)*0x1e35a7bd)>>20,
  *   match if prev pos with equal 4 bytes and offset < 65536; extend forward;
  *   emit copies of <=64 bytes, first copy uses the 1-byte-offset form when
  *   len in [4,11] and offset < 2048; literals flushed before each copy.
@@ -332,7 +332,7 @@ size_t orc_snappy_compress(const uint8_t* in, size_t n, uint8_t* out) {
   uint8_t* op = out;
   op += orc_varint32_put(op, (uint32_t)n);
   if (n == 0) return (size_t)(op - out);
-  enum { HBITS = 14 };
+  enum { HBITS = 12 }; /* 4 KiB-block-sized table; also the GPU encoder's LDS table */
   static const uint32_t HMUL = 0x1e35a7bdu;
   uint32_t* tab = (uint32_t*)malloc(sizeof(uint32_t) << HBITS);
   memset(tab, 0xff, sizeof(uint32_t) << HBITS);

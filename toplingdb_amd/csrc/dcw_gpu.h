@@ -1,0 +1,110 @@
+// dcw_gpu.h — interface of the device pipeline (implemented in
+// dcw_kernels.hip).  The worker (dcw_worker.cpp) drives one GpuJob per
+// compaction job:
+//   stage() -> decode() -> merge() -> dedup() -> [per file: plan on host,
+//   emit_blocks() -> pack_and_fetch()/block keys/gathers] -> host tail.
+// All heavy bytes stay on the device; the host sees only per-entry plan
+// metadata, per-block keys, and final file images.
+#pragma once
+#include <cstdint>
+#include <string>
+#include <vector>
+
+#include "../../include/dcw.h"
+#include "dcw_host.h"
+
+namespace dcw {
+
+struct BlockRef {
+  uint64_t off;  // absolute offset into the staged input blob
+  uint32_t size; // on-disk (possibly compressed) block size
+};
+
+struct GpuInputs {
+  const uint8_t* blob = nullptr; // pinned host buffer: input files concatenated
+  size_t blob_size = 0;
+  std::vector<BlockRef> blocks;         // grouped by run, run-major order
+  std::vector<uint32_t> run_block_begin; // size num_runs+1
+  uint32_t checksum_type = 4;
+};
+
+// device-resident staged inputs (bench: "inputs already in HBM");
+// owned by the worker, shared across GpuJob runs.
+struct StagedInput {
+  void* d_blob = nullptr;
+  void* d_boff = nullptr;  // uint64*
+  void* d_bsize = nullptr; // uint32*
+  uint32_t n_blocks = 0;
+  uint32_t checksum_type = 4;
+  std::vector<uint32_t> run_block_begin;
+  ~StagedInput();
+};
+
+class GpuJob {
+ public:
+  GpuJob();
+  ~GpuJob();
+  GpuJob(const GpuJob&) = delete;
+  GpuJob& operator=(const GpuJob&) = delete;
+
+  // H2D of the input blob + block table; or adopt an existing staged input
+  // (no H2D, buffers borrowed — not freed by this job).
+  int stage(const GpuInputs& in, std::string* err);
+  int stage_adopt(const StagedInput& s, std::string* err);
+  // move this job's staged buffers out into `s` (for dcw_stage_inputs)
+  int stage_release(StagedInput* s, std::string* err);
+  // checksum-verify, decompress, parse blocks -> per-run entry arrays
+  int decode(std::string* err);
+  // k-way merge (pairwise merge-path rounds) -> one sorted entry array
+  int merge(std::string* err);
+  // visibility/dedup FSM + survivor compaction + plan metadata D2H
+  int dedup(const dcw_job_desc* d, std::string* err);
+
+  uint64_t num_input_entries() const { return n_entries_; }
+  size_t num_survivors() const { return n_surv_; }
+  // plan metadata (host copies, valid after dedup)
+  const std::vector<uint8_t>& plan_shared() const { return h_shared_; }
+  const std::vector<uint8_t>& plan_klen() const { return h_klen_; }
+  const std::vector<uint32_t>& plan_vlen() const { return h_vlen_; }
+
+  // Emit planned blocks (uncompressed encode + optional snappy + checksum).
+  // Returns per-block final body sizes in comp_sizes.
+  int emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts& o,
+                  std::vector<uint32_t>* comp_sizes, std::string* err);
+  // Pack blocks [b0,b1) of the last emit_blocks call into one contiguous
+  // [body|trailer]* image and copy to host.
+  int pack_and_fetch(size_t b0, size_t b1, const TableOpts& o, std::string* out,
+                     std::vector<SstIndexEntry>* handles, std::string* err);
+  // first/last internal keys of blocks [b0,b1) of the last emit
+  int fetch_block_keys(size_t b0, size_t b1, std::vector<std::string>* first_keys,
+                       std::vector<std::string>* last_keys, std::string* err);
+  // survivor range -> (ikey, value) pairs on host (partial blocks, file meta)
+  int gather_entries(uint64_t first, uint32_t count,
+                     std::vector<std::pair<std::string, std::string>>* kvs,
+                     std::string* err);
+  // min/max sequence + tombstone count over survivor range (post zeroing)
+  int seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* mx,
+                 uint64_t* n_tombstones, std::string* err);
+
+  double ms_decode = 0, ms_merge = 0, ms_dedup = 0, ms_emit = 0, ms_h2d = 0,
+         ms_d2h = 0;
+  // set by decode(): uniform user key length of the job's entries
+  uint32_t ukey_len = 0;
+
+  struct Impl; // implementation detail (dcw_kernels.hip)
+
+ private:
+  Impl* p_;
+  uint64_t n_entries_ = 0;
+  size_t n_surv_ = 0;
+  std::vector<uint8_t> h_shared_, h_klen_;
+  std::vector<uint32_t> h_vlen_;
+  std::vector<uint32_t> run_blocks_;
+};
+
+// device management
+int gpu_init(int device_ordinal, std::string* err);
+void gpu_shutdown();
+bool gpu_available();
+
+} // namespace dcw

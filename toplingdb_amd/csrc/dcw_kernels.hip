@@ -1,0 +1,1617 @@
+// dcw_kernels.hip — MI355X (gfx950) device pipeline of the dcompact worker.
+//
+// GPU counterparts of the reference CPU hot loops:
+//  - k_verify_usize / k_decompress / k_count_entries / k_decode_entries:
+//      BlockFetcher::ReadBlockContents + DataBlockIter::ParseNextDataKey
+//      (table/block_fetcher.cc:242, table/block_based/block.cc:37-139,667) —
+//      restart-interval-parallel decode of prefix-compressed blocks.
+//  - k_merge_pair: the k-way merge (table/compaction_merging_iterator.cc)
+//      recast as merge-path partitioned pairwise merges over 32-byte
+//      normalized entries (ulong4), stable on ties (lower run first, like
+//      the heap's child order).
+//  - k_mark_heads / k_group_fsm / gather: CompactionIterator::NextFromInput
+//      + PrepareOutput (db/compaction/compaction_iterator.cc:475-1341) as a
+//      per-user-key-group FSM.
+//  - k_emit / k_compress / k_checksum / k_pack: BlockBuilder +
+//      BlockBasedTableBuilder write path (block_builder.cc:189-253,
+//      block_based_table_builder.cc:1277-1330) — block-parallel encode,
+//      DCW-deterministic snappy, XXH3/CRC32C trailers.
+//
+// Design notes (MI355X): integer/byte, HBM-bound work — no MFMA.  Loads are
+// vectorized where layout permits (ulong4 = 32 B/entry merge moves); the
+// snappy encoder keeps its 16 KiB hash table in LDS; launches are
+// grid-strided and sized ≫256 workgroups to fill 8 XCDs.
+#include <hip/hip_runtime.h>
+
+#include <cstdio>
+#include <cstring>
+
+#include "dcw_gpu.h"
+
+#define WAVE 64
+
+namespace dcw {
+
+// ------------------------------------------------------------------
+// error plumbing
+// ------------------------------------------------------------------
+#define HIPCHK(x)                                                      \
+  do {                                                                 \
+    hipError_t _e = (x);                                               \
+    if (_e != hipSuccess) {                                            \
+      if (err) *err = std::string("HIP error: ") + hipGetErrorString(_e) + \
+                      " at " #x;                                       \
+      return -1;                                                       \
+    }                                                                  \
+  } while (0)
+
+enum DevErr : uint32_t {
+  DE_OK = 0,
+  DE_CHECKSUM = 1,
+  DE_SNAPPY = 2,
+  DE_BLOCK_PARSE = 3,
+  DE_UKEY_LEN = 4,     // non-uniform or >16B user keys (round-1 envelope)
+  DE_TYPE = 5,         // value type outside {Put, Delete, SingleDelete}
+  DE_SD_CONTRACT = 6,  // SingleDelete + Delete mix (Corruption in reference)
+  DE_COMPRESS_TYPE = 7,
+};
+
+static int g_device = -1;
+
+int gpu_init(int ordinal, std::string* err) {
+  hipError_t e = hipSetDevice(ordinal);
+  if (e != hipSuccess) {
+    if (err) *err = std::string("hipSetDevice: ") + hipGetErrorString(e);
+    return -1;
+  }
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess || n <= ordinal) {
+    if (err) *err = "no HIP device";
+    return -1;
+  }
+  hipDeviceProp_t prop;
+  e = hipGetDeviceProperties(&prop, ordinal);
+  if (e != hipSuccess) {
+    if (err) *err = std::string("hipGetDeviceProperties: ") + hipGetErrorString(e);
+    return -1;
+  }
+  g_device = ordinal;
+  return 0;
+}
+void gpu_shutdown() { g_device = -1; }
+bool gpu_available() { return g_device >= 0; }
+
+// ------------------------------------------------------------------
+// device-side helpers
+// ------------------------------------------------------------------
+__device__ __forceinline__ bool ent_le(const ulong4& a, const ulong4& b) {
+  // (k0,k1,k2) lexicographic; ties -> true (A-side/lower-run wins: stable)
+  if (a.x != b.x) return a.x < b.x;
+  if (a.y != b.y) return a.y < b.y;
+  return a.z <= b.z;
+}
+__device__ __forceinline__ void set_err(uint32_t* err_flag, uint32_t code) {
+  atomicCAS(err_flag, DE_OK, code);
+}
+
+// ------------------------------------------------------------------
+// decode-side kernels
+// ------------------------------------------------------------------
+__global__ void k_verify_usize(const uint8_t* __restrict__ blob,
+                               const uint64_t* __restrict__ boff,
+                               const uint32_t* __restrict__ bsize, uint32_t nblocks,
+                               uint32_t checksum_type,
+                               const Crc32cTables* __restrict__ crc_tt,
+                               uint32_t* __restrict__ usize,
+                               uint8_t* __restrict__ btype, uint32_t* err_flag) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nblocks;
+       i += gridDim.x * blockDim.x) {
+    const uint8_t* p = blob + boff[i];
+    uint32_t sz = bsize[i];
+    uint8_t type = p[sz];
+    uint32_t stored;
+    memcpy(&stored, p + sz + 1, 4);
+    if (checksum_type != 0) {
+      uint32_t actual = block_checksum(checksum_type, crc_tt, p, sz, type);
+      if (actual != stored) {
+        set_err(err_flag, DE_CHECKSUM);
+        return;
+      }
+    }
+    if (type == 0) {
+      usize[i] = sz;
+    } else if (type == 1) {
+      size_t ul = snappy_uncompressed_len(p, sz);
+      if (ul == (size_t)-1) {
+        set_err(err_flag, DE_SNAPPY);
+        return;
+      }
+      usize[i] = (uint32_t)ul;
+    } else {
+      set_err(err_flag, DE_COMPRESS_TYPE);
+      return;
+    }
+    btype[i] = type;
+  }
+}
+
+// one wave per block: raw -> cooperative copy; snappy -> lane 0 decodes
+__global__ void k_decompress(const uint8_t* __restrict__ blob,
+                             const uint64_t* __restrict__ boff,
+                             const uint32_t* __restrict__ bsize,
+                             const uint8_t* __restrict__ btype,
+                             const uint64_t* __restrict__ uoff,
+                             const uint32_t* __restrict__ usize, uint32_t nblocks,
+                             uint8_t* __restrict__ ublob, uint32_t* err_flag) {
+  uint32_t waves_per_wg = blockDim.x / WAVE;
+  uint32_t wave = blockIdx.x * waves_per_wg + threadIdx.x / WAVE;
+  uint32_t lane = threadIdx.x % WAVE;
+  uint32_t stride = gridDim.x * waves_per_wg;
+  for (uint32_t i = wave; i < nblocks; i += stride) {
+    const uint8_t* src = blob + boff[i];
+    uint8_t* dst = ublob + uoff[i];
+    if (btype[i] == 0) {
+      uint32_t n = bsize[i];
+      uint32_t pos = lane * 16;
+      for (; pos + 16 <= n; pos += WAVE * 16) {
+        ulong2 v;
+        memcpy(&v, src + pos, 16);
+        memcpy(dst + pos, &v, 16);
+      }
+      // tail bytes by lane 0
+      if (lane == 0)
+        for (uint32_t t = n & ~15u; t < n; t++) dst[t] = src[t];
+    } else {
+      if (lane == 0) {
+        if (snappy_uncompress(src, bsize[i], dst, usize[i]) != usize[i])
+          set_err(err_flag, DE_SNAPPY);
+      }
+    }
+  }
+}
+
+__global__ void k_num_restarts(const uint8_t* __restrict__ ublob,
+                               const uint64_t* __restrict__ uoff,
+                               const uint32_t* __restrict__ usize, uint32_t nblocks,
+                               uint32_t* __restrict__ nrestarts, uint32_t* err_flag) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nblocks;
+       i += gridDim.x * blockDim.x) {
+    uint32_t us = usize[i];
+    if (us < 8) {
+      set_err(err_flag, DE_BLOCK_PARSE);
+      return;
+    }
+    uint32_t footer;
+    memcpy(&footer, ublob + uoff[i] + us - 4, 4);
+    uint32_t nr = footer & 0x7fffffffu;
+    if (4ull + 4ull * nr + 4ull > us) {
+      set_err(err_flag, DE_BLOCK_PARSE);
+      return;
+    }
+    nrestarts[i] = nr;
+  }
+}
+
+// walk one restart interval: count entries (pass 1) or decode them (pass 2)
+__device__ __forceinline__ int interval_bounds(
+    const uint8_t* ublk, uint32_t usize, uint32_t nr, uint32_t j, uint32_t* beg,
+    uint32_t* end) {
+  const uint8_t* rst = ublk + usize - 4 - 4 * nr;
+  uint32_t b, e;
+  memcpy(&b, rst + 4 * j, 4);
+  if (j + 1 < nr)
+    memcpy(&e, rst + 4 * (j + 1), 4);
+  else
+    e = (uint32_t)(usize - 4 - 4 * nr);
+  if (b > e || e > usize) return -1;
+  *beg = b;
+  *end = e;
+  return 0;
+}
+
+__global__ void k_count_entries(const uint8_t* __restrict__ ublob,
+                                const uint64_t* __restrict__ uoff,
+                                const uint32_t* __restrict__ usize,
+                                const uint32_t* __restrict__ nrestarts,
+                                const uint32_t* __restrict__ iv_block,
+                                const uint32_t* __restrict__ iv_local,
+                                uint32_t nintervals, uint32_t* __restrict__ iv_cnt,
+                                uint32_t* err_flag) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nintervals;
+       i += gridDim.x * blockDim.x) {
+    uint32_t b = iv_block[i];
+    const uint8_t* ublk = ublob + uoff[b];
+    uint32_t beg, end;
+    if (interval_bounds(ublk, usize[b], nrestarts[b], iv_local[i], &beg, &end) != 0) {
+      set_err(err_flag, DE_BLOCK_PARSE);
+      return;
+    }
+    const uint8_t* p = ublk + beg;
+    const uint8_t* lim = ublk + end;
+    uint32_t n = 0;
+    while (p < lim) {
+      uint32_t shared, non_shared, vlen;
+      int a = varint32_get(p, lim, &shared);
+      if (a < 0) break;
+      p += a;
+      a = varint32_get(p, lim, &non_shared);
+      if (a < 0) break;
+      p += a;
+      a = varint32_get(p, lim, &vlen);
+      if (a < 0) break;
+      p += a;
+      p += non_shared + vlen;
+      if (p > lim) {
+        set_err(err_flag, DE_BLOCK_PARSE);
+        return;
+      }
+      n++;
+    }
+    iv_cnt[i] = n;
+  }
+}
+
+__global__ void k_decode_entries(
+    const uint8_t* __restrict__ ublob, const uint64_t* __restrict__ uoff,
+    const uint32_t* __restrict__ usize, const uint32_t* __restrict__ nrestarts,
+    const uint32_t* __restrict__ iv_block, const uint32_t* __restrict__ iv_local,
+    const uint32_t* __restrict__ iv_base, uint32_t nintervals,
+    ulong4* __restrict__ ents, uint64_t* __restrict__ voff,
+    uint32_t* __restrict__ vlen_out, uint8_t* __restrict__ klen_out,
+    uint32_t* __restrict__ ukey_len_probe, uint32_t* err_flag) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nintervals;
+       i += gridDim.x * blockDim.x) {
+    uint32_t b = iv_block[i];
+    const uint8_t* ublk = ublob + uoff[b];
+    uint32_t beg, end;
+    if (interval_bounds(ublk, usize[b], nrestarts[b], iv_local[i], &beg, &end) != 0)
+      return;
+    const uint8_t* p = ublk + beg;
+    const uint8_t* lim = ublk + end;
+    uint8_t key[32];
+    uint32_t klen = 0;
+    uint32_t out = iv_base[i];
+    while (p < lim) {
+      uint32_t shared, non_shared, vl;
+      int a = varint32_get(p, lim, &shared);
+      if (a < 0) break;
+      p += a;
+      a = varint32_get(p, lim, &non_shared);
+      if (a < 0) break;
+      p += a;
+      a = varint32_get(p, lim, &vl);
+      if (a < 0) break;
+      p += a;
+      if (shared > klen || shared + non_shared > 32) {
+        set_err(err_flag, DE_BLOCK_PARSE);
+        return;
+      }
+      for (uint32_t t = 0; t < non_shared; t++) key[shared + t] = p[t];
+      klen = shared + non_shared;
+      p += non_shared;
+      if (klen < 9) {
+        set_err(err_flag, DE_BLOCK_PARSE);
+        return;
+      }
+      uint32_t ulen = klen - 8;
+      if (ulen > 16) {
+        set_err(err_flag, DE_UKEY_LEN);
+        return;
+      }
+      uint32_t expect = atomicCAS(ukey_len_probe, 0xffffffffu, ulen);
+      if (expect != 0xffffffffu && expect != ulen) {
+        set_err(err_flag, DE_UKEY_LEN);
+        return;
+      }
+      uint64_t tag;
+      memcpy(&tag, key + ulen, 8);
+      uint8_t vt = (uint8_t)tag;
+      if (!(vt == kTypeValue || vt == kTypeDeletion || vt == kTypeSingleDeletion)) {
+        set_err(err_flag, DE_TYPE);
+        return;
+      }
+      uint64_t k0, k1, k2;
+      make_normkey(key, ulen, tag, &k0, &k1, &k2);
+      ents[out] = make_ulong4(k0, k1, k2, out);
+      voff[out] = uoff[b] + (uint64_t)(p - ublk);
+      vlen_out[out] = vl;
+      klen_out[out] = (uint8_t)klen;
+      p += vl;
+      out++;
+    }
+  }
+}
+
+// ------------------------------------------------------------------
+// merge
+// ------------------------------------------------------------------
+__device__ uint64_t merge_diag(const ulong4* A, uint64_t nA, const ulong4* B,
+                               uint64_t nB, uint64_t d) {
+  // largest a in [max(0,d-nB), min(d,nA)] s.t. A[0..a) all <= B from b=d-a on
+  uint64_t lo = d > nB ? d - nB : 0;
+  uint64_t hi = d < nA ? d : nA;
+  while (lo < hi) {
+    uint64_t mid = (lo + hi + 1) / 2;
+    // A[mid-1] vs B[d-mid]: A goes first on ties
+    if (ent_le(A[mid - 1], B[d - mid]))
+      lo = mid;
+    else
+      hi = mid - 1;
+  }
+  return lo;
+}
+
+template <int ITEMS>
+__global__ void k_merge_pair(const ulong4* __restrict__ A, uint64_t nA,
+                             const ulong4* __restrict__ B, uint64_t nB,
+                             ulong4* __restrict__ out) {
+  uint64_t total = nA + nB;
+  uint64_t nchunk = (total + ITEMS - 1) / ITEMS;
+  for (uint64_t c = blockIdx.x * blockDim.x + threadIdx.x; c < nchunk;
+       c += (uint64_t)gridDim.x * blockDim.x) {
+    uint64_t d0 = c * ITEMS;
+    uint64_t d1 = d0 + ITEMS < total ? d0 + ITEMS : total;
+    uint64_t a = merge_diag(A, nA, B, nB, d0);
+    uint64_t b = d0 - a;
+    for (uint64_t d = d0; d < d1; d++) {
+      bool takeA = a < nA && (b >= nB || ent_le(A[a], B[b]));
+      out[d] = takeA ? A[a++] : B[b++];
+    }
+  }
+}
+
+// ------------------------------------------------------------------
+// dedup / visibility
+// ------------------------------------------------------------------
+__global__ void k_mark_heads(const ulong4* __restrict__ e, uint64_t n,
+                             uint8_t* __restrict__ head) {
+  for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    head[i] = (i == 0) || e[i].x != e[i - 1].x || e[i].y != e[i - 1].y;
+  }
+}
+
+struct FsmParams {
+  uint64_t earliest_snapshot;
+  uint64_t ewcs;
+  const uint64_t* snapshots;
+  uint32_t num_snapshots;
+  uint32_t visible_at_tip;
+  uint32_t bottommost;
+  uint32_t levels_below_valid;
+  // levels-below normkey ranges, concatenated; per-level [begin,end) offsets
+  const uint64_t* lb_sm_k0;
+  const uint64_t* lb_sm_k1;
+  const uint64_t* lb_lg_k0;
+  const uint64_t* lb_lg_k1;
+  const uint32_t* lb_level_beg; // num_levels+1
+  uint32_t num_levels;
+};
+
+__device__ uint64_t fsm_find_earliest(const FsmParams& P, uint64_t seq,
+                                      uint64_t* prev) {
+  uint32_t lo = 0, hi = P.num_snapshots;
+  while (lo < hi) {
+    uint32_t mid = (lo + hi) / 2;
+    if (P.snapshots[mid] < seq)
+      lo = mid + 1;
+    else
+      hi = mid;
+  }
+  *prev = lo > 0 ? P.snapshots[lo - 1] : 0;
+  return lo < P.num_snapshots ? P.snapshots[lo] : kMaxSeq;
+}
+
+__device__ bool fsm_key_not_exists_beyond(const FsmParams& P, uint64_t k0,
+                                          uint64_t k1) {
+  if (P.bottommost) return true;
+  if (!P.levels_below_valid) return false; // reference worker branch
+  for (uint32_t lvl = 0; lvl < P.num_levels; lvl++) {
+    uint32_t b = P.lb_level_beg[lvl], e = P.lb_level_beg[lvl + 1];
+    // first file with largest >= key
+    uint32_t lo = b, hi = e;
+    while (lo < hi) {
+      uint32_t mid = (lo + hi) / 2;
+      bool lg_lt = P.lb_lg_k0[mid] < k0 ||
+                   (P.lb_lg_k0[mid] == k0 && P.lb_lg_k1[mid] < k1);
+      if (lg_lt)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    if (lo < e) {
+      bool sm_le = P.lb_sm_k0[lo] < k0 ||
+                   (P.lb_sm_k0[lo] == k0 && P.lb_sm_k1[lo] <= k1);
+      if (sm_le) return false;
+    }
+  }
+  return true;
+}
+
+// group flags
+enum : uint8_t { GF_PRODUCED = 1, GF_LAG_SENSITIVE = 2 };
+
+// One thread per user-key group.  Mirrors CompactionIterator::NextFromInput
+// (:475-1082) + PrepareOutput seq-zeroing (:1286-1328), restricted to one
+// user key (all FSM state resets at user-key change, :568-583).
+// is_first_group_mode: 0 = normal pass (has_outputted = outputs>=1),
+// group `lag_group` (if != ~0) uses outputs>=2 (SeekToFirst lag,
+// compaction_iterator.cc:156-159 + 223-226).
+__global__ void k_group_fsm(const ulong4* __restrict__ e, uint64_t n,
+                            const uint64_t* __restrict__ head_idx,
+                            uint64_t ngroups, FsmParams P,
+                            uint8_t* __restrict__ survive,
+                            uint64_t* __restrict__ newtag,
+                            uint8_t* __restrict__ clearv,
+                            uint8_t* __restrict__ gflags, uint64_t lag_group,
+                            uint32_t* err_flag) {
+  for (uint64_t g = blockIdx.x * blockDim.x + threadIdx.x; g < ngroups;
+       g += (uint64_t)gridDim.x * blockDim.x) {
+    if (lag_group != ~0ull && g != lag_group) continue;
+    uint64_t g0 = head_idx[g];
+    uint64_t g1 = g + 1 < ngroups ? head_idx[g + 1] : n;
+    uint64_t k0 = e[g0].x, k1 = e[g0].y;
+    bool is_lag = (g == lag_group);
+    int outputs = 0;
+    bool clear_next = false, last_zeroed = false, lag_sensitive = false;
+    uint64_t cukSnap = 0;
+    bool first_entry = true;
+    uint64_t pos = g0;
+    while (pos < g1) {
+      uint64_t tag = ~e[pos].z;
+      uint64_t seq = tag >> 8;
+      uint8_t type = (uint8_t)tag;
+      uint64_t last_snapshot = first_entry ? 0 : cukSnap;
+      first_entry = false;
+      uint64_t prev_snapshot = 0;
+      cukSnap = P.visible_at_tip ? P.earliest_snapshot
+                                 : fsm_find_earliest(P, seq, &prev_snapshot);
+      survive[pos] = 0;
+      clearv[pos] = 0;
+      newtag[pos] = tag;
+      bool has_outputted = is_lag ? outputs >= 2 : outputs >= 1;
+      bool out_this = false, clear_this = false;
+      uint64_t consumed = 1;
+      if (clear_next) {
+        if (type != kTypeValue) {
+          set_err(err_flag, DE_TYPE);
+          return;
+        }
+        out_this = true;
+        clear_this = true;
+        clear_next = false;
+      } else if (type == kTypeSingleDeletion) {
+        if (outputs == 1) lag_sensitive = true; // decision may depend on lag
+        if (pos + 1 < g1) {
+          uint64_t ntag = ~e[pos + 1].z;
+          uint64_t nseq = ntag >> 8;
+          uint8_t ntype = (uint8_t)ntag;
+          if (last_zeroed) {
+            consumed = 2; // drop SD and the next version
+          } else if (prev_snapshot == 0 || nseq > prev_snapshot) {
+            if (ntype == kTypeSingleDeletion) {
+              consumed = 1; // skip the first SD; reprocess the second
+            } else if (ntype == kTypeDeletion) {
+              set_err(err_flag, DE_SD_CONTRACT);
+              return;
+            } else if (has_outputted || seq <= P.ewcs ||
+                       (P.earliest_snapshot < P.ewcs &&
+                        seq <= P.earliest_snapshot)) {
+              consumed = 2; // drop both SD and value
+            } else {
+              out_this = true; // kKeepSDForConflictCheck
+              clear_next = true;
+            }
+          } else {
+            out_this = true; // kKeepSDForSnapshot
+          }
+        } else {
+          if (seq <= P.earliest_snapshot && fsm_key_not_exists_beyond(P, k0, k1)) {
+            // drop (fallthrough SD)
+          } else if (last_zeroed) {
+            // drop
+          } else {
+            out_this = true; // kKeepSD
+          }
+        }
+      } else if (last_snapshot == cukSnap ||
+                 (last_snapshot > 0 && last_snapshot < cukSnap)) {
+        // rule (A): hidden by newer entry in the same snapshot stripe
+      } else if (type == kTypeDeletion && seq <= P.earliest_snapshot &&
+                 fsm_key_not_exists_beyond(P, k0, k1)) {
+        // obsolete delete
+      } else if (type == kTypeDeletion && P.bottommost) {
+        // skip versions in the same snapshot range
+        uint64_t j = pos + 1;
+        while (j < g1) {
+          uint64_t jseq = (~e[j].z) >> 8;
+          if (!(prev_snapshot == 0 || jseq > prev_snapshot)) break;
+          j++;
+        }
+        if (j < g1) {
+          out_this = true; // kKeepDel
+          consumed = j - pos;
+        } else {
+          consumed = g1 - pos; // drop delete and all covered versions
+        }
+      } else {
+        out_this = true; // kNewUserKey / plain keep
+      }
+      if (out_this) {
+        uint64_t otag = tag;
+        // PrepareOutput seq-zeroing (bottommost, visible in every snapshot)
+        if (P.bottommost && seq <= P.earliest_snapshot && type == kTypeValue) {
+          otag = (uint64_t)type;
+          last_zeroed = true;
+        }
+        survive[pos] = 1;
+        newtag[pos] = otag;
+        clearv[pos] = clear_this ? 1 : 0;
+        outputs++;
+      }
+      pos += consumed;
+    }
+    if (gflags)
+      gflags[g] = (outputs > 0 ? GF_PRODUCED : 0) |
+                  (lag_sensitive ? GF_LAG_SENSITIVE : 0);
+  }
+}
+
+// ------------------------------------------------------------------
+// scans (u32 exclusive scan, two-level with host combining the block sums)
+// ------------------------------------------------------------------
+__global__ void k_scan_partial(const uint8_t* __restrict__ in, uint64_t n,
+                               uint32_t* __restrict__ out,
+                               uint32_t* __restrict__ block_sums) {
+  __shared__ uint32_t tmp[1024];
+  uint64_t base = (uint64_t)blockIdx.x * 1024;
+  uint32_t t = threadIdx.x;
+  uint32_t v = (base + t < n) ? in[base + t] : 0;
+  tmp[t] = v;
+  __syncthreads();
+  // Hillis-Steele in LDS
+  for (uint32_t off = 1; off < 1024; off *= 2) {
+    uint32_t x = (t >= off) ? tmp[t - off] : 0;
+    __syncthreads();
+    tmp[t] += x;
+    __syncthreads();
+  }
+  if (base + t < n) out[base + t] = tmp[t] - v; // exclusive
+  if (t == 1023) block_sums[blockIdx.x] = tmp[t];
+}
+__global__ void k_scan_add_base(uint32_t* __restrict__ out, uint64_t n,
+                                const uint32_t* __restrict__ base_per_block) {
+  uint64_t i = (uint64_t)blockIdx.x * 1024 + threadIdx.x;
+  if (i < n) out[i] += base_per_block[blockIdx.x];
+}
+
+// ------------------------------------------------------------------
+// survivor gather + shared-prefix
+// ------------------------------------------------------------------
+__global__ void k_gather_survivors(
+    const ulong4* __restrict__ e, uint64_t n, const uint8_t* __restrict__ survive,
+    const uint32_t* __restrict__ pos, const uint64_t* __restrict__ newtag,
+    const uint8_t* __restrict__ clearv, const uint64_t* __restrict__ voff,
+    const uint32_t* __restrict__ vlen, const uint8_t* __restrict__ klen,
+    uint64_t* __restrict__ s_k0, uint64_t* __restrict__ s_k1,
+    uint64_t* __restrict__ s_tag, uint64_t* __restrict__ s_voff,
+    uint32_t* __restrict__ s_vlen, uint8_t* __restrict__ s_klen) {
+  for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    if (!survive[i]) continue;
+    uint32_t o = pos[i];
+    uint64_t ref = e[i].w;
+    s_k0[o] = e[i].x;
+    s_k1[o] = e[i].y;
+    s_tag[o] = newtag[i];
+    s_voff[o] = voff[ref];
+    s_vlen[o] = clearv[i] ? 0 : vlen[ref];
+    s_klen[o] = klen[ref];
+  }
+}
+
+__device__ __forceinline__ void build_ikey(uint64_t k0, uint64_t k1, uint64_t tag,
+                                           uint32_t klen, uint8_t* out) {
+  uint64_t b0 = __builtin_bswap64(k0), b1 = __builtin_bswap64(k1);
+  memcpy(out, &b0, 8);
+  memcpy(out + 8, &b1, 8);
+  memcpy(out + (klen - 8), &tag, 8);
+}
+
+__global__ void k_shared_prefix(const uint64_t* __restrict__ s_k0,
+                                const uint64_t* __restrict__ s_k1,
+                                const uint64_t* __restrict__ s_tag,
+                                const uint8_t* __restrict__ s_klen, uint64_t n,
+                                uint8_t* __restrict__ s_shared) {
+  for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    if (i == 0) {
+      s_shared[0] = 0;
+      continue;
+    }
+    uint8_t a[24], b[24];
+    uint32_t ka = s_klen[i - 1], kb = s_klen[i];
+    build_ikey(s_k0[i - 1], s_k1[i - 1], s_tag[i - 1], ka, a);
+    build_ikey(s_k0[i], s_k1[i], s_tag[i], kb, b);
+    uint32_t m = ka < kb ? ka : kb;
+    uint32_t s = 0;
+    while (s < m && a[s] == b[s]) s++;
+    s_shared[i] = (uint8_t)s;
+  }
+}
+
+// ------------------------------------------------------------------
+// emit / compress / checksum / pack
+// ------------------------------------------------------------------
+struct EmitBlockDesc {
+  uint32_t first;     // survivor index of first entry (relative to chunk base added on host)
+  uint32_t count;
+  uint32_t unc_size;  // total uncompressed block size (with restarts+footer)
+  uint32_t num_restarts;
+  uint64_t uout;      // offset into ucblob
+};
+
+// one workgroup per block; thread t encodes entries t, t+bd, ...
+__global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
+                       const uint64_t* __restrict__ s_k0,
+                       const uint64_t* __restrict__ s_k1,
+                       const uint64_t* __restrict__ s_tag,
+                       const uint64_t* __restrict__ s_voff,
+                       const uint32_t* __restrict__ s_vlen,
+                       const uint8_t* __restrict__ s_klen,
+                       const uint32_t* __restrict__ eoff, // per-entry in-block offset
+                       const uint8_t* __restrict__ ublob,
+                       uint8_t* __restrict__ ucblob, uint32_t restart_interval) {
+  for (uint32_t b = blockIdx.x; b < nblocks; b += gridDim.x) {
+    EmitBlockDesc d = bds[b];
+    uint8_t* out = ucblob + d.uout;
+    for (uint32_t li = threadIdx.x; li < d.count; li += blockDim.x) {
+      uint32_t i = d.first + li;
+      uint32_t klen = s_klen[i];
+      uint8_t key[24];
+      build_ikey(s_k0[i], s_k1[i], s_tag[i], klen, key);
+      uint32_t shared = 0;
+      if (li % restart_interval != 0) {
+        // shared prefix with previous survivor == previous entry in block
+        uint32_t kp = s_klen[i - 1];
+        uint8_t prev[24];
+        build_ikey(s_k0[i - 1], s_k1[i - 1], s_tag[i - 1], kp, prev);
+        uint32_t m = kp < klen ? kp : klen;
+        while (shared < m && prev[shared] == key[shared]) shared++;
+      }
+      uint32_t non_shared = klen - shared;
+      uint32_t vl = s_vlen[i];
+      uint8_t* p = out + eoff[i];
+      p += varint32_put(p, shared);
+      p += varint32_put(p, non_shared);
+      p += varint32_put(p, vl);
+      for (uint32_t t = 0; t < non_shared; t++) p[t] = key[shared + t];
+      p += non_shared;
+      const uint8_t* src = ublob + s_voff[i];
+      for (uint32_t t = 0; t < vl; t++) p[t] = src[t];
+    }
+    if (threadIdx.x == 0) {
+      // restart array + packed footer (data_block_footer.cc:24-39)
+      uint32_t body = d.unc_size - 4 - 4 * d.num_restarts;
+      uint8_t* tail = out + body;
+      uint32_t r0 = 0;
+      memcpy(tail, &r0, 4);
+      for (uint32_t j = 1; j < d.num_restarts; j++) {
+        uint32_t off = eoff[d.first + j * restart_interval];
+        memcpy(tail + 4 * j, &off, 4);
+      }
+      uint32_t footer = d.num_restarts; // kDataBlockBinarySearch
+      memcpy(tail + 4 * d.num_restarts, &footer, 4);
+    }
+  }
+}
+
+// one workgroup (64 threads) per block; snappy hash table in LDS; lane 0
+// encodes; all lanes clear the table.
+__global__ void k_compress(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
+                           const uint8_t* __restrict__ ucblob,
+                           uint8_t* __restrict__ cblob, uint64_t ccap_per_block,
+                           uint32_t* __restrict__ bsize,
+                           uint8_t* __restrict__ btype) {
+  __shared__ uint32_t tab[1u << kSnapHashBits];
+  for (uint32_t b = blockIdx.x; b < nblocks; b += gridDim.x) {
+    for (uint32_t t = threadIdx.x; t < (1u << kSnapHashBits); t += blockDim.x)
+      tab[t] = 0xffffffffu;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      EmitBlockDesc d = bds[b];
+      const uint8_t* in = ucblob + d.uout;
+      uint8_t* out = cblob + (uint64_t)b * ccap_per_block;
+      size_t cn = snappy_compress_block(in, d.unc_size, out, tab);
+      // GoodCompressionRatio, default max_compressed_bytes_per_kb=896
+      if (cn <= (((uint64_t)896 * d.unc_size) >> 10)) {
+        bsize[b] = (uint32_t)cn;
+        btype[b] = 1;
+      } else {
+        bsize[b] = d.unc_size;
+        btype[b] = 0;
+      }
+    }
+    __syncthreads();
+  }
+}
+
+__global__ void k_checksum(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
+                           const uint8_t* __restrict__ ucblob,
+                           const uint8_t* __restrict__ cblob, uint64_t ccap,
+                           const uint32_t* __restrict__ bsize,
+                           const uint8_t* __restrict__ btype,
+                           uint32_t checksum_type,
+                           const Crc32cTables* __restrict__ crc_tt,
+                           uint32_t* __restrict__ csum) {
+  for (uint32_t b = blockIdx.x * blockDim.x + threadIdx.x; b < nblocks;
+       b += gridDim.x * blockDim.x) {
+    const uint8_t* body =
+        btype[b] ? cblob + (uint64_t)b * ccap : ucblob + bds[b].uout;
+    csum[b] = block_checksum(checksum_type, crc_tt, body, bsize[b], btype[b]);
+  }
+}
+
+// pack [body|trailer]* into a contiguous image at host-provided offsets
+__global__ void k_pack(const EmitBlockDesc* __restrict__ bds, uint32_t b0,
+                       uint32_t b1, const uint8_t* __restrict__ ucblob,
+                       const uint8_t* __restrict__ cblob, uint64_t ccap,
+                       const uint32_t* __restrict__ bsize,
+                       const uint8_t* __restrict__ btype,
+                       const uint32_t* __restrict__ csum,
+                       const uint64_t* __restrict__ outoff,
+                       uint8_t* __restrict__ out) {
+  uint32_t waves_per_wg = blockDim.x / WAVE;
+  uint32_t wave = blockIdx.x * waves_per_wg + threadIdx.x / WAVE;
+  uint32_t lane = threadIdx.x % WAVE;
+  for (uint32_t b = b0 + wave; b < b1; b += gridDim.x * waves_per_wg) {
+    const uint8_t* body = btype[b] ? cblob + (uint64_t)b * ccap : ucblob + bds[b].uout;
+    uint8_t* dst = out + outoff[b - b0];
+    uint32_t n = bsize[b];
+    for (uint32_t posn = lane * 16; posn < n; posn += WAVE * 16) {
+      uint32_t chunk = n - posn < 16 ? n - posn : 16;
+      for (uint32_t t = 0; t < chunk; t++) dst[posn + t] = body[posn + t];
+    }
+    if (lane == 0) {
+      dst[n] = btype[b];
+      uint32_t cs = csum[b];
+      memcpy(dst + n + 1, &cs, 4);
+    }
+  }
+}
+
+__global__ void k_block_keys(const EmitBlockDesc* __restrict__ bds, uint32_t b0,
+                             uint32_t b1, const uint64_t* __restrict__ s_k0,
+                             const uint64_t* __restrict__ s_k1,
+                             const uint64_t* __restrict__ s_tag,
+                             const uint8_t* __restrict__ s_klen,
+                             uint8_t* __restrict__ out /*64B per block: first|last*/) {
+  for (uint32_t b = b0 + blockIdx.x * blockDim.x + threadIdx.x; b < b1;
+       b += gridDim.x * blockDim.x) {
+    uint8_t* o = out + (uint64_t)(b - b0) * 64;
+    uint32_t f = bds[b].first, l = bds[b].first + bds[b].count - 1;
+    memset(o, 0, 64);
+    o[0] = s_klen[f];
+    build_ikey(s_k0[f], s_k1[f], s_tag[f], s_klen[f], o + 1);
+    o[32] = s_klen[l];
+    build_ikey(s_k0[l], s_k1[l], s_tag[l], s_klen[l], o + 33);
+  }
+}
+
+// pack survivor range into [klen u8 | key | vlen u32 | value]* records
+__global__ void k_gather_range(const uint64_t* __restrict__ s_k0,
+                               const uint64_t* __restrict__ s_k1,
+                               const uint64_t* __restrict__ s_tag,
+                               const uint64_t* __restrict__ s_voff,
+                               const uint32_t* __restrict__ s_vlen,
+                               const uint8_t* __restrict__ s_klen,
+                               const uint8_t* __restrict__ ublob, uint64_t first,
+                               uint32_t count, const uint64_t* __restrict__ recoff,
+                               uint8_t* __restrict__ out) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < count;
+       i += gridDim.x * blockDim.x) {
+    uint64_t idx = first + i;
+    uint8_t* p = out + recoff[i];
+    uint32_t klen = s_klen[idx];
+    p[0] = (uint8_t)klen;
+    build_ikey(s_k0[idx], s_k1[idx], s_tag[idx], klen, p + 1);
+    uint32_t vl = s_vlen[idx];
+    memcpy(p + 1 + klen, &vl, 4);
+    const uint8_t* src = ublob + s_voff[idx];
+    for (uint32_t t = 0; t < vl; t++) p[5 + klen + t] = src[t];
+  }
+}
+
+__global__ void k_build_headidx(const uint8_t* __restrict__ head,
+                                const uint32_t* __restrict__ pos, uint64_t n,
+                                uint64_t* __restrict__ headidx) {
+  for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (uint64_t)gridDim.x * blockDim.x)
+    if (head[i]) headidx[pos[i]] = i;
+}
+
+__global__ void k_sizes_nocomp(const EmitBlockDesc* __restrict__ bds, uint32_t nb,
+                               uint32_t* __restrict__ bsize,
+                               uint8_t* __restrict__ btype) {
+  for (uint32_t b = blockIdx.x * blockDim.x + threadIdx.x; b < nb;
+       b += gridDim.x * blockDim.x) {
+    bsize[b] = bds[b].unc_size;
+    btype[b] = 0;
+  }
+}
+
+__global__ void k_seq_minmax(const uint64_t* __restrict__ s_tag, uint64_t first,
+                             uint64_t count, unsigned long long* mn,
+                             unsigned long long* mx,
+                             unsigned long long* n_tombstones) {
+  for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < count;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    uint64_t tag = s_tag[first + i];
+    unsigned long long seq = tag >> 8;
+    atomicMin(mn, seq);
+    atomicMax(mx, seq);
+    uint8_t vt = (uint8_t)tag;
+    if (vt == kTypeDeletion || vt == kTypeSingleDeletion)
+      atomicAdd(n_tombstones, 1ull);
+  }
+}
+
+// ------------------------------------------------------------------
+// host-side Impl
+// ------------------------------------------------------------------
+struct GpuJob::Impl {
+  hipStream_t stream = nullptr;
+  bool borrowed_stage = false; // staged buffers owned by a StagedInput
+  // staged input
+  uint8_t* d_blob = nullptr;
+  uint64_t* d_boff = nullptr;
+  uint32_t* d_bsize = nullptr;
+  uint32_t n_blocks = 0;
+  uint32_t checksum_type = 4;
+  // decode
+  uint32_t* d_usize = nullptr;
+  uint8_t* d_btype_in = nullptr;
+  uint64_t* d_uoff = nullptr;
+  uint8_t* d_ublob = nullptr;
+  uint64_t ublob_size = 0;
+  uint32_t* d_nrestarts = nullptr;
+  uint32_t* d_iv_block = nullptr;
+  uint32_t* d_iv_local = nullptr;
+  uint32_t* d_iv_cnt = nullptr;
+  uint32_t* d_iv_base = nullptr;
+  uint32_t n_intervals = 0;
+  uint32_t* d_err = nullptr;
+  uint32_t* d_uklen_probe = nullptr;
+  Crc32cTables* d_crc = nullptr;
+  // entries
+  ulong4* d_ent[2] = {nullptr, nullptr};
+  uint64_t* d_voff = nullptr;
+  uint32_t* d_vlen = nullptr;
+  uint8_t* d_klen = nullptr;
+  uint64_t n_entries = 0;
+  std::vector<uint64_t> run_entry_begin;
+  int final_buf = 0;
+  // dedup
+  uint8_t* d_head = nullptr;
+  uint64_t* d_headidx = nullptr;
+  uint64_t n_groups = 0;
+  uint8_t* d_survive = nullptr;
+  uint64_t* d_newtag = nullptr;
+  uint8_t* d_clearv = nullptr;
+  uint8_t* d_gflags = nullptr;
+  uint32_t* d_pos = nullptr;
+  // levels-below device copies
+  uint64_t *d_lb_sm0 = nullptr, *d_lb_sm1 = nullptr, *d_lb_lg0 = nullptr,
+           *d_lb_lg1 = nullptr;
+  uint32_t* d_lb_beg = nullptr;
+  // survivors
+  uint64_t *d_sk0 = nullptr, *d_sk1 = nullptr, *d_stag = nullptr,
+           *d_svoff = nullptr;
+  uint32_t* d_svlen = nullptr;
+  uint8_t *d_sklen = nullptr, *d_sshared = nullptr;
+  uint64_t n_surv = 0;
+  // emit chunk state
+  EmitBlockDesc* d_bds = nullptr;
+  size_t bds_cap = 0;
+  uint32_t* d_eoff = nullptr;
+  size_t eoff_cap = 0;
+  uint8_t* d_ucblob = nullptr;
+  size_t ucblob_cap = 0;
+  uint8_t* d_cblob = nullptr;
+  size_t cblob_cap = 0;
+  uint32_t* d_ebsize = nullptr;
+  uint8_t* d_ebtype = nullptr;
+  uint32_t* d_ecsum = nullptr;
+  size_t eb_cap = 0;
+  uint64_t ccap_per_block = 0;
+  uint32_t emit_nblocks = 0;
+  uint64_t emit_base = 0; // survivor index offset of this chunk's eoff array
+  // misc scratch
+  uint64_t* d_scratch64 = nullptr;
+  size_t scratch64_cap = 0;
+  uint8_t* d_out_img = nullptr;
+  size_t out_img_cap = 0;
+
+  hipError_t ensure(void** p, size_t* cap, size_t need) {
+    if (*cap >= need) return hipSuccess;
+    if (*p) (void)hipFree(*p);
+    *p = nullptr;
+    size_t cap2 = need + need / 4;
+    hipError_t e = hipMalloc(p, cap2);
+    if (e == hipSuccess) *cap = cap2;
+    else *cap = 0;
+    return e;
+  }
+};
+
+#define ENSURE(ptr, cap, need) HIPCHK(p->ensure((void**)&(ptr), &(cap), (need)))
+
+static double ms_between(hipEvent_t a, hipEvent_t b) {
+  float ms = 0;
+  hipEventElapsedTime(&ms, a, b);
+  return (double)ms;
+}
+
+GpuJob::GpuJob() : p_(new Impl) { (void)hipStreamCreate(&p_->stream); }
+GpuJob::~GpuJob() {
+  Impl* p = p_;
+  auto F = [](void* x) {
+    if (x) (void)hipFree(x);
+  };
+  if (!p->borrowed_stage) {
+    F(p->d_blob);
+    F(p->d_boff);
+    F(p->d_bsize);
+  }
+  F(p->d_usize); F(p->d_btype_in);
+  F(p->d_uoff); F(p->d_ublob); F(p->d_nrestarts); F(p->d_iv_block);
+  F(p->d_iv_local); F(p->d_iv_cnt); F(p->d_iv_base); F(p->d_err);
+  F(p->d_uklen_probe); F(p->d_crc); F(p->d_ent[0]); F(p->d_ent[1]);
+  F(p->d_voff); F(p->d_vlen); F(p->d_klen); F(p->d_head); F(p->d_headidx);
+  F(p->d_survive); F(p->d_newtag); F(p->d_clearv); F(p->d_gflags); F(p->d_pos);
+  F(p->d_lb_sm0); F(p->d_lb_sm1); F(p->d_lb_lg0); F(p->d_lb_lg1); F(p->d_lb_beg);
+  F(p->d_sk0); F(p->d_sk1); F(p->d_stag); F(p->d_svoff); F(p->d_svlen);
+  F(p->d_sklen); F(p->d_sshared); F(p->d_bds); F(p->d_eoff); F(p->d_ucblob);
+  F(p->d_cblob); F(p->d_ebsize); F(p->d_ebtype); F(p->d_ecsum);
+  F(p->d_scratch64); F(p->d_out_img);
+  if (p->stream) (void)hipStreamDestroy(p->stream);
+  delete p;
+}
+
+static uint32_t grid_for(uint64_t work, uint32_t block = 256) {
+  uint64_t g = (work + block - 1) / block;
+  if (g > 4096) g = 4096; // grid-stride beyond (≫256 WGs fills 8 XCDs)
+  if (g == 0) g = 1;
+  return (uint32_t)g;
+}
+
+int GpuJob::stage(const GpuInputs& in, std::string* err) {
+  Impl* p = p_;
+  hipEvent_t t0, t1;
+  hipEventCreate(&t0);
+  hipEventCreate(&t1);
+  hipEventRecord(t0, p->stream);
+  HIPCHK(hipMalloc(&p->d_blob, in.blob_size));
+  HIPCHK(hipMemcpyAsync(p->d_blob, in.blob, in.blob_size, hipMemcpyHostToDevice,
+                        p->stream));
+  p->n_blocks = (uint32_t)in.blocks.size();
+  p->checksum_type = in.checksum_type;
+  std::vector<uint64_t> boff(p->n_blocks);
+  std::vector<uint32_t> bsize(p->n_blocks);
+  for (uint32_t i = 0; i < p->n_blocks; i++) {
+    boff[i] = in.blocks[i].off;
+    bsize[i] = in.blocks[i].size;
+  }
+  HIPCHK(hipMalloc(&p->d_boff, sizeof(uint64_t) * p->n_blocks));
+  HIPCHK(hipMalloc(&p->d_bsize, sizeof(uint32_t) * p->n_blocks));
+  HIPCHK(hipMemcpyAsync(p->d_boff, boff.data(), sizeof(uint64_t) * p->n_blocks,
+                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMemcpyAsync(p->d_bsize, bsize.data(), sizeof(uint32_t) * p->n_blocks,
+                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMalloc(&p->d_crc, sizeof(Crc32cTables)));
+  HIPCHK(hipMemcpyAsync(p->d_crc, &g_crc, sizeof(Crc32cTables),
+                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMalloc(&p->d_err, 8));
+  HIPCHK(hipMemsetAsync(p->d_err, 0, 8, p->stream));
+  HIPCHK(hipMalloc(&p->d_uklen_probe, 4));
+  HIPCHK(hipMemsetAsync(p->d_uklen_probe, 0xff, 4, p->stream));
+  hipEventRecord(t1, p->stream);
+  HIPCHK(hipStreamSynchronize(p->stream));
+  ms_h2d += ms_between(t0, t1);
+  hipEventDestroy(t0);
+  hipEventDestroy(t1);
+  // remember run boundaries (translated to entries later)
+  run_blocks_ = in.run_block_begin;
+  return 0;
+}
+
+StagedInput::~StagedInput() {
+  if (d_blob) (void)hipFree(d_blob);
+  if (d_boff) (void)hipFree(d_boff);
+  if (d_bsize) (void)hipFree(d_bsize);
+}
+
+int GpuJob::stage_adopt(const StagedInput& s, std::string* err) {
+  Impl* p = p_;
+  p->borrowed_stage = true;
+  p->d_blob = (uint8_t*)s.d_blob;
+  p->d_boff = (uint64_t*)s.d_boff;
+  p->d_bsize = (uint32_t*)s.d_bsize;
+  p->n_blocks = s.n_blocks;
+  p->checksum_type = s.checksum_type;
+  run_blocks_ = s.run_block_begin;
+  HIPCHK(hipMalloc(&p->d_crc, sizeof(Crc32cTables)));
+  HIPCHK(hipMemcpyAsync(p->d_crc, &g_crc, sizeof(Crc32cTables),
+                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMalloc(&p->d_err, 8));
+  HIPCHK(hipMemsetAsync(p->d_err, 0, 8, p->stream));
+  HIPCHK(hipMalloc(&p->d_uklen_probe, 4));
+  HIPCHK(hipMemsetAsync(p->d_uklen_probe, 0xff, 4, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  return 0;
+}
+
+int GpuJob::stage_release(StagedInput* s, std::string* err) {
+  Impl* p = p_;
+  (void)err;
+  s->d_blob = p->d_blob;
+  s->d_boff = p->d_boff;
+  s->d_bsize = p->d_bsize;
+  s->n_blocks = p->n_blocks;
+  s->checksum_type = p->checksum_type;
+  s->run_block_begin = run_blocks_;
+  p->borrowed_stage = true; // dtor must not free them now
+  return 0;
+}
+
+int GpuJob::decode(std::string* err) {
+  Impl* p = p_;
+  hipEvent_t t0, t1;
+  hipEventCreate(&t0);
+  hipEventCreate(&t1);
+  hipEventRecord(t0, p->stream);
+  uint32_t nb = p->n_blocks;
+  HIPCHK(hipMalloc(&p->d_usize, sizeof(uint32_t) * nb));
+  HIPCHK(hipMalloc(&p->d_btype_in, nb));
+  hipLaunchKernelGGL(k_verify_usize, dim3(grid_for(nb)), dim3(256), 0, p->stream,
+                     p->d_blob, p->d_boff, p->d_bsize, nb, p->checksum_type,
+                     p->d_crc, p->d_usize, p->d_btype_in, p->d_err);
+  // host scan of usize -> uoff
+  std::vector<uint32_t> usize(nb);
+  HIPCHK(hipMemcpyAsync(usize.data(), p->d_usize, sizeof(uint32_t) * nb,
+                        hipMemcpyDeviceToHost, p->stream));
+  uint32_t err_host = 0;
+  HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  if (err_host) {
+    if (err) *err = "input block verify failed, code " + std::to_string(err_host);
+    return -1;
+  }
+  std::vector<uint64_t> uoff(nb);
+  uint64_t acc = 0;
+  for (uint32_t i = 0; i < nb; i++) {
+    uoff[i] = acc;
+    acc += usize[i];
+  }
+  p->ublob_size = acc;
+  HIPCHK(hipMalloc(&p->d_uoff, sizeof(uint64_t) * nb));
+  HIPCHK(hipMemcpyAsync(p->d_uoff, uoff.data(), sizeof(uint64_t) * nb,
+                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMalloc(&p->d_ublob, acc ? acc : 1));
+  hipLaunchKernelGGL(k_decompress, dim3(grid_for(nb * 4ull)), dim3(256), 0,
+                     p->stream, p->d_blob, p->d_boff, p->d_bsize, p->d_btype_in,
+                     p->d_uoff, p->d_usize, nb, p->d_ublob, p->d_err);
+  HIPCHK(hipMalloc(&p->d_nrestarts, sizeof(uint32_t) * nb));
+  hipLaunchKernelGGL(k_num_restarts, dim3(grid_for(nb)), dim3(256), 0, p->stream,
+                     p->d_ublob, p->d_uoff, p->d_usize, nb, p->d_nrestarts,
+                     p->d_err);
+  std::vector<uint32_t> nrestarts(nb);
+  HIPCHK(hipMemcpyAsync(nrestarts.data(), p->d_nrestarts, sizeof(uint32_t) * nb,
+                        hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  if (err_host) {
+    if (err) *err = "block decompress/parse failed, code " + std::to_string(err_host);
+    return -1;
+  }
+  // interval tables
+  std::vector<uint32_t> iv_block, iv_local;
+  std::vector<uint64_t> blk_iv_base(nb);
+  uint64_t niv = 0;
+  for (uint32_t b = 0; b < nb; b++) {
+    blk_iv_base[b] = niv;
+    for (uint32_t j = 0; j < nrestarts[b]; j++) {
+      iv_block.push_back(b);
+      iv_local.push_back(j);
+    }
+    niv += nrestarts[b];
+  }
+  p->n_intervals = (uint32_t)niv;
+  HIPCHK(hipMalloc(&p->d_iv_block, sizeof(uint32_t) * niv));
+  HIPCHK(hipMalloc(&p->d_iv_local, sizeof(uint32_t) * niv));
+  HIPCHK(hipMalloc(&p->d_iv_cnt, sizeof(uint32_t) * niv));
+  HIPCHK(hipMalloc(&p->d_iv_base, sizeof(uint32_t) * niv));
+  HIPCHK(hipMemcpyAsync(p->d_iv_block, iv_block.data(), sizeof(uint32_t) * niv,
+                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMemcpyAsync(p->d_iv_local, iv_local.data(), sizeof(uint32_t) * niv,
+                        hipMemcpyHostToDevice, p->stream));
+  hipLaunchKernelGGL(k_count_entries, dim3(grid_for(niv)), dim3(256), 0, p->stream,
+                     p->d_ublob, p->d_uoff, p->d_usize, p->d_nrestarts,
+                     p->d_iv_block, p->d_iv_local, (uint32_t)niv, p->d_iv_cnt,
+                     p->d_err);
+  std::vector<uint32_t> iv_cnt(niv);
+  HIPCHK(hipMemcpyAsync(iv_cnt.data(), p->d_iv_cnt, sizeof(uint32_t) * niv,
+                        hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  if (err_host) {
+    if (err) *err = "entry count failed, code " + std::to_string(err_host);
+    return -1;
+  }
+  std::vector<uint32_t> iv_base(niv);
+  uint64_t total_entries = 0;
+  for (uint64_t i = 0; i < niv; i++) {
+    iv_base[i] = (uint32_t)total_entries;
+    total_entries += iv_cnt[i];
+  }
+  p->n_entries = total_entries;
+  n_entries_ = total_entries;
+  // run entry boundaries
+  p->run_entry_begin.clear();
+  for (size_t r = 0; r + 1 < run_blocks_.size(); r++) {
+    uint32_t first_blk = run_blocks_[r];
+    p->run_entry_begin.push_back(first_blk < nb ? (first_blk == 0 ? 0 : iv_base[blk_iv_base[first_blk]])
+                                                : total_entries);
+  }
+  p->run_entry_begin.push_back(total_entries);
+  HIPCHK(hipMemcpyAsync(p->d_iv_base, iv_base.data(), sizeof(uint32_t) * niv,
+                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMalloc(&p->d_ent[0], sizeof(ulong4) * total_entries));
+  HIPCHK(hipMalloc(&p->d_ent[1], sizeof(ulong4) * total_entries));
+  HIPCHK(hipMalloc(&p->d_voff, sizeof(uint64_t) * total_entries));
+  HIPCHK(hipMalloc(&p->d_vlen, sizeof(uint32_t) * total_entries));
+  HIPCHK(hipMalloc(&p->d_klen, total_entries));
+  hipLaunchKernelGGL(k_decode_entries, dim3(grid_for(niv)), dim3(256), 0,
+                     p->stream, p->d_ublob, p->d_uoff, p->d_usize, p->d_nrestarts,
+                     p->d_iv_block, p->d_iv_local, p->d_iv_base, (uint32_t)niv,
+                     p->d_ent[0], p->d_voff, p->d_vlen, p->d_klen,
+                     p->d_uklen_probe, p->d_err);
+  uint32_t uklen = 0;
+  HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipMemcpyAsync(&uklen, p->d_uklen_probe, 4, hipMemcpyDeviceToHost,
+                        p->stream));
+  hipEventRecord(t1, p->stream);
+  HIPCHK(hipStreamSynchronize(p->stream));
+  ms_decode += ms_between(t0, t1);
+  hipEventDestroy(t0);
+  hipEventDestroy(t1);
+  if (err_host) {
+    if (err) *err = "entry decode failed, code " + std::to_string(err_host) +
+                    (err_host == DE_UKEY_LEN
+                         ? " (non-uniform or >16B user keys: outside the round-1 worker envelope)"
+                         : "");
+    return -1;
+  }
+  ukey_len = uklen;
+  return 0;
+}
+
+int GpuJob::merge(std::string* err) {
+  Impl* p = p_;
+  hipEvent_t t0, t1;
+  hipEventCreate(&t0);
+  hipEventCreate(&t1);
+  hipEventRecord(t0, p->stream);
+  std::vector<uint64_t> bounds = p->run_entry_begin; // size k+1
+  int cur = 0;
+  while (bounds.size() > 2) {
+    std::vector<uint64_t> nbounds;
+    nbounds.push_back(0);
+    size_t k = bounds.size() - 1;
+    for (size_t i = 0; i + 1 < k; i += 2) {
+      uint64_t a0 = bounds[i], a1 = bounds[i + 1], b1 = bounds[i + 2];
+      uint64_t nA = a1 - a0, nB = b1 - a1;
+      hipLaunchKernelGGL(k_merge_pair<16>, dim3(grid_for((nA + nB) / 16 + 1)),
+                         dim3(256), 0, p->stream, p->d_ent[cur] + a0, nA,
+                         p->d_ent[cur] + a1, nB, p->d_ent[cur ^ 1] + a0);
+      nbounds.push_back(b1);
+    }
+    if (k % 2) { // odd leftover run: copy through
+      uint64_t a0 = bounds[k - 1], a1 = bounds[k];
+      HIPCHK(hipMemcpyAsync(p->d_ent[cur ^ 1] + a0, p->d_ent[cur] + a0,
+                            sizeof(ulong4) * (a1 - a0), hipMemcpyDeviceToDevice,
+                            p->stream));
+      nbounds.push_back(a1);
+    }
+    bounds.swap(nbounds);
+    cur ^= 1;
+  }
+  p->final_buf = cur;
+  hipEventRecord(t1, p->stream);
+  HIPCHK(hipStreamSynchronize(p->stream));
+  ms_merge += ms_between(t0, t1);
+  hipEventDestroy(t0);
+  hipEventDestroy(t1);
+  return 0;
+}
+
+// device exclusive scan of a u8 array into u32 positions; returns total
+static int scan_u8(GpuJob::Impl* p, const uint8_t* d_in, uint64_t n,
+                   uint32_t* d_out, uint64_t* total, std::string* err) {
+  uint64_t nblk = (n + 1023) / 1024;
+  ENSURE(p->d_scratch64, p->scratch64_cap, sizeof(uint32_t) * (nblk + 1));
+  uint32_t* d_bs = (uint32_t*)p->d_scratch64;
+  hipLaunchKernelGGL(k_scan_partial, dim3((uint32_t)nblk), dim3(1024), 0,
+                     p->stream, d_in, n, d_out, d_bs);
+  std::vector<uint32_t> bs(nblk);
+  HIPCHK(hipMemcpyAsync(bs.data(), d_bs, sizeof(uint32_t) * nblk,
+                        hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  uint64_t acc = 0;
+  for (uint64_t i = 0; i < nblk; i++) {
+    uint32_t v = bs[i];
+    bs[i] = (uint32_t)acc;
+    acc += v;
+  }
+  HIPCHK(hipMemcpyAsync(d_bs, bs.data(), sizeof(uint32_t) * nblk,
+                        hipMemcpyHostToDevice, p->stream));
+  hipLaunchKernelGGL(k_scan_add_base, dim3((uint32_t)nblk), dim3(1024), 0,
+                     p->stream, d_out, n, d_bs);
+  *total = acc;
+  return 0;
+}
+
+int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
+  Impl* p = p_;
+  hipEvent_t t0, t1;
+  hipEventCreate(&t0);
+  hipEventCreate(&t1);
+  hipEventRecord(t0, p->stream);
+  uint64_t n = p->n_entries;
+  const ulong4* ents = p->d_ent[p->final_buf];
+  HIPCHK(hipMalloc(&p->d_head, n));
+  hipLaunchKernelGGL(k_mark_heads, dim3(grid_for(n)), dim3(256), 0, p->stream,
+                     ents, n, p->d_head);
+  // head positions via scan, then gather head indices on host
+  HIPCHK(hipMalloc(&p->d_pos, sizeof(uint32_t) * n));
+  uint64_t ngroups = 0;
+  if (scan_u8(p, p->d_head, n, p->d_pos, &ngroups, err) != 0) return -1;
+  p->n_groups = ngroups;
+  // build head_idx on device: head[i] -> headidx[pos[i]] = i
+  HIPCHK(hipMalloc(&p->d_headidx, sizeof(uint64_t) * (ngroups ? ngroups : 1)));
+  hipLaunchKernelGGL(k_build_headidx, dim3(grid_for(n)), dim3(256), 0, p->stream,
+                     p->d_head, p->d_pos, n, p->d_headidx);
+  // FSM params
+  FsmParams P;
+  memset(&P, 0, sizeof(P));
+  P.num_snapshots = d->num_snapshots;
+  P.visible_at_tip = d->num_snapshots == 0;
+  P.earliest_snapshot = d->num_snapshots ? d->snapshots[0] : kMaxSeq;
+  P.ewcs = d->earliest_write_conflict_snapshot;
+  P.bottommost = d->bottommost_level ? 1 : 0;
+  P.levels_below_valid = d->levels_below_valid ? 1 : 0;
+  uint64_t* d_snaps = nullptr;
+  if (d->num_snapshots) {
+    HIPCHK(hipMalloc(&d_snaps, sizeof(uint64_t) * d->num_snapshots));
+    HIPCHK(hipMemcpyAsync(d_snaps, d->snapshots, sizeof(uint64_t) * d->num_snapshots,
+                          hipMemcpyHostToDevice, p->stream));
+  }
+  P.snapshots = d_snaps;
+  // levels below -> normkeys
+  std::vector<uint64_t> sm0, sm1, lg0, lg1;
+  std::vector<uint32_t> lbeg{0};
+  for (uint32_t l = 0; l < d->num_levels_below; l++) {
+    const dcw_level_files* lf = &d->levels_below[l];
+    for (uint32_t f = 0; f < lf->num_files; f++) {
+      uint64_t a, b, c;
+      make_normkey(lf->files[f].smallest_ukey, lf->files[f].smallest_len, 0, &a,
+                   &b, &c);
+      sm0.push_back(a);
+      sm1.push_back(b);
+      make_normkey(lf->files[f].largest_ukey, lf->files[f].largest_len, 0, &a, &b,
+                   &c);
+      lg0.push_back(a);
+      lg1.push_back(b);
+    }
+    lbeg.push_back((uint32_t)sm0.size());
+  }
+  P.num_levels = d->num_levels_below;
+  if (!sm0.empty()) {
+    size_t nb = sm0.size() * 8;
+    HIPCHK(hipMalloc(&p->d_lb_sm0, nb));
+    HIPCHK(hipMalloc(&p->d_lb_sm1, nb));
+    HIPCHK(hipMalloc(&p->d_lb_lg0, nb));
+    HIPCHK(hipMalloc(&p->d_lb_lg1, nb));
+    HIPCHK(hipMemcpyAsync(p->d_lb_sm0, sm0.data(), nb, hipMemcpyHostToDevice, p->stream));
+    HIPCHK(hipMemcpyAsync(p->d_lb_sm1, sm1.data(), nb, hipMemcpyHostToDevice, p->stream));
+    HIPCHK(hipMemcpyAsync(p->d_lb_lg0, lg0.data(), nb, hipMemcpyHostToDevice, p->stream));
+    HIPCHK(hipMemcpyAsync(p->d_lb_lg1, lg1.data(), nb, hipMemcpyHostToDevice, p->stream));
+  }
+  HIPCHK(hipMalloc(&p->d_lb_beg, sizeof(uint32_t) * lbeg.size()));
+  HIPCHK(hipMemcpyAsync(p->d_lb_beg, lbeg.data(), sizeof(uint32_t) * lbeg.size(),
+                        hipMemcpyHostToDevice, p->stream));
+  P.lb_sm_k0 = p->d_lb_sm0;
+  P.lb_sm_k1 = p->d_lb_sm1;
+  P.lb_lg_k0 = p->d_lb_lg0;
+  P.lb_lg_k1 = p->d_lb_lg1;
+  P.lb_level_beg = p->d_lb_beg;
+
+  HIPCHK(hipMalloc(&p->d_survive, n));
+  HIPCHK(hipMalloc(&p->d_newtag, sizeof(uint64_t) * n));
+  HIPCHK(hipMalloc(&p->d_clearv, n));
+  HIPCHK(hipMalloc(&p->d_gflags, ngroups ? ngroups : 1));
+  hipLaunchKernelGGL(k_group_fsm, dim3(grid_for(ngroups)), dim3(256), 0,
+                     p->stream, ents, n, p->d_headidx, ngroups, P, p->d_survive,
+                     p->d_newtag, p->d_clearv, p->d_gflags, ~0ull, p->d_err);
+  // SeekToFirst lag: re-run the first group that produced output if its SD
+  // decisions were has_outputted-sensitive
+  std::vector<uint8_t> gflags(ngroups);
+  HIPCHK(hipMemcpyAsync(gflags.data(), p->d_gflags, ngroups,
+                        hipMemcpyDeviceToHost, p->stream));
+  uint32_t err_host = 0;
+  HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  if (err_host) {
+    if (err) *err = "dedup FSM failed, code " + std::to_string(err_host);
+    if (d_snaps) (void)hipFree(d_snaps);
+    return -1;
+  }
+  for (uint64_t g = 0; g < ngroups; g++) {
+    if (gflags[g] & GF_PRODUCED) {
+      if (gflags[g] & GF_LAG_SENSITIVE) {
+        hipLaunchKernelGGL(k_group_fsm, dim3(1), dim3(64), 0, p->stream, ents, n,
+                           p->d_headidx, ngroups, P, p->d_survive, p->d_newtag,
+                           p->d_clearv, (uint8_t*)nullptr, g, p->d_err);
+      }
+      break;
+    }
+  }
+  // survivor compaction
+  uint64_t nsurv = 0;
+  if (scan_u8(p, p->d_survive, n, p->d_pos, &nsurv, err) != 0) return -1;
+  p->n_surv = nsurv;
+  n_surv_ = nsurv;
+  HIPCHK(hipMalloc(&p->d_sk0, sizeof(uint64_t) * (nsurv + 1)));
+  HIPCHK(hipMalloc(&p->d_sk1, sizeof(uint64_t) * (nsurv + 1)));
+  HIPCHK(hipMalloc(&p->d_stag, sizeof(uint64_t) * (nsurv + 1)));
+  HIPCHK(hipMalloc(&p->d_svoff, sizeof(uint64_t) * (nsurv + 1)));
+  HIPCHK(hipMalloc(&p->d_svlen, sizeof(uint32_t) * (nsurv + 1)));
+  HIPCHK(hipMalloc(&p->d_sklen, nsurv + 1));
+  HIPCHK(hipMalloc(&p->d_sshared, nsurv + 1));
+  hipLaunchKernelGGL(k_gather_survivors, dim3(grid_for(n)), dim3(256), 0,
+                     p->stream, ents, n, p->d_survive, p->d_pos, p->d_newtag,
+                     p->d_clearv, p->d_voff, p->d_vlen, p->d_klen, p->d_sk0,
+                     p->d_sk1, p->d_stag, p->d_svoff, p->d_svlen, p->d_sklen);
+  hipLaunchKernelGGL(k_shared_prefix, dim3(grid_for(nsurv)), dim3(256), 0,
+                     p->stream, p->d_sk0, p->d_sk1, p->d_stag, p->d_sklen, nsurv,
+                     p->d_sshared);
+  // plan metadata D2H
+  h_shared_.resize(nsurv);
+  h_klen_.resize(nsurv);
+  h_vlen_.resize(nsurv);
+  if (nsurv) {
+    HIPCHK(hipMemcpyAsync(h_shared_.data(), p->d_sshared, nsurv,
+                          hipMemcpyDeviceToHost, p->stream));
+    HIPCHK(hipMemcpyAsync(h_klen_.data(), p->d_sklen, nsurv,
+                          hipMemcpyDeviceToHost, p->stream));
+    HIPCHK(hipMemcpyAsync(h_vlen_.data(), p->d_svlen, sizeof(uint32_t) * nsurv,
+                          hipMemcpyDeviceToHost, p->stream));
+  }
+  hipEventRecord(t1, p->stream);
+  HIPCHK(hipStreamSynchronize(p->stream));
+  ms_dedup += ms_between(t0, t1);
+  hipEventDestroy(t0);
+  hipEventDestroy(t1);
+  if (d_snaps) (void)hipFree(d_snaps);
+  return 0;
+}
+
+int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts& o,
+                        std::vector<uint32_t>* comp_sizes, std::string* err) {
+  Impl* p = p_;
+  hipEvent_t t0, t1;
+  hipEventCreate(&t0);
+  hipEventCreate(&t1);
+  hipEventRecord(t0, p->stream);
+  uint32_t nb = (uint32_t)blocks.size();
+  if (nb == 0) {
+    comp_sizes->clear();
+    return 0;
+  }
+  // per-entry in-block offsets + block descs
+  uint64_t first = blocks.front().first;
+  uint64_t last = blocks.back().first + blocks.back().count;
+  uint64_t nent = last - first;
+  std::vector<uint32_t> eoff(nent);
+  std::vector<EmitBlockDesc> bds(nb);
+  uint64_t uout = 0;
+  for (uint32_t b = 0; b < nb; b++) {
+    const PlannedBlock& pb = blocks[b];
+    uint32_t off = 0;
+    for (uint32_t li = 0; li < pb.count; li++) {
+      uint64_t i = pb.first + li;
+      uint32_t shared = (li % o.block_restart_interval == 0) ? 0 : h_shared_[i];
+      uint32_t klen = h_klen_[i];
+      uint32_t vl = h_vlen_[i];
+      eoff[i - first] = off;
+      off += varint_len(shared) + varint_len(klen - shared) + varint_len(vl) +
+             (klen - shared) + vl;
+    }
+    bds[b] = {(uint32_t)pb.first, pb.count, pb.unc_size, pb.num_restarts, uout};
+    uout += pb.unc_size;
+  }
+  p->emit_base = first;
+  p->emit_nblocks = nb;
+  ENSURE(p->d_bds, p->bds_cap, sizeof(EmitBlockDesc) * nb);
+  ENSURE(p->d_eoff, p->eoff_cap, sizeof(uint32_t) * nent);
+  ENSURE(p->d_ucblob, p->ucblob_cap, uout);
+  ENSURE(p->d_ebsize, p->eb_cap, sizeof(uint32_t) * nb * 2 + nb); // bsize+csum+btype
+  p->d_ecsum = p->d_ebsize + nb;
+  p->d_ebtype = (uint8_t*)(p->d_ecsum + nb);
+  HIPCHK(hipMemcpyAsync(p->d_bds, bds.data(), sizeof(EmitBlockDesc) * nb,
+                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMemcpyAsync(p->d_eoff, eoff.data(), sizeof(uint32_t) * nent,
+                        hipMemcpyHostToDevice, p->stream));
+  // NOTE: k_emit indexes eoff by absolute survivor index minus chunk base.
+  hipLaunchKernelGGL(k_emit, dim3(nb < 4096 ? nb : 4096), dim3(256), 0, p->stream,
+                     p->d_bds, nb, p->d_sk0, p->d_sk1, p->d_stag, p->d_svoff,
+                     p->d_svlen, p->d_sklen, p->d_eoff - first, p->d_ublob,
+                     p->d_ucblob, o.block_restart_interval);
+  if (o.compression == 1) {
+    p->ccap_per_block = snappy_max_compressed(o.block_size + 1024);
+    ENSURE(p->d_cblob, p->cblob_cap, p->ccap_per_block * nb);
+    hipLaunchKernelGGL(k_compress, dim3(nb < 4096 ? nb : 4096), dim3(64), 0,
+                       p->stream, p->d_bds, nb, p->d_ucblob, p->d_cblob,
+                       p->ccap_per_block, p->d_ebsize, p->d_ebtype);
+  } else {
+    hipLaunchKernelGGL(k_sizes_nocomp, dim3(grid_for(nb)), dim3(256), 0,
+                       p->stream, p->d_bds, nb, p->d_ebsize, p->d_ebtype);
+  }
+  hipLaunchKernelGGL(k_checksum, dim3(grid_for(nb)), dim3(256), 0, p->stream,
+                     p->d_bds, nb, p->d_ucblob, p->d_cblob, p->ccap_per_block,
+                     p->d_ebsize, p->d_ebtype, o.checksum_type, p->d_crc,
+                     p->d_ecsum);
+  comp_sizes->resize(nb);
+  HIPCHK(hipMemcpyAsync(comp_sizes->data(), p->d_ebsize, sizeof(uint32_t) * nb,
+                        hipMemcpyDeviceToHost, p->stream));
+  hipEventRecord(t1, p->stream);
+  HIPCHK(hipStreamSynchronize(p->stream));
+  ms_emit += ms_between(t0, t1);
+  hipEventDestroy(t0);
+  hipEventDestroy(t1);
+  return 0;
+}
+
+int GpuJob::pack_and_fetch(size_t b0, size_t b1, const TableOpts& o,
+                           std::string* out, std::vector<SstIndexEntry>* handles,
+                           std::string* err) {
+  Impl* p = p_;
+  (void)o;
+  uint32_t nb = (uint32_t)(b1 - b0);
+  if (nb == 0) {
+    out->clear();
+    return 0;
+  }
+  std::vector<uint32_t> bsize(p->emit_nblocks);
+  HIPCHK(hipMemcpyAsync(bsize.data(), p->d_ebsize, sizeof(uint32_t) * p->emit_nblocks,
+                        hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  std::vector<uint64_t> outoff(nb);
+  uint64_t acc = 0;
+  handles->clear();
+  for (uint32_t i = 0; i < nb; i++) {
+    outoff[i] = acc;
+    handles->push_back({acc, (uint64_t)bsize[b0 + i]});
+    acc += bsize[b0 + i] + kTrailerSize;
+  }
+  uint64_t* d_outoff;
+  HIPCHK(hipMalloc(&d_outoff, sizeof(uint64_t) * nb));
+  HIPCHK(hipMemcpyAsync(d_outoff, outoff.data(), sizeof(uint64_t) * nb,
+                        hipMemcpyHostToDevice, p->stream));
+  ENSURE(p->d_out_img, p->out_img_cap, acc);
+  hipLaunchKernelGGL(k_pack, dim3(grid_for(nb * 4ull)), dim3(256), 0, p->stream,
+                     p->d_bds, (uint32_t)b0, (uint32_t)b1, p->d_ucblob, p->d_cblob,
+                     p->ccap_per_block, p->d_ebsize, p->d_ebtype, p->d_ecsum,
+                     d_outoff, p->d_out_img);
+  out->resize(acc);
+  hipEvent_t t0, t1;
+  hipEventCreate(&t0);
+  hipEventCreate(&t1);
+  hipEventRecord(t0, p->stream);
+  HIPCHK(hipMemcpyAsync(out->data(), p->d_out_img, acc, hipMemcpyDeviceToHost,
+                        p->stream));
+  hipEventRecord(t1, p->stream);
+  HIPCHK(hipStreamSynchronize(p->stream));
+  ms_d2h += ms_between(t0, t1);
+  hipEventDestroy(t0);
+  hipEventDestroy(t1);
+  (void)hipFree(d_outoff);
+  return 0;
+}
+
+int GpuJob::fetch_block_keys(size_t b0, size_t b1,
+                             std::vector<std::string>* first_keys,
+                             std::vector<std::string>* last_keys,
+                             std::string* err) {
+  Impl* p = p_;
+  uint32_t nb = (uint32_t)(b1 - b0);
+  first_keys->clear();
+  last_keys->clear();
+  if (!nb) return 0;
+  uint8_t* d_keys;
+  HIPCHK(hipMalloc(&d_keys, (uint64_t)nb * 64));
+  hipLaunchKernelGGL(k_block_keys, dim3(grid_for(nb)), dim3(256), 0, p->stream,
+                     p->d_bds, (uint32_t)b0, (uint32_t)b1, p->d_sk0, p->d_sk1,
+                     p->d_stag, p->d_sklen, d_keys);
+  std::vector<uint8_t> h((uint64_t)nb * 64);
+  HIPCHK(hipMemcpyAsync(h.data(), d_keys, h.size(), hipMemcpyDeviceToHost,
+                        p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  (void)hipFree(d_keys);
+  for (uint32_t i = 0; i < nb; i++) {
+    const uint8_t* o = h.data() + (uint64_t)i * 64;
+    first_keys->emplace_back((const char*)o + 1, o[0]);
+    last_keys->emplace_back((const char*)o + 33, o[32]);
+  }
+  return 0;
+}
+
+int GpuJob::gather_entries(uint64_t first, uint32_t count,
+                           std::vector<std::pair<std::string, std::string>>* kvs,
+                           std::string* err) {
+  Impl* p = p_;
+  kvs->clear();
+  if (!count) return 0;
+  // record layout: [klen u8][key][vlen u32][value]
+  std::vector<uint64_t> recoff(count);
+  uint64_t acc = 0;
+  for (uint32_t i = 0; i < count; i++) {
+    recoff[i] = acc;
+    acc += 1 + h_klen_[first + i] + 4 + h_vlen_[first + i];
+  }
+  uint64_t* d_recoff;
+  uint8_t* d_out;
+  HIPCHK(hipMalloc(&d_recoff, sizeof(uint64_t) * count));
+  HIPCHK(hipMalloc(&d_out, acc));
+  HIPCHK(hipMemcpyAsync(d_recoff, recoff.data(), sizeof(uint64_t) * count,
+                        hipMemcpyHostToDevice, p->stream));
+  hipLaunchKernelGGL(k_gather_range, dim3(grid_for(count)), dim3(256), 0,
+                     p->stream, p->d_sk0, p->d_sk1, p->d_stag, p->d_svoff,
+                     p->d_svlen, p->d_sklen, p->d_ublob, first, count, d_recoff,
+                     d_out);
+  std::vector<uint8_t> h(acc);
+  HIPCHK(hipMemcpyAsync(h.data(), d_out, acc, hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  (void)hipFree(d_recoff);
+  (void)hipFree(d_out);
+  for (uint32_t i = 0; i < count; i++) {
+    const uint8_t* r = h.data() + recoff[i];
+    uint32_t klen = r[0];
+    uint32_t vl;
+    memcpy(&vl, r + 1 + klen, 4);
+    kvs->emplace_back(std::string((const char*)r + 1, klen),
+                      std::string((const char*)r + 5 + klen, vl));
+  }
+  return 0;
+}
+
+int GpuJob::seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* mx,
+                       uint64_t* n_tombstones, std::string* err) {
+  Impl* p = p_;
+  unsigned long long* d;
+  HIPCHK(hipMalloc(&d, 24));
+  unsigned long long init[3] = {~0ull, 0, 0};
+  HIPCHK(hipMemcpyAsync(d, init, 24, hipMemcpyHostToDevice, p->stream));
+  hipLaunchKernelGGL(k_seq_minmax, dim3(grid_for(count)), dim3(256), 0, p->stream,
+                     p->d_stag, first, count, d, d + 1, d + 2);
+  unsigned long long out[3];
+  HIPCHK(hipMemcpyAsync(out, d, 24, hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  (void)hipFree(d);
+  *mn = out[0];
+  *mx = out[1];
+  *n_tombstones = out[2];
+  return 0;
+}
+
+} // namespace dcw

@@ -1,0 +1,431 @@
+// dcw_worker.cpp — job orchestration + the C ABI (include/dcw.h).
+//
+// dcw_execute mirrors the worker side of ToplingDB's dcompact seam
+// (CompactionExecutor::Execute, db/compaction/compaction_executor.h:165-171;
+// worker lifecycle like DBImplSecondary::CompactWithoutInstallation,
+// db/db_impl/db_impl_secondary.cc:805): deserialize job -> open input SSTs
+// -> GPU hot loop (decode/merge/dedup/encode) -> write SSTs -> return file
+// metas.  The hot path runs on the GPU (dcw_kernels.hip); the host does I/O,
+// the block/file-cut plan FSM over per-entry size metadata
+// (SURVEY.md §7 "plan pass"), and the per-file meta tail (<0.1% of bytes).
+// There is NO CPU fallback: without a GPU, dcw_init fails.
+#include <inttypes.h>
+#include <sys/stat.h>
+
+#include <cstdio>
+#include <cstring>
+#include <mutex>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+#include "../../include/dcw.h"
+#include "dcw_gpu.h"
+#include "dcw_host.h"
+
+namespace dcw {
+namespace {
+
+std::mutex g_mu;
+bool g_inited = false;
+uint64_t g_next_stage_handle = 1;
+
+struct StagedJob {
+  StagedInput dev;
+  uint64_t in_bytes = 0;
+};
+std::unordered_map<uint64_t, StagedJob*> g_staged;
+
+uint64_t now_usec() {
+  struct timespec ts;
+  clock_gettime(CLOCK_MONOTONIC, &ts);
+  return (uint64_t)ts.tv_sec * 1000000 + ts.tv_nsec / 1000;
+}
+
+int read_file(const std::string& path, std::string* out, std::string* err) {
+  FILE* f = fopen(path.c_str(), "rb");
+  if (!f) {
+    *err = "cannot open " + path;
+    return -1;
+  }
+  fseek(f, 0, SEEK_END);
+  long sz = ftell(f);
+  fseek(f, 0, SEEK_SET);
+  out->resize((size_t)sz);
+  if (fread(&(*out)[0], 1, (size_t)sz, f) != (size_t)sz) {
+    fclose(f);
+    *err = "short read " + path;
+    return -1;
+  }
+  fclose(f);
+  return 0;
+}
+
+struct LoadedInputs {
+  std::string blob; // all files concatenated
+  GpuInputs gi;
+  uint64_t in_bytes = 0;
+};
+
+int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err) {
+  L->gi.run_block_begin.push_back(0);
+  uint32_t cstype = 0xffffffff;
+  for (uint32_t r = 0; r < d->num_runs; r++) {
+    for (uint32_t f = 0; f < d->runs[r].num_files; f++) {
+      std::string img;
+      if (read_file(d->runs[r].files[f], &img, err) != 0) return -1;
+      uint64_t base = L->blob.size();
+      ParsedSst ps = parse_sst((const uint8_t*)img.data(), img.size());
+      if (!ps.ok) {
+        *err = std::string(d->runs[r].files[f]) + ": " + ps.error;
+        return -1;
+      }
+      if (cstype == 0xffffffff) cstype = ps.checksum_type;
+      if (cstype != ps.checksum_type) {
+        *err = "mixed input checksum types unsupported";
+        return -1;
+      }
+      for (auto& h : ps.data_blocks)
+        L->gi.blocks.push_back({base + h.off, (uint32_t)h.size});
+      L->blob += img;
+      L->in_bytes += img.size();
+    }
+    L->gi.run_block_begin.push_back((uint32_t)L->gi.blocks.size());
+  }
+  L->gi.blob = (const uint8_t*)L->blob.data();
+  L->gi.blob_size = L->blob.size();
+  L->gi.checksum_type = cstype == 0xffffffff ? 4 : cstype;
+  return 0;
+}
+
+TableOpts opts_from_desc(const dcw_job_desc* d) {
+  TableOpts o;
+  if (d->block_size) o.block_size = d->block_size;
+  if (d->block_restart_interval) o.block_restart_interval = d->block_restart_interval;
+  if (d->format_version) o.format_version = d->format_version;
+  o.checksum_type = d->checksum_type;
+  o.compression = d->compression;
+  if (d->block_size_deviation) o.block_size_deviation = d->block_size_deviation;
+  o.db_id = d->db_id ? d->db_id : "";
+  o.db_session_id = d->db_session_id ? d->db_session_id : "";
+  o.db_host_id = d->db_host_id ? d->db_host_id : "";
+  o.cf_name = d->cf_name ? d->cf_name : "default";
+  o.cf_id = d->cf_id;
+  o.creation_time = d->oldest_ancester_time ? d->oldest_ancester_time : d->current_time;
+  o.file_creation_time = d->current_time;
+  o.oldest_key_time = 0;
+  o.level_at_creation = d->output_level;
+  return o;
+}
+
+int fail(dcw_job_result* res, int code, const std::string& msg) {
+  res->status = code;
+  snprintf(res->error, sizeof(res->error), "%s", msg.c_str());
+  return code;
+}
+
+} // namespace
+} // namespace dcw
+
+using namespace dcw;
+
+extern "C" {
+
+const char* dcw_version(void) { return "toplingdb_amd dcompact worker r1 (gfx950)"; }
+
+int32_t dcw_init(int32_t device_ordinal) {
+  std::lock_guard<std::mutex> lk(g_mu);
+  std::string err;
+  if (gpu_init(device_ordinal, &err) != 0) {
+    fprintf(stderr, "dcw_init: %s\n", err.c_str());
+    return -1;
+  }
+  g_inited = true;
+  return 0;
+}
+
+void dcw_shutdown(void) {
+  std::lock_guard<std::mutex> lk(g_mu);
+  for (auto& kv : g_staged) delete kv.second;
+  g_staged.clear();
+  gpu_shutdown();
+  g_inited = false;
+}
+
+void dcw_free_result(dcw_job_result* res) {
+  free(res->files);
+  res->files = nullptr;
+  res->num_files = 0;
+}
+
+uint64_t dcw_stage_inputs(const dcw_job_desc* d) {
+  std::lock_guard<std::mutex> lk(g_mu);
+  if (!g_inited) return 0;
+  std::string err;
+  LoadedInputs L;
+  if (load_inputs(d, &L, &err) != 0) {
+    fprintf(stderr, "dcw_stage_inputs: %s\n", err.c_str());
+    return 0;
+  }
+  GpuJob job;
+  if (job.stage(L.gi, &err) != 0) {
+    fprintf(stderr, "dcw_stage_inputs: %s\n", err.c_str());
+    return 0;
+  }
+  StagedJob* sj = new StagedJob;
+  if (job.stage_release(&sj->dev, &err) != 0) {
+    delete sj;
+    return 0;
+  }
+  sj->in_bytes = L.in_bytes;
+  uint64_t h = g_next_stage_handle++;
+  g_staged[h] = sj;
+  return h;
+}
+
+void dcw_release_staged(uint64_t handle) {
+  std::lock_guard<std::mutex> lk(g_mu);
+  auto it = g_staged.find(handle);
+  if (it != g_staged.end()) {
+    delete it->second;
+    g_staged.erase(it);
+  }
+}
+
+int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
+  memset(res, 0, sizeof(*res));
+  std::lock_guard<std::mutex> lk(g_mu); // one GPU job slot per process, round 1
+  if (!g_inited)
+    return fail(res, 10, "dcw_init not called or no gfx950 device (no CPU fallback)");
+  if (d->struct_size != sizeof(dcw_job_desc))
+    return fail(res, 11, "ABI mismatch: dcw_job_desc size");
+  if (d->comparator_name &&
+      strcmp(d->comparator_name, "leveldb.BytewiseComparator") != 0)
+    return fail(res, 12, "unsupported comparator (bytewise only)");
+  if (d->compression > 1)
+    return fail(res, 13, "unsupported compression (none/snappy only)");
+  if (d->num_grandparents > 0)
+    return fail(res, 14, "grandparent-aware file cuts not yet in the GPU worker envelope");
+
+  uint64_t t_start = now_usec();
+  std::string err;
+
+  // ---- read + parse inputs (host I/O) ----
+  uint64_t t0 = now_usec();
+  GpuJob job;
+  uint64_t in_bytes = 0;
+  LoadedInputs L;
+  StagedJob* staged = nullptr;
+  if (d->staged_handle) {
+    auto it = g_staged.find(d->staged_handle);
+    if (it == g_staged.end()) return fail(res, 15, "bad staged handle");
+    staged = it->second;
+    in_bytes = staged->in_bytes;
+  } else {
+    if (load_inputs(d, &L, &err) != 0) return fail(res, 16, err);
+    in_bytes = L.in_bytes;
+  }
+  res->t_read_usec = now_usec() - t0;
+
+  // ---- GPU pipeline ----
+  if (staged) {
+    if (job.stage_adopt(staged->dev, &err) != 0) return fail(res, 17, err);
+  } else {
+    if (job.stage(L.gi, &err) != 0) return fail(res, 17, err);
+  }
+  if (job.decode(&err) != 0) return fail(res, 18, err);
+  if (job.merge(&err) != 0) return fail(res, 19, err);
+  if (job.dedup(d, &err) != 0) return fail(res, 20, err);
+
+  size_t nsurv = job.num_survivors();
+  const auto& klen = job.plan_klen();
+  const auto& vlen = job.plan_vlen();
+  PlanIn pin{job.plan_shared().data(), klen.data(), vlen.data(), nullptr, nsurv};
+
+  TableOpts base = opts_from_desc(d);
+  uint64_t next_file_number = d->next_file_number;
+  std::vector<dcw_output_file> out_files;
+  uint64_t total_out_bytes = 0, total_out_entries = 0;
+  uint64_t plan_usec = 0, write_usec = 0;
+
+  size_t s = 0;
+  while (s < nsurv) {
+    TableOpts o = base;
+    o.orig_file_number = next_file_number++;
+    std::string image; // data-block region accumulated on host
+    std::vector<SstIndexEntry> handles;
+    std::vector<std::string> first_keys, last_keys;
+    std::vector<uint32_t> block_counts;
+    size_t cur = s;
+    bool cut = false;
+    uint64_t cut_entry = 0;
+    // chunked: plan -> GPU emit -> fetch, until the file-size cut fires
+    while (!cut && cur < nsurv) {
+      uint64_t tp0 = now_usec();
+      uint64_t already = image.size();
+      uint64_t want = d->target_file_size > already
+                          ? d->target_file_size - already
+                          : (64 << 10);
+      // cushion: compressed bytes per uncompressed byte unknown; assume
+      // >= 0.45 for snappy inputs, 1.0 otherwise, plus a block of slack
+      uint64_t min_unc = o.compression == 1 ? want * 23 / 10 : want + 2 * o.block_size;
+      std::vector<PlannedBlock> blocks = plan_blocks(pin, cur, o, min_unc);
+      plan_usec += now_usec() - tp0;
+      if (blocks.empty()) break;
+      std::vector<uint32_t> csizes;
+      if (job.emit_blocks(blocks, o, &csizes, &err) != 0) return fail(res, 21, err);
+      // fetch everything emitted; excess past the cut is discarded
+      std::string chunk;
+      std::vector<SstIndexEntry> chandles;
+      if (job.pack_and_fetch(0, blocks.size(), o, &chunk, &chandles, &err) != 0)
+        return fail(res, 22, err);
+      std::vector<std::string> cfirst, clast;
+      if (job.fetch_block_keys(0, blocks.size(), &cfirst, &clast, &err) != 0)
+        return fail(res, 23, err);
+      // walk blocks: find the flush that crosses the target
+      // (cut = first entry after it; the open block then holds exactly one
+      // entry — AddToOutput/FileSize interplay, compaction_outputs.cc:356-420)
+      uint64_t tw0 = now_usec();
+      size_t take = blocks.size();
+      for (size_t b = 0; b < blocks.size(); b++) {
+        uint64_t off_after = image.size() + chandles[b].off + chandles[b].size + kTrailerSize;
+        bool more_entries = blocks[b].first + blocks[b].count < nsurv;
+        if (off_after >= d->target_file_size && more_entries) {
+          take = b + 1;
+          cut = true;
+          cut_entry = blocks[b].first + blocks[b].count; // e0 = first of next block
+          break;
+        }
+      }
+      uint64_t base_off = image.size();
+      for (size_t b = 0; b < take; b++) {
+        handles.push_back({base_off + chandles[b].off, chandles[b].size});
+        first_keys.push_back(cfirst[b]);
+        last_keys.push_back(clast[b]);
+        block_counts.push_back(blocks[b].count);
+      }
+      uint64_t take_bytes =
+          take ? chandles[take - 1].off + chandles[take - 1].size + kTrailerSize : 0;
+      image.append(chunk.data(), take_bytes);
+      cur = take ? blocks[take - 1].first + blocks[take - 1].count : cur;
+      write_usec += now_usec() - tw0;
+      if (take < blocks.size()) break; // cut decided inside this chunk
+    }
+    // partial single-entry block after a size cut
+    uint64_t file_first = s;
+    uint64_t file_count = cur - s;
+    if (cut) {
+      std::vector<std::pair<std::string, std::string>> kvs;
+      if (job.gather_entries(cut_entry, 1, &kvs, &err) != 0) return fail(res, 24, err);
+      BlockBuilder bb(o.block_restart_interval, false);
+      bb.AddWithLastKey((const uint8_t*)kvs[0].first.data(), kvs[0].first.size(),
+                        (const uint8_t*)kvs[0].second.data(), kvs[0].second.size(),
+                        nullptr, 0);
+      std::string contents = bb.Finish();
+      SstIndexEntry h = append_block(image, o, (const uint8_t*)contents.data(),
+                                     contents.size(), true);
+      handles.push_back(h);
+      first_keys.push_back(kvs[0].first);
+      last_keys.push_back(kvs[0].first);
+      block_counts.push_back(1);
+      file_count += 1;
+      cur = cut_entry + 1;
+    }
+    if (handles.empty()) break; // nothing left
+    // separators (FindShortestInternalKeySeparator between adjacent blocks;
+    // last block keeps its last key — default kShortenSeparators mode)
+    uint64_t tw1 = now_usec();
+    std::vector<std::string> seps(handles.size());
+    bool sep_key_plus_seq = false;
+    for (size_t b = 0; b < handles.size(); b++) {
+      std::string sep = last_keys[b];
+      if (b + 1 < handles.size()) {
+        shorten_separator(sep, (const uint8_t*)first_keys[b + 1].data(),
+                          first_keys[b + 1].size());
+        size_t su = sep.size() - 8, nu = first_keys[b + 1].size() - 8;
+        if (su == nu && memcmp(sep.data(), first_keys[b + 1].data(), su) == 0)
+          sep_key_plus_seq = true;
+      }
+      seps[b] = sep;
+    }
+    // stats
+    TailStats st;
+    st.num_data_blocks = handles.size();
+    st.data_size = image.size();
+    st.num_entries = file_count;
+    for (uint64_t i = file_first; i < file_first + file_count; i++) {
+      st.raw_key_size += klen[i];
+      st.raw_value_size += vlen[i];
+    }
+    uint64_t mn_seq = 0, mx_seq = 0, n_tomb = 0;
+    if (job.seq_minmax(file_first, file_count, &mn_seq, &mx_seq, &n_tomb, &err) != 0)
+      return fail(res, 25, err);
+    st.num_deletions = n_tomb;
+    uint64_t tail_start = image.size();
+    image += build_tail(o, st, handles, seps, !sep_key_plus_seq, tail_start);
+    // write the file
+    char path[600];
+    snprintf(path, sizeof(path), "%s/%06" PRIu64 ".sst", d->output_dir,
+             o.orig_file_number);
+    FILE* f = fopen(path, "wb");
+    if (!f) return fail(res, 26, std::string("cannot write ") + path);
+    fwrite(image.data(), 1, image.size(), f);
+    fclose(f);
+    write_usec += now_usec() - tw1;
+
+    dcw_output_file of;
+    memset(&of, 0, sizeof(of));
+    snprintf(of.path, sizeof(of.path), "%s", path);
+    of.file_number = o.orig_file_number;
+    of.file_size = image.size();
+    std::string smallest = first_keys.front();
+    std::string largest = last_keys.back();
+    of.smallest_len = (uint32_t)smallest.size();
+    memcpy(of.smallest_ikey, smallest.data(), std::min<size_t>(64, smallest.size()));
+    of.largest_len = (uint32_t)largest.size();
+    memcpy(of.largest_ikey, largest.data(), std::min<size_t>(64, largest.size()));
+    of.smallest_seqno = mn_seq == ~0ull ? 0 : mn_seq;
+    of.largest_seqno = mx_seq;
+    of.num_entries = file_count;
+    out_files.push_back(of);
+    total_out_bytes += image.size();
+    total_out_entries += file_count;
+    s = cur;
+  }
+
+  res->num_files = (uint32_t)out_files.size();
+  res->files = (dcw_output_file*)malloc(sizeof(dcw_output_file) * out_files.size());
+  memcpy(res->files, out_files.data(), sizeof(dcw_output_file) * out_files.size());
+  res->in_bytes = in_bytes;
+  res->out_bytes = total_out_bytes;
+  res->in_entries = job.num_input_entries();
+  res->out_entries = total_out_entries;
+  res->t_h2d_usec = (uint64_t)(job.ms_h2d * 1000);
+  res->t_gpu_usec =
+      (uint64_t)((job.ms_decode + job.ms_merge + job.ms_dedup + job.ms_emit) * 1000);
+  res->t_plan_usec = plan_usec;
+  res->t_d2h_usec = (uint64_t)(job.ms_d2h * 1000);
+  res->t_write_usec = write_usec;
+  res->work_time_usec = now_usec() - t_start;
+  res->status = 0;
+  return 0;
+}
+
+int32_t dcw_gen_sst(const char* path, uint64_t seed, uint64_t num_entries,
+                    uint32_t key_len, uint32_t value_len, uint64_t seq_base,
+                    uint32_t compression, uint32_t checksum_type,
+                    uint64_t file_number, const char* db_id,
+                    const char* db_session_id, uint64_t current_time) {
+  TableOpts o;
+  o.compression = compression;
+  o.checksum_type = checksum_type;
+  o.db_id = db_id ? db_id : "";
+  o.db_session_id = db_session_id ? db_session_id : "";
+  o.db_host_id = "dcw-host";
+  o.orig_file_number = file_number;
+  o.creation_time = current_time;
+  o.file_creation_time = current_time;
+  return gen_sst_file(path, seed, num_entries, key_len, value_len, seq_base, o);
+}
+
+} // extern "C"
