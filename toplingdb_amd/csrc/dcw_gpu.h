@@ -72,9 +72,10 @@ class GpuJob {
   int emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts& o,
                   std::vector<uint32_t>* comp_sizes, std::string* err);
   // Pack blocks [b0,b1) of the last emit_blocks call into one contiguous
-  // [body|trailer]* image and copy to host.
-  int pack_and_fetch(size_t b0, size_t b1, const TableOpts& o, std::string* out,
-                     std::vector<SstIndexEntry>* handles, std::string* err);
+  // [body|trailer]* image on device and D2H it straight into host_dst
+  // (total_bytes = sum of (csize+5)); outoff[i] = image offset of block b0+i.
+  int pack_into(size_t b0, size_t b1, const std::vector<uint64_t>& outoff,
+                uint8_t* host_dst, size_t total_bytes, std::string* err);
   // first/last internal keys of blocks [b0,b1) of the last emit
   int fetch_block_keys(size_t b0, size_t b1, std::vector<std::string>* first_keys,
                        std::vector<std::string>* last_keys, std::string* err);

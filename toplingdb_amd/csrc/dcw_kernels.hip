@@ -25,6 +25,8 @@
 
 #include <cstdio>
 #include <cstring>
+#include <map>
+#include <mutex>
 
 #include "dcw_gpu.h"
 
@@ -859,8 +861,52 @@ __global__ void k_seq_minmax(const uint64_t* __restrict__ s_tag, uint64_t first,
 // ------------------------------------------------------------------
 // host-side Impl
 // ------------------------------------------------------------------
+// ------------------------------------------------------------------
+// per-kernel timing (HIP events on the pipeline stream) + algorithmic
+// byte accounting, for bench.py's roofline object
+// ------------------------------------------------------------------
+struct KStat {
+  uint64_t launches = 0;
+  double ms = 0;
+  double alg_bytes = 0;
+};
+static std::map<std::string, KStat>& kstats() {
+  static std::map<std::string, KStat> m;
+  return m;
+}
+static std::mutex g_kmu;
+
+struct KEv {
+  const char* name;
+  double bytes;
+  hipEvent_t a, b;
+};
+
 struct GpuJob::Impl {
   hipStream_t stream = nullptr;
+  std::vector<KEv> kpending;
+  void kbegin(const char* n, double bytes) {
+    KEv e{n, bytes, nullptr, nullptr};
+    (void)hipEventCreate(&e.a);
+    (void)hipEventCreate(&e.b);
+    (void)hipEventRecord(e.a, stream);
+    kpending.push_back(e);
+  }
+  void kend() { (void)hipEventRecord(kpending.back().b, stream); }
+  void kresolve() { // call after a stream sync
+    std::lock_guard<std::mutex> lk(g_kmu);
+    for (auto& e : kpending) {
+      float ms = 0;
+      (void)hipEventElapsedTime(&ms, e.a, e.b);
+      KStat& s = kstats()[e.name];
+      s.launches++;
+      s.ms += ms;
+      s.alg_bytes += e.bytes;
+      (void)hipEventDestroy(e.a);
+      (void)hipEventDestroy(e.b);
+    }
+    kpending.clear();
+  }
   bool borrowed_stage = false; // staged buffers owned by a StagedInput
   // staged input
   uint8_t* d_blob = nullptr;
@@ -1073,9 +1119,12 @@ int GpuJob::decode(std::string* err) {
   uint32_t nb = p->n_blocks;
   HIPCHK(hipMalloc(&p->d_usize, sizeof(uint32_t) * nb));
   HIPCHK(hipMalloc(&p->d_btype_in, nb));
+  double in_block_bytes = 0; // filled below from bsize D2H; verify reads them
+  p->kbegin("verify_checksum", 0);
   hipLaunchKernelGGL(k_verify_usize, dim3(grid_for(nb)), dim3(256), 0, p->stream,
                      p->d_blob, p->d_boff, p->d_bsize, nb, p->checksum_type,
                      p->d_crc, p->d_usize, p->d_btype_in, p->d_err);
+  p->kend();
   // host scan of usize -> uoff
   std::vector<uint32_t> usize(nb);
   HIPCHK(hipMemcpyAsync(usize.data(), p->d_usize, sizeof(uint32_t) * nb,
@@ -1092,15 +1141,22 @@ int GpuJob::decode(std::string* err) {
   for (uint32_t i = 0; i < nb; i++) {
     uoff[i] = acc;
     acc += usize[i];
+    in_block_bytes += usize[i]; // ~= compressed size; close enough for alg accounting
   }
   p->ublob_size = acc;
   HIPCHK(hipMalloc(&p->d_uoff, sizeof(uint64_t) * nb));
   HIPCHK(hipMemcpyAsync(p->d_uoff, uoff.data(), sizeof(uint64_t) * nb,
                         hipMemcpyHostToDevice, p->stream));
   HIPCHK(hipMalloc(&p->d_ublob, acc ? acc : 1));
+  {
+    std::lock_guard<std::mutex> lk(g_kmu);
+    kstats()["verify_checksum"].alg_bytes += in_block_bytes;
+  }
+  p->kbegin("decompress", in_block_bytes + (double)acc);
   hipLaunchKernelGGL(k_decompress, dim3(grid_for(nb * 4ull)), dim3(256), 0,
                      p->stream, p->d_blob, p->d_boff, p->d_bsize, p->d_btype_in,
                      p->d_uoff, p->d_usize, nb, p->d_ublob, p->d_err);
+  p->kend();
   HIPCHK(hipMalloc(&p->d_nrestarts, sizeof(uint32_t) * nb));
   hipLaunchKernelGGL(k_num_restarts, dim3(grid_for(nb)), dim3(256), 0, p->stream,
                      p->d_ublob, p->d_uoff, p->d_usize, nb, p->d_nrestarts,
@@ -1135,10 +1191,12 @@ int GpuJob::decode(std::string* err) {
                         hipMemcpyHostToDevice, p->stream));
   HIPCHK(hipMemcpyAsync(p->d_iv_local, iv_local.data(), sizeof(uint32_t) * niv,
                         hipMemcpyHostToDevice, p->stream));
+  p->kbegin("count_entries", (double)p->ublob_size);
   hipLaunchKernelGGL(k_count_entries, dim3(grid_for(niv)), dim3(256), 0, p->stream,
                      p->d_ublob, p->d_uoff, p->d_usize, p->d_nrestarts,
                      p->d_iv_block, p->d_iv_local, (uint32_t)niv, p->d_iv_cnt,
                      p->d_err);
+  p->kend();
   std::vector<uint32_t> iv_cnt(niv);
   HIPCHK(hipMemcpyAsync(iv_cnt.data(), p->d_iv_cnt, sizeof(uint32_t) * niv,
                         hipMemcpyDeviceToHost, p->stream));
@@ -1171,11 +1229,13 @@ int GpuJob::decode(std::string* err) {
   HIPCHK(hipMalloc(&p->d_voff, sizeof(uint64_t) * total_entries));
   HIPCHK(hipMalloc(&p->d_vlen, sizeof(uint32_t) * total_entries));
   HIPCHK(hipMalloc(&p->d_klen, total_entries));
+  p->kbegin("decode_entries", (double)p->ublob_size + 45.0 * total_entries);
   hipLaunchKernelGGL(k_decode_entries, dim3(grid_for(niv)), dim3(256), 0,
                      p->stream, p->d_ublob, p->d_uoff, p->d_usize, p->d_nrestarts,
                      p->d_iv_block, p->d_iv_local, p->d_iv_base, (uint32_t)niv,
                      p->d_ent[0], p->d_voff, p->d_vlen, p->d_klen,
                      p->d_uklen_probe, p->d_err);
+  p->kend();
   uint32_t uklen = 0;
   HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost, p->stream));
   HIPCHK(hipMemcpyAsync(&uklen, p->d_uklen_probe, 4, hipMemcpyDeviceToHost,
@@ -1183,6 +1243,7 @@ int GpuJob::decode(std::string* err) {
   hipEventRecord(t1, p->stream);
   HIPCHK(hipStreamSynchronize(p->stream));
   ms_decode += ms_between(t0, t1);
+  p->kresolve();
   hipEventDestroy(t0);
   hipEventDestroy(t1);
   if (err_host) {
@@ -1211,9 +1272,11 @@ int GpuJob::merge(std::string* err) {
     for (size_t i = 0; i + 1 < k; i += 2) {
       uint64_t a0 = bounds[i], a1 = bounds[i + 1], b1 = bounds[i + 2];
       uint64_t nA = a1 - a0, nB = b1 - a1;
+      p->kbegin("merge_pair", 64.0 * (double)(nA + nB));
       hipLaunchKernelGGL(k_merge_pair<16>, dim3(grid_for((nA + nB) / 16 + 1)),
                          dim3(256), 0, p->stream, p->d_ent[cur] + a0, nA,
                          p->d_ent[cur] + a1, nB, p->d_ent[cur ^ 1] + a0);
+      p->kend();
       nbounds.push_back(b1);
     }
     if (k % 2) { // odd leftover run: copy through
@@ -1230,6 +1293,7 @@ int GpuJob::merge(std::string* err) {
   hipEventRecord(t1, p->stream);
   HIPCHK(hipStreamSynchronize(p->stream));
   ms_merge += ms_between(t0, t1);
+  p->kresolve();
   hipEventDestroy(t0);
   hipEventDestroy(t1);
   return 0;
@@ -1270,8 +1334,10 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   uint64_t n = p->n_entries;
   const ulong4* ents = p->d_ent[p->final_buf];
   HIPCHK(hipMalloc(&p->d_head, n));
+  p->kbegin("mark_heads", 17.0 * n);
   hipLaunchKernelGGL(k_mark_heads, dim3(grid_for(n)), dim3(256), 0, p->stream,
                      ents, n, p->d_head);
+  p->kend();
   // head positions via scan, then gather head indices on host
   HIPCHK(hipMalloc(&p->d_pos, sizeof(uint32_t) * n));
   uint64_t ngroups = 0;
@@ -1340,9 +1406,15 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   HIPCHK(hipMalloc(&p->d_newtag, sizeof(uint64_t) * n));
   HIPCHK(hipMalloc(&p->d_clearv, n));
   HIPCHK(hipMalloc(&p->d_gflags, ngroups ? ngroups : 1));
+  // entries skipped by multi-consume FSM paths never store their slot:
+  // they must read as "dropped"
+  HIPCHK(hipMemsetAsync(p->d_survive, 0, n, p->stream));
+  HIPCHK(hipMemsetAsync(p->d_clearv, 0, n, p->stream));
+  p->kbegin("group_fsm", 42.0 * n);
   hipLaunchKernelGGL(k_group_fsm, dim3(grid_for(ngroups)), dim3(256), 0,
                      p->stream, ents, n, p->d_headidx, ngroups, P, p->d_survive,
                      p->d_newtag, p->d_clearv, p->d_gflags, ~0ull, p->d_err);
+  p->kend();
   // SeekToFirst lag: re-run the first group that produced output if its SD
   // decisions were has_outputted-sensitive
   std::vector<uint8_t> gflags(ngroups);
@@ -1378,13 +1450,17 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   HIPCHK(hipMalloc(&p->d_svlen, sizeof(uint32_t) * (nsurv + 1)));
   HIPCHK(hipMalloc(&p->d_sklen, nsurv + 1));
   HIPCHK(hipMalloc(&p->d_sshared, nsurv + 1));
+  p->kbegin("gather_survivors", 90.0 * n);
   hipLaunchKernelGGL(k_gather_survivors, dim3(grid_for(n)), dim3(256), 0,
                      p->stream, ents, n, p->d_survive, p->d_pos, p->d_newtag,
                      p->d_clearv, p->d_voff, p->d_vlen, p->d_klen, p->d_sk0,
                      p->d_sk1, p->d_stag, p->d_svoff, p->d_svlen, p->d_sklen);
+  p->kend();
+  p->kbegin("shared_prefix", 50.0 * nsurv);
   hipLaunchKernelGGL(k_shared_prefix, dim3(grid_for(nsurv)), dim3(256), 0,
                      p->stream, p->d_sk0, p->d_sk1, p->d_stag, p->d_sklen, nsurv,
                      p->d_sshared);
+  p->kend();
   // plan metadata D2H
   h_shared_.resize(nsurv);
   h_klen_.resize(nsurv);
@@ -1400,6 +1476,7 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   hipEventRecord(t1, p->stream);
   HIPCHK(hipStreamSynchronize(p->stream));
   ms_dedup += ms_between(t0, t1);
+  p->kresolve();
   hipEventDestroy(t0);
   hipEventDestroy(t1);
   if (d_snaps) (void)hipFree(d_snaps);
@@ -1453,79 +1530,71 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   HIPCHK(hipMemcpyAsync(p->d_eoff, eoff.data(), sizeof(uint32_t) * nent,
                         hipMemcpyHostToDevice, p->stream));
   // NOTE: k_emit indexes eoff by absolute survivor index minus chunk base.
+  p->kbegin("emit", 2.0 * (double)uout);
   hipLaunchKernelGGL(k_emit, dim3(nb < 4096 ? nb : 4096), dim3(256), 0, p->stream,
                      p->d_bds, nb, p->d_sk0, p->d_sk1, p->d_stag, p->d_svoff,
                      p->d_svlen, p->d_sklen, p->d_eoff - first, p->d_ublob,
                      p->d_ucblob, o.block_restart_interval);
+  p->kend();
   if (o.compression == 1) {
     p->ccap_per_block = snappy_max_compressed(o.block_size + 1024);
     ENSURE(p->d_cblob, p->cblob_cap, p->ccap_per_block * nb);
+    p->kbegin("compress", 1.6 * (double)uout);
     hipLaunchKernelGGL(k_compress, dim3(nb < 4096 ? nb : 4096), dim3(64), 0,
                        p->stream, p->d_bds, nb, p->d_ucblob, p->d_cblob,
                        p->ccap_per_block, p->d_ebsize, p->d_ebtype);
+    p->kend();
   } else {
     hipLaunchKernelGGL(k_sizes_nocomp, dim3(grid_for(nb)), dim3(256), 0,
                        p->stream, p->d_bds, nb, p->d_ebsize, p->d_ebtype);
   }
+  p->kbegin("checksum", (double)uout);
   hipLaunchKernelGGL(k_checksum, dim3(grid_for(nb)), dim3(256), 0, p->stream,
                      p->d_bds, nb, p->d_ucblob, p->d_cblob, p->ccap_per_block,
                      p->d_ebsize, p->d_ebtype, o.checksum_type, p->d_crc,
                      p->d_ecsum);
+  p->kend();
   comp_sizes->resize(nb);
   HIPCHK(hipMemcpyAsync(comp_sizes->data(), p->d_ebsize, sizeof(uint32_t) * nb,
                         hipMemcpyDeviceToHost, p->stream));
   hipEventRecord(t1, p->stream);
   HIPCHK(hipStreamSynchronize(p->stream));
   ms_emit += ms_between(t0, t1);
+  p->kresolve();
   hipEventDestroy(t0);
   hipEventDestroy(t1);
   return 0;
 }
 
-int GpuJob::pack_and_fetch(size_t b0, size_t b1, const TableOpts& o,
-                           std::string* out, std::vector<SstIndexEntry>* handles,
-                           std::string* err) {
+int GpuJob::pack_into(size_t b0, size_t b1, const std::vector<uint64_t>& outoff,
+                      uint8_t* host_dst, size_t total_bytes, std::string* err) {
   Impl* p = p_;
-  (void)o;
   uint32_t nb = (uint32_t)(b1 - b0);
-  if (nb == 0) {
-    out->clear();
-    return 0;
-  }
-  std::vector<uint32_t> bsize(p->emit_nblocks);
-  HIPCHK(hipMemcpyAsync(bsize.data(), p->d_ebsize, sizeof(uint32_t) * p->emit_nblocks,
-                        hipMemcpyDeviceToHost, p->stream));
-  HIPCHK(hipStreamSynchronize(p->stream));
-  std::vector<uint64_t> outoff(nb);
-  uint64_t acc = 0;
-  handles->clear();
-  for (uint32_t i = 0; i < nb; i++) {
-    outoff[i] = acc;
-    handles->push_back({acc, (uint64_t)bsize[b0 + i]});
-    acc += bsize[b0 + i] + kTrailerSize;
-  }
+  if (nb == 0) return 0;
   uint64_t* d_outoff;
-  HIPCHK(hipMalloc(&d_outoff, sizeof(uint64_t) * nb));
+  HIPCHK(hipMallocAsync(&d_outoff, sizeof(uint64_t) * nb, p->stream));
   HIPCHK(hipMemcpyAsync(d_outoff, outoff.data(), sizeof(uint64_t) * nb,
                         hipMemcpyHostToDevice, p->stream));
-  ENSURE(p->d_out_img, p->out_img_cap, acc);
+  ENSURE(p->d_out_img, p->out_img_cap, total_bytes);
+  p->kbegin("pack", 2.0 * (double)total_bytes);
   hipLaunchKernelGGL(k_pack, dim3(grid_for(nb * 4ull)), dim3(256), 0, p->stream,
                      p->d_bds, (uint32_t)b0, (uint32_t)b1, p->d_ucblob, p->d_cblob,
                      p->ccap_per_block, p->d_ebsize, p->d_ebtype, p->d_ecsum,
                      d_outoff, p->d_out_img);
-  out->resize(acc);
+  p->kend();
   hipEvent_t t0, t1;
   hipEventCreate(&t0);
   hipEventCreate(&t1);
   hipEventRecord(t0, p->stream);
-  HIPCHK(hipMemcpyAsync(out->data(), p->d_out_img, acc, hipMemcpyDeviceToHost,
-                        p->stream));
+  HIPCHK(hipMemcpyAsync(host_dst, p->d_out_img, total_bytes,
+                        hipMemcpyDeviceToHost, p->stream));
   hipEventRecord(t1, p->stream);
+  HIPCHK(hipFreeAsync(d_outoff, p->stream));
   HIPCHK(hipStreamSynchronize(p->stream));
   ms_d2h += ms_between(t0, t1);
+  p->kresolve();
   hipEventDestroy(t0);
   hipEventDestroy(t1);
-  (void)hipFree(d_outoff);
   return 0;
 }
 
@@ -1615,3 +1684,30 @@ int GpuJob::seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* m
 }
 
 } // namespace dcw
+
+// ------------------------------------------------------------------
+// kernel stats ABI (per-process cumulative; bench.py roofline input)
+// ------------------------------------------------------------------
+extern "C" int32_t dcw_kernel_stats_json(char* buf, uint32_t cap) {
+  std::lock_guard<std::mutex> lk(dcw::g_kmu);
+  std::string s = "{";
+  bool first = true;
+  for (auto& kv : dcw::kstats()) {
+    char line[256];
+    snprintf(line, sizeof(line),
+             "%s\"%s\": {\"launches\": %llu, \"ms\": %.6f, \"alg_bytes\": %.0f}",
+             first ? "" : ", ", kv.first.c_str(),
+             (unsigned long long)kv.second.launches, kv.second.ms,
+             kv.second.alg_bytes);
+    s += line;
+    first = false;
+  }
+  s += "}";
+  if (s.size() + 1 > cap) return -1;
+  memcpy(buf, s.c_str(), s.size() + 1);
+  return (int32_t)s.size();
+}
+extern "C" void dcw_kernel_stats_reset(void) {
+  std::lock_guard<std::mutex> lk(dcw::g_kmu);
+  dcw::kstats().clear();
+}

@@ -14,6 +14,7 @@
 
 #include <cstdio>
 #include <cstring>
+#include <future>
 #include <mutex>
 #include <string>
 #include <unordered_map>
@@ -245,6 +246,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   TableOpts base = opts_from_desc(d);
   uint64_t next_file_number = d->next_file_number;
   std::vector<dcw_output_file> out_files;
+  std::vector<std::future<int>> writers;
   uint64_t total_out_bytes = 0, total_out_entries = 0;
   uint64_t plan_usec = 0, write_usec = 0;
 
@@ -274,10 +276,17 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       if (blocks.empty()) break;
       std::vector<uint32_t> csizes;
       if (job.emit_blocks(blocks, o, &csizes, &err) != 0) return fail(res, 21, err);
-      // fetch everything emitted; excess past the cut is discarded
-      std::string chunk;
-      std::vector<SstIndexEntry> chandles;
-      if (job.pack_and_fetch(0, blocks.size(), o, &chunk, &chandles, &err) != 0)
+      // image layout for the whole chunk (excess past the cut is dropped)
+      std::vector<uint64_t> outoff(blocks.size());
+      uint64_t acc = 0;
+      for (size_t b = 0; b < blocks.size(); b++) {
+        outoff[b] = acc;
+        acc += csizes[b] + kTrailerSize;
+      }
+      size_t old = image.size();
+      image.resize(old + acc);
+      if (job.pack_into(0, blocks.size(), outoff, (uint8_t*)&image[old], acc,
+                        &err) != 0)
         return fail(res, 22, err);
       std::vector<std::string> cfirst, clast;
       if (job.fetch_block_keys(0, blocks.size(), &cfirst, &clast, &err) != 0)
@@ -285,10 +294,9 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       // walk blocks: find the flush that crosses the target
       // (cut = first entry after it; the open block then holds exactly one
       // entry — AddToOutput/FileSize interplay, compaction_outputs.cc:356-420)
-      uint64_t tw0 = now_usec();
       size_t take = blocks.size();
       for (size_t b = 0; b < blocks.size(); b++) {
-        uint64_t off_after = image.size() + chandles[b].off + chandles[b].size + kTrailerSize;
+        uint64_t off_after = old + outoff[b] + csizes[b] + kTrailerSize;
         bool more_entries = blocks[b].first + blocks[b].count < nsurv;
         if (off_after >= d->target_file_size && more_entries) {
           take = b + 1;
@@ -297,18 +305,16 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
           break;
         }
       }
-      uint64_t base_off = image.size();
       for (size_t b = 0; b < take; b++) {
-        handles.push_back({base_off + chandles[b].off, chandles[b].size});
+        handles.push_back({old + outoff[b], csizes[b]});
         first_keys.push_back(cfirst[b]);
         last_keys.push_back(clast[b]);
         block_counts.push_back(blocks[b].count);
       }
       uint64_t take_bytes =
-          take ? chandles[take - 1].off + chandles[take - 1].size + kTrailerSize : 0;
-      image.append(chunk.data(), take_bytes);
+          take ? outoff[take - 1] + csizes[take - 1] + kTrailerSize : 0;
+      image.resize(old + take_bytes);
       cur = take ? blocks[take - 1].first + blocks[take - 1].count : cur;
-      write_usec += now_usec() - tw0;
       if (take < blocks.size()) break; // cut decided inside this chunk
     }
     // partial single-entry block after a size cut
@@ -363,21 +369,29 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     st.num_deletions = n_tomb;
     uint64_t tail_start = image.size();
     image += build_tail(o, st, handles, seps, !sep_key_plus_seq, tail_start);
-    // write the file
+    // write the file on a background thread (overlaps the next file's GPU
+    // work; joined before returning)
     char path[600];
     snprintf(path, sizeof(path), "%s/%06" PRIu64 ".sst", d->output_dir,
              o.orig_file_number);
-    FILE* f = fopen(path, "wb");
-    if (!f) return fail(res, 26, std::string("cannot write ") + path);
-    fwrite(image.data(), 1, image.size(), f);
-    fclose(f);
+    uint64_t image_size = image.size();
+    writers.emplace_back(std::async(
+        std::launch::async,
+        [](std::string pth, std::string img) -> int {
+          FILE* f = fopen(pth.c_str(), "wb");
+          if (!f) return -1;
+          size_t w = fwrite(img.data(), 1, img.size(), f);
+          fclose(f);
+          return w == img.size() ? 0 : -1;
+        },
+        std::string(path), std::move(image)));
     write_usec += now_usec() - tw1;
 
     dcw_output_file of;
     memset(&of, 0, sizeof(of));
     snprintf(of.path, sizeof(of.path), "%s", path);
     of.file_number = o.orig_file_number;
-    of.file_size = image.size();
+    of.file_size = image_size;
     std::string smallest = first_keys.front();
     std::string largest = last_keys.back();
     of.smallest_len = (uint32_t)smallest.size();
@@ -388,9 +402,15 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     of.largest_seqno = mx_seq;
     of.num_entries = file_count;
     out_files.push_back(of);
-    total_out_bytes += image.size();
+    total_out_bytes += image_size;
     total_out_entries += file_count;
     s = cur;
+  }
+  {
+    uint64_t tj = now_usec();
+    for (auto& w : writers)
+      if (w.get() != 0) return fail(res, 26, "output file write failed");
+    write_usec += now_usec() - tj;
   }
 
   res->num_files = (uint32_t)out_files.size();
