@@ -1321,6 +1321,9 @@ static int scan_u8(GpuJob::Impl* p, const uint8_t* d_in, uint64_t n,
                         hipMemcpyHostToDevice, p->stream));
   hipLaunchKernelGGL(k_scan_add_base, dim3((uint32_t)nblk), dim3(1024), 0,
                      p->stream, d_out, n, d_bs);
+  // `bs` is a local: the async H2D must complete before it dies (ROCm may
+  // read pageable sources lazily at stream-execution time)
+  HIPCHK(hipStreamSynchronize(p->stream));
   *total = acc;
   return 0;
 }

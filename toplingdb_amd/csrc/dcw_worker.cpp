@@ -288,6 +288,25 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       if (job.pack_into(0, blocks.size(), outoff, (uint8_t*)&image[old], acc,
                         &err) != 0)
         return fail(res, 22, err);
+      if (getenv("DCW_PARANOID")) {
+        // validate every packed block's trailer checksum host-side
+        for (size_t b = 0; b < blocks.size(); b++) {
+          const uint8_t* body = (const uint8_t*)&image[old] + outoff[b];
+          uint32_t stored;
+          memcpy(&stored, body + csizes[b] + 1, 4);
+          uint8_t ty = body[csizes[b]];
+          uint32_t actual =
+              block_checksum(o.checksum_type, &g_crc, body, csizes[b], ty);
+          if (stored != actual) {
+            fprintf(stderr,
+                    "[paranoid] chunk@%zu block %zu/%zu first=%u count=%u "
+                    "csize=%u type=%u stored=%08x actual=%08x\n",
+                    s, b, blocks.size(), blocks[b].first, blocks[b].count,
+                    csizes[b], ty, stored, actual);
+            if (b > 2) break;
+          }
+        }
+      }
       std::vector<std::string> cfirst, clast;
       if (job.fetch_block_keys(0, blocks.size(), &cfirst, &clast, &err) != 0)
         return fail(res, 23, err);
