@@ -975,6 +975,8 @@ struct GpuJob::Impl {
   // misc scratch
   uint64_t* d_scratch64 = nullptr;
   size_t scratch64_cap = 0;
+  void* d_outoff = nullptr;
+  size_t outoff_cap = 0;
   uint8_t* d_out_img = nullptr;
   size_t out_img_cap = 0;
 
@@ -1019,7 +1021,7 @@ GpuJob::~GpuJob() {
   F(p->d_sk0); F(p->d_sk1); F(p->d_stag); F(p->d_svoff); F(p->d_svlen);
   F(p->d_sklen); F(p->d_sshared); F(p->d_bds); F(p->d_eoff); F(p->d_ucblob);
   F(p->d_cblob); F(p->d_ebsize); F(p->d_ebtype); F(p->d_ecsum);
-  F(p->d_scratch64); F(p->d_out_img);
+  F(p->d_scratch64); F(p->d_out_img); F(p->d_outoff);
   if (p->stream) (void)hipStreamDestroy(p->stream);
   delete p;
 }
@@ -1050,10 +1052,10 @@ int GpuJob::stage(const GpuInputs& in, std::string* err) {
   }
   HIPCHK(hipMalloc(&p->d_boff, sizeof(uint64_t) * p->n_blocks));
   HIPCHK(hipMalloc(&p->d_bsize, sizeof(uint32_t) * p->n_blocks));
-  HIPCHK(hipMemcpyAsync(p->d_boff, boff.data(), sizeof(uint64_t) * p->n_blocks,
-                        hipMemcpyHostToDevice, p->stream));
-  HIPCHK(hipMemcpyAsync(p->d_bsize, bsize.data(), sizeof(uint32_t) * p->n_blocks,
-                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMemcpy(p->d_boff, boff.data(), sizeof(uint64_t) * p->n_blocks,
+                   hipMemcpyHostToDevice));
+  HIPCHK(hipMemcpy(p->d_bsize, bsize.data(), sizeof(uint32_t) * p->n_blocks,
+                   hipMemcpyHostToDevice));
   HIPCHK(hipMalloc(&p->d_crc, sizeof(Crc32cTables)));
   HIPCHK(hipMemcpyAsync(p->d_crc, &g_crc, sizeof(Crc32cTables),
                         hipMemcpyHostToDevice, p->stream));
@@ -1145,8 +1147,8 @@ int GpuJob::decode(std::string* err) {
   }
   p->ublob_size = acc;
   HIPCHK(hipMalloc(&p->d_uoff, sizeof(uint64_t) * nb));
-  HIPCHK(hipMemcpyAsync(p->d_uoff, uoff.data(), sizeof(uint64_t) * nb,
-                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMemcpy(p->d_uoff, uoff.data(), sizeof(uint64_t) * nb,
+                   hipMemcpyHostToDevice));
   HIPCHK(hipMalloc(&p->d_ublob, acc ? acc : 1));
   {
     std::lock_guard<std::mutex> lk(g_kmu);
@@ -1187,10 +1189,10 @@ int GpuJob::decode(std::string* err) {
   HIPCHK(hipMalloc(&p->d_iv_local, sizeof(uint32_t) * niv));
   HIPCHK(hipMalloc(&p->d_iv_cnt, sizeof(uint32_t) * niv));
   HIPCHK(hipMalloc(&p->d_iv_base, sizeof(uint32_t) * niv));
-  HIPCHK(hipMemcpyAsync(p->d_iv_block, iv_block.data(), sizeof(uint32_t) * niv,
-                        hipMemcpyHostToDevice, p->stream));
-  HIPCHK(hipMemcpyAsync(p->d_iv_local, iv_local.data(), sizeof(uint32_t) * niv,
-                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMemcpy(p->d_iv_block, iv_block.data(), sizeof(uint32_t) * niv,
+                   hipMemcpyHostToDevice));
+  HIPCHK(hipMemcpy(p->d_iv_local, iv_local.data(), sizeof(uint32_t) * niv,
+                   hipMemcpyHostToDevice));
   p->kbegin("count_entries", (double)p->ublob_size);
   hipLaunchKernelGGL(k_count_entries, dim3(grid_for(niv)), dim3(256), 0, p->stream,
                      p->d_ublob, p->d_uoff, p->d_usize, p->d_nrestarts,
@@ -1222,8 +1224,8 @@ int GpuJob::decode(std::string* err) {
                                                 : total_entries);
   }
   p->run_entry_begin.push_back(total_entries);
-  HIPCHK(hipMemcpyAsync(p->d_iv_base, iv_base.data(), sizeof(uint32_t) * niv,
-                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMemcpy(p->d_iv_base, iv_base.data(), sizeof(uint32_t) * niv,
+                   hipMemcpyHostToDevice));
   HIPCHK(hipMalloc(&p->d_ent[0], sizeof(ulong4) * total_entries));
   HIPCHK(hipMalloc(&p->d_ent[1], sizeof(ulong4) * total_entries));
   HIPCHK(hipMalloc(&p->d_voff, sizeof(uint64_t) * total_entries));
@@ -1317,8 +1319,8 @@ static int scan_u8(GpuJob::Impl* p, const uint8_t* d_in, uint64_t n,
     bs[i] = (uint32_t)acc;
     acc += v;
   }
-  HIPCHK(hipMemcpyAsync(d_bs, bs.data(), sizeof(uint32_t) * nblk,
-                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMemcpy(d_bs, bs.data(), sizeof(uint32_t) * nblk,
+                   hipMemcpyHostToDevice));
   hipLaunchKernelGGL(k_scan_add_base, dim3((uint32_t)nblk), dim3(1024), 0,
                      p->stream, d_out, n, d_bs);
   // `bs` is a local: the async H2D must complete before it dies (ROCm may
@@ -1362,8 +1364,8 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   uint64_t* d_snaps = nullptr;
   if (d->num_snapshots) {
     HIPCHK(hipMalloc(&d_snaps, sizeof(uint64_t) * d->num_snapshots));
-    HIPCHK(hipMemcpyAsync(d_snaps, d->snapshots, sizeof(uint64_t) * d->num_snapshots,
-                          hipMemcpyHostToDevice, p->stream));
+    HIPCHK(hipMemcpy(d_snaps, d->snapshots, sizeof(uint64_t) * d->num_snapshots,
+                     hipMemcpyHostToDevice));
   }
   P.snapshots = d_snaps;
   // levels below -> normkeys
@@ -1391,14 +1393,14 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
     HIPCHK(hipMalloc(&p->d_lb_sm1, nb));
     HIPCHK(hipMalloc(&p->d_lb_lg0, nb));
     HIPCHK(hipMalloc(&p->d_lb_lg1, nb));
-    HIPCHK(hipMemcpyAsync(p->d_lb_sm0, sm0.data(), nb, hipMemcpyHostToDevice, p->stream));
-    HIPCHK(hipMemcpyAsync(p->d_lb_sm1, sm1.data(), nb, hipMemcpyHostToDevice, p->stream));
-    HIPCHK(hipMemcpyAsync(p->d_lb_lg0, lg0.data(), nb, hipMemcpyHostToDevice, p->stream));
-    HIPCHK(hipMemcpyAsync(p->d_lb_lg1, lg1.data(), nb, hipMemcpyHostToDevice, p->stream));
+    HIPCHK(hipMemcpy(p->d_lb_sm0, sm0.data(), nb, hipMemcpyHostToDevice));
+    HIPCHK(hipMemcpy(p->d_lb_sm1, sm1.data(), nb, hipMemcpyHostToDevice));
+    HIPCHK(hipMemcpy(p->d_lb_lg0, lg0.data(), nb, hipMemcpyHostToDevice));
+    HIPCHK(hipMemcpy(p->d_lb_lg1, lg1.data(), nb, hipMemcpyHostToDevice));
   }
   HIPCHK(hipMalloc(&p->d_lb_beg, sizeof(uint32_t) * lbeg.size()));
-  HIPCHK(hipMemcpyAsync(p->d_lb_beg, lbeg.data(), sizeof(uint32_t) * lbeg.size(),
-                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMemcpy(p->d_lb_beg, lbeg.data(), sizeof(uint32_t) * lbeg.size(),
+                   hipMemcpyHostToDevice));
   P.lb_sm_k0 = p->d_lb_sm0;
   P.lb_sm_k1 = p->d_lb_sm1;
   P.lb_lg_k0 = p->d_lb_lg0;
@@ -1528,10 +1530,10 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   ENSURE(p->d_ebsize, p->eb_cap, sizeof(uint32_t) * nb * 2 + nb); // bsize+csum+btype
   p->d_ecsum = p->d_ebsize + nb;
   p->d_ebtype = (uint8_t*)(p->d_ecsum + nb);
-  HIPCHK(hipMemcpyAsync(p->d_bds, bds.data(), sizeof(EmitBlockDesc) * nb,
-                        hipMemcpyHostToDevice, p->stream));
-  HIPCHK(hipMemcpyAsync(p->d_eoff, eoff.data(), sizeof(uint32_t) * nent,
-                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMemcpy(p->d_bds, bds.data(), sizeof(EmitBlockDesc) * nb,
+                   hipMemcpyHostToDevice));
+  HIPCHK(hipMemcpy(p->d_eoff, eoff.data(), sizeof(uint32_t) * nent,
+                   hipMemcpyHostToDevice));
   // NOTE: k_emit indexes eoff by absolute survivor index minus chunk base.
   p->kbegin("emit", 2.0 * (double)uout);
   hipLaunchKernelGGL(k_emit, dim3(nb < 4096 ? nb : 4096), dim3(256), 0, p->stream,
@@ -1574,10 +1576,12 @@ int GpuJob::pack_into(size_t b0, size_t b1, const std::vector<uint64_t>& outoff,
   Impl* p = p_;
   uint32_t nb = (uint32_t)(b1 - b0);
   if (nb == 0) return 0;
-  uint64_t* d_outoff;
-  HIPCHK(hipMallocAsync(&d_outoff, sizeof(uint64_t) * nb, p->stream));
-  HIPCHK(hipMemcpyAsync(d_outoff, outoff.data(), sizeof(uint64_t) * nb,
-                        hipMemcpyHostToDevice, p->stream));
+  ENSURE(p->d_outoff, p->outoff_cap, sizeof(uint64_t) * nb);
+  uint64_t* d_outoff = (uint64_t*)p->d_outoff;
+  // synchronous copy: NULL-stream ordering with the (blocking) pipeline
+  // stream, no pageable-async lifetime/ordering hazards
+  HIPCHK(hipMemcpy(d_outoff, outoff.data(), sizeof(uint64_t) * nb,
+                   hipMemcpyHostToDevice));
   ENSURE(p->d_out_img, p->out_img_cap, total_bytes);
   p->kbegin("pack", 2.0 * (double)total_bytes);
   hipLaunchKernelGGL(k_pack, dim3(grid_for(nb * 4ull)), dim3(256), 0, p->stream,
@@ -1592,7 +1596,6 @@ int GpuJob::pack_into(size_t b0, size_t b1, const std::vector<uint64_t>& outoff,
   HIPCHK(hipMemcpyAsync(host_dst, p->d_out_img, total_bytes,
                         hipMemcpyDeviceToHost, p->stream));
   hipEventRecord(t1, p->stream);
-  HIPCHK(hipFreeAsync(d_outoff, p->stream));
   HIPCHK(hipStreamSynchronize(p->stream));
   ms_d2h += ms_between(t0, t1);
   p->kresolve();
@@ -1645,8 +1648,8 @@ int GpuJob::gather_entries(uint64_t first, uint32_t count,
   uint8_t* d_out;
   HIPCHK(hipMalloc(&d_recoff, sizeof(uint64_t) * count));
   HIPCHK(hipMalloc(&d_out, acc));
-  HIPCHK(hipMemcpyAsync(d_recoff, recoff.data(), sizeof(uint64_t) * count,
-                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMemcpy(d_recoff, recoff.data(), sizeof(uint64_t) * count,
+                   hipMemcpyHostToDevice));
   hipLaunchKernelGGL(k_gather_range, dim3(grid_for(count)), dim3(256), 0,
                      p->stream, p->d_sk0, p->d_sk1, p->d_stag, p->d_svoff,
                      p->d_svlen, p->d_sklen, p->d_ublob, first, count, d_recoff,
@@ -1673,7 +1676,7 @@ int GpuJob::seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* m
   unsigned long long* d;
   HIPCHK(hipMalloc(&d, 24));
   unsigned long long init[3] = {~0ull, 0, 0};
-  HIPCHK(hipMemcpyAsync(d, init, 24, hipMemcpyHostToDevice, p->stream));
+  HIPCHK(hipMemcpy(d, init, 24, hipMemcpyHostToDevice));
   hipLaunchKernelGGL(k_seq_minmax, dim3(grid_for(count)), dim3(256), 0, p->stream,
                      p->d_stag, first, count, d, d + 1, d + 2);
   unsigned long long out[3];
