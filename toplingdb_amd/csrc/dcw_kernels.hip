@@ -137,35 +137,63 @@ __global__ void k_verify_usize(const uint8_t* __restrict__ blob,
   }
 }
 
-// one wave per block: raw -> cooperative copy; snappy -> lane 0 decodes
-__global__ void k_decompress(const uint8_t* __restrict__ blob,
-                             const uint64_t* __restrict__ boff,
-                             const uint32_t* __restrict__ bsize,
-                             const uint8_t* __restrict__ btype,
-                             const uint64_t* __restrict__ uoff,
-                             const uint32_t* __restrict__ usize, uint32_t nblocks,
-                             uint8_t* __restrict__ ublob, uint32_t* err_flag) {
+// one wave per block: raw -> cooperative copy; snappy -> staged through LDS
+// (compressed in + decoded out both in LDS, lane 0 runs the serial decoder,
+// all lanes copy in/out); oversized blocks fall back to the direct path.
+#define DEC_MAX 5376
+struct DecLds {
+  uint8_t in[DEC_MAX];
+  uint8_t out[DEC_MAX];
+};
+__device__ __forceinline__ void wave_lds_sync2() {
+  __builtin_amdgcn_s_waitcnt(0);
+  __builtin_amdgcn_wave_barrier();
+}
+__global__ __launch_bounds__(256) void k_decompress(
+    const uint8_t* __restrict__ blob, const uint64_t* __restrict__ boff,
+    const uint32_t* __restrict__ bsize, const uint8_t* __restrict__ btype,
+    const uint64_t* __restrict__ uoff, const uint32_t* __restrict__ usize,
+    uint32_t nblocks, uint8_t* __restrict__ ublob, uint32_t* err_flag) {
+  __shared__ DecLds lds[4];
   uint32_t waves_per_wg = blockDim.x / WAVE;
-  uint32_t wave = blockIdx.x * waves_per_wg + threadIdx.x / WAVE;
+  uint32_t wid = threadIdx.x / WAVE;
+  uint32_t wave = blockIdx.x * waves_per_wg + wid;
   uint32_t lane = threadIdx.x % WAVE;
   uint32_t stride = gridDim.x * waves_per_wg;
+  DecLds& L = lds[wid];
   for (uint32_t i = wave; i < nblocks; i += stride) {
     const uint8_t* src = blob + boff[i];
     uint8_t* dst = ublob + uoff[i];
+    uint32_t n = bsize[i];
     if (btype[i] == 0) {
-      uint32_t n = bsize[i];
       uint32_t pos = lane * 16;
       for (; pos + 16 <= n; pos += WAVE * 16) {
         ulong2 v;
         memcpy(&v, src + pos, 16);
         memcpy(dst + pos, &v, 16);
       }
-      // tail bytes by lane 0
       if (lane == 0)
         for (uint32_t t = n & ~15u; t < n; t++) dst[t] = src[t];
+    } else if (n <= DEC_MAX && usize[i] <= DEC_MAX) {
+      for (uint32_t t = lane * 4; t < n; t += WAVE * 4) {
+        uint32_t chunk = n - t < 4 ? n - t : 4;
+        for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = src[t + x];
+      }
+      wave_lds_sync2();
+      if (lane == 0) {
+        if (snappy_uncompress(L.in, n, L.out, usize[i]) != usize[i])
+          set_err(err_flag, DE_SNAPPY);
+      }
+      wave_lds_sync2();
+      uint32_t un = usize[i];
+      for (uint32_t t = lane * 4; t < un; t += WAVE * 4) {
+        uint32_t chunk = un - t < 4 ? un - t : 4;
+        for (uint32_t x = 0; x < chunk; x++) dst[t + x] = L.out[t + x];
+      }
+      wave_lds_sync2();
     } else {
       if (lane == 0) {
-        if (snappy_uncompress(src, bsize[i], dst, usize[i]) != usize[i])
+        if (snappy_uncompress(src, n, dst, usize[i]) != usize[i])
           set_err(err_flag, DE_SNAPPY);
       }
     }
@@ -253,6 +281,25 @@ __global__ void k_count_entries(const uint8_t* __restrict__ ublob,
   }
 }
 
+// Register-resident key state: the internal key (user key <= 16 B + 8 B tag)
+// lives in three u64s — kb0/kb1 = key bytes 0..7 / 8..15 in MEMORY order
+// (little-endian u64 of the byte string), tail = bytes 16..klen.  Dynamic
+// byte indexing through shifts only: no per-thread scratch array.
+__device__ __forceinline__ void key_set_byte(uint64_t& kb0, uint64_t& kb1,
+                                             uint64_t& tail, uint32_t j,
+                                             uint8_t v) {
+  uint64_t m = 0xffull;
+  uint64_t x = (uint64_t)v;
+  if (j < 8) {
+    kb0 = (kb0 & ~(m << (8 * j))) | (x << (8 * j));
+  } else if (j < 16) {
+    kb0 = kb0; // no-op to keep branches balanced
+    kb1 = (kb1 & ~(m << (8 * (j - 8)))) | (x << (8 * (j - 8)));
+  } else {
+    tail = (tail & ~(m << (8 * (j - 16)))) | (x << (8 * (j - 16)));
+  }
+}
+
 __global__ void k_decode_entries(
     const uint8_t* __restrict__ ublob, const uint64_t* __restrict__ uoff,
     const uint32_t* __restrict__ usize, const uint32_t* __restrict__ nrestarts,
@@ -270,9 +317,10 @@ __global__ void k_decode_entries(
       return;
     const uint8_t* p = ublk + beg;
     const uint8_t* lim = ublk + end;
-    uint8_t key[32];
+    uint64_t kb0 = 0, kb1 = 0, ktail = 0; // key bytes in registers
     uint32_t klen = 0;
     uint32_t out = iv_base[i];
+    uint32_t probed_ulen = 0xffffffffu;
     while (p < lim) {
       uint32_t shared, non_shared, vl;
       int a = varint32_get(p, lim, &shared);
@@ -284,11 +332,13 @@ __global__ void k_decode_entries(
       a = varint32_get(p, lim, &vl);
       if (a < 0) break;
       p += a;
-      if (shared > klen || shared + non_shared > 32) {
+      if (shared > klen || shared + non_shared > 24 ||
+          p + non_shared + vl > lim) {
         set_err(err_flag, DE_BLOCK_PARSE);
         return;
       }
-      for (uint32_t t = 0; t < non_shared; t++) key[shared + t] = p[t];
+      for (uint32_t t = 0; t < non_shared; t++)
+        key_set_byte(kb0, kb1, ktail, shared + t, p[t]);
       klen = shared + non_shared;
       p += non_shared;
       if (klen < 9) {
@@ -300,21 +350,42 @@ __global__ void k_decode_entries(
         set_err(err_flag, DE_UKEY_LEN);
         return;
       }
-      uint32_t expect = atomicCAS(ukey_len_probe, 0xffffffffu, ulen);
-      if (expect != 0xffffffffu && expect != ulen) {
-        set_err(err_flag, DE_UKEY_LEN);
-        return;
+      if (probed_ulen != ulen) {
+        uint32_t expect = atomicCAS(ukey_len_probe, 0xffffffffu, ulen);
+        if (expect != 0xffffffffu && expect != ulen) {
+          set_err(err_flag, DE_UKEY_LEN);
+          return;
+        }
+        probed_ulen = ulen;
       }
+      // tag = 8 LE bytes starting at byte ulen of the key
       uint64_t tag;
-      memcpy(&tag, key + ulen, 8);
+      if (ulen == 16) {
+        tag = ktail;
+      } else if (ulen >= 8) {
+        uint32_t sh = 8 * (ulen - 8);
+        tag = sh ? ((kb1 >> sh) | (ktail << (64 - sh))) : kb1;
+      } else {
+        uint32_t sh = 8 * ulen;
+        tag = sh ? ((kb0 >> sh) | (kb1 << (64 - sh))) : kb0;
+      }
       uint8_t vt = (uint8_t)tag;
       if (!(vt == kTypeValue || vt == kTypeDeletion || vt == kTypeSingleDeletion)) {
         set_err(err_flag, DE_TYPE);
         return;
       }
-      uint64_t k0, k1, k2;
-      make_normkey(key, ulen, tag, &k0, &k1, &k2);
-      ents[out] = make_ulong4(k0, k1, k2, out);
+      // normkey: big-endian words of the zero-padded user key
+      uint64_t u0, u1;
+      if (ulen >= 8) {
+        u0 = kb0;
+        uint32_t rem = ulen - 8; // bytes of user key in kb1
+        u1 = rem ? (kb1 & ((rem == 8) ? ~0ull : ((1ull << (8 * rem)) - 1))) : 0;
+      } else {
+        u0 = ulen ? (kb0 & ((1ull << (8 * ulen)) - 1)) : 0;
+        u1 = 0;
+      }
+      ents[out] = make_ulong4(__builtin_bswap64(u0), __builtin_bswap64(u1),
+                              ~tag, out);
       voff[out] = uoff[b] + (uint64_t)(p - ublk);
       vlen_out[out] = vl;
       klen_out[out] = (uint8_t)klen;
@@ -708,33 +779,105 @@ __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
   }
 }
 
-// one workgroup (64 threads) per block; snappy hash table in LDS; lane 0
-// encodes; all lanes clear the table.
-__global__ void k_compress(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
-                           const uint8_t* __restrict__ ucblob,
-                           uint8_t* __restrict__ cblob, uint64_t ccap_per_block,
-                           uint32_t* __restrict__ bsize,
-                           uint8_t* __restrict__ btype) {
-  __shared__ uint32_t tab[1u << kSnapHashBits];
-  for (uint32_t b = blockIdx.x; b < nblocks; b += gridDim.x) {
-    for (uint32_t t = threadIdx.x; t < (1u << kSnapHashBits); t += blockDim.x)
-      tab[t] = 0xffffffffu;
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      EmitBlockDesc d = bds[b];
-      const uint8_t* in = ucblob + d.uout;
-      uint8_t* out = cblob + (uint64_t)b * ccap_per_block;
-      size_t cn = snappy_compress_block(in, d.unc_size, out, tab);
-      // GoodCompressionRatio, default max_compressed_bytes_per_kb=896
-      if (cn <= (((uint64_t)896 * d.unc_size) >> 10)) {
-        bsize[b] = (uint32_t)cn;
-        btype[b] = 1;
-      } else {
+// snappy encode, one wave (64 threads) per 4 KiB-class block:
+// the block is staged into LDS by all lanes, the u16 hash table and the
+// output live in LDS too, lane 0 runs the serial matcher over LDS (the
+// greedy match chain is inherently sequential — the DETERMINISM of the
+// self-pinned codec spec is the point), all lanes copy the result out.
+// LDS per wave: in 5 KiB + tab 8 KiB + out 6.2 KiB ≈ 19.2 KiB -> with
+// 4 waves/WG ≈ 77 KiB/WG, 2 WGs/CU = 8 concurrent encoders per CU.
+#define SNAP_MAX_UNC 5120 // block_size + slack; host guards this bound
+#define SNAP_MAX_OUT (32 + SNAP_MAX_UNC + SNAP_MAX_UNC / 6)
+struct SnapLds {
+  uint8_t in[SNAP_MAX_UNC];
+  uint8_t out[SNAP_MAX_OUT + 8];
+  uint16_t tab[1u << kSnapHashBits];
+  uint32_t cn;
+};
+
+// wave-internal LDS ordering: drain DS ops + stop compiler reordering
+__device__ __forceinline__ void wave_lds_sync() {
+  __builtin_amdgcn_s_waitcnt(0); // lgkmcnt(0) & vmcnt(0)
+  __builtin_amdgcn_wave_barrier();
+}
+
+// u16-table variant of dcw::snappy_compress_block — byte-identical output
+// (positions < 65536 in a block; 0xffff = empty, position 0xffff cannot
+// occur since SNAP_MAX_UNC < 0xffff).
+__device__ static size_t snap_encode_lds(const uint8_t* in, size_t n,
+                                         uint8_t* out, uint16_t* tab) {
+  uint8_t* op = out;
+  op += varint32_put(op, (uint32_t)n);
+  if (n == 0) return (size_t)(op - out);
+  size_t lit_start = 0, pz = 0;
+  while (pz + 4 <= n) {
+    uint32_t w = load32(in + pz);
+    uint32_t h = (w * kSnapHashMul) >> (32 - kSnapHashBits);
+    uint32_t cand = tab[h];
+    tab[h] = (uint16_t)pz;
+    if (cand != 0xffffu && cand < pz && load32(in + cand) == w) {
+      size_t mlen = 4;
+      while (pz + mlen < n && in[cand + mlen] == in[pz + mlen]) mlen++;
+      op = snap_emit_literal(op, in + lit_start, pz - lit_start);
+      op = snap_emit_copy(op, pz - cand, mlen);
+      pz += mlen;
+      lit_start = pz;
+    } else {
+      pz++;
+    }
+  }
+  op = snap_emit_literal(op, in + lit_start, n - lit_start);
+  return (size_t)(op - out);
+}
+
+__global__ __launch_bounds__(256) void k_compress(
+    const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
+    const uint8_t* __restrict__ ucblob, uint8_t* __restrict__ cblob,
+    uint64_t ccap_per_block, uint32_t* __restrict__ bsize,
+    uint8_t* __restrict__ btype, uint32_t* err_flag) {
+  __shared__ SnapLds lds[4];
+  uint32_t wid = threadIdx.x / WAVE;  // wave within workgroup
+  uint32_t lane = threadIdx.x % WAVE;
+  uint32_t waves = blockDim.x / WAVE;
+  SnapLds& L = lds[wid];
+  for (uint32_t b = blockIdx.x * waves + wid; b < nblocks;
+       b += gridDim.x * waves) {
+    EmitBlockDesc d = bds[b];
+    if (d.unc_size > SNAP_MAX_UNC) {
+      if (lane == 0) {
+        set_err(err_flag, DE_BLOCK_PARSE);
         bsize[b] = d.unc_size;
         btype[b] = 0;
       }
+      continue;
     }
-    __syncthreads();
+    const uint8_t* gin = ucblob + d.uout;
+    for (uint32_t t = lane * 4; t < d.unc_size; t += WAVE * 4) {
+      uint32_t chunk = d.unc_size - t < 4 ? d.unc_size - t : 4;
+      for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = gin[t + x];
+    }
+    for (uint32_t t = lane; t < (1u << kSnapHashBits); t += WAVE)
+      L.tab[t] = 0xffffu;
+    wave_lds_sync();
+    if (lane == 0) L.cn = (uint32_t)snap_encode_lds(L.in, d.unc_size, L.out, L.tab);
+    wave_lds_sync();
+    uint32_t cn = L.cn;
+    // GoodCompressionRatio, default max_compressed_bytes_per_kb=896
+    if (cn <= (((uint64_t)896 * d.unc_size) >> 10)) {
+      uint8_t* gout = cblob + (uint64_t)b * ccap_per_block;
+      for (uint32_t t = lane * 4; t < cn; t += WAVE * 4) {
+        uint32_t chunk = (uint32_t)cn - t < 4 ? (uint32_t)cn - t : 4;
+        for (uint32_t x = 0; x < chunk; x++) gout[t + x] = L.out[t + x];
+      }
+      if (lane == 0) {
+        bsize[b] = (uint32_t)cn;
+        btype[b] = 1;
+      }
+    } else if (lane == 0) {
+      bsize[b] = d.unc_size;
+      btype[b] = 0;
+    }
+    wave_lds_sync();
   }
 }
 
@@ -1542,12 +1685,20 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
                      p->d_ucblob, o.block_restart_interval);
   p->kend();
   if (o.compression == 1) {
+    for (auto& pb : blocks)
+      if (pb.unc_size > SNAP_MAX_UNC) {
+        if (err)
+          *err = "data block exceeds the GPU snappy staging bound (" +
+                 std::to_string(pb.unc_size) + " B); job outside envelope";
+        return -1;
+      }
     p->ccap_per_block = snappy_max_compressed(o.block_size + 1024);
     ENSURE(p->d_cblob, p->cblob_cap, p->ccap_per_block * nb);
     p->kbegin("compress", 1.6 * (double)uout);
-    hipLaunchKernelGGL(k_compress, dim3(nb < 4096 ? nb : 4096), dim3(64), 0,
-                       p->stream, p->d_bds, nb, p->d_ucblob, p->d_cblob,
-                       p->ccap_per_block, p->d_ebsize, p->d_ebtype);
+    uint32_t cgrid = (nb + 3) / 4;
+    hipLaunchKernelGGL(k_compress, dim3(cgrid < 4096 ? cgrid : 4096), dim3(256),
+                       0, p->stream, p->d_bds, nb, p->d_ucblob, p->d_cblob,
+                       p->ccap_per_block, p->d_ebsize, p->d_ebtype, p->d_err);
     p->kend();
   } else {
     hipLaunchKernelGGL(k_sizes_nocomp, dim3(grid_for(nb)), dim3(256), 0,

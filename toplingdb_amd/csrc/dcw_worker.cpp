@@ -119,6 +119,30 @@ TableOpts opts_from_desc(const dcw_job_desc* d) {
   return o;
 }
 
+// grow-without-zero-fill byte buffer for output file images (a 64 MiB
+// std::string::resize zero-fills; the D2H overwrites it anyway)
+struct RawBuf {
+  uint8_t* p = nullptr;
+  size_t len = 0, cap = 0;
+  void reserve(size_t n) {
+    if (cap >= n) return;
+    size_t c = n + n / 4;
+    p = (uint8_t*)realloc(p, c);
+    cap = c;
+  }
+  void resize_uninit(size_t n) {
+    reserve(n);
+    len = n;
+  }
+  void append(const void* src, size_t n) {
+    reserve(len + n);
+    memcpy(p + len, src, n);
+    len += n;
+  }
+  void release() { p = nullptr; len = cap = 0; }
+  ~RawBuf() { free(p); }
+};
+
 int fail(dcw_job_result* res, int code, const std::string& msg) {
   res->status = code;
   snprintf(res->error, sizeof(res->error), "%s", msg.c_str());
@@ -251,47 +275,68 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   uint64_t plan_usec = 0, write_usec = 0;
 
   size_t s = 0;
+  double comp_ratio = 0.62; // adaptive: refined from the first chunk
   while (s < nsurv) {
     TableOpts o = base;
     o.orig_file_number = next_file_number++;
-    std::string image; // data-block region accumulated on host
+    RawBuf image; // data-block region accumulated on host
     std::vector<SstIndexEntry> handles;
     std::vector<std::string> first_keys, last_keys;
     std::vector<uint32_t> block_counts;
     size_t cur = s;
     bool cut = false;
     uint64_t cut_entry = 0;
-    // chunked: plan -> GPU emit -> fetch, until the file-size cut fires
+    // chunked: plan -> GPU emit -> cut walk on csizes -> fetch only what the
+    // file keeps
     while (!cut && cur < nsurv) {
       uint64_t tp0 = now_usec();
-      uint64_t already = image.size();
+      uint64_t already = image.len;
       uint64_t want = d->target_file_size > already
                           ? d->target_file_size - already
                           : (64 << 10);
-      // cushion: compressed bytes per uncompressed byte unknown; assume
-      // >= 0.45 for snappy inputs, 1.0 otherwise, plus a block of slack
-      uint64_t min_unc = o.compression == 1 ? want * 23 / 10 : want + 2 * o.block_size;
+      uint64_t min_unc = o.compression == 1
+                             ? (uint64_t)(want / comp_ratio) + 8 * o.block_size
+                             : want + 2 * o.block_size;
       std::vector<PlannedBlock> blocks = plan_blocks(pin, cur, o, min_unc);
       plan_usec += now_usec() - tp0;
       if (blocks.empty()) break;
       std::vector<uint32_t> csizes;
       if (job.emit_blocks(blocks, o, &csizes, &err) != 0) return fail(res, 21, err);
-      // image layout for the whole chunk (excess past the cut is dropped)
-      std::vector<uint64_t> outoff(blocks.size());
+      // walk csizes first: find the flush that crosses the target
+      // (cut = first entry after it; the open block then holds exactly one
+      // entry — AddToOutput/FileSize interplay, compaction_outputs.cc:356-420)
+      uint64_t old = image.len;
+      size_t take = blocks.size();
+      {
+        uint64_t off = old;
+        uint64_t unc_sum = 0, c_sum = 0;
+        for (size_t b = 0; b < blocks.size(); b++) {
+          off += csizes[b] + kTrailerSize;
+          unc_sum += blocks[b].unc_size;
+          c_sum += csizes[b];
+          bool more_entries = blocks[b].first + blocks[b].count < nsurv;
+          if (off >= d->target_file_size && more_entries) {
+            take = b + 1;
+            cut = true;
+            cut_entry = blocks[b].first + blocks[b].count;
+            break;
+          }
+        }
+        if (o.compression == 1 && unc_sum > 0)
+          comp_ratio = 0.5 * comp_ratio + 0.5 * ((double)c_sum / unc_sum);
+      }
+      std::vector<uint64_t> outoff(take);
       uint64_t acc = 0;
-      for (size_t b = 0; b < blocks.size(); b++) {
+      for (size_t b = 0; b < take; b++) {
         outoff[b] = acc;
         acc += csizes[b] + kTrailerSize;
       }
-      size_t old = image.size();
-      image.resize(old + acc);
-      if (job.pack_into(0, blocks.size(), outoff, (uint8_t*)&image[old], acc,
-                        &err) != 0)
+      image.resize_uninit(old + acc);
+      if (job.pack_into(0, take, outoff, image.p + old, acc, &err) != 0)
         return fail(res, 22, err);
       if (getenv("DCW_PARANOID")) {
-        // validate every packed block's trailer checksum host-side
-        for (size_t b = 0; b < blocks.size(); b++) {
-          const uint8_t* body = (const uint8_t*)&image[old] + outoff[b];
+        for (size_t b = 0; b < take; b++) {
+          const uint8_t* body = image.p + old + outoff[b];
           uint32_t stored;
           memcpy(&stored, body + csizes[b] + 1, 4);
           uint8_t ty = body[csizes[b]];
@@ -301,38 +346,21 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
             fprintf(stderr,
                     "[paranoid] chunk@%zu block %zu/%zu first=%u count=%u "
                     "csize=%u type=%u stored=%08x actual=%08x\n",
-                    s, b, blocks.size(), blocks[b].first, blocks[b].count,
-                    csizes[b], ty, stored, actual);
+                    s, b, take, blocks[b].first, blocks[b].count, csizes[b],
+                    ty, stored, actual);
             if (b > 2) break;
           }
         }
       }
       std::vector<std::string> cfirst, clast;
-      if (job.fetch_block_keys(0, blocks.size(), &cfirst, &clast, &err) != 0)
+      if (job.fetch_block_keys(0, take, &cfirst, &clast, &err) != 0)
         return fail(res, 23, err);
-      // walk blocks: find the flush that crosses the target
-      // (cut = first entry after it; the open block then holds exactly one
-      // entry — AddToOutput/FileSize interplay, compaction_outputs.cc:356-420)
-      size_t take = blocks.size();
-      for (size_t b = 0; b < blocks.size(); b++) {
-        uint64_t off_after = old + outoff[b] + csizes[b] + kTrailerSize;
-        bool more_entries = blocks[b].first + blocks[b].count < nsurv;
-        if (off_after >= d->target_file_size && more_entries) {
-          take = b + 1;
-          cut = true;
-          cut_entry = blocks[b].first + blocks[b].count; // e0 = first of next block
-          break;
-        }
-      }
       for (size_t b = 0; b < take; b++) {
         handles.push_back({old + outoff[b], csizes[b]});
         first_keys.push_back(cfirst[b]);
         last_keys.push_back(clast[b]);
         block_counts.push_back(blocks[b].count);
       }
-      uint64_t take_bytes =
-          take ? outoff[take - 1] + csizes[take - 1] + kTrailerSize : 0;
-      image.resize(old + take_bytes);
       cur = take ? blocks[take - 1].first + blocks[take - 1].count : cur;
       if (take < blocks.size()) break; // cut decided inside this chunk
     }
@@ -347,8 +375,11 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
                         (const uint8_t*)kvs[0].second.data(), kvs[0].second.size(),
                         nullptr, 0);
       std::string contents = bb.Finish();
-      SstIndexEntry h = append_block(image, o, (const uint8_t*)contents.data(),
+      std::string pblk;
+      SstIndexEntry h = append_block(pblk, o, (const uint8_t*)contents.data(),
                                      contents.size(), true);
+      h.off += image.len;
+      image.append(pblk.data(), pblk.size());
       handles.push_back(h);
       first_keys.push_back(kvs[0].first);
       last_keys.push_back(kvs[0].first);
@@ -376,7 +407,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     // stats
     TailStats st;
     st.num_data_blocks = handles.size();
-    st.data_size = image.size();
+    st.data_size = image.len;
     st.num_entries = file_count;
     for (uint64_t i = file_first; i < file_first + file_count; i++) {
       st.raw_key_size += klen[i];
@@ -386,24 +417,31 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     if (job.seq_minmax(file_first, file_count, &mn_seq, &mx_seq, &n_tomb, &err) != 0)
       return fail(res, 25, err);
     st.num_deletions = n_tomb;
-    uint64_t tail_start = image.size();
-    image += build_tail(o, st, handles, seps, !sep_key_plus_seq, tail_start);
+    uint64_t tail_start = image.len;
+    std::string tail = build_tail(o, st, handles, seps, !sep_key_plus_seq, tail_start);
+    image.append(tail.data(), tail.size());
     // write the file on a background thread (overlaps the next file's GPU
     // work; joined before returning)
     char path[600];
     snprintf(path, sizeof(path), "%s/%06" PRIu64 ".sst", d->output_dir,
              o.orig_file_number);
-    uint64_t image_size = image.size();
+    uint64_t image_size = image.len;
+    uint8_t* img_ptr = image.p;
+    image.release(); // ownership moves to the writer thread
     writers.emplace_back(std::async(
         std::launch::async,
-        [](std::string pth, std::string img) -> int {
+        [](std::string pth, uint8_t* ptr, size_t n) -> int {
           FILE* f = fopen(pth.c_str(), "wb");
-          if (!f) return -1;
-          size_t w = fwrite(img.data(), 1, img.size(), f);
+          if (!f) {
+            free(ptr);
+            return -1;
+          }
+          size_t w = fwrite(ptr, 1, n, f);
           fclose(f);
-          return w == img.size() ? 0 : -1;
+          free(ptr);
+          return w == n ? 0 : -1;
         },
-        std::string(path), std::move(image)));
+        std::string(path), img_ptr, (size_t)image_size));
     write_usec += now_usec() - tw1;
 
     dcw_output_file of;
