@@ -288,7 +288,7 @@ uint32_t orc_block_checksum(uint32_t type, const void* data, size_t n, uint8_t l
 /* ---------------- snappy-format codec ----------------
  * Decoder: full public snappy format (any compliant producer).
  * Encoder: DCW-DETERMINISTIC spec: greedy matcher,
- *   hash table 1<<12 over 4-byte windows, hash = (load32(p)*0x1e35a7bd)>>20,
+ *   hash table 1<<12 over 4-byte windows, hash = (load32(p)*0x1e35a7bd)>>20,\n *   skip-ahead: on a miss advance by (skip>>5) with skip starting at 32\n *   and incrementing per miss, reset to 32 on every match (spec v2),
  *   match if prev pos with equal 4 bytes and offset < 65536; extend forward;
  *   emit copies of <=64 bytes, first copy uses the 1-byte-offset form when
  *   len in [4,11] and offset < 2048; literals flushed before each copy.
@@ -337,6 +337,7 @@ size_t orc_snappy_compress(const uint8_t* in, size_t n, uint8_t* out) {
   uint32_t* tab = (uint32_t*)malloc(sizeof(uint32_t) << HBITS);
   memset(tab, 0xff, sizeof(uint32_t) << HBITS);
   size_t lit_start = 0, p = 0;
+  uint32_t skip = 32; /* spec v2: snappy-style skip acceleration */
   while (p + 4 <= n) {
     uint32_t w = rd32(in + p);
     uint32_t h = (w * HMUL) >> (32 - HBITS);
@@ -349,8 +350,10 @@ size_t orc_snappy_compress(const uint8_t* in, size_t n, uint8_t* out) {
       op = sn_emit_copy(op, p - cand, mlen);
       p += mlen;
       lit_start = p;
+      skip = 32;
     } else {
-      p++;
+      p += skip >> 5;
+      skip++;
     }
   }
   op = sn_emit_literal(op, in + lit_start, n - lit_start);

@@ -373,6 +373,7 @@ DCW_HD size_t snappy_compress_block(const uint8_t* in, size_t n, uint8_t* out,
   op += varint32_put(op, (uint32_t)n);
   if (n == 0) return (size_t)(op - out);
   size_t lit_start = 0, p = 0;
+  uint32_t skip = 32; // spec v2: snappy-style skip acceleration
   while (p + 4 <= n) {
     uint32_t w = load32(in + p);
     uint32_t h = (w * kSnapHashMul) >> (32 - kSnapHashBits);
@@ -385,8 +386,10 @@ DCW_HD size_t snappy_compress_block(const uint8_t* in, size_t n, uint8_t* out,
       op = snap_emit_copy(op, p - cand, mlen);
       p += mlen;
       lit_start = p;
+      skip = 32;
     } else {
-      p++;
+      p += skip >> 5;
+      skip++;
     }
   }
   op = snap_emit_literal(op, in + lit_start, n - lit_start);

@@ -810,6 +810,7 @@ __device__ static size_t snap_encode_lds(const uint8_t* in, size_t n,
   op += varint32_put(op, (uint32_t)n);
   if (n == 0) return (size_t)(op - out);
   size_t lit_start = 0, pz = 0;
+  uint32_t skip = 32; // spec v2: snappy-style skip acceleration
   while (pz + 4 <= n) {
     uint32_t w = load32(in + pz);
     uint32_t h = (w * kSnapHashMul) >> (32 - kSnapHashBits);
@@ -822,8 +823,10 @@ __device__ static size_t snap_encode_lds(const uint8_t* in, size_t n,
       op = snap_emit_copy(op, pz - cand, mlen);
       pz += mlen;
       lit_start = pz;
+      skip = 32;
     } else {
-      pz++;
+      pz += skip >> 5;
+      skip++;
     }
   }
   op = snap_emit_literal(op, in + lit_start, n - lit_start);
