@@ -423,7 +423,14 @@ DCW_HD size_t snappy_uncompress(const uint8_t* in, size_t n, uint8_t* out,
         ip += nb;
       }
       if (ip + len > iend || op + len > oend) return 0;
-      for (size_t i = 0; i < len; i++) op[i] = ip[i];
+      {
+        size_t i = 0;
+        for (; i + 4 <= len; i += 4) {
+          uint32_t v = load32(ip + i);
+          memcpy(op + i, &v, 4);
+        }
+        for (; i < len; i++) op[i] = ip[i];
+      }
       ip += len;
       op += len;
     } else {
@@ -445,7 +452,16 @@ DCW_HD size_t snappy_uncompress(const uint8_t* in, size_t n, uint8_t* out,
       }
       if (offset == 0 || (size_t)(op - out) < offset || op + len > oend) return 0;
       const uint8_t* src = op - offset;
-      for (size_t i = 0; i < len; i++) op[i] = src[i];
+      if (offset >= 4) {
+        size_t i = 0;
+        for (; i + 4 <= len; i += 4) {
+          uint32_t v = load32(src + i);
+          memcpy(op + i, &v, 4);
+        }
+        for (; i < len; i++) op[i] = src[i];
+      } else {
+        for (size_t i = 0; i < len; i++) op[i] = src[i]; // overlapping run
+      }
       op += len;
     }
   }

@@ -790,10 +790,9 @@ __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
 #define SNAP_MAX_OUT (32 + SNAP_MAX_UNC + SNAP_MAX_UNC / 6)
 struct SnapLds {
   uint8_t in[SNAP_MAX_UNC];
-  uint8_t out[SNAP_MAX_OUT + 8];
   uint16_t tab[1u << kSnapHashBits];
   uint32_t cn;
-};
+}; // ~13.1 KiB per encoder; output goes straight to global (write-only)
 
 // wave-internal LDS ordering: drain DS ops + stop compiler reordering
 __device__ __forceinline__ void wave_lds_sync() {
@@ -803,7 +802,35 @@ __device__ __forceinline__ void wave_lds_sync() {
 
 // u16-table variant of dcw::snappy_compress_block — byte-identical output
 // (positions < 65536 in a block; 0xffff = empty, position 0xffff cannot
-// occur since SNAP_MAX_UNC < 0xffff).
+// occur since SNAP_MAX_UNC < 0xffff).  Match extension and literal copies
+// run 4 bytes at a time (same output; LDS 32-bit reads are
+// alignment-penalty-free, unlike _b64+).
+__device__ __forceinline__ uint8_t* snap_emit_literal4(uint8_t* op,
+                                                       const uint8_t* lit,
+                                                       size_t len) {
+  if (len == 0) return op;
+  size_t n = len - 1;
+  if (n < 60) {
+    *op++ = (uint8_t)(n << 2);
+  } else {
+    uint8_t tmp[4];
+    int count = 0;
+    size_t x = n;
+    while (x > 0) {
+      tmp[count++] = (uint8_t)(x & 0xff);
+      x >>= 8;
+    }
+    *op++ = (uint8_t)((59 + count) << 2);
+    for (int i = 0; i < count; i++) *op++ = tmp[i];
+  }
+  size_t t = 0;
+  for (; t + 4 <= len; t += 4) {
+    uint32_t v = load32(lit + t);
+    memcpy(op + t, &v, 4);
+  }
+  for (; t < len; t++) op[t] = lit[t];
+  return op + len;
+}
 __device__ static size_t snap_encode_lds(const uint8_t* in, size_t n,
                                          uint8_t* out, uint16_t* tab) {
   uint8_t* op = out;
@@ -818,8 +845,19 @@ __device__ static size_t snap_encode_lds(const uint8_t* in, size_t n,
     tab[h] = (uint16_t)pz;
     if (cand != 0xffffu && cand < pz && load32(in + cand) == w) {
       size_t mlen = 4;
+      while (pz + mlen + 4 <= n) {
+        uint32_t a = load32(in + cand + mlen);
+        uint32_t bz = load32(in + pz + mlen);
+        uint32_t x = a ^ bz;
+        if (x) {
+          mlen += __builtin_ctz(x) >> 3;
+          goto done_ext;
+        }
+        mlen += 4;
+      }
       while (pz + mlen < n && in[cand + mlen] == in[pz + mlen]) mlen++;
-      op = snap_emit_literal(op, in + lit_start, pz - lit_start);
+    done_ext:
+      op = snap_emit_literal4(op, in + lit_start, pz - lit_start);
       op = snap_emit_copy(op, pz - cand, mlen);
       pz += mlen;
       lit_start = pz;
@@ -829,7 +867,7 @@ __device__ static size_t snap_encode_lds(const uint8_t* in, size_t n,
       skip++;
     }
   }
-  op = snap_emit_literal(op, in + lit_start, n - lit_start);
+  op = snap_emit_literal4(op, in + lit_start, n - lit_start);
   return (size_t)(op - out);
 }
 
@@ -838,7 +876,7 @@ __global__ __launch_bounds__(256) void k_compress(
     const uint8_t* __restrict__ ucblob, uint8_t* __restrict__ cblob,
     uint64_t ccap_per_block, uint32_t* __restrict__ bsize,
     uint8_t* __restrict__ btype, uint32_t* err_flag) {
-  __shared__ SnapLds lds[4];
+  __shared__ SnapLds lds[4]; // 4 encoders/WG ~= 53 KiB -> 3 WGs/CU resident
   uint32_t wid = threadIdx.x / WAVE;  // wave within workgroup
   uint32_t lane = threadIdx.x % WAVE;
   uint32_t waves = blockDim.x / WAVE;
@@ -862,23 +900,17 @@ __global__ __launch_bounds__(256) void k_compress(
     for (uint32_t t = lane; t < (1u << kSnapHashBits); t += WAVE)
       L.tab[t] = 0xffffu;
     wave_lds_sync();
-    if (lane == 0) L.cn = (uint32_t)snap_encode_lds(L.in, d.unc_size, L.out, L.tab);
-    wave_lds_sync();
-    uint32_t cn = L.cn;
-    // GoodCompressionRatio, default max_compressed_bytes_per_kb=896
-    if (cn <= (((uint64_t)896 * d.unc_size) >> 10)) {
+    if (lane == 0) {
       uint8_t* gout = cblob + (uint64_t)b * ccap_per_block;
-      for (uint32_t t = lane * 4; t < cn; t += WAVE * 4) {
-        uint32_t chunk = (uint32_t)cn - t < 4 ? (uint32_t)cn - t : 4;
-        for (uint32_t x = 0; x < chunk; x++) gout[t + x] = L.out[t + x];
-      }
-      if (lane == 0) {
-        bsize[b] = (uint32_t)cn;
+      uint32_t cn = (uint32_t)snap_encode_lds(L.in, d.unc_size, gout, L.tab);
+      // GoodCompressionRatio, default max_compressed_bytes_per_kb=896
+      if (cn <= (((uint64_t)896 * d.unc_size) >> 10)) {
+        bsize[b] = cn;
         btype[b] = 1;
+      } else {
+        bsize[b] = d.unc_size;
+        btype[b] = 0;
       }
-    } else if (lane == 0) {
-      bsize[b] = d.unc_size;
-      btype[b] = 0;
     }
     wave_lds_sync();
   }
