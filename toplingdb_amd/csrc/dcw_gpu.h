@@ -46,6 +46,8 @@ class GpuJob {
   ~GpuJob();
   GpuJob(const GpuJob&) = delete;
   GpuJob& operator=(const GpuJob&) = delete;
+  // reuse the object (and its device buffers, grow-only) for a new job
+  void reset();
 
   // H2D of the input blob + block table; or adopt an existing staged input
   // (no H2D, buffers borrowed — not freed by this job).

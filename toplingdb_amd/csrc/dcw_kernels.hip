@@ -1158,6 +1158,18 @@ struct GpuJob::Impl {
   uint8_t* d_out_img = nullptr;
   size_t out_img_cap = 0;
 
+  std::map<void*, size_t> caps; // capacity per buffer slot (&member)
+  // named-slot ensure: grow-only reuse across jobs
+  hipError_t ens(void** slot, size_t need) {
+    size_t& cap = caps[(void*)slot];
+    if (cap >= need && *slot) return hipSuccess;
+    if (*slot) (void)hipFree(*slot);
+    *slot = nullptr;
+    size_t c = need + need / 4 + 64;
+    hipError_t e = hipMalloc(slot, c);
+    cap = (e == hipSuccess) ? c : 0;
+    return e;
+  }
   hipError_t ensure(void** p, size_t* cap, size_t need) {
     if (*cap >= need) return hipSuccess;
     if (*p) (void)hipFree(*p);
@@ -1179,6 +1191,37 @@ static double ms_between(hipEvent_t a, hipEvent_t b) {
 }
 
 GpuJob::GpuJob() : p_(new Impl) { (void)hipStreamCreate(&p_->stream); }
+
+// Reuse this job object for a new job: keep device buffers (grow-only),
+// reset the per-job state.  Staged-input pointers are dropped (borrowed
+// buffers belong to their StagedInput; owned ones are freed).
+void GpuJob::reset() {
+  Impl* p = p_;
+  if (!p->borrowed_stage) {
+    if (p->d_blob) (void)hipFree(p->d_blob);
+    if (p->d_boff) (void)hipFree(p->d_boff);
+    if (p->d_bsize) (void)hipFree(p->d_bsize);
+  }
+  p->d_blob = nullptr;
+  p->d_boff = nullptr;
+  p->d_bsize = nullptr;
+  p->borrowed_stage = false;
+  p->n_blocks = 0;
+  p->n_entries = 0;
+  p->n_surv = 0;
+  p->run_entry_begin.clear();
+  p->final_buf = 0;
+  p->emit_nblocks = 0;
+  p->emit_base = 0;
+  n_entries_ = 0;
+  n_surv_ = 0;
+  ukey_len = 0;
+  h_shared_.clear();
+  h_klen_.clear();
+  h_vlen_.clear();
+  run_blocks_.clear();
+  ms_decode = ms_merge = ms_dedup = ms_emit = ms_h2d = ms_d2h = 0;
+}
 GpuJob::~GpuJob() {
   Impl* p = p_;
   auto F = [](void* x) {
@@ -1234,12 +1277,12 @@ int GpuJob::stage(const GpuInputs& in, std::string* err) {
                    hipMemcpyHostToDevice));
   HIPCHK(hipMemcpy(p->d_bsize, bsize.data(), sizeof(uint32_t) * p->n_blocks,
                    hipMemcpyHostToDevice));
-  HIPCHK(hipMalloc(&p->d_crc, sizeof(Crc32cTables)));
+  if (!p->d_crc) HIPCHK(hipMalloc(&p->d_crc, sizeof(Crc32cTables)));
   HIPCHK(hipMemcpyAsync(p->d_crc, &g_crc, sizeof(Crc32cTables),
                         hipMemcpyHostToDevice, p->stream));
-  HIPCHK(hipMalloc(&p->d_err, 8));
+  if (!p->d_err) HIPCHK(hipMalloc(&p->d_err, 8));
   HIPCHK(hipMemsetAsync(p->d_err, 0, 8, p->stream));
-  HIPCHK(hipMalloc(&p->d_uklen_probe, 4));
+  if (!p->d_uklen_probe) HIPCHK(hipMalloc(&p->d_uklen_probe, 4));
   HIPCHK(hipMemsetAsync(p->d_uklen_probe, 0xff, 4, p->stream));
   hipEventRecord(t1, p->stream);
   HIPCHK(hipStreamSynchronize(p->stream));
@@ -1266,12 +1309,12 @@ int GpuJob::stage_adopt(const StagedInput& s, std::string* err) {
   p->n_blocks = s.n_blocks;
   p->checksum_type = s.checksum_type;
   run_blocks_ = s.run_block_begin;
-  HIPCHK(hipMalloc(&p->d_crc, sizeof(Crc32cTables)));
+  if (!p->d_crc) HIPCHK(hipMalloc(&p->d_crc, sizeof(Crc32cTables)));
   HIPCHK(hipMemcpyAsync(p->d_crc, &g_crc, sizeof(Crc32cTables),
                         hipMemcpyHostToDevice, p->stream));
-  HIPCHK(hipMalloc(&p->d_err, 8));
+  if (!p->d_err) HIPCHK(hipMalloc(&p->d_err, 8));
   HIPCHK(hipMemsetAsync(p->d_err, 0, 8, p->stream));
-  HIPCHK(hipMalloc(&p->d_uklen_probe, 4));
+  if (!p->d_uklen_probe) HIPCHK(hipMalloc(&p->d_uklen_probe, 4));
   HIPCHK(hipMemsetAsync(p->d_uklen_probe, 0xff, 4, p->stream));
   HIPCHK(hipStreamSynchronize(p->stream));
   return 0;
@@ -1297,8 +1340,8 @@ int GpuJob::decode(std::string* err) {
   hipEventCreate(&t1);
   hipEventRecord(t0, p->stream);
   uint32_t nb = p->n_blocks;
-  HIPCHK(hipMalloc(&p->d_usize, sizeof(uint32_t) * nb));
-  HIPCHK(hipMalloc(&p->d_btype_in, nb));
+  HIPCHK(p->ens((void**)&p->d_usize, sizeof(uint32_t) * nb));
+  HIPCHK(p->ens((void**)&p->d_btype_in, nb));
   double in_block_bytes = 0; // filled below from bsize D2H; verify reads them
   p->kbegin("verify_checksum", 0);
   hipLaunchKernelGGL(k_verify_usize, dim3(grid_for(nb)), dim3(256), 0, p->stream,
@@ -1324,10 +1367,10 @@ int GpuJob::decode(std::string* err) {
     in_block_bytes += usize[i]; // ~= compressed size; close enough for alg accounting
   }
   p->ublob_size = acc;
-  HIPCHK(hipMalloc(&p->d_uoff, sizeof(uint64_t) * nb));
+  HIPCHK(p->ens((void**)&p->d_uoff, sizeof(uint64_t) * nb));
   HIPCHK(hipMemcpy(p->d_uoff, uoff.data(), sizeof(uint64_t) * nb,
                    hipMemcpyHostToDevice));
-  HIPCHK(hipMalloc(&p->d_ublob, acc ? acc : 1));
+  HIPCHK(p->ens((void**)&p->d_ublob, acc ? acc : 1));
   {
     std::lock_guard<std::mutex> lk(g_kmu);
     kstats()["verify_checksum"].alg_bytes += in_block_bytes;
@@ -1337,7 +1380,7 @@ int GpuJob::decode(std::string* err) {
                      p->stream, p->d_blob, p->d_boff, p->d_bsize, p->d_btype_in,
                      p->d_uoff, p->d_usize, nb, p->d_ublob, p->d_err);
   p->kend();
-  HIPCHK(hipMalloc(&p->d_nrestarts, sizeof(uint32_t) * nb));
+  HIPCHK(p->ens((void**)&p->d_nrestarts, sizeof(uint32_t) * nb));
   hipLaunchKernelGGL(k_num_restarts, dim3(grid_for(nb)), dim3(256), 0, p->stream,
                      p->d_ublob, p->d_uoff, p->d_usize, nb, p->d_nrestarts,
                      p->d_err);
@@ -1363,10 +1406,10 @@ int GpuJob::decode(std::string* err) {
     niv += nrestarts[b];
   }
   p->n_intervals = (uint32_t)niv;
-  HIPCHK(hipMalloc(&p->d_iv_block, sizeof(uint32_t) * niv));
-  HIPCHK(hipMalloc(&p->d_iv_local, sizeof(uint32_t) * niv));
-  HIPCHK(hipMalloc(&p->d_iv_cnt, sizeof(uint32_t) * niv));
-  HIPCHK(hipMalloc(&p->d_iv_base, sizeof(uint32_t) * niv));
+  HIPCHK(p->ens((void**)&p->d_iv_block, sizeof(uint32_t) * niv));
+  HIPCHK(p->ens((void**)&p->d_iv_local, sizeof(uint32_t) * niv));
+  HIPCHK(p->ens((void**)&p->d_iv_cnt, sizeof(uint32_t) * niv));
+  HIPCHK(p->ens((void**)&p->d_iv_base, sizeof(uint32_t) * niv));
   HIPCHK(hipMemcpy(p->d_iv_block, iv_block.data(), sizeof(uint32_t) * niv,
                    hipMemcpyHostToDevice));
   HIPCHK(hipMemcpy(p->d_iv_local, iv_local.data(), sizeof(uint32_t) * niv,
@@ -1404,11 +1447,11 @@ int GpuJob::decode(std::string* err) {
   p->run_entry_begin.push_back(total_entries);
   HIPCHK(hipMemcpy(p->d_iv_base, iv_base.data(), sizeof(uint32_t) * niv,
                    hipMemcpyHostToDevice));
-  HIPCHK(hipMalloc(&p->d_ent[0], sizeof(ulong4) * total_entries));
-  HIPCHK(hipMalloc(&p->d_ent[1], sizeof(ulong4) * total_entries));
-  HIPCHK(hipMalloc(&p->d_voff, sizeof(uint64_t) * total_entries));
-  HIPCHK(hipMalloc(&p->d_vlen, sizeof(uint32_t) * total_entries));
-  HIPCHK(hipMalloc(&p->d_klen, total_entries));
+  HIPCHK(p->ens((void**)&p->d_ent[0], sizeof(ulong4) * total_entries));
+  HIPCHK(p->ens((void**)&p->d_ent[1], sizeof(ulong4) * total_entries));
+  HIPCHK(p->ens((void**)&p->d_voff, sizeof(uint64_t) * total_entries));
+  HIPCHK(p->ens((void**)&p->d_vlen, sizeof(uint32_t) * total_entries));
+  HIPCHK(p->ens((void**)&p->d_klen, total_entries));
   p->kbegin("decode_entries", (double)p->ublob_size + 45.0 * total_entries);
   hipLaunchKernelGGL(k_decode_entries, dim3(grid_for(niv)), dim3(256), 0,
                      p->stream, p->d_ublob, p->d_uoff, p->d_usize, p->d_nrestarts,
@@ -1516,18 +1559,18 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   hipEventRecord(t0, p->stream);
   uint64_t n = p->n_entries;
   const ulong4* ents = p->d_ent[p->final_buf];
-  HIPCHK(hipMalloc(&p->d_head, n));
+  HIPCHK(p->ens((void**)&p->d_head, n));
   p->kbegin("mark_heads", 17.0 * n);
   hipLaunchKernelGGL(k_mark_heads, dim3(grid_for(n)), dim3(256), 0, p->stream,
                      ents, n, p->d_head);
   p->kend();
   // head positions via scan, then gather head indices on host
-  HIPCHK(hipMalloc(&p->d_pos, sizeof(uint32_t) * n));
+  HIPCHK(p->ens((void**)&p->d_pos, sizeof(uint32_t) * n));
   uint64_t ngroups = 0;
   if (scan_u8(p, p->d_head, n, p->d_pos, &ngroups, err) != 0) return -1;
   p->n_groups = ngroups;
   // build head_idx on device: head[i] -> headidx[pos[i]] = i
-  HIPCHK(hipMalloc(&p->d_headidx, sizeof(uint64_t) * (ngroups ? ngroups : 1)));
+  HIPCHK(p->ens((void**)&p->d_headidx, sizeof(uint64_t) * (ngroups ? ngroups : 1)));
   hipLaunchKernelGGL(k_build_headidx, dim3(grid_for(n)), dim3(256), 0, p->stream,
                      p->d_head, p->d_pos, n, p->d_headidx);
   // FSM params
@@ -1567,16 +1610,16 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   P.num_levels = d->num_levels_below;
   if (!sm0.empty()) {
     size_t nb = sm0.size() * 8;
-    HIPCHK(hipMalloc(&p->d_lb_sm0, nb));
-    HIPCHK(hipMalloc(&p->d_lb_sm1, nb));
-    HIPCHK(hipMalloc(&p->d_lb_lg0, nb));
-    HIPCHK(hipMalloc(&p->d_lb_lg1, nb));
+    HIPCHK(p->ens((void**)&p->d_lb_sm0, nb));
+    HIPCHK(p->ens((void**)&p->d_lb_sm1, nb));
+    HIPCHK(p->ens((void**)&p->d_lb_lg0, nb));
+    HIPCHK(p->ens((void**)&p->d_lb_lg1, nb));
     HIPCHK(hipMemcpy(p->d_lb_sm0, sm0.data(), nb, hipMemcpyHostToDevice));
     HIPCHK(hipMemcpy(p->d_lb_sm1, sm1.data(), nb, hipMemcpyHostToDevice));
     HIPCHK(hipMemcpy(p->d_lb_lg0, lg0.data(), nb, hipMemcpyHostToDevice));
     HIPCHK(hipMemcpy(p->d_lb_lg1, lg1.data(), nb, hipMemcpyHostToDevice));
   }
-  HIPCHK(hipMalloc(&p->d_lb_beg, sizeof(uint32_t) * lbeg.size()));
+  HIPCHK(p->ens((void**)&p->d_lb_beg, sizeof(uint32_t) * lbeg.size()));
   HIPCHK(hipMemcpy(p->d_lb_beg, lbeg.data(), sizeof(uint32_t) * lbeg.size(),
                    hipMemcpyHostToDevice));
   P.lb_sm_k0 = p->d_lb_sm0;
@@ -1585,10 +1628,10 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   P.lb_lg_k1 = p->d_lb_lg1;
   P.lb_level_beg = p->d_lb_beg;
 
-  HIPCHK(hipMalloc(&p->d_survive, n));
-  HIPCHK(hipMalloc(&p->d_newtag, sizeof(uint64_t) * n));
-  HIPCHK(hipMalloc(&p->d_clearv, n));
-  HIPCHK(hipMalloc(&p->d_gflags, ngroups ? ngroups : 1));
+  HIPCHK(p->ens((void**)&p->d_survive, n));
+  HIPCHK(p->ens((void**)&p->d_newtag, sizeof(uint64_t) * n));
+  HIPCHK(p->ens((void**)&p->d_clearv, n));
+  HIPCHK(p->ens((void**)&p->d_gflags, ngroups ? ngroups : 1));
   // entries skipped by multi-consume FSM paths never store their slot:
   // they must read as "dropped"
   HIPCHK(hipMemsetAsync(p->d_survive, 0, n, p->stream));
@@ -1626,13 +1669,13 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   if (scan_u8(p, p->d_survive, n, p->d_pos, &nsurv, err) != 0) return -1;
   p->n_surv = nsurv;
   n_surv_ = nsurv;
-  HIPCHK(hipMalloc(&p->d_sk0, sizeof(uint64_t) * (nsurv + 1)));
-  HIPCHK(hipMalloc(&p->d_sk1, sizeof(uint64_t) * (nsurv + 1)));
-  HIPCHK(hipMalloc(&p->d_stag, sizeof(uint64_t) * (nsurv + 1)));
-  HIPCHK(hipMalloc(&p->d_svoff, sizeof(uint64_t) * (nsurv + 1)));
-  HIPCHK(hipMalloc(&p->d_svlen, sizeof(uint32_t) * (nsurv + 1)));
-  HIPCHK(hipMalloc(&p->d_sklen, nsurv + 1));
-  HIPCHK(hipMalloc(&p->d_sshared, nsurv + 1));
+  HIPCHK(p->ens((void**)&p->d_sk0, sizeof(uint64_t) * (nsurv + 1)));
+  HIPCHK(p->ens((void**)&p->d_sk1, sizeof(uint64_t) * (nsurv + 1)));
+  HIPCHK(p->ens((void**)&p->d_stag, sizeof(uint64_t) * (nsurv + 1)));
+  HIPCHK(p->ens((void**)&p->d_svoff, sizeof(uint64_t) * (nsurv + 1)));
+  HIPCHK(p->ens((void**)&p->d_svlen, sizeof(uint32_t) * (nsurv + 1)));
+  HIPCHK(p->ens((void**)&p->d_sklen, nsurv + 1));
+  HIPCHK(p->ens((void**)&p->d_sshared, nsurv + 1));
   p->kbegin("gather_survivors", 90.0 * n);
   hipLaunchKernelGGL(k_gather_survivors, dim3(grid_for(n)), dim3(256), 0,
                      p->stream, ents, n, p->d_survive, p->d_pos, p->d_newtag,

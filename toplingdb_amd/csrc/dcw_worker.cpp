@@ -283,7 +283,12 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
 
   // ---- read + parse inputs (host I/O) ----
   uint64_t t0 = now_usec();
-  GpuJob job;
+  // one cached GpuJob per process: device buffers are reused across jobs
+  // (grow-only), avoiding ~30 hipMalloc/hipFree pairs per job
+  static GpuJob* g_job = nullptr;
+  if (!g_job) g_job = new GpuJob();
+  g_job->reset();
+  GpuJob& job = *g_job;
   uint64_t in_bytes = 0;
   LoadedInputs L;
   StagedJob* staged = nullptr;
