@@ -28,6 +28,7 @@ namespace dcw {
 namespace {
 
 std::mutex g_mu;
+std::mutex g_out_mu; // guards out_files slots filled by tail threads
 bool g_inited = false;
 uint64_t g_next_stage_handle = 1;
 
@@ -321,6 +322,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   TableOpts base = opts_from_desc(d);
   uint64_t next_file_number = d->next_file_number;
   std::vector<dcw_output_file> out_files;
+  out_files.reserve(1024); // slots are written by tail threads; no realloc
   std::vector<std::future<int>> writers;
   uint64_t total_out_bytes = 0, total_out_entries = 0;
   uint64_t plan_usec = 0, write_usec = 0;
@@ -439,87 +441,120 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       cur = cut_entry + 1;
     }
     if (handles.empty()) break; // nothing left
-    // separators (FindShortestInternalKeySeparator between adjacent blocks;
-    // last block keeps its last key — default kShortenSeparators mode)
     uint64_t tw1 = now_usec();
-    std::vector<std::string> seps(handles.size());
-    bool sep_key_plus_seq = false;
-    for (size_t b = 0; b < handles.size(); b++) {
-      std::string sep = last_keys[b];
-      if (b + 1 < handles.size()) {
-        shorten_separator(sep, (const uint8_t*)first_keys[b + 1].data(),
-                          first_keys[b + 1].size());
-        size_t su = sep.size() - 8, nu = first_keys[b + 1].size() - 8;
-        if (su == nu && memcmp(sep.data(), first_keys[b + 1].data(), su) == 0)
-          sep_key_plus_seq = true;
-      }
-      seps[b] = sep;
-    }
-    // stats
-    TailStats st;
-    st.num_data_blocks = handles.size();
-    st.data_size = image.len;
-    st.num_entries = file_count;
-    for (uint64_t i = file_first; i < file_first + file_count; i++) {
-      st.raw_key_size += klen[i];
-      st.raw_value_size += vlen[i];
-    }
     uint64_t mn_seq = 0, mx_seq = 0, n_tomb = 0;
     if (job.seq_minmax(file_first, file_count, &mn_seq, &mx_seq, &n_tomb, &err) != 0)
       return fail(res, 25, err);
-    st.num_deletions = n_tomb;
-    uint64_t tail_start = image.len;
-    std::string tail = build_tail(o, st, handles, seps, !sep_key_plus_seq, tail_start);
-    image.append(tail.data(), tail.size());
-    // write the file on a background thread (overlaps the next file's GPU
-    // work; joined before returning)
+    // hand the whole per-file tail (separators, stats, meta tail, file
+    // write) to a background thread — the GPU starts the next file now.
     char path[600];
     snprintf(path, sizeof(path), "%s/%06" PRIu64 ".sst", d->output_dir,
              o.orig_file_number);
-    uint64_t image_size = image.len;
-    uint8_t* img_ptr = image.p;
-    size_t img_cap = image.cap;
-    image.release(); // ownership moves to the writer thread
-    writers.emplace_back(std::async(
-        std::launch::async,
-        [](std::string pth, uint8_t* ptr, size_t n, size_t cp) -> int {
-          FILE* f = fopen(pth.c_str(), "wb");
-          if (!f) {
-            g_pin_pool.release(ptr, cp);
-            return -1;
-          }
-          size_t w = fwrite(ptr, 1, n, f);
-          fclose(f);
-          g_pin_pool.release(ptr, cp);
-          return w == n ? 0 : -1;
-        },
-        std::string(path), img_ptr, (size_t)image_size, img_cap));
-    write_usec += now_usec() - tw1;
-
-    dcw_output_file of;
-    memset(&of, 0, sizeof(of));
-    snprintf(of.path, sizeof(of.path), "%s", path);
-    of.file_number = o.orig_file_number;
-    of.file_size = image_size;
-    std::string smallest = first_keys.front();
-    std::string largest = last_keys.back();
-    of.smallest_len = (uint32_t)smallest.size();
-    memcpy(of.smallest_ikey, smallest.data(), std::min<size_t>(64, smallest.size()));
-    of.largest_len = (uint32_t)largest.size();
-    memcpy(of.largest_ikey, largest.data(), std::min<size_t>(64, largest.size()));
-    of.smallest_seqno = mn_seq == ~0ull ? 0 : mn_seq;
-    of.largest_seqno = mx_seq;
-    of.num_entries = file_count;
-    out_files.push_back(of);
-    total_out_bytes += image_size;
+    if (out_files.size() + 1 >= out_files.capacity()) {
+      // join tail threads before the vector reallocates (slots are written
+      // by threads through indices into this vector)
+      for (auto& w : writers)
+        if (w.get() != 0) return fail(res, 26, "output file tail/write failed");
+      writers.clear();
+      out_files.reserve(out_files.capacity() * 2);
+    }
+    size_t slot = out_files.size();
+    out_files.emplace_back();
     total_out_entries += file_count;
+    struct TailJob {
+      TableOpts o;
+      RawBuf image;
+      std::vector<SstIndexEntry> handles;
+      std::vector<std::string> first_keys, last_keys;
+      uint64_t file_first, file_count, mn_seq, mx_seq, n_tomb;
+      std::string path;
+    };
+    auto tj = std::make_shared<TailJob>();
+    tj->o = o;
+    tj->image.p = image.p;
+    tj->image.len = image.len;
+    tj->image.cap = image.cap;
+    image.release();
+    tj->handles = std::move(handles);
+    tj->first_keys = std::move(first_keys);
+    tj->last_keys = std::move(last_keys);
+    tj->file_first = file_first;
+    tj->file_count = file_count;
+    tj->mn_seq = mn_seq;
+    tj->mx_seq = mx_seq;
+    tj->n_tomb = n_tomb;
+    tj->path = path;
+    const uint8_t* klen_p = klen.data();
+    const uint32_t* vlen_p = vlen.data();
+    uint64_t* out_bytes_p = &total_out_bytes;
+    std::mutex* ob_mu = &g_out_mu;
+    writers.emplace_back(std::async(std::launch::async, [tj, klen_p, vlen_p,
+                                                         slot, &out_files,
+                                                         out_bytes_p, ob_mu]() -> int {
+      // separators (FindShortestInternalKeySeparator between adjacent
+      // blocks; last block keeps its last key — kShortenSeparators mode)
+      size_t nb = tj->handles.size();
+      std::vector<std::string> seps(nb);
+      bool sep_key_plus_seq = false;
+      for (size_t b = 0; b < nb; b++) {
+        std::string sep = tj->last_keys[b];
+        if (b + 1 < nb) {
+          shorten_separator(sep, (const uint8_t*)tj->first_keys[b + 1].data(),
+                            tj->first_keys[b + 1].size());
+          size_t su = sep.size() - 8, nu = tj->first_keys[b + 1].size() - 8;
+          if (su == nu &&
+              memcmp(sep.data(), tj->first_keys[b + 1].data(), su) == 0)
+            sep_key_plus_seq = true;
+        }
+        seps[b] = sep;
+      }
+      TailStats st;
+      st.num_data_blocks = nb;
+      st.data_size = tj->image.len;
+      st.num_entries = tj->file_count;
+      for (uint64_t i = tj->file_first; i < tj->file_first + tj->file_count; i++) {
+        st.raw_key_size += klen_p[i];
+        st.raw_value_size += vlen_p[i];
+      }
+      st.num_deletions = tj->n_tomb;
+      uint64_t tail_start = tj->image.len;
+      std::string tail =
+          build_tail(tj->o, st, tj->handles, seps, !sep_key_plus_seq, tail_start);
+      tj->image.append(tail.data(), tail.size());
+      FILE* f = fopen(tj->path.c_str(), "wb");
+      if (!f) return -1;
+      size_t w = fwrite(tj->image.p, 1, tj->image.len, f);
+      fclose(f);
+      if (w != tj->image.len) return -1;
+      dcw_output_file of;
+      memset(&of, 0, sizeof(of));
+      snprintf(of.path, sizeof(of.path), "%s", tj->path.c_str());
+      of.file_number = tj->o.orig_file_number;
+      of.file_size = tj->image.len;
+      const std::string& smallest = tj->first_keys.front();
+      const std::string& largest = tj->last_keys.back();
+      of.smallest_len = (uint32_t)smallest.size();
+      memcpy(of.smallest_ikey, smallest.data(), std::min<size_t>(64, smallest.size()));
+      of.largest_len = (uint32_t)largest.size();
+      memcpy(of.largest_ikey, largest.data(), std::min<size_t>(64, largest.size()));
+      of.smallest_seqno = tj->mn_seq == ~0ull ? 0 : tj->mn_seq;
+      of.largest_seqno = tj->mx_seq;
+      of.num_entries = tj->file_count;
+      {
+        std::lock_guard<std::mutex> lk(*ob_mu);
+        out_files[slot] = of;
+        *out_bytes_p += of.file_size;
+      }
+      return 0;
+    }));
+    write_usec += now_usec() - tw1;
     s = cur;
   }
   {
-    uint64_t tj = now_usec();
+    uint64_t tj2 = now_usec();
     for (auto& w : writers)
-      if (w.get() != 0) return fail(res, 26, "output file write failed");
-    write_usec += now_usec() - tj;
+      if (w.get() != 0) return fail(res, 26, "output file tail/write failed");
+    write_usec += now_usec() - tj2;
   }
 
   res->num_files = (uint32_t)out_files.size();
