@@ -142,86 +142,9 @@ __global__ void k_verify_usize(const uint8_t* __restrict__ blob,
 // all lanes copy in/out); oversized blocks fall back to the direct path.
 #define DEC_MAX 5376
 struct DecLds {
-  uint32_t in32[DEC_MAX / 4 + 2];
+  uint8_t in[DEC_MAX];
   uint8_t out[DEC_MAX];
 };
-
-__device__ __forceinline__ uint32_t dec_ld32(const uint32_t* w32, uint32_t pos) {
-  uint32_t i = pos >> 2, sh = (pos & 3) * 8;
-  uint64_t pair = ((uint64_t)w32[i + 1] << 32) | w32[i];
-  return (uint32_t)(pair >> sh);
-}
-
-// device snappy decoder: source reads through the aligned-u32 funnel view
-// (same output as dcw::snappy_uncompress)
-__device__ static size_t snap_decode_lds(const uint32_t* in32, size_t n,
-                                         uint8_t* out, size_t cap) {
-  const uint8_t* in = (const uint8_t*)in32;
-  uint32_t ulen;
-  int k = varint32_get(in, in + n, &ulen);
-  if (k < 0 || ulen > cap) return 0;
-  uint32_t ip = (uint32_t)k;
-  uint8_t* op = out;
-  uint8_t* oend = out + ulen;
-  while (ip < n) {
-    uint8_t tag = in[ip++];
-    if ((tag & 3) == 0) {
-      size_t len = (size_t)(tag >> 2) + 1;
-      if (len > 60) {
-        int nb = (int)len - 60;
-        if (ip + nb > n) return 0;
-        len = 0;
-        for (int i = 0; i < nb; i++) len |= (size_t)in[ip + i] << (8 * i);
-        len += 1;
-        ip += nb;
-      }
-      if (ip + len > n || op + len > oend) return 0;
-      size_t t = 0;
-      for (; t + 4 <= len; t += 4) {
-        uint32_t v = dec_ld32(in32, ip + (uint32_t)t);
-        op[t] = (uint8_t)v;
-        op[t + 1] = (uint8_t)(v >> 8);
-        op[t + 2] = (uint8_t)(v >> 16);
-        op[t + 3] = (uint8_t)(v >> 24);
-      }
-      for (; t < len; t++) op[t] = in[ip + t];
-      ip += (uint32_t)len;
-      op += len;
-    } else {
-      size_t len, offset;
-      if ((tag & 3) == 1) {
-        len = ((tag >> 2) & 7) + 4;
-        if (ip >= n) return 0;
-        offset = ((size_t)(tag >> 5) << 8) | in[ip++];
-      } else if ((tag & 3) == 2) {
-        len = (size_t)(tag >> 2) + 1;
-        if (ip + 2 > n) return 0;
-        offset = (size_t)in[ip] | ((size_t)in[ip + 1] << 8);
-        ip += 2;
-      } else {
-        len = (size_t)(tag >> 2) + 1;
-        if (ip + 4 > n) return 0;
-        offset = dec_ld32(in32, ip);
-        ip += 4;
-      }
-      if (offset == 0 || (size_t)(op - out) < offset || op + len > oend) return 0;
-      const uint8_t* src = op - offset;
-      if (offset >= 8) {
-        size_t i = 0;
-        for (; i + 8 <= len; i += 8) {
-          uint64_t v;
-          memcpy(&v, src + i, 8);
-          memcpy(op + i, &v, 8);
-        }
-        for (; i < len; i++) op[i] = src[i];
-      } else {
-        for (size_t i = 0; i < len; i++) op[i] = src[i];
-      }
-      op += len;
-    }
-  }
-  return (size_t)(op - out) == ulen ? ulen : 0;
-}
 __device__ __forceinline__ void wave_lds_sync2() {
   __builtin_amdgcn_s_waitcnt(0);
   __builtin_amdgcn_wave_barrier();
@@ -252,17 +175,13 @@ __global__ __launch_bounds__(256) void k_decompress(
       if (lane == 0)
         for (uint32_t t = n & ~15u; t < n; t++) dst[t] = src[t];
     } else if (n <= DEC_MAX && usize[i] <= DEC_MAX) {
-      uint32_t nw = (n + 3) / 4;
-      for (uint32_t t = lane; t < nw; t += WAVE) {
-        uint32_t v = 0;
-        uint32_t nbytes = n - t * 4 < 4 ? n - t * 4 : 4;
-        memcpy(&v, src + t * 4, nbytes);
-        L.in32[t] = v;
+      for (uint32_t t = lane * 4; t < n; t += WAVE * 4) {
+        uint32_t chunk = n - t < 4 ? n - t : 4;
+        for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = src[t + x];
       }
-      if (lane == 0) L.in32[nw] = 0;
       wave_lds_sync2();
       if (lane == 0) {
-        if (snap_decode_lds(L.in32, n, L.out, usize[i]) != usize[i])
+        if (snappy_uncompress(L.in, n, L.out, usize[i]) != usize[i])
           set_err(err_flag, DE_SNAPPY);
       }
       wave_lds_sync2();
@@ -936,19 +855,10 @@ __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
 #define SNAP_MAX_UNC 5120 // block_size + slack; host guards this bound
 #define SNAP_MAX_OUT (32 + SNAP_MAX_UNC + SNAP_MAX_UNC / 6)
 struct SnapLds {
-  uint32_t in32[SNAP_MAX_UNC / 4 + 2]; // byte view via (const uint8_t*)in32
+  uint8_t in[SNAP_MAX_UNC];
   uint16_t tab[1u << kSnapHashBits];
   uint32_t cn;
 }; // ~13.1 KiB per encoder; output goes straight to global (write-only)
-
-// unaligned 32-bit read from an LDS byte position via two ALIGNED dword
-// reads + funnel shift (a memcpy at unknown alignment lowers to 4 byte
-// loads on LDS — ~4x the latency chain)
-__device__ __forceinline__ uint32_t lds_ld32(const uint32_t* w32, uint32_t pos) {
-  uint32_t i = pos >> 2, sh = (pos & 3) * 8;
-  uint64_t pair = ((uint64_t)w32[i + 1] << 32) | w32[i];
-  return (uint32_t)(pair >> sh);
-}
 
 // wave-internal LDS ordering: drain DS ops + stop compiler reordering
 __device__ __forceinline__ void wave_lds_sync() {
@@ -987,52 +897,23 @@ __device__ __forceinline__ uint8_t* snap_emit_literal4(uint8_t* op,
   for (; t < len; t++) op[t] = lit[t];
   return op + len;
 }
-// literal emission reading from the LDS u32 view with funnel loads
-__device__ __forceinline__ uint8_t* snap_emit_literal_lds(
-    uint8_t* op, const uint32_t* in32, uint32_t lit_pos, size_t len) {
-  if (len == 0) return op;
-  size_t n = len - 1;
-  if (n < 60) {
-    *op++ = (uint8_t)(n << 2);
-  } else {
-    uint8_t tmp[4];
-    int count = 0;
-    size_t x = n;
-    while (x > 0) {
-      tmp[count++] = (uint8_t)(x & 0xff);
-      x >>= 8;
-    }
-    *op++ = (uint8_t)((59 + count) << 2);
-    for (int i = 0; i < count; i++) *op++ = tmp[i];
-  }
-  size_t t = 0;
-  for (; t + 4 <= len; t += 4) {
-    uint32_t v = lds_ld32(in32, lit_pos + (uint32_t)t);
-    memcpy(op + t, &v, 4);
-  }
-  const uint8_t* in = (const uint8_t*)in32;
-  for (; t < len; t++) op[t] = in[lit_pos + t];
-  return op + len;
-}
-
-__device__ static size_t snap_encode_lds(const uint32_t* in32, size_t n,
+__device__ static size_t snap_encode_lds(const uint8_t* in, size_t n,
                                          uint8_t* out, uint16_t* tab) {
-  const uint8_t* in = (const uint8_t*)in32;
   uint8_t* op = out;
   op += varint32_put(op, (uint32_t)n);
   if (n == 0) return (size_t)(op - out);
   size_t lit_start = 0, pz = 0;
   uint32_t skip = 32; // spec v2: snappy-style skip acceleration
   while (pz + 4 <= n) {
-    uint32_t w = lds_ld32(in32, (uint32_t)pz);
+    uint32_t w = load32(in + pz);
     uint32_t h = (w * kSnapHashMul) >> (32 - kSnapHashBits);
     uint32_t cand = tab[h];
     tab[h] = (uint16_t)pz;
-    if (cand != 0xffffu && cand < pz && lds_ld32(in32, cand) == w) {
+    if (cand != 0xffffu && cand < pz && load32(in + cand) == w) {
       size_t mlen = 4;
       while (pz + mlen + 4 <= n) {
-        uint32_t a = lds_ld32(in32, cand + (uint32_t)mlen);
-        uint32_t bz = lds_ld32(in32, (uint32_t)(pz + mlen));
+        uint32_t a = load32(in + cand + mlen);
+        uint32_t bz = load32(in + pz + mlen);
         uint32_t x = a ^ bz;
         if (x) {
           mlen += __builtin_ctz(x) >> 3;
@@ -1042,7 +923,7 @@ __device__ static size_t snap_encode_lds(const uint32_t* in32, size_t n,
       }
       while (pz + mlen < n && in[cand + mlen] == in[pz + mlen]) mlen++;
     done_ext:
-      op = snap_emit_literal_lds(op, in32, (uint32_t)lit_start, pz - lit_start);
+      op = snap_emit_literal4(op, in + lit_start, pz - lit_start);
       op = snap_emit_copy(op, pz - cand, mlen);
       pz += mlen;
       lit_start = pz;
@@ -1052,7 +933,7 @@ __device__ static size_t snap_encode_lds(const uint32_t* in32, size_t n,
       skip++;
     }
   }
-  op = snap_emit_literal_lds(op, in32, (uint32_t)lit_start, n - lit_start);
+  op = snap_emit_literal4(op, in + lit_start, n - lit_start);
   return (size_t)(op - out);
 }
 
@@ -1078,20 +959,16 @@ __global__ __launch_bounds__(256) void k_compress(
       continue;
     }
     const uint8_t* gin = ucblob + d.uout;
-    uint32_t nw = (d.unc_size + 3) / 4;
-    for (uint32_t t = lane; t < nw; t += WAVE) {
-      uint32_t v = 0;
-      uint32_t nbytes = d.unc_size - t * 4 < 4 ? d.unc_size - t * 4 : 4;
-      memcpy(&v, gin + t * 4, nbytes);
-      L.in32[t] = v;
+    for (uint32_t t = lane * 4; t < d.unc_size; t += WAVE * 4) {
+      uint32_t chunk = d.unc_size - t < 4 ? d.unc_size - t : 4;
+      for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = gin[t + x];
     }
-    if (lane == 0) L.in32[nw] = 0;
     for (uint32_t t = lane; t < (1u << kSnapHashBits); t += WAVE)
       L.tab[t] = 0xffffu;
     wave_lds_sync();
     if (lane == 0) {
       uint8_t* gout = cblob + (uint64_t)b * ccap_per_block;
-      uint32_t cn = (uint32_t)snap_encode_lds(L.in32, d.unc_size, gout, L.tab);
+      uint32_t cn = (uint32_t)snap_encode_lds(L.in, d.unc_size, gout, L.tab);
       // GoodCompressionRatio, default max_compressed_bytes_per_kb=896
       if (cn <= (((uint64_t)896 * d.unc_size) >> 10)) {
         bsize[b] = cn;
