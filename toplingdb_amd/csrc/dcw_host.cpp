@@ -361,8 +361,10 @@ void shorten_separator(std::string& start, const uint8_t* limit, size_t limit_le
 
 // ---------------- plan FSM: pure block planning ----------------
 std::vector<PlannedBlock> plan_blocks(const PlanIn& in, size_t from,
-                                      const TableOpts& o, uint64_t min_bytes) {
+                                      const TableOpts& o, uint64_t min_bytes,
+                                      std::vector<uint32_t>* eoff_out) {
   std::vector<PlannedBlock> out;
+  if (eoff_out) eoff_out->clear();
   uint64_t dev_limit = ((o.block_size * (100 - o.block_size_deviation)) + 99) / 100;
   uint64_t produced = 0;
   size_t i = from;
@@ -389,6 +391,7 @@ std::vector<PlannedBlock> plan_blocks(const PlanIn& in, size_t from,
         shared = 0; // first entry of the block: last_key truncated to empty
       }
       size_t non_shared = klen - shared;
+      if (eoff_out) eoff_out->push_back((uint32_t)bytes);
       bytes += varint_len(shared) + varint_len(non_shared) + varint_len(vlen) +
                non_shared + vlen;
       counter++;

@@ -125,9 +125,12 @@ struct PlannedBlock {
   uint32_t num_restarts;
 };
 // Pure block FSM (no file cuts): plan blocks from `from` until covering
-// `min_bytes` of uncompressed output or entries run out.
+// `min_bytes` of uncompressed output or entries run out.  If eoff_out is
+// non-null it receives each planned entry's in-block byte offset (indexed
+// from `from`), saving the emit path a second walk.
 std::vector<PlannedBlock> plan_blocks(const PlanIn& in, size_t from,
-                                      const TableOpts& o, uint64_t min_bytes);
+                                      const TableOpts& o, uint64_t min_bytes,
+                                      std::vector<uint32_t>* eoff_out = nullptr);
 
 // grandparent accounting state (compaction_outputs.cc:121-230), driven by
 // per-entry boundary positions precomputed on the GPU:

@@ -1237,6 +1237,10 @@ struct GpuJob::Impl {
   size_t scratch64_cap = 0;
   void* d_outoff = nullptr;
   size_t outoff_cap = 0;
+  void* d_scratch_keys = nullptr;
+  void* d_scratch_recoff = nullptr;
+  void* d_scratch_gather = nullptr;
+  void* d_scratch_mm = nullptr;
   uint8_t* d_out_img = nullptr;
   size_t out_img_cap = 0;
 
@@ -1325,6 +1329,8 @@ GpuJob::~GpuJob() {
   F(p->d_sklen); F(p->d_sshared); F(p->d_bds); F(p->d_eoff); F(p->d_ucblob);
   F(p->d_cblob); F(p->d_ebsize); F(p->d_ebtype); F(p->d_ecsum);
   F(p->d_scratch64); F(p->d_out_img); F(p->d_outoff);
+  F(p->d_scratch_keys); F(p->d_scratch_recoff); F(p->d_scratch_gather);
+  F(p->d_scratch_mm);
   if (p->stream) (void)hipStreamDestroy(p->stream);
   delete p;
 }
@@ -1794,7 +1800,8 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
 }
 
 int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts& o,
-                        std::vector<uint32_t>* comp_sizes, std::string* err) {
+                        std::vector<uint32_t>* comp_sizes, std::string* err,
+                        const std::vector<uint32_t>* eoff_in) {
   Impl* p = p_;
   hipEvent_t t0, t1;
   hipEventCreate(&t0);
@@ -1809,21 +1816,29 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   uint64_t first = blocks.front().first;
   uint64_t last = blocks.back().first + blocks.back().count;
   uint64_t nent = last - first;
-  std::vector<uint32_t> eoff(nent);
+  std::vector<uint32_t> eoff_local;
+  const std::vector<uint32_t>* eoff = eoff_in;
+  if (!eoff) {
+    eoff_local.resize(nent);
+    for (uint32_t b = 0; b < nb; b++) {
+      const PlannedBlock& pb = blocks[b];
+      uint32_t off = 0;
+      for (uint32_t li = 0; li < pb.count; li++) {
+        uint64_t i = pb.first + li;
+        uint32_t shared = (li % o.block_restart_interval == 0) ? 0 : h_shared_[i];
+        uint32_t klen = h_klen_[i];
+        uint32_t vl = h_vlen_[i];
+        eoff_local[i - first] = off;
+        off += varint_len(shared) + varint_len(klen - shared) + varint_len(vl) +
+               (klen - shared) + vl;
+      }
+    }
+    eoff = &eoff_local;
+  }
   std::vector<EmitBlockDesc> bds(nb);
   uint64_t uout = 0;
   for (uint32_t b = 0; b < nb; b++) {
     const PlannedBlock& pb = blocks[b];
-    uint32_t off = 0;
-    for (uint32_t li = 0; li < pb.count; li++) {
-      uint64_t i = pb.first + li;
-      uint32_t shared = (li % o.block_restart_interval == 0) ? 0 : h_shared_[i];
-      uint32_t klen = h_klen_[i];
-      uint32_t vl = h_vlen_[i];
-      eoff[i - first] = off;
-      off += varint_len(shared) + varint_len(klen - shared) + varint_len(vl) +
-             (klen - shared) + vl;
-    }
     bds[b] = {(uint32_t)pb.first, pb.count, pb.unc_size, pb.num_restarts, uout};
     uout += pb.unc_size;
   }
@@ -1837,7 +1852,7 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   p->d_ebtype = (uint8_t*)(p->d_ecsum + nb);
   HIPCHK(hipMemcpy(p->d_bds, bds.data(), sizeof(EmitBlockDesc) * nb,
                    hipMemcpyHostToDevice));
-  HIPCHK(hipMemcpy(p->d_eoff, eoff.data(), sizeof(uint32_t) * nent,
+  HIPCHK(hipMemcpy(p->d_eoff, eoff->data(), sizeof(uint32_t) * nent,
                    hipMemcpyHostToDevice));
   // NOTE: k_emit indexes eoff by absolute survivor index minus chunk base.
   p->kbegin("emit", 2.0 * (double)uout);
@@ -1926,8 +1941,8 @@ int GpuJob::fetch_block_keys(size_t b0, size_t b1,
   first_keys->clear();
   last_keys->clear();
   if (!nb) return 0;
-  uint8_t* d_keys;
-  HIPCHK(hipMalloc(&d_keys, (uint64_t)nb * 64));
+  HIPCHK(p->ens((void**)&p->d_scratch_keys, (uint64_t)nb * 64));
+  uint8_t* d_keys = (uint8_t*)p->d_scratch_keys;
   hipLaunchKernelGGL(k_block_keys, dim3(grid_for(nb)), dim3(256), 0, p->stream,
                      p->d_bds, (uint32_t)b0, (uint32_t)b1, p->d_sk0, p->d_sk1,
                      p->d_stag, p->d_sklen, d_keys);
@@ -1935,7 +1950,6 @@ int GpuJob::fetch_block_keys(size_t b0, size_t b1,
   HIPCHK(hipMemcpyAsync(h.data(), d_keys, h.size(), hipMemcpyDeviceToHost,
                         p->stream));
   HIPCHK(hipStreamSynchronize(p->stream));
-  (void)hipFree(d_keys);
   for (uint32_t i = 0; i < nb; i++) {
     const uint8_t* o = h.data() + (uint64_t)i * 64;
     first_keys->emplace_back((const char*)o + 1, o[0]);
@@ -1957,10 +1971,10 @@ int GpuJob::gather_entries(uint64_t first, uint32_t count,
     recoff[i] = acc;
     acc += 1 + h_klen_[first + i] + 4 + h_vlen_[first + i];
   }
-  uint64_t* d_recoff;
-  uint8_t* d_out;
-  HIPCHK(hipMalloc(&d_recoff, sizeof(uint64_t) * count));
-  HIPCHK(hipMalloc(&d_out, acc));
+  HIPCHK(p->ens(&p->d_scratch_recoff, sizeof(uint64_t) * count));
+  HIPCHK(p->ens(&p->d_scratch_gather, acc));
+  uint64_t* d_recoff = (uint64_t*)p->d_scratch_recoff;
+  uint8_t* d_out = (uint8_t*)p->d_scratch_gather;
   HIPCHK(hipMemcpy(d_recoff, recoff.data(), sizeof(uint64_t) * count,
                    hipMemcpyHostToDevice));
   hipLaunchKernelGGL(k_gather_range, dim3(grid_for(count)), dim3(256), 0,
@@ -1970,8 +1984,6 @@ int GpuJob::gather_entries(uint64_t first, uint32_t count,
   std::vector<uint8_t> h(acc);
   HIPCHK(hipMemcpyAsync(h.data(), d_out, acc, hipMemcpyDeviceToHost, p->stream));
   HIPCHK(hipStreamSynchronize(p->stream));
-  (void)hipFree(d_recoff);
-  (void)hipFree(d_out);
   for (uint32_t i = 0; i < count; i++) {
     const uint8_t* r = h.data() + recoff[i];
     uint32_t klen = r[0];
@@ -1986,8 +1998,8 @@ int GpuJob::gather_entries(uint64_t first, uint32_t count,
 int GpuJob::seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* mx,
                        uint64_t* n_tombstones, std::string* err) {
   Impl* p = p_;
-  unsigned long long* d;
-  HIPCHK(hipMalloc(&d, 24));
+  HIPCHK(p->ens(&p->d_scratch_mm, 24));
+  unsigned long long* d = (unsigned long long*)p->d_scratch_mm;
   unsigned long long init[3] = {~0ull, 0, 0};
   HIPCHK(hipMemcpy(d, init, 24, hipMemcpyHostToDevice));
   hipLaunchKernelGGL(k_seq_minmax, dim3(grid_for(count)), dim3(256), 0, p->stream,
@@ -1995,7 +2007,6 @@ int GpuJob::seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* m
   unsigned long long out[3];
   HIPCHK(hipMemcpyAsync(out, d, 24, hipMemcpyDeviceToHost, p->stream));
   HIPCHK(hipStreamSynchronize(p->stream));
-  (void)hipFree(d);
   *mn = out[0];
   *mx = out[1];
   *n_tombstones = out[2];

@@ -350,11 +350,13 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       uint64_t min_unc = o.compression == 1
                              ? (uint64_t)(want / comp_ratio) + 8 * o.block_size
                              : want + 2 * o.block_size;
-      std::vector<PlannedBlock> blocks = plan_blocks(pin, cur, o, min_unc);
+      std::vector<uint32_t> eoff_plan;
+      std::vector<PlannedBlock> blocks = plan_blocks(pin, cur, o, min_unc, &eoff_plan);
       plan_usec += now_usec() - tp0;
       if (blocks.empty()) break;
       std::vector<uint32_t> csizes;
-      if (job.emit_blocks(blocks, o, &csizes, &err) != 0) return fail(res, 21, err);
+      if (job.emit_blocks(blocks, o, &csizes, &err, &eoff_plan) != 0)
+        return fail(res, 21, err);
       // walk csizes first: find the flush that crosses the target
       // (cut = first entry after it; the open block then holds exactly one
       // entry — AddToOutput/FileSize interplay, compaction_outputs.cc:356-420)
