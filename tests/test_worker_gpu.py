@@ -179,3 +179,11 @@ def test_refuses_merge_operand(tmp_path):
     jd = dcw.make_job([[p]], str(out))
     with pytest.raises(RuntimeError):
         dcw.execute(jd)
+
+
+def test_config2_exact_shape(tmp_path):
+    # BASELINE.json configs[1]: 2-way merge of two ~64 MiB SSTs, 16B keys,
+    # no compression, 1 GPU — full size, bit-exact vs oracle.
+    runs = gen_runs(tmp_path, 2, 530_000)
+    rg, ro = run_both(tmp_path, runs, compression=0, target_file_size=64 << 20)
+    assert_identical(rg, ro)
