@@ -150,3 +150,27 @@ def test_worker_refuses_merge_operands(tmp_path):
     job = o.make_job([[p]], str(outd))
     with pytest.raises(RuntimeError):
         o.execute(job)
+
+
+def test_multi_file_level_run(tmp_path):
+    # a level>=1 run = ordered, non-overlapping FILES forming one sorted
+    # stream (VersionSet::MakeInputIterator LevelIterator, version_set.cc:7332)
+    outd = tmp_path / "out"
+    outd.mkdir()
+    level_files = []
+    for part in range(3):
+        kvs = [("k%08d" % i, 100 + i, "V", "L1-%d" % i)
+               for i in range(part * 1000, (part + 1) * 1000)]
+        level_files.append(write_run(tmp_path, "l1_%d.sst" % part, kvs))
+    l0 = write_run(tmp_path, "l0.sst",
+                   [("k%08d" % i, 5000 + i, "V", "L0-%d" % i)
+                    for i in range(0, 3000, 3)])
+    job = o.make_job([[l0], level_files], str(outd))
+    res = o.execute(job)
+    assert res["in_entries"] == 4000
+    assert res["out_entries"] == 3000
+    kvs = []
+    for f in res["files"]:
+        with open(f["path"], "rb") as fh:
+            kvs += o.read_sst(fh.read())
+    assert [v for _, v in kvs[:3]] == [b"L0-0", b"L1-1", b"L1-2"]

@@ -187,3 +187,54 @@ def test_config2_exact_shape(tmp_path):
     runs = gen_runs(tmp_path, 2, 530_000)
     rg, ro = run_both(tmp_path, runs, compression=0, target_file_size=64 << 20)
     assert_identical(rg, ro)
+
+
+def test_multi_file_level_run_gpu(tmp_path):
+    # one run made of several non-overlapping files (a level-N run)
+    import oracle as o
+    level_files = []
+    for part in range(3):
+        kvs = [(b"k%08d" % i, 100 + i, 1, b"L1-%d" % i)
+               for i in range(part * 5000, (part + 1) * 5000)]
+        es = [(o.make_ikey(k, s, t), v) for k, s, t, v in kvs]
+        p = str(tmp_path / ("l1_%d.sst" % part))
+        with open(p, "wb") as f:
+            f.write(o.build_sst(es))
+        level_files.append(p)
+    l0 = gen_runs(tmp_path, 1, 8000, seed0=99)[0]
+    rg, ro = run_both(tmp_path, [l0, level_files])
+    assert_identical(rg, ro)
+
+
+def test_grandparent_cuts_match_oracle(tmp_path):
+    # grandparent-aware file cutting (ShouldStopBefore boundary rules,
+    # compaction_outputs.cc:231-352): GPU worker vs oracle, bit-exact,
+    # including mid-block cuts.  Dense grandparents with huge sizes force
+    # max_compaction_bytes cuts; smaller ones exercise the dynamic-size
+    # rules.
+    runs = gen_runs(tmp_path, 2, 60000)
+    import random
+    rnd = random.Random(5)
+    # grandparent ranges over the 16B uniform key space
+    gps = []
+    lo = b"\x00" * 16
+    for g in range(64):
+        hi = bytes([4 * g + rnd.randrange(1, 4)]) + bytes(
+            rnd.randrange(256) for _ in range(15))
+        if hi <= lo:
+            continue
+        gps.append((lo, hi, rnd.choice([1 << 20, 1 << 26, 1 << 30])))
+        lo = hi + b"\x01"
+        lo = lo[:16]
+    for mcb in [1 << 30, 8 << 30]:
+        og = tmp_path / ("g%d" % (mcb >> 30))
+        oo = tmp_path / ("o%d" % (mcb >> 30))
+        og.mkdir()
+        oo.mkdir()
+        jd = dcw.make_job(runs, str(og), target_file_size=2 << 20,
+                          max_compaction_bytes=mcb, grandparents=gps)
+        jo = oracle.make_job(runs, str(oo), target_file_size=2 << 20,
+                             max_compaction_bytes=mcb, grandparents=gps)
+        rg = dcw.execute(jd)
+        ro = oracle.execute(jo)
+        assert_identical(rg, ro)

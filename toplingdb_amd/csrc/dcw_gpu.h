@@ -88,6 +88,10 @@ class GpuJob {
   int gather_entries(uint64_t first, uint32_t count,
                      std::vector<std::pair<std::string, std::string>>* kvs,
                      std::string* err);
+  // per-survivor grandparent boundary positions (host file-cut FSM input);
+  // gp user keys must match the job's uniform ukey length
+  int gp_positions(const dcw_job_desc* d, std::vector<uint32_t>* pos,
+                   std::vector<uint8_t>* nback, std::string* err);
   // min/max sequence + tombstone count over survivor range (post zeroing)
   int seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* mx,
                  uint64_t* n_tombstones, std::string* err);

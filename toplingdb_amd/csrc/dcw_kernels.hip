@@ -1068,6 +1068,62 @@ __global__ void k_gather_range(const uint64_t* __restrict__ s_k0,
   }
 }
 
+// Grandparent boundary positions (CompactionOutputs::
+// UpdateGrandparentBoundaryInfo, compaction_outputs.cc:121-230, recast as a
+// pure per-key function): for survivor user key u,
+//   A(u) = #files with smallest <= u           (enter transitions)
+//   B(u) = #files the walk has fully left:
+//          #{i: largest_i < u} + the prefix of the equal-largest run at u
+//          whose tie_next flag is set (tie_next_i = smallest_{i+1} ==
+//          largest_i, the multi-file same-user-key case, :157-166)
+//   pos(u) = A(u)+B(u); odd pos = inside file B(u); nback(u) = #files j <
+//   B(u) with largest_j == u (GetCurrentKeyGrandparentOverlappedBytes's
+//   backward tie loop, :218-227).
+// The host file-cut FSM consumes pos/nback per survivor.
+__global__ void k_gp_positions(const uint64_t* __restrict__ s_k0,
+                               const uint64_t* __restrict__ s_k1, uint64_t n,
+                               const uint64_t* __restrict__ gsm0,
+                               const uint64_t* __restrict__ gsm1,
+                               const uint64_t* __restrict__ glg0,
+                               const uint64_t* __restrict__ glg1,
+                               const uint8_t* __restrict__ tie_next,
+                               uint32_t ng, uint32_t* __restrict__ pos_out,
+                               uint8_t* __restrict__ nback_out) {
+  for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    uint64_t k0 = s_k0[i], k1 = s_k1[i];
+    // A = upper_bound over smallest (count sm <= u)
+    uint32_t lo = 0, hi = ng;
+    while (lo < hi) {
+      uint32_t mid = (lo + hi) / 2;
+      bool sm_le = gsm0[mid] < k0 || (gsm0[mid] == k0 && gsm1[mid] <= k1);
+      if (sm_le)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    uint32_t A = lo;
+    // lb = lower_bound over largest (first lg >= u)
+    lo = 0;
+    hi = ng;
+    while (lo < hi) {
+      uint32_t mid = (lo + hi) / 2;
+      bool lg_lt = glg0[mid] < k0 || (glg0[mid] == k0 && glg1[mid] < k1);
+      if (lg_lt)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    uint32_t B = lo;
+    uint32_t lb = lo;
+    while (B < ng && glg0[B] == k0 && glg1[B] == k1 && tie_next[B]) B++;
+    uint32_t nb = 0;
+    if (B > lb && glg0[lb] == k0 && glg1[lb] == k1) nb = B - lb;
+    pos_out[i] = A + B;
+    nback_out[i] = (uint8_t)(nb > 255 ? 255 : nb);
+  }
+}
+
 __global__ void k_build_headidx(const uint8_t* __restrict__ head,
                                 const uint32_t* __restrict__ pos, uint64_t n,
                                 uint64_t* __restrict__ headidx) {
@@ -1241,6 +1297,9 @@ struct GpuJob::Impl {
   void* d_scratch_recoff = nullptr;
   void* d_scratch_gather = nullptr;
   void* d_scratch_mm = nullptr;
+  void *d_gp_sm0 = nullptr, *d_gp_sm1 = nullptr, *d_gp_lg0 = nullptr,
+       *d_gp_lg1 = nullptr, *d_gp_tie = nullptr, *d_gp_pos = nullptr,
+       *d_gp_nback = nullptr;
   uint8_t* d_out_img = nullptr;
   size_t out_img_cap = 0;
 
@@ -1331,6 +1390,8 @@ GpuJob::~GpuJob() {
   F(p->d_scratch64); F(p->d_out_img); F(p->d_outoff);
   F(p->d_scratch_keys); F(p->d_scratch_recoff); F(p->d_scratch_gather);
   F(p->d_scratch_mm);
+  F(p->d_gp_sm0); F(p->d_gp_sm1); F(p->d_gp_lg0); F(p->d_gp_lg1);
+  F(p->d_gp_tie); F(p->d_gp_pos); F(p->d_gp_nback);
   if (p->stream) (void)hipStreamDestroy(p->stream);
   delete p;
 }
@@ -1992,6 +2053,53 @@ int GpuJob::gather_entries(uint64_t first, uint32_t count,
     kvs->emplace_back(std::string((const char*)r + 1, klen),
                       std::string((const char*)r + 5 + klen, vl));
   }
+  return 0;
+}
+
+int GpuJob::gp_positions(const dcw_job_desc* d, std::vector<uint32_t>* pos,
+                         std::vector<uint8_t>* nback, std::string* err) {
+  Impl* p = p_;
+  uint32_t ng = d->num_grandparents;
+  uint64_t n = p->n_surv;
+  pos->resize(n);
+  nback->resize(n);
+  if (ng == 0 || n == 0) return 0;
+  std::vector<uint64_t> sm0(ng), sm1(ng), lg0(ng), lg1(ng);
+  std::vector<uint8_t> tie(ng, 0);
+  for (uint32_t g = 0; g < ng; g++) {
+    const dcw_grandparent& f = d->grandparents[g];
+    if (f.smallest_len != ukey_len || f.largest_len != ukey_len) {
+      if (err) *err = "grandparent user-key length differs from job keys";
+      return -1;
+    }
+    uint64_t c;
+    make_normkey(f.smallest_ukey, f.smallest_len, 0, &sm0[g], &sm1[g], &c);
+    make_normkey(f.largest_ukey, f.largest_len, 0, &lg0[g], &lg1[g], &c);
+  }
+  for (uint32_t g = 0; g + 1 < ng; g++)
+    tie[g] = (lg0[g] == sm0[g + 1] && lg1[g] == sm1[g + 1]) ? 1 : 0;
+  HIPCHK(p->ens(&p->d_gp_sm0, ng * 8));
+  HIPCHK(p->ens(&p->d_gp_sm1, ng * 8));
+  HIPCHK(p->ens(&p->d_gp_lg0, ng * 8));
+  HIPCHK(p->ens(&p->d_gp_lg1, ng * 8));
+  HIPCHK(p->ens(&p->d_gp_tie, ng));
+  HIPCHK(p->ens(&p->d_gp_pos, n * 4));
+  HIPCHK(p->ens(&p->d_gp_nback, n));
+  HIPCHK(hipMemcpy(p->d_gp_sm0, sm0.data(), ng * 8, hipMemcpyHostToDevice));
+  HIPCHK(hipMemcpy(p->d_gp_sm1, sm1.data(), ng * 8, hipMemcpyHostToDevice));
+  HIPCHK(hipMemcpy(p->d_gp_lg0, lg0.data(), ng * 8, hipMemcpyHostToDevice));
+  HIPCHK(hipMemcpy(p->d_gp_lg1, lg1.data(), ng * 8, hipMemcpyHostToDevice));
+  HIPCHK(hipMemcpy(p->d_gp_tie, tie.data(), ng, hipMemcpyHostToDevice));
+  hipLaunchKernelGGL(k_gp_positions, dim3(grid_for(n)), dim3(256), 0, p->stream,
+                     p->d_sk0, p->d_sk1, n, (const uint64_t*)p->d_gp_sm0,
+                     (const uint64_t*)p->d_gp_sm1, (const uint64_t*)p->d_gp_lg0,
+                     (const uint64_t*)p->d_gp_lg1, (const uint8_t*)p->d_gp_tie,
+                     ng, (uint32_t*)p->d_gp_pos, (uint8_t*)p->d_gp_nback);
+  HIPCHK(hipMemcpyAsync(pos->data(), p->d_gp_pos, n * 4, hipMemcpyDeviceToHost,
+                        p->stream));
+  HIPCHK(hipMemcpyAsync(nback->data(), p->d_gp_nback, n, hipMemcpyDeviceToHost,
+                        p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
   return 0;
 }
 

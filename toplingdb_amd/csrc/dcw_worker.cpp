@@ -276,8 +276,6 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     return fail(res, 12, "unsupported comparator (bytewise only)");
   if (d->compression > 1)
     return fail(res, 13, "unsupported compression (none/snappy only)");
-  if (d->num_grandparents > 0)
-    return fail(res, 14, "grandparent-aware file cuts not yet in the GPU worker envelope");
 
   uint64_t t_start = now_usec();
   std::string err;
@@ -319,6 +317,31 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   const auto& vlen = job.plan_vlen();
   PlanIn pin{job.plan_shared().data(), klen.data(), vlen.data(), nullptr, nsurv};
 
+  // grandparent-aware file cutting (compaction_outputs.cc:231-352): per-
+  // survivor boundary positions from the GPU + host FSM state
+  const bool use_gp = d->num_grandparents > 0;
+  std::vector<uint32_t> gp_pos;
+  std::vector<uint8_t> gp_nback;
+  std::vector<uint64_t> gp_psum(d->num_grandparents + 1, 0);
+  if (use_gp) {
+    if (job.gp_positions(d, &gp_pos, &gp_nback, &err) != 0) return fail(res, 27, err);
+    for (uint32_t g = 0; g < d->num_grandparents; g++)
+      gp_psum[g + 1] = gp_psum[g] + d->grandparents[g].file_size;
+  }
+  struct GpWalk {
+    bool seen_key = false;
+    uint64_t last_pos = 0;
+    uint64_t overlapped = 0;   // grandparent_overlapped_bytes_
+    uint64_t switched = 0;     // grandparent_boundary_switched_num_
+  } gpw;
+  auto gp_cur_overlap = [&](uint64_t e) -> uint64_t {
+    uint32_t p = gp_pos[e];
+    if ((p & 1) == 0) return 0; // in gap
+    uint32_t idx = (p - 1) / 2;
+    uint32_t nb = gp_nback[e];
+    return gp_psum[idx + 1] - gp_psum[idx - nb];
+  };
+
   TableOpts base = opts_from_desc(d);
   uint64_t next_file_number = d->next_file_number;
   std::vector<dcw_output_file> out_files;
@@ -339,6 +362,8 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     size_t cur = s;
     bool cut = false;
     uint64_t cut_entry = 0;
+    uint64_t partial_first = 0; // mid-block remainder [partial_first, cut_entry)
+    uint32_t partial_count = 0;
     // chunked: plan -> GPU emit -> cut walk on csizes -> fetch only what the
     // file keeps
     while (!cut && cur < nsurv) {
@@ -357,28 +382,96 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       std::vector<uint32_t> csizes;
       if (job.emit_blocks(blocks, o, &csizes, &err, &eoff_plan) != 0)
         return fail(res, 21, err);
-      // walk csizes first: find the flush that crosses the target
-      // (cut = first entry after it; the open block then holds exactly one
-      // entry — AddToOutput/FileSize interplay, compaction_outputs.cc:356-420)
+      // walk csizes: find the cut (ShouldStopBefore semantics,
+      // compaction_outputs.cc:231-352).  Without grandparents the file size
+      // only changes at block flushes, so the walk is block-level and the
+      // cut lands one entry after the crossing flush (the open block then
+      // holds exactly one entry).  With grandparents every entry is checked
+      // against the boundary-crossing rules and the cut may land mid-block.
       uint64_t old = image.len;
-      size_t take = blocks.size();
+      size_t take = blocks.size();      // full blocks kept by this file
       {
-        uint64_t off = old;
         uint64_t unc_sum = 0, c_sum = 0;
+        for (auto& pb : blocks) unc_sum += pb.unc_size;
+        for (auto cz : csizes) c_sum += cz;
+        if (o.compression == 1 && unc_sum > 0)
+          comp_ratio = 0.5 * comp_ratio + 0.5 * ((double)c_sum / unc_sum);
+      }
+      if (!use_gp) {
+        uint64_t off = old;
         for (size_t b = 0; b < blocks.size(); b++) {
           off += csizes[b] + kTrailerSize;
-          unc_sum += blocks[b].unc_size;
-          c_sum += csizes[b];
           bool more_entries = blocks[b].first + blocks[b].count < nsurv;
           if (off >= d->target_file_size && more_entries) {
             take = b + 1;
             cut = true;
-            cut_entry = blocks[b].first + blocks[b].count;
+            partial_first = blocks[b].first + blocks[b].count; // e0
+            partial_count = 1;
+            cut_entry = partial_first + 1; // next file starts after e0
             break;
           }
         }
-        if (o.compression == 1 && unc_sum > 0)
-          comp_ratio = 0.5 * comp_ratio + 0.5 * ((double)c_sum / unc_sum);
+      } else {
+        uint64_t off = old;
+        bool dyn = d->level_compaction_dynamic_file_size != 0;
+        for (size_t b = 0; b < blocks.size() && !cut; b++) {
+          for (uint32_t li = 0; li < blocks[b].count; li++) {
+            uint64_t e = blocks[b].first + li;
+            // (1) ShouldStopBefore(e): grandparent update, then cut rules
+            uint32_t newpos = gp_pos[e];
+            uint64_t crossings = 0, prev_overlapped = gpw.overlapped;
+            if (gpw.seen_key) {
+              crossings = newpos - gpw.last_pos;
+              if (newpos >= 1 && crossings > 0) {
+                uint64_t f_lo = (gpw.last_pos + 1) / 2; // files entered in span
+                uint64_t f_hi = (newpos - 1) / 2;
+                if (f_hi >= f_lo && f_lo < d->num_grandparents) {
+                  if (f_hi >= d->num_grandparents) f_hi = d->num_grandparents - 1;
+                  gpw.overlapped += gp_psum[f_hi + 1] - gp_psum[f_lo];
+                }
+              }
+              gpw.switched += crossings;
+            } else if (newpos & 1) {
+              gpw.overlapped = gp_cur_overlap(e); // first key lands mid-file
+            }
+            gpw.seen_key = true;
+            gpw.last_pos = newpos;
+            bool has_builder = e != s; // builder empty before the file's first Add
+            if (has_builder && e < nsurv) {
+              bool docut = false;
+              if (off >= d->target_file_size) {
+                docut = true;
+              } else if (crossings > 0) {
+                if (gpw.overlapped + off > d->max_compaction_bytes) {
+                  docut = true;
+                } else {
+                  bool in_gap = (newpos & 1) == 0;
+                  uint64_t skippable = in_gap ? 2 : 3;
+                  uint64_t sw5 = gpw.switched * 5 < 40 ? gpw.switched * 5 : 40;
+                  if (dyn && crossings >= skippable &&
+                      gpw.overlapped - prev_overlapped > d->target_file_size / 8)
+                    docut = true;
+                  else if (dyn && off >= ((d->target_file_size + 99) / 100) *
+                                             (50 + sw5))
+                    docut = true;
+                }
+              }
+              if (docut) {
+                cut = true;
+                cut_entry = e;           // next file starts AT e
+                take = b;                // full blocks before the current one
+                partial_first = blocks[b].first;
+                partial_count = li;      // may be 0 (cut at a block boundary)
+                // close-file resets (compaction_outputs.cc:374-380)
+                gpw.switched = 0;
+                gpw.overlapped = gp_cur_overlap(e);
+                break;
+              }
+            }
+            // (2) Add(e): the flush of the previous block happens here
+            if (li == 0 && b > 0) off += csizes[b - 1] + kTrailerSize;
+          }
+        }
       }
       std::vector<uint64_t> outoff(take);
       uint64_t acc = 0;
@@ -419,16 +512,21 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       cur = take ? blocks[take - 1].first + blocks[take - 1].count : cur;
       if (take < blocks.size()) break; // cut decided inside this chunk
     }
-    // partial single-entry block after a size cut
+    // partial block after a cut (1 entry for pure size cuts; up to a full
+    // block's worth for grandparent-rule cuts)
     uint64_t file_first = s;
-    uint64_t file_count = cur - s;
-    if (cut) {
+    if (cut && partial_count > 0) {
       std::vector<std::pair<std::string, std::string>> kvs;
-      if (job.gather_entries(cut_entry, 1, &kvs, &err) != 0) return fail(res, 24, err);
+      if (job.gather_entries(partial_first, partial_count, &kvs, &err) != 0)
+        return fail(res, 24, err);
       BlockBuilder bb(o.block_restart_interval, false);
-      bb.AddWithLastKey((const uint8_t*)kvs[0].first.data(), kvs[0].first.size(),
-                        (const uint8_t*)kvs[0].second.data(), kvs[0].second.size(),
-                        nullptr, 0);
+      std::string lastk;
+      for (auto& kv : kvs) {
+        bb.AddWithLastKey((const uint8_t*)kv.first.data(), kv.first.size(),
+                          (const uint8_t*)kv.second.data(), kv.second.size(),
+                          (const uint8_t*)lastk.data(), lastk.size());
+        lastk = kv.first;
+      }
       std::string contents = bb.Finish();
       std::string pblk;
       SstIndexEntry h = append_block(pblk, o, (const uint8_t*)contents.data(),
@@ -436,12 +534,12 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       h.off += image.len;
       image.append(pblk.data(), pblk.size());
       handles.push_back(h);
-      first_keys.push_back(kvs[0].first);
-      last_keys.push_back(kvs[0].first);
-      block_counts.push_back(1);
-      file_count += 1;
-      cur = cut_entry + 1;
+      first_keys.push_back(kvs.front().first);
+      last_keys.push_back(kvs.back().first);
+      block_counts.push_back(partial_count);
     }
+    if (cut) cur = cut_entry;
+    uint64_t file_count = cur - s;
     if (handles.empty()) break; // nothing left
     uint64_t tw1 = now_usec();
     uint64_t mn_seq = 0, mx_seq = 0, n_tomb = 0;
