@@ -194,7 +194,8 @@ def test_multi_file_level_run_gpu(tmp_path):
     import oracle as o
     level_files = []
     for part in range(3):
-        kvs = [(b"k%08d" % i, 100 + i, 1, b"L1-%d" % i)
+        # 16-byte user keys (must match the generator's uniform key length)
+        kvs = [(b"k%015d" % i, 100 + i, 1, b"L1-%d" % i)
                for i in range(part * 5000, (part + 1) * 5000)]
         es = [(o.make_ikey(k, s, t), v) for k, s, t, v in kvs]
         p = str(tmp_path / ("l1_%d.sst" % part))

@@ -199,6 +199,21 @@ def make_job(runs, output_dir, **kw) -> JobDesc:
     d.db_host_id = kw.pop("db_host_id", "dcw-host").encode()
     d.current_time = kw.pop("current_time", 1757900000)
     d.oldest_ancester_time = kw.pop("oldest_ancester_time", 1757800000)
+    gps = kw.pop("grandparents", [])
+    if gps:
+        garr = (Grandparent * len(gps))()
+        for i, (sm, lg, fsz) in enumerate(gps):
+            smb = (C.c_uint8 * len(sm)).from_buffer_copy(sm)
+            lgb = (C.c_uint8 * len(lg)).from_buffer_copy(lg)
+            d._keep += [smb, lgb]
+            garr[i].smallest_ukey = smb
+            garr[i].smallest_len = len(sm)
+            garr[i].largest_ukey = lgb
+            garr[i].largest_len = len(lg)
+            garr[i].file_size = fsz
+        d._keep.append(garr)
+        d.grandparents = garr
+        d.num_grandparents = len(gps)
     lvls = kw.pop("levels_below", [])
     d.levels_below_valid = kw.pop("levels_below_valid", 1)
     larr = (LevelFiles * max(len(lvls), 1))()
