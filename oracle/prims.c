@@ -332,7 +332,7 @@ size_t orc_snappy_compress(const uint8_t* in, size_t n, uint8_t* out) {
   uint8_t* op = out;
   op += orc_varint32_put(op, (uint32_t)n);
   if (n == 0) return (size_t)(op - out);
-  enum { HBITS = 12 }; /* 4 KiB-block-sized table; also the GPU encoder's LDS table */
+  enum { HBITS = 11 }; /* spec v3: 2 KiB-entry table (GPU LDS residency) */
   static const uint32_t HMUL = 0x1e35a7bdu;
   uint32_t* tab = (uint32_t*)malloc(sizeof(uint32_t) << HBITS);
   memset(tab, 0xff, sizeof(uint32_t) << HBITS);

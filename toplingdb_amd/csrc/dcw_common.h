@@ -324,7 +324,7 @@ DCW_HD uint32_t block_checksum(uint32_t type, const Crc32cTables* crc_tt,
 }
 
 // ---------- DCW-deterministic snappy codec (spec at file top) ----------
-static const int kSnapHashBits = 12;
+static const int kSnapHashBits = 11; // spec v3
 static const uint32_t kSnapHashMul = 0x1e35a7bdu;
 
 DCW_HD size_t snappy_max_compressed(size_t n) { return 32 + n + n / 6; }

@@ -897,41 +897,73 @@ __device__ __forceinline__ uint8_t* snap_emit_literal4(uint8_t* op,
   for (; t < len; t++) op[t] = lit[t];
   return op + len;
 }
-__device__ static size_t snap_encode_lds(const uint8_t* in, size_t n,
+// Exact-output encoder with a 2-position software pipeline: on the common
+// miss path two probe positions' loads and table reads overlap their LDS
+// latencies.  Semantics identical to the sequential spec: the second
+// probe's candidate accounts for the first probe's table insert
+// (cand2 = pz1 when the hashes collide), and speculation is discarded on a
+// match.
+__device__ static size_t snap_encode_lds(const uint8_t* in, uint32_t n,
                                          uint8_t* out, uint16_t* tab) {
   uint8_t* op = out;
-  op += varint32_put(op, (uint32_t)n);
+  op += varint32_put(op, n);
   if (n == 0) return (size_t)(op - out);
-  size_t lit_start = 0, pz = 0;
+  uint32_t lit_start = 0, pz = 0;
   uint32_t skip = 32; // spec v2: snappy-style skip acceleration
   while (pz + 4 <= n) {
-    uint32_t w = load32(in + pz);
-    uint32_t h = (w * kSnapHashMul) >> (32 - kSnapHashBits);
-    uint32_t cand = tab[h];
-    tab[h] = (uint16_t)pz;
-    if (cand != 0xffffu && cand < pz && load32(in + cand) == w) {
-      size_t mlen = 4;
-      while (pz + mlen + 4 <= n) {
-        uint32_t a = load32(in + cand + mlen);
-        uint32_t bz = load32(in + pz + mlen);
-        uint32_t x = a ^ bz;
-        if (x) {
-          mlen += __builtin_ctz(x) >> 3;
-          goto done_ext;
-        }
-        mlen += 4;
-      }
-      while (pz + mlen < n && in[cand + mlen] == in[pz + mlen]) mlen++;
-    done_ext:
-      op = snap_emit_literal4(op, in + lit_start, pz - lit_start);
-      op = snap_emit_copy(op, pz - cand, mlen);
-      pz += mlen;
-      lit_start = pz;
-      skip = 32;
+    uint32_t pz2 = pz + (skip >> 5);
+    bool have2 = pz2 + 4 <= n;
+    uint32_t w1 = load32(in + pz);
+    uint32_t w2 = have2 ? load32(in + pz2) : 0;
+    uint32_t h1 = (w1 * kSnapHashMul) >> (32 - kSnapHashBits);
+    uint32_t h2 = (w2 * kSnapHashMul) >> (32 - kSnapHashBits);
+    uint32_t cand1 = tab[h1];
+    uint32_t cand2_raw = have2 ? tab[h2] : 0xffffu;
+    tab[h1] = (uint16_t)pz;
+    uint32_t mpos, mcand, w;
+    if (cand1 != 0xffffu && cand1 < pz && load32(in + cand1) == w1) {
+      mpos = pz;
+      mcand = cand1;
+      w = w1;
     } else {
-      pz += skip >> 5;
+      // miss at pz
       skip++;
+      if (!have2) {
+        pz = pz2;
+        break;
+      }
+      uint32_t cand2 = (h2 == h1) ? pz : cand2_raw;
+      tab[h2] = (uint16_t)pz2;
+      if (cand2 != 0xffffu && cand2 < pz2 && load32(in + cand2) == w2) {
+        mpos = pz2;
+        mcand = cand2;
+        w = w2;
+      } else {
+        pz = pz2 + (skip >> 5);
+        skip++;
+        continue;
+      }
     }
+    // match at mpos against mcand
+    (void)w;
+    uint32_t mlen = 4;
+    while (mpos + mlen + 4 <= n) {
+      uint32_t a = load32(in + mcand + mlen);
+      uint32_t bz = load32(in + mpos + mlen);
+      uint32_t x = a ^ bz;
+      if (x) {
+        mlen += __builtin_ctz(x) >> 3;
+        goto done_ext;
+      }
+      mlen += 4;
+    }
+    while (mpos + mlen < n && in[mcand + mlen] == in[mpos + mlen]) mlen++;
+  done_ext:
+    op = snap_emit_literal4(op, in + lit_start, mpos - lit_start);
+    op = snap_emit_copy(op, mpos - mcand, mlen);
+    pz = mpos + mlen;
+    lit_start = pz;
+    skip = 32;
   }
   op = snap_emit_literal4(op, in + lit_start, n - lit_start);
   return (size_t)(op - out);
