@@ -790,7 +790,11 @@ struct EmitBlockDesc {
   uint64_t uout;      // offset into ucblob
 };
 
-// one workgroup per block; thread t encodes entries t, t+bd, ...
+// one workgroup per block; thread 0 walks the block's entry sizes into an
+// LDS offset table (<=512 entries per 4 KiB-class block), then all threads
+// encode their entries.  s_shared is the adjacent-survivor prefix from
+// k_shared_prefix (shared forced 0 at restart points).
+#define EMIT_MAX_ENTRIES 512
 __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
                        const uint64_t* __restrict__ s_k0,
                        const uint64_t* __restrict__ s_k1,
@@ -798,29 +802,42 @@ __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
                        const uint64_t* __restrict__ s_voff,
                        const uint32_t* __restrict__ s_vlen,
                        const uint8_t* __restrict__ s_klen,
-                       const uint32_t* __restrict__ eoff, // per-entry in-block offset
+                       const uint8_t* __restrict__ s_shared,
                        const uint8_t* __restrict__ ublob,
-                       uint8_t* __restrict__ ucblob, uint32_t restart_interval) {
+                       uint8_t* __restrict__ ucblob, uint32_t restart_interval,
+                       uint32_t* err_flag) {
+  __shared__ uint32_t offs[EMIT_MAX_ENTRIES];
   for (uint32_t b = blockIdx.x; b < nblocks; b += gridDim.x) {
     EmitBlockDesc d = bds[b];
     uint8_t* out = ucblob + d.uout;
+    if (d.count > EMIT_MAX_ENTRIES) {
+      if (threadIdx.x == 0) set_err(err_flag, DE_BLOCK_PARSE);
+      __syncthreads();
+      continue;
+    }
+    if (threadIdx.x == 0) {
+      uint32_t off = 0;
+      for (uint32_t li = 0; li < d.count; li++) {
+        uint32_t i = d.first + li;
+        offs[li] = off;
+        uint32_t shared =
+            (li % restart_interval == 0) ? 0 : s_shared[i];
+        uint32_t klen = s_klen[i];
+        uint32_t vl = s_vlen[i];
+        off += varint_len(shared) + varint_len(klen - shared) +
+               varint_len(vl) + (klen - shared) + vl;
+      }
+    }
+    __syncthreads();
     for (uint32_t li = threadIdx.x; li < d.count; li += blockDim.x) {
       uint32_t i = d.first + li;
       uint32_t klen = s_klen[i];
       uint8_t key[24];
       build_ikey(s_k0[i], s_k1[i], s_tag[i], klen, key);
-      uint32_t shared = 0;
-      if (li % restart_interval != 0) {
-        // shared prefix with previous survivor == previous entry in block
-        uint32_t kp = s_klen[i - 1];
-        uint8_t prev[24];
-        build_ikey(s_k0[i - 1], s_k1[i - 1], s_tag[i - 1], kp, prev);
-        uint32_t m = kp < klen ? kp : klen;
-        while (shared < m && prev[shared] == key[shared]) shared++;
-      }
+      uint32_t shared = (li % restart_interval == 0) ? 0 : s_shared[i];
       uint32_t non_shared = klen - shared;
       uint32_t vl = s_vlen[i];
-      uint8_t* p = out + eoff[i];
+      uint8_t* p = out + offs[li];
       p += varint32_put(p, shared);
       p += varint32_put(p, non_shared);
       p += varint32_put(p, vl);
@@ -836,12 +853,13 @@ __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
       uint32_t r0 = 0;
       memcpy(tail, &r0, 4);
       for (uint32_t j = 1; j < d.num_restarts; j++) {
-        uint32_t off = eoff[d.first + j * restart_interval];
+        uint32_t off = offs[j * restart_interval];
         memcpy(tail + 4 * j, &off, 4);
       }
       uint32_t footer = d.num_restarts; // kDataBlockBinarySearch
       memcpy(tail + 4 * d.num_restarts, &footer, 4);
     }
+    __syncthreads();
   }
 }
 
@@ -1100,6 +1118,53 @@ __global__ void k_gather_range(const uint64_t* __restrict__ s_k0,
   }
 }
 
+// Per-survivor block planning: for every survivor e, simulate the
+// BlockBuilder flush-policy FSM (flush_block_policy.cc:37-52 +
+// block_builder.cc estimate accounting) for a block STARTING at e, and
+// record where the next block would start plus the block's size/restarts.
+// The host then walks the chain from any position — block boundaries after
+// a file cut come for free (every entry is a potential block start).
+__global__ void k_plan_next(const uint8_t* __restrict__ s_shared,
+                            const uint8_t* __restrict__ s_klen,
+                            const uint32_t* __restrict__ s_vlen, uint64_t n,
+                            uint32_t block_size, uint32_t restart_interval,
+                            uint64_t dev_limit, uint32_t* __restrict__ next_out,
+                            uint32_t* __restrict__ meta_out) {
+  for (uint64_t e = blockIdx.x * blockDim.x + threadIdx.x; e < n;
+       e += (uint64_t)gridDim.x * blockDim.x) {
+    uint64_t i = e;
+    uint64_t bytes = 0;
+    uint32_t nrestarts = 1, counter = 0;
+    while (i < n) {
+      uint32_t klen = s_klen[i], vlen = s_vlen[i];
+      uint64_t curr = 8 + bytes + 4 * (nrestarts - 1);
+      if (bytes > 0) {
+        if (curr >= block_size) break;
+        uint64_t after = curr + klen + vlen + 4 + varint_len(klen) +
+                         varint_len(vlen) +
+                         (counter >= restart_interval ? 4 : 0);
+        if (after > block_size && curr > dev_limit && dev_limit != 0) break;
+      }
+      uint32_t shared = s_shared[i];
+      if (counter >= restart_interval) {
+        nrestarts++;
+        counter = 0;
+        shared = 0;
+      } else if (bytes == 0) {
+        shared = 0;
+      }
+      uint32_t non_shared = klen - shared;
+      bytes += varint_len(shared) + varint_len(non_shared) + varint_len(vlen) +
+               non_shared + vlen;
+      counter++;
+      i++;
+    }
+    next_out[e] = (uint32_t)i;
+    uint32_t unc = (uint32_t)(bytes + 4 * nrestarts + 4);
+    meta_out[e] = (unc & 0xffffffu) | (nrestarts << 24); // unc<=~4.3k, nr<=255
+  }
+}
+
 // Grandparent boundary positions (CompactionOutputs::
 // UpdateGrandparentBoundaryInfo, compaction_outputs.cc:121-230, recast as a
 // pure per-key function): for survivor user key u,
@@ -1332,6 +1397,10 @@ struct GpuJob::Impl {
   void *d_gp_sm0 = nullptr, *d_gp_sm1 = nullptr, *d_gp_lg0 = nullptr,
        *d_gp_lg1 = nullptr, *d_gp_tie = nullptr, *d_gp_pos = nullptr,
        *d_gp_nback = nullptr;
+  void *d_plan_next = nullptr, *d_plan_meta = nullptr;
+  void* h_plan = nullptr; // pinned host landing for next+meta
+  size_t h_plan_cap = 0;
+  bool h_plan_pageable = false;
   uint8_t* d_out_img = nullptr;
   size_t out_img_cap = 0;
 
@@ -1424,6 +1493,13 @@ GpuJob::~GpuJob() {
   F(p->d_scratch_mm);
   F(p->d_gp_sm0); F(p->d_gp_sm1); F(p->d_gp_lg0); F(p->d_gp_lg1);
   F(p->d_gp_tie); F(p->d_gp_pos); F(p->d_gp_nback);
+  F(p->d_plan_next); F(p->d_plan_meta);
+  if (p->h_plan) {
+    if (p->h_plan_pageable)
+      free(p->h_plan);
+    else
+      (void)hipHostFree(p->h_plan);
+  }
   if (p->stream) (void)hipStreamDestroy(p->stream);
   delete p;
 }
@@ -1892,9 +1968,48 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   return 0;
 }
 
+int GpuJob::plan_all(const TableOpts& o, const uint32_t** next,
+                     const uint32_t** meta, std::string* err) {
+  Impl* p = p_;
+  uint64_t n = p->n_surv;
+  // pinned host landing buffers (grow-only)
+  if (p->h_plan_cap < n * 8) {
+    if (p->h_plan) (void)hipHostFree(p->h_plan);
+    p->h_plan = nullptr;
+    size_t c = n * 8 + (n * 8) / 4 + 64;
+    if (hipHostMalloc(&p->h_plan, c, hipHostMallocDefault) != hipSuccess) {
+      p->h_plan = malloc(c); // pageable fallback, still correct
+      p->h_plan_pageable = true;
+    } else {
+      p->h_plan_pageable = false;
+    }
+    p->h_plan_cap = c;
+  }
+  uint32_t* h_next = (uint32_t*)p->h_plan;
+  uint32_t* h_meta = h_next + n;
+  HIPCHK(p->ens(&p->d_plan_next, n * 4 + 4));
+  HIPCHK(p->ens(&p->d_plan_meta, n * 4 + 4));
+  uint64_t dev_limit =
+      ((o.block_size * (100 - o.block_size_deviation)) + 99) / 100;
+  p->kbegin("plan_next", 40.0 * n);
+  hipLaunchKernelGGL(k_plan_next, dim3(grid_for(n)), dim3(256), 0, p->stream,
+                     p->d_sshared, p->d_sklen, p->d_svlen, n, o.block_size,
+                     o.block_restart_interval, dev_limit,
+                     (uint32_t*)p->d_plan_next, (uint32_t*)p->d_plan_meta);
+  p->kend();
+  HIPCHK(hipMemcpyAsync(h_next, p->d_plan_next, n * 4, hipMemcpyDeviceToHost,
+                        p->stream));
+  HIPCHK(hipMemcpyAsync(h_meta, p->d_plan_meta, n * 4, hipMemcpyDeviceToHost,
+                        p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  p->kresolve();
+  *next = h_next;
+  *meta = h_meta;
+  return 0;
+}
+
 int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts& o,
-                        std::vector<uint32_t>* comp_sizes, std::string* err,
-                        const std::vector<uint32_t>* eoff_in) {
+                        std::vector<uint32_t>* comp_sizes, std::string* err) {
   Impl* p = p_;
   hipEvent_t t0, t1;
   hipEventCreate(&t0);
@@ -1909,25 +2024,7 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   uint64_t first = blocks.front().first;
   uint64_t last = blocks.back().first + blocks.back().count;
   uint64_t nent = last - first;
-  std::vector<uint32_t> eoff_local;
-  const std::vector<uint32_t>* eoff = eoff_in;
-  if (!eoff) {
-    eoff_local.resize(nent);
-    for (uint32_t b = 0; b < nb; b++) {
-      const PlannedBlock& pb = blocks[b];
-      uint32_t off = 0;
-      for (uint32_t li = 0; li < pb.count; li++) {
-        uint64_t i = pb.first + li;
-        uint32_t shared = (li % o.block_restart_interval == 0) ? 0 : h_shared_[i];
-        uint32_t klen = h_klen_[i];
-        uint32_t vl = h_vlen_[i];
-        eoff_local[i - first] = off;
-        off += varint_len(shared) + varint_len(klen - shared) + varint_len(vl) +
-               (klen - shared) + vl;
-      }
-    }
-    eoff = &eoff_local;
-  }
+  (void)nent;
   std::vector<EmitBlockDesc> bds(nb);
   uint64_t uout = 0;
   for (uint32_t b = 0; b < nb; b++) {
@@ -1938,21 +2035,17 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   p->emit_base = first;
   p->emit_nblocks = nb;
   ENSURE(p->d_bds, p->bds_cap, sizeof(EmitBlockDesc) * nb);
-  ENSURE(p->d_eoff, p->eoff_cap, sizeof(uint32_t) * nent);
   ENSURE(p->d_ucblob, p->ucblob_cap, uout);
   ENSURE(p->d_ebsize, p->eb_cap, sizeof(uint32_t) * nb * 2 + nb); // bsize+csum+btype
   p->d_ecsum = p->d_ebsize + nb;
   p->d_ebtype = (uint8_t*)(p->d_ecsum + nb);
   HIPCHK(hipMemcpy(p->d_bds, bds.data(), sizeof(EmitBlockDesc) * nb,
                    hipMemcpyHostToDevice));
-  HIPCHK(hipMemcpy(p->d_eoff, eoff->data(), sizeof(uint32_t) * nent,
-                   hipMemcpyHostToDevice));
-  // NOTE: k_emit indexes eoff by absolute survivor index minus chunk base.
   p->kbegin("emit", 2.0 * (double)uout);
   hipLaunchKernelGGL(k_emit, dim3(nb < 4096 ? nb : 4096), dim3(256), 0, p->stream,
                      p->d_bds, nb, p->d_sk0, p->d_sk1, p->d_stag, p->d_svoff,
-                     p->d_svlen, p->d_sklen, p->d_eoff - first, p->d_ublob,
-                     p->d_ucblob, o.block_restart_interval);
+                     p->d_svlen, p->d_sklen, p->d_sshared, p->d_ublob,
+                     p->d_ucblob, o.block_restart_interval, p->d_err);
   p->kend();
   if (o.compression == 1) {
     for (auto& pb : blocks)

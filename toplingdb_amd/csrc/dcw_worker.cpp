@@ -317,6 +317,14 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   const auto& vlen = job.plan_vlen();
   PlanIn pin{job.plan_shared().data(), klen.data(), vlen.data(), nullptr, nsurv};
 
+  // GPU block plan: per-survivor next-block-start chain (k_plan_next)
+  TableOpts base_for_plan = opts_from_desc(d);
+  const uint32_t* plan_next = nullptr;
+  const uint32_t* plan_meta = nullptr;
+  if (nsurv > 0 &&
+      job.plan_all(base_for_plan, &plan_next, &plan_meta, &err) != 0)
+    return fail(res, 28, err);
+
   // grandparent-aware file cutting (compaction_outputs.cc:231-352): per-
   // survivor boundary positions from the GPU + host FSM state
   const bool use_gp = d->num_grandparents > 0;
@@ -375,12 +383,28 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       uint64_t min_unc = o.compression == 1
                              ? (uint64_t)(want / comp_ratio) + 8 * o.block_size
                              : want + 2 * o.block_size;
-      std::vector<uint32_t> eoff_plan;
-      std::vector<PlannedBlock> blocks = plan_blocks(pin, cur, o, min_unc, &eoff_plan);
+      // chain walk over the GPU plan (identical FSM to plan_blocks)
+      std::vector<PlannedBlock> blocks;
+      {
+        uint64_t produced = 0;
+        size_t i = cur;
+        while (i < nsurv && produced < min_unc) {
+          uint32_t nx = plan_next[i];
+          uint32_t m = plan_meta[i];
+          PlannedBlock pb;
+          pb.first = (uint32_t)i;
+          pb.count = nx - (uint32_t)i;
+          pb.unc_size = m & 0xffffffu;
+          pb.num_restarts = m >> 24;
+          blocks.push_back(pb);
+          produced += pb.unc_size + kTrailerSize;
+          i = nx;
+        }
+      }
       plan_usec += now_usec() - tp0;
       if (blocks.empty()) break;
       std::vector<uint32_t> csizes;
-      if (job.emit_blocks(blocks, o, &csizes, &err, &eoff_plan) != 0)
+      if (job.emit_blocks(blocks, o, &csizes, &err) != 0)
         return fail(res, 21, err);
       // walk csizes: find the cut (ShouldStopBefore semantics,
       // compaction_outputs.cc:231-352).  Without grandparents the file size
