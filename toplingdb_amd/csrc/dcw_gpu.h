@@ -81,8 +81,15 @@ class GpuJob {
   // Pack blocks [b0,b1) of the last emit_blocks call into one contiguous
   // [body|trailer]* image on device and D2H it straight into host_dst
   // (total_bytes = sum of (csize+5)); outoff[i] = image offset of block b0+i.
+  // The D2H runs on a second stream overlapping later kernels; *done_event
+  // (opaque hipEvent_t) signals host_dst completeness — wait with
+  // wait_event() (e.g. from the per-file tail thread) before reading.
   int pack_into(size_t b0, size_t b1, const std::vector<uint64_t>& outoff,
-                uint8_t* host_dst, size_t total_bytes, std::string* err);
+                uint8_t* host_dst, size_t total_bytes, void** done_event,
+                std::string* err);
+  static void wait_event(void* done_event);
+  // wait all in-flight output D2H transfers and fold their time into ms_d2h
+  void drain_d2h();
   // first/last internal keys of blocks [b0,b1) of the last emit
   int fetch_block_keys(size_t b0, size_t b1, std::vector<std::string>* first_keys,
                        std::vector<std::string>* last_keys, std::string* err);

@@ -293,7 +293,6 @@ __device__ __forceinline__ void key_set_byte(uint64_t& kb0, uint64_t& kb1,
   if (j < 8) {
     kb0 = (kb0 & ~(m << (8 * j))) | (x << (8 * j));
   } else if (j < 16) {
-    kb0 = kb0; // no-op to keep branches balanced
     kb1 = (kb1 & ~(m << (8 * (j - 8)))) | (x << (8 * (j - 8)));
   } else {
     tail = (tail & ~(m << (8 * (j - 16)))) | (x << (8 * (j - 16)));
@@ -1297,6 +1296,17 @@ struct KEv {
 
 struct GpuJob::Impl {
   hipStream_t stream = nullptr;
+  hipStream_t d2h_stream = nullptr;
+  // double-buffered pack->D2H: out_img slots with completion events so the
+  // next file's emit kernels overlap the previous file's output transfer
+  struct OutSlot {
+    void* img = nullptr;
+    size_t cap = 0;
+    hipEvent_t t0 = nullptr; // D2H start (on d2h_stream)
+    hipEvent_t done = nullptr;
+    bool pending = false;
+  } outslots[2];
+  int cur_outslot = 0;
   std::vector<KEv> kpending;
   void kbegin(const char* n, double bytes) {
     KEv e{n, bytes, nullptr, nullptr};
@@ -1432,11 +1442,31 @@ struct GpuJob::Impl {
 
 static double ms_between(hipEvent_t a, hipEvent_t b) {
   float ms = 0;
-  hipEventElapsedTime(&ms, a, b);
+  (void)hipEventElapsedTime(&ms, a, b);
   return (double)ms;
 }
 
-GpuJob::GpuJob() : p_(new Impl) { (void)hipStreamCreate(&p_->stream); }
+GpuJob::GpuJob() : p_(new Impl) {
+  (void)hipStreamCreate(&p_->stream);
+  (void)hipStreamCreate(&p_->d2h_stream);
+  for (auto& s : p_->outslots) {
+    (void)hipEventCreate(&s.t0);
+    (void)hipEventCreate(&s.done);
+  }
+}
+
+void GpuJob::wait_event(void* done_event) {
+  if (done_event) (void)hipEventSynchronize((hipEvent_t)done_event);
+}
+
+void GpuJob::drain_d2h() {
+  for (auto& s : p_->outslots) {
+    if (!s.pending) continue;
+    (void)hipEventSynchronize(s.done);
+    ms_d2h += ms_between(s.t0, s.done);
+    s.pending = false;
+  }
+}
 
 // Reuse this job object for a new job: keep device buffers (grow-only),
 // reset the per-job state.  Staged-input pointers are dropped (borrowed
@@ -1500,6 +1530,13 @@ GpuJob::~GpuJob() {
     else
       (void)hipHostFree(p->h_plan);
   }
+  for (auto& s : p->outslots) {
+    if (s.pending) (void)hipEventSynchronize(s.done);
+    if (s.img) (void)hipFree(s.img);
+    (void)hipEventDestroy(s.t0);
+    (void)hipEventDestroy(s.done);
+  }
+  if (p->d2h_stream) (void)hipStreamDestroy(p->d2h_stream);
   if (p->stream) (void)hipStreamDestroy(p->stream);
   delete p;
 }
@@ -2086,35 +2123,53 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
 }
 
 int GpuJob::pack_into(size_t b0, size_t b1, const std::vector<uint64_t>& outoff,
-                      uint8_t* host_dst, size_t total_bytes, std::string* err) {
+                      uint8_t* host_dst, size_t total_bytes, void** done_event,
+                      std::string* err) {
   Impl* p = p_;
   uint32_t nb = (uint32_t)(b1 - b0);
-  if (nb == 0) return 0;
+  if (nb == 0) {
+    *done_event = nullptr;
+    return 0;
+  }
   ENSURE(p->d_outoff, p->outoff_cap, sizeof(uint64_t) * nb);
   uint64_t* d_outoff = (uint64_t*)p->d_outoff;
   // synchronous copy: NULL-stream ordering with the (blocking) pipeline
   // stream, no pageable-async lifetime/ordering hazards
   HIPCHK(hipMemcpy(d_outoff, outoff.data(), sizeof(uint64_t) * nb,
                    hipMemcpyHostToDevice));
-  ENSURE(p->d_out_img, p->out_img_cap, total_bytes);
+  Impl::OutSlot& slot = p->outslots[p->cur_outslot];
+  p->cur_outslot ^= 1;
+  if (slot.pending) { // slot reused two files later; transfer long done
+    HIPCHK(hipEventSynchronize(slot.done));
+    ms_d2h += ms_between(slot.t0, slot.done);
+    slot.pending = false;
+  }
+  if (slot.cap < total_bytes) {
+    if (slot.img) (void)hipFree(slot.img);
+    slot.img = nullptr;
+    size_t c = total_bytes + total_bytes / 4;
+    HIPCHK(hipMalloc(&slot.img, c));
+    slot.cap = c;
+  }
   p->kbegin("pack", 2.0 * (double)total_bytes);
   hipLaunchKernelGGL(k_pack, dim3(grid_for(nb * 4ull)), dim3(256), 0, p->stream,
                      p->d_bds, (uint32_t)b0, (uint32_t)b1, p->d_ucblob, p->d_cblob,
                      p->ccap_per_block, p->d_ebsize, p->d_ebtype, p->d_ecsum,
-                     d_outoff, p->d_out_img);
+                     d_outoff, (uint8_t*)slot.img);
   p->kend();
-  hipEvent_t t0, t1;
-  hipEventCreate(&t0);
-  hipEventCreate(&t1);
-  hipEventRecord(t0, p->stream);
-  HIPCHK(hipMemcpyAsync(host_dst, p->d_out_img, total_bytes,
-                        hipMemcpyDeviceToHost, p->stream));
-  hipEventRecord(t1, p->stream);
-  HIPCHK(hipStreamSynchronize(p->stream));
-  ms_d2h += ms_between(t0, t1);
-  p->kresolve();
-  hipEventDestroy(t0);
-  hipEventDestroy(t1);
+  // D2H on the copy stream, ordered after the pack kernel via an event; the
+  // main stream is free for the next file's kernels immediately
+  hipEvent_t packed;
+  (void)hipEventCreate(&packed);
+  (void)hipEventRecord(packed, p->stream);
+  HIPCHK(hipStreamWaitEvent(p->d2h_stream, packed, 0));
+  (void)hipEventRecord(slot.t0, p->d2h_stream);
+  HIPCHK(hipMemcpyAsync(host_dst, slot.img, total_bytes, hipMemcpyDeviceToHost,
+                        p->d2h_stream));
+  (void)hipEventRecord(slot.done, p->d2h_stream);
+  (void)hipEventDestroy(packed);
+  slot.pending = true;
+  *done_event = (void*)slot.done;
   return 0;
 }
 

@@ -315,7 +315,6 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   size_t nsurv = job.num_survivors();
   const auto& klen = job.plan_klen();
   const auto& vlen = job.plan_vlen();
-  PlanIn pin{job.plan_shared().data(), klen.data(), vlen.data(), nullptr, nsurv};
 
   // GPU block plan: per-survivor next-block-start chain (k_plan_next)
   TableOpts base_for_plan = opts_from_desc(d);
@@ -364,6 +363,10 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     TableOpts o = base;
     o.orig_file_number = next_file_number++;
     RawBuf image; // data-block region accumulated on host
+    // output D2H transfers land in `image` asynchronously (second HIP
+    // stream); every reader of image bytes waits these events first
+    std::vector<void*> pend;
+    image.reserve(d->target_file_size + (d->target_file_size >> 2) + (2u << 20));
     std::vector<SstIndexEntry> handles;
     std::vector<std::string> first_keys, last_keys;
     std::vector<uint32_t> block_counts;
@@ -503,10 +506,19 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
         outoff[b] = acc;
         acc += csizes[b] + kTrailerSize;
       }
+      if (old + acc > image.cap) {
+        // realloc would memcpy regions still being written by in-flight
+        // D2H transfers — settle them first (rare: image is pre-reserved)
+        for (void* e : pend) GpuJob::wait_event(e);
+        pend.clear();
+      }
       image.resize_uninit(old + acc);
-      if (job.pack_into(0, take, outoff, image.p + old, acc, &err) != 0)
+      void* ev = nullptr;
+      if (job.pack_into(0, take, outoff, image.p + old, acc, &ev, &err) != 0)
         return fail(res, 22, err);
+      if (ev) pend.push_back(ev);
       if (getenv("DCW_PARANOID")) {
+        for (void* e : pend) GpuJob::wait_event(e);
         for (size_t b = 0; b < take; b++) {
           const uint8_t* body = image.p + old + outoff[b];
           uint32_t stored;
@@ -556,6 +568,10 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       SstIndexEntry h = append_block(pblk, o, (const uint8_t*)contents.data(),
                                      contents.size(), true);
       h.off += image.len;
+      if (image.len + pblk.size() > image.cap) {
+        for (void* e : pend) GpuJob::wait_event(e);
+        pend.clear();
+      }
       image.append(pblk.data(), pblk.size());
       handles.push_back(h);
       first_keys.push_back(kvs.front().first);
@@ -588,6 +604,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     struct TailJob {
       TableOpts o;
       RawBuf image;
+      std::vector<void*> pend; // D2H completion events for image regions
       std::vector<SstIndexEntry> handles;
       std::vector<std::string> first_keys, last_keys;
       uint64_t file_first, file_count, mn_seq, mx_seq, n_tomb;
@@ -599,6 +616,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     tj->image.len = image.len;
     tj->image.cap = image.cap;
     image.release();
+    tj->pend = std::move(pend);
     tj->handles = std::move(handles);
     tj->first_keys = std::move(first_keys);
     tj->last_keys = std::move(last_keys);
@@ -617,6 +635,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
                                                          out_bytes_p, ob_mu]() -> int {
       // separators (FindShortestInternalKeySeparator between adjacent
       // blocks; last block keeps its last key — kShortenSeparators mode)
+      for (void* e : tj->pend) GpuJob::wait_event(e); // image bytes complete
       size_t nb = tj->handles.size();
       std::vector<std::string> seps(nb);
       bool sep_key_plus_seq = false;
@@ -680,6 +699,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       if (w.get() != 0) return fail(res, 26, "output file tail/write failed");
     write_usec += now_usec() - tj2;
   }
+  job.drain_d2h();
 
   res->num_files = (uint32_t)out_files.size();
   res->files = (dcw_output_file*)malloc(sizeof(dcw_output_file) * out_files.size());
