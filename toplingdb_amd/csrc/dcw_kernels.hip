@@ -140,7 +140,7 @@ __global__ void k_verify_usize(const uint8_t* __restrict__ blob,
 // one wave per block: raw -> cooperative copy; snappy -> staged through LDS
 // (compressed in + decoded out both in LDS, lane 0 runs the serial decoder,
 // all lanes copy in/out); oversized blocks fall back to the direct path.
-#define DEC_MAX 5376
+#define DEC_MAX 4992 // 2*4992*4 waves = 39.9 KB LDS/WG -> 4 WGs (16 decoders)/CU
 struct DecLds {
   uint8_t in[DEC_MAX];
   uint8_t out[DEC_MAX];
