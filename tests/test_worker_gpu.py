@@ -207,6 +207,30 @@ def test_multi_file_level_run_gpu(tmp_path):
     assert_identical(rg, ro)
 
 
+def test_all_tombstones_bottommost_zero_files(tmp_path):
+    # bottommost + no snapshots: every Delete is dropped (CompactionIterator
+    # bottommost-delete rule) -> zero survivors -> zero output files
+    kvs = [(b"k%014d" % i, 1 + i, 0, b"") for i in range(5000)]
+    p = _write_kv_run(tmp_path, "d.sst", kvs)
+    rg, ro = run_both(tmp_path, [[p]], bottommost_level=1)
+    assert rg["out_entries"] == 0 and ro["out_entries"] == 0
+    assert len(rg["files"]) == 0 and len(ro["files"]) == 0
+
+
+def test_large_values_multi_chunk(tmp_path):
+    # 1 KiB values force multiple plan->emit chunks per output file and
+    # exercise the image-buffer pre-reserve/realloc guard under async D2H
+    runs = []
+    for r in range(2):
+        p = str(tmp_path / ("lv%d.sst" % r))
+        dcw.gen_sst(p, seed=7 + r, num_entries=40000, value_len=1024,
+                    seq_base=1 + r * 40000, compression=1)
+        runs.append([p])
+    rg, ro = run_both(tmp_path, runs, compression=1, target_file_size=16 << 20)
+    assert_identical(rg, ro)
+    assert len(rg["files"]) > 1
+
+
 def test_grandparent_cuts_match_oracle(tmp_path):
     # grandparent-aware file cutting (ShouldStopBefore boundary rules,
     # compaction_outputs.cc:231-352): GPU worker vs oracle, bit-exact,
