@@ -806,6 +806,7 @@ __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
                        uint8_t* __restrict__ ucblob, uint32_t restart_interval,
                        uint32_t* err_flag) {
   __shared__ uint32_t offs[EMIT_MAX_ENTRIES];
+  __shared__ uint32_t esz[EMIT_MAX_ENTRIES];
   for (uint32_t b = blockIdx.x; b < nblocks; b += gridDim.x) {
     EmitBlockDesc d = bds[b];
     uint8_t* out = ucblob + d.uout;
@@ -814,17 +815,23 @@ __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
       __syncthreads();
       continue;
     }
+    // per-entry encoded sizes in parallel (the only cross-entry state,
+    // the shared prefix, is already in s_shared); thread 0 then runs the
+    // short LDS-only prefix walk
+    for (uint32_t li = threadIdx.x; li < d.count; li += blockDim.x) {
+      uint32_t i = d.first + li;
+      uint32_t shared = (li % restart_interval == 0) ? 0 : s_shared[i];
+      uint32_t klen = s_klen[i];
+      uint32_t vl = s_vlen[i];
+      esz[li] = varint_len(shared) + varint_len(klen - shared) +
+                varint_len(vl) + (klen - shared) + vl;
+    }
+    __syncthreads();
     if (threadIdx.x == 0) {
       uint32_t off = 0;
       for (uint32_t li = 0; li < d.count; li++) {
-        uint32_t i = d.first + li;
         offs[li] = off;
-        uint32_t shared =
-            (li % restart_interval == 0) ? 0 : s_shared[i];
-        uint32_t klen = s_klen[i];
-        uint32_t vl = s_vlen[i];
-        off += varint_len(shared) + varint_len(klen - shared) +
-               varint_len(vl) + (klen - shared) + vl;
+        off += esz[li];
       }
     }
     __syncthreads();
