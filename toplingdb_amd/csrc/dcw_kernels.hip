@@ -933,43 +933,50 @@ __device__ static size_t snap_encode_lds(const uint8_t* in, uint32_t n,
   op += varint32_put(op, n);
   if (n == 0) return (size_t)(op - out);
   uint32_t lit_start = 0, pz = 0;
-  uint32_t skip = 32; // spec v2: snappy-style skip acceleration
+  uint32_t skip = 32; // spec v3: snappy-style skip acceleration
   while (pz + 4 <= n) {
-    uint32_t pz2 = pz + (skip >> 5);
-    bool have2 = pz2 + 4 <= n;
-    uint32_t w1 = load32(in + pz);
-    uint32_t w2 = have2 ? load32(in + pz2) : 0;
-    uint32_t h1 = (w1 * kSnapHashMul) >> (32 - kSnapHashBits);
-    uint32_t h2 = (w2 * kSnapHashMul) >> (32 - kSnapHashBits);
-    uint32_t cand1 = tab[h1];
-    uint32_t cand2_raw = have2 ? tab[h2] : 0xffffu;
-    tab[h1] = (uint16_t)pz;
-    uint32_t mpos, mcand, w;
-    if (cand1 != 0xffffu && cand1 < pz && load32(in + cand1) == w1) {
-      mpos = pz;
-      mcand = cand1;
-      w = w1;
-    } else {
-      // miss at pz
-      skip++;
-      if (!have2) {
-        pz = pz2;
-        break;
-      }
-      uint32_t cand2 = (h2 == h1) ? pz : cand2_raw;
-      tab[h2] = (uint16_t)pz2;
-      if (cand2 != 0xffffu && cand2 < pz2 && load32(in + cand2) == w2) {
-        mpos = pz2;
-        mcand = cand2;
-        w = w2;
-      } else {
-        pz = pz2 + (skip >> 5);
-        skip++;
-        continue;
+    // 4-position speculative probe group.  The serial spec probes one
+    // position at a time: probe p (read tab, then insert p), on miss
+    // skip++ and advance by (skip>>5) computed BEFORE the increment.
+    // The group issues the 4 independent LDS load chains together and
+    // reconstructs serial semantics: a candidate read at p[j] sees the
+    // in-group insertion of the latest p[i<j] with the same hash slot,
+    // and insertions commit only up to the first matching position.
+    uint32_t ps[4];
+    ps[0] = pz;
+    ps[1] = ps[0] + (skip >> 5);
+    ps[2] = ps[1] + ((skip + 1) >> 5);
+    ps[3] = ps[2] + ((skip + 2) >> 5);
+    int np = 4;
+    if (ps[1] + 4 > n) np = 1;
+    else if (ps[2] + 4 > n) np = 2;
+    else if (ps[3] + 4 > n) np = 3;
+    uint32_t w4[4], h4[4], cr[4];
+    for (int i = 0; i < np; i++) w4[i] = load32(in + ps[i]);
+    for (int i = 0; i < np; i++)
+      h4[i] = (w4[i] * kSnapHashMul) >> (32 - kSnapHashBits);
+    for (int i = 0; i < np; i++) cr[i] = tab[h4[i]];
+    int jm = -1;
+    uint32_t mcand = 0;
+    for (int j = 0; j < np && jm < 0; j++) {
+      uint32_t c = cr[j];
+      for (int i = j - 1; i >= 0; i--)
+        if (h4[i] == h4[j]) { c = ps[i]; break; }
+      if (c != 0xffffu && c < ps[j] && load32(in + c) == w4[j]) {
+        jm = j;
+        mcand = c;
       }
     }
+    int lastins = jm >= 0 ? jm : np - 1;
+    for (int i = 0; i <= lastins; i++) tab[h4[i]] = (uint16_t)ps[i];
+    if (jm < 0) {
+      if (np < 4) break; // next probe would run past the block
+      pz = ps[3] + ((skip + 3) >> 5);
+      skip += 4;
+      continue;
+    }
+    uint32_t mpos = ps[jm];
     // match at mpos against mcand
-    (void)w;
     uint32_t mlen = 4;
     while (mpos + mlen + 4 <= n) {
       uint32_t a = load32(in + mcand + mlen);

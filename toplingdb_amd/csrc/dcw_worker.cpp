@@ -384,7 +384,9 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
                           ? d->target_file_size - already
                           : (64 << 10);
       uint64_t min_unc = o.compression == 1
-                             ? (uint64_t)(want / comp_ratio) + 8 * o.block_size
+                             ? (uint64_t)(want / comp_ratio) +
+                                   (uint64_t)(want / comp_ratio) / 32 +
+                                   16 * o.block_size
                              : want + 2 * o.block_size;
       // chain walk over the GPU plan (identical FSM to plan_blocks)
       std::vector<PlannedBlock> blocks;
