@@ -938,44 +938,66 @@ __device__ static size_t snap_encode_lds(const uint8_t* in, uint32_t n,
     // 4-position speculative probe group.  The serial spec probes one
     // position at a time: probe p (read tab, then insert p), on miss
     // skip++ and advance by (skip>>5) computed BEFORE the increment.
-    // The group issues the 4 independent LDS load chains together and
-    // reconstructs serial semantics: a candidate read at p[j] sees the
-    // in-group insertion of the latest p[i<j] with the same hash slot,
-    // and insertions commit only up to the first matching position.
-    uint32_t ps[4];
-    ps[0] = pz;
-    ps[1] = ps[0] + (skip >> 5);
-    ps[2] = ps[1] + ((skip + 1) >> 5);
-    ps[3] = ps[2] + ((skip + 2) >> 5);
-    int np = 4;
-    if (ps[1] + 4 > n) np = 1;
-    else if (ps[2] + 4 > n) np = 2;
-    else if (ps[3] + 4 > n) np = 3;
-    uint32_t w4[4], h4[4], cr[4];
-    for (int i = 0; i < np; i++) w4[i] = load32(in + ps[i]);
-    for (int i = 0; i < np; i++)
-      h4[i] = (w4[i] * kSnapHashMul) >> (32 - kSnapHashBits);
-    for (int i = 0; i < np; i++) cr[i] = tab[h4[i]];
-    int jm = -1;
-    uint32_t mcand = 0;
-    for (int j = 0; j < np && jm < 0; j++) {
-      uint32_t c = cr[j];
-      for (int i = j - 1; i >= 0; i--)
-        if (h4[i] == h4[j]) { c = ps[i]; break; }
-      if (c != 0xffffu && c < ps[j] && load32(in + c) == w4[j]) {
-        jm = j;
-        mcand = c;
-      }
-    }
-    int lastins = jm >= 0 ? jm : np - 1;
-    for (int i = 0; i <= lastins; i++) tab[h4[i]] = (uint16_t)ps[i];
-    if (jm < 0) {
+    // All four probe chains (in-load -> hash -> tab-load -> candidate
+    // compare-load) are issued as independent scalar chains so their LDS
+    // latencies overlap; serial semantics are reconstructed afterwards:
+    // a candidate read at p_j sees the in-group insertion of the latest
+    // p_i (i<j) hashing to the same slot, and insertions commit only up
+    // to the first matching position.
+    uint32_t p0 = pz;
+    uint32_t p1 = p0 + (skip >> 5);
+    uint32_t p2 = p1 + ((skip + 1) >> 5);
+    uint32_t p3 = p2 + ((skip + 2) >> 5);
+    int np = p3 + 4 <= n ? 4 : (p2 + 4 <= n ? 3 : (p1 + 4 <= n ? 2 : 1));
+    uint32_t w0 = load32(in + p0);
+    uint32_t w1 = np > 1 ? load32(in + p1) : 0;
+    uint32_t w2 = np > 2 ? load32(in + p2) : 0;
+    uint32_t w3 = np > 3 ? load32(in + p3) : 0;
+    uint32_t h0 = (w0 * kSnapHashMul) >> (32 - kSnapHashBits);
+    uint32_t h1 = (w1 * kSnapHashMul) >> (32 - kSnapHashBits);
+    uint32_t h2 = (w2 * kSnapHashMul) >> (32 - kSnapHashBits);
+    uint32_t h3 = (w3 * kSnapHashMul) >> (32 - kSnapHashBits);
+    uint32_t c0 = tab[h0];
+    uint32_t c1 = tab[h1];
+    uint32_t c2 = tab[h2];
+    uint32_t c3 = tab[h3];
+    // in-group overwrites (latest earlier position with the same slot wins)
+    if (h1 == h0) c1 = p0;
+    if (h2 == h1) c2 = p1; else if (h2 == h0) c2 = p0;
+    if (h3 == h2) c3 = p2; else if (h3 == h1) c3 = p1; else if (h3 == h0) c3 = p0;
+    // speculative candidate loads with sanitized offsets (0xffff = empty)
+    uint32_t a0 = load32(in + (c0 == 0xffffu ? 0 : c0));
+    uint32_t a1 = load32(in + (c1 == 0xffffu ? 0 : c1));
+    uint32_t a2 = load32(in + (c2 == 0xffffu ? 0 : c2));
+    uint32_t a3 = load32(in + (c3 == 0xffffu ? 0 : c3));
+    bool m0 = c0 != 0xffffu && c0 < p0 && a0 == w0;
+    bool m1 = np > 1 && c1 != 0xffffu && c1 < p1 && a1 == w1;
+    bool m2 = np > 2 && c2 != 0xffffu && c2 < p2 && a2 == w2;
+    bool m3 = np > 3 && c3 != 0xffffu && c3 < p3 && a3 == w3;
+    uint32_t mpos, mcand;
+    if (m0) {
+      tab[h0] = (uint16_t)p0;
+      mpos = p0; mcand = c0;
+    } else if (m1) {
+      tab[h0] = (uint16_t)p0; tab[h1] = (uint16_t)p1;
+      mpos = p1; mcand = c1;
+    } else if (m2) {
+      tab[h0] = (uint16_t)p0; tab[h1] = (uint16_t)p1; tab[h2] = (uint16_t)p2;
+      mpos = p2; mcand = c2;
+    } else if (m3) {
+      tab[h0] = (uint16_t)p0; tab[h1] = (uint16_t)p1; tab[h2] = (uint16_t)p2;
+      tab[h3] = (uint16_t)p3;
+      mpos = p3; mcand = c3;
+    } else {
+      tab[h0] = (uint16_t)p0;
+      if (np > 1) tab[h1] = (uint16_t)p1;
+      if (np > 2) tab[h2] = (uint16_t)p2;
+      if (np > 3) tab[h3] = (uint16_t)p3;
       if (np < 4) break; // next probe would run past the block
-      pz = ps[3] + ((skip + 3) >> 5);
+      pz = p3 + ((skip + 3) >> 5);
       skip += 4;
       continue;
     }
-    uint32_t mpos = ps[jm];
     // match at mpos against mcand
     uint32_t mlen = 4;
     while (mpos + mlen + 4 <= n) {
