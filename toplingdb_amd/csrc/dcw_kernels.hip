@@ -879,10 +879,11 @@ __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
 #define SNAP_MAX_UNC 5120 // block_size + slack; host guards this bound
 #define SNAP_MAX_OUT (32 + SNAP_MAX_UNC + SNAP_MAX_UNC / 6)
 struct SnapLds {
-  uint8_t in[SNAP_MAX_UNC];
   uint16_t tab[1u << kSnapHashBits];
-  uint32_t cn;
-}; // ~13.1 KiB per encoder; output goes straight to global (write-only)
+}; // 4 KiB per encoder: hash table only.  Input is read straight from
+// global (the 4 KiB block stays hot in L1/L2 for the single encoding
+// lane); output goes straight to global (write-only).  Keeping LDS per
+// encoder small lets ~9 workgroups (36 encoder waves) reside per CU.
 
 // wave-internal LDS ordering: drain DS ops + stop compiler reordering
 __device__ __forceinline__ void wave_lds_sync() {
@@ -1027,7 +1028,7 @@ __global__ __launch_bounds__(256) void k_compress(
     const uint8_t* __restrict__ ucblob, uint8_t* __restrict__ cblob,
     uint64_t ccap_per_block, uint32_t* __restrict__ bsize,
     uint8_t* __restrict__ btype, uint32_t* err_flag) {
-  __shared__ SnapLds lds[4]; // 4 encoders/WG ~= 53 KiB -> 3 WGs/CU resident
+  __shared__ SnapLds lds[4]; // 4 encoders/WG = 16 KiB
   uint32_t wid = threadIdx.x / WAVE;  // wave within workgroup
   uint32_t lane = threadIdx.x % WAVE;
   uint32_t waves = blockDim.x / WAVE;
@@ -1044,16 +1045,12 @@ __global__ __launch_bounds__(256) void k_compress(
       continue;
     }
     const uint8_t* gin = ucblob + d.uout;
-    for (uint32_t t = lane * 4; t < d.unc_size; t += WAVE * 4) {
-      uint32_t chunk = d.unc_size - t < 4 ? d.unc_size - t : 4;
-      for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = gin[t + x];
-    }
     for (uint32_t t = lane; t < (1u << kSnapHashBits); t += WAVE)
       L.tab[t] = 0xffffu;
     wave_lds_sync();
     if (lane == 0) {
       uint8_t* gout = cblob + (uint64_t)b * ccap_per_block;
-      uint32_t cn = (uint32_t)snap_encode_lds(L.in, d.unc_size, gout, L.tab);
+      uint32_t cn = (uint32_t)snap_encode_lds(gin, d.unc_size, gout, L.tab);
       // GoodCompressionRatio, default max_compressed_bytes_per_kb=896
       if (cn <= (((uint64_t)896 * d.unc_size) >> 10)) {
         bsize[b] = cn;
