@@ -14,8 +14,8 @@ struct CrcInit {
 } g_crc_init;
 
 void put_fixed32(std::string& s, uint32_t v) { s.append((const char*)&v, 4); }
-void put_fixed64(std::string& s, uint64_t v) { s.append((const char*)&v, 8); }
-void put_varint32(std::string& s, uint32_t v) {
+[[maybe_unused]] void put_fixed64(std::string& s, uint64_t v) { s.append((const char*)&v, 8); }
+[[maybe_unused]] void put_varint32(std::string& s, uint32_t v) {
   uint8_t t[5];
   s.append((const char*)t, varint32_put(t, v));
 }

@@ -1584,9 +1584,9 @@ static uint32_t grid_for(uint64_t work, uint32_t block = 256) {
 int GpuJob::stage(const GpuInputs& in, std::string* err) {
   Impl* p = p_;
   hipEvent_t t0, t1;
-  hipEventCreate(&t0);
-  hipEventCreate(&t1);
-  hipEventRecord(t0, p->stream);
+  (void)hipEventCreate(&t0);
+  (void)hipEventCreate(&t1);
+  (void)hipEventRecord(t0, p->stream);
   HIPCHK(hipMalloc(&p->d_blob, in.blob_size));
   HIPCHK(hipMemcpyAsync(p->d_blob, in.blob, in.blob_size, hipMemcpyHostToDevice,
                         p->stream));
@@ -1611,11 +1611,11 @@ int GpuJob::stage(const GpuInputs& in, std::string* err) {
   HIPCHK(hipMemsetAsync(p->d_err, 0, 8, p->stream));
   if (!p->d_uklen_probe) HIPCHK(hipMalloc(&p->d_uklen_probe, 4));
   HIPCHK(hipMemsetAsync(p->d_uklen_probe, 0xff, 4, p->stream));
-  hipEventRecord(t1, p->stream);
+  (void)hipEventRecord(t1, p->stream);
   HIPCHK(hipStreamSynchronize(p->stream));
   ms_h2d += ms_between(t0, t1);
-  hipEventDestroy(t0);
-  hipEventDestroy(t1);
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
   // remember run boundaries (translated to entries later)
   run_blocks_ = in.run_block_begin;
   return 0;
@@ -1663,9 +1663,9 @@ int GpuJob::stage_release(StagedInput* s, std::string* err) {
 int GpuJob::decode(std::string* err) {
   Impl* p = p_;
   hipEvent_t t0, t1;
-  hipEventCreate(&t0);
-  hipEventCreate(&t1);
-  hipEventRecord(t0, p->stream);
+  (void)hipEventCreate(&t0);
+  (void)hipEventCreate(&t1);
+  (void)hipEventRecord(t0, p->stream);
   uint32_t nb = p->n_blocks;
   HIPCHK(p->ens((void**)&p->d_usize, sizeof(uint32_t) * nb));
   HIPCHK(p->ens((void**)&p->d_btype_in, nb));
@@ -1790,12 +1790,12 @@ int GpuJob::decode(std::string* err) {
   HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost, p->stream));
   HIPCHK(hipMemcpyAsync(&uklen, p->d_uklen_probe, 4, hipMemcpyDeviceToHost,
                         p->stream));
-  hipEventRecord(t1, p->stream);
+  (void)hipEventRecord(t1, p->stream);
   HIPCHK(hipStreamSynchronize(p->stream));
   ms_decode += ms_between(t0, t1);
   p->kresolve();
-  hipEventDestroy(t0);
-  hipEventDestroy(t1);
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
   if (err_host) {
     if (err) *err = "entry decode failed, code " + std::to_string(err_host) +
                     (err_host == DE_UKEY_LEN
@@ -1810,9 +1810,9 @@ int GpuJob::decode(std::string* err) {
 int GpuJob::merge(std::string* err) {
   Impl* p = p_;
   hipEvent_t t0, t1;
-  hipEventCreate(&t0);
-  hipEventCreate(&t1);
-  hipEventRecord(t0, p->stream);
+  (void)hipEventCreate(&t0);
+  (void)hipEventCreate(&t1);
+  (void)hipEventRecord(t0, p->stream);
   std::vector<uint64_t> bounds = p->run_entry_begin; // size k+1
   int cur = 0;
   while (bounds.size() > 2) {
@@ -1842,12 +1842,12 @@ int GpuJob::merge(std::string* err) {
     cur ^= 1;
   }
   p->final_buf = cur;
-  hipEventRecord(t1, p->stream);
+  (void)hipEventRecord(t1, p->stream);
   HIPCHK(hipStreamSynchronize(p->stream));
   ms_merge += ms_between(t0, t1);
   p->kresolve();
-  hipEventDestroy(t0);
-  hipEventDestroy(t1);
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
   return 0;
 }
 
@@ -1883,9 +1883,9 @@ static int scan_u8(GpuJob::Impl* p, const uint8_t* d_in, uint64_t n,
 int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   Impl* p = p_;
   hipEvent_t t0, t1;
-  hipEventCreate(&t0);
-  hipEventCreate(&t1);
-  hipEventRecord(t0, p->stream);
+  (void)hipEventCreate(&t0);
+  (void)hipEventCreate(&t1);
+  (void)hipEventRecord(t0, p->stream);
   uint64_t n = p->n_entries;
   const ulong4* ents = p->d_ent[p->final_buf];
   HIPCHK(p->ens((void**)&p->d_head, n));
@@ -2028,12 +2028,12 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
     HIPCHK(hipMemcpyAsync(h_vlen_.data(), p->d_svlen, sizeof(uint32_t) * nsurv,
                           hipMemcpyDeviceToHost, p->stream));
   }
-  hipEventRecord(t1, p->stream);
+  (void)hipEventRecord(t1, p->stream);
   HIPCHK(hipStreamSynchronize(p->stream));
   ms_dedup += ms_between(t0, t1);
   p->kresolve();
-  hipEventDestroy(t0);
-  hipEventDestroy(t1);
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
   if (d_snaps) (void)hipFree(d_snaps);
   return 0;
 }
@@ -2082,9 +2082,9 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
                         std::vector<uint32_t>* comp_sizes, std::string* err) {
   Impl* p = p_;
   hipEvent_t t0, t1;
-  hipEventCreate(&t0);
-  hipEventCreate(&t1);
-  hipEventRecord(t0, p->stream);
+  (void)hipEventCreate(&t0);
+  (void)hipEventCreate(&t1);
+  (void)hipEventRecord(t0, p->stream);
   uint32_t nb = (uint32_t)blocks.size();
   if (nb == 0) {
     comp_sizes->clear();
@@ -2146,12 +2146,12 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   comp_sizes->resize(nb);
   HIPCHK(hipMemcpyAsync(comp_sizes->data(), p->d_ebsize, sizeof(uint32_t) * nb,
                         hipMemcpyDeviceToHost, p->stream));
-  hipEventRecord(t1, p->stream);
+  (void)hipEventRecord(t1, p->stream);
   HIPCHK(hipStreamSynchronize(p->stream));
   ms_emit += ms_between(t0, t1);
   p->kresolve();
-  hipEventDestroy(t0);
-  hipEventDestroy(t1);
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
   return 0;
 }
 
