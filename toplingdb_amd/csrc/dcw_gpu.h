@@ -102,6 +102,8 @@ class GpuJob {
   int gp_positions(const dcw_job_desc* d, std::vector<uint32_t>* pos,
                    std::vector<uint8_t>* nback, std::string* err);
   // min/max sequence + tombstone count over survivor range (post zeroing)
+  // per-block seq stats from the records prefetched by emit_blocks
+  void block_stats(size_t b, uint64_t* mn, uint64_t* mx, uint64_t* tomb);
   int seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* mx,
                  uint64_t* n_tombstones, std::string* err);
 

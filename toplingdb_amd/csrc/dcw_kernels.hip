@@ -1108,21 +1108,37 @@ __global__ void k_pack(const EmitBlockDesc* __restrict__ bds, uint32_t b0,
   }
 }
 
-__global__ void k_block_keys(const EmitBlockDesc* __restrict__ bds, uint32_t b0,
-                             uint32_t b1, const uint64_t* __restrict__ s_k0,
-                             const uint64_t* __restrict__ s_k1,
-                             const uint64_t* __restrict__ s_tag,
-                             const uint8_t* __restrict__ s_klen,
-                             uint8_t* __restrict__ out /*64B per block: first|last*/) {
+// per-block metadata record, fetched once per chunk together with the
+// compressed sizes: boundary keys (for index separators) + seq-range and
+// tombstone count (for file properties; replaces a per-file reduce pass)
+#define BLKSTAT_STRIDE 96
+__global__ void k_block_stats(const EmitBlockDesc* __restrict__ bds, uint32_t b0,
+                              uint32_t b1, const uint64_t* __restrict__ s_k0,
+                              const uint64_t* __restrict__ s_k1,
+                              const uint64_t* __restrict__ s_tag,
+                              const uint8_t* __restrict__ s_klen,
+                              uint8_t* __restrict__ out) {
   for (uint32_t b = b0 + blockIdx.x * blockDim.x + threadIdx.x; b < b1;
        b += gridDim.x * blockDim.x) {
-    uint8_t* o = out + (uint64_t)(b - b0) * 64;
+    uint8_t* o = out + (uint64_t)(b - b0) * BLKSTAT_STRIDE;
     uint32_t f = bds[b].first, l = bds[b].first + bds[b].count - 1;
     memset(o, 0, 64);
     o[0] = s_klen[f];
     build_ikey(s_k0[f], s_k1[f], s_tag[f], s_klen[f], o + 1);
     o[32] = s_klen[l];
     build_ikey(s_k0[l], s_k1[l], s_tag[l], s_klen[l], o + 33);
+    uint64_t mn = ~0ull, mx = 0, tomb = 0;
+    for (uint32_t i = f; i <= l; i++) {
+      uint64_t tag = s_tag[i];
+      uint64_t seq = tag >> 8;
+      if (seq < mn) mn = seq;
+      if (seq > mx) mx = seq;
+      uint8_t vt = (uint8_t)tag;
+      if (vt == kTypeDeletion || vt == kTypeSingleDeletion) tomb++;
+    }
+    memcpy(o + 64, &mn, 8);
+    memcpy(o + 72, &mx, 8);
+    memcpy(o + 80, &tomb, 8);
   }
 }
 
@@ -1340,6 +1356,8 @@ struct GpuJob::Impl {
     bool pending = false;
   } outslots[2];
   int cur_outslot = 0;
+  void* h_keys = nullptr; // pinned block-stats records (BLKSTAT_STRIDE each)
+  uint64_t h_keys_cap = 0;
   std::vector<KEv> kpending;
   void kbegin(const char* n, double bytes) {
     KEv e{n, bytes, nullptr, nullptr};
@@ -1569,6 +1587,7 @@ GpuJob::~GpuJob() {
     (void)hipEventDestroy(s.t0);
     (void)hipEventDestroy(s.done);
   }
+  if (p->h_keys) (void)hipHostFree(p->h_keys);
   if (p->d2h_stream) (void)hipStreamDestroy(p->d2h_stream);
   if (p->stream) (void)hipStreamDestroy(p->stream);
   delete p;
@@ -2143,9 +2162,23 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
                      p->d_ebsize, p->d_ebtype, o.checksum_type, p->d_crc,
                      p->d_ecsum);
   p->kend();
+  // per-block boundary keys + seq stats, prefetched with the sizes so the
+  // host needs no further GPU round trip for them
+  HIPCHK(p->ens((void**)&p->d_scratch_keys, (uint64_t)nb * BLKSTAT_STRIDE));
+  if (p->h_keys_cap < (uint64_t)nb * BLKSTAT_STRIDE) {
+    if (p->h_keys) (void)hipHostFree(p->h_keys);
+    p->h_keys_cap = (uint64_t)nb * BLKSTAT_STRIDE * 5 / 4;
+    HIPCHK(hipHostMalloc(&p->h_keys, p->h_keys_cap));
+  }
+  hipLaunchKernelGGL(k_block_stats, dim3(grid_for(nb)), dim3(256), 0, p->stream,
+                     p->d_bds, 0u, nb, p->d_sk0, p->d_sk1, p->d_stag,
+                     p->d_sklen, (uint8_t*)p->d_scratch_keys);
   comp_sizes->resize(nb);
   HIPCHK(hipMemcpyAsync(comp_sizes->data(), p->d_ebsize, sizeof(uint32_t) * nb,
                         hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipMemcpyAsync(p->h_keys, p->d_scratch_keys,
+                        (uint64_t)nb * BLKSTAT_STRIDE, hipMemcpyDeviceToHost,
+                        p->stream));
   (void)hipEventRecord(t1, p->stream);
   HIPCHK(hipStreamSynchronize(p->stream));
   ms_emit += ms_between(t0, t1);
@@ -2210,26 +2243,29 @@ int GpuJob::fetch_block_keys(size_t b0, size_t b1,
                              std::vector<std::string>* first_keys,
                              std::vector<std::string>* last_keys,
                              std::string* err) {
+  // served from the block-stats records prefetched by emit_blocks (same
+  // chunk); no GPU round trip
   Impl* p = p_;
-  uint32_t nb = (uint32_t)(b1 - b0);
   first_keys->clear();
   last_keys->clear();
-  if (!nb) return 0;
-  HIPCHK(p->ens((void**)&p->d_scratch_keys, (uint64_t)nb * 64));
-  uint8_t* d_keys = (uint8_t*)p->d_scratch_keys;
-  hipLaunchKernelGGL(k_block_keys, dim3(grid_for(nb)), dim3(256), 0, p->stream,
-                     p->d_bds, (uint32_t)b0, (uint32_t)b1, p->d_sk0, p->d_sk1,
-                     p->d_stag, p->d_sklen, d_keys);
-  std::vector<uint8_t> h((uint64_t)nb * 64);
-  HIPCHK(hipMemcpyAsync(h.data(), d_keys, h.size(), hipMemcpyDeviceToHost,
-                        p->stream));
-  HIPCHK(hipStreamSynchronize(p->stream));
-  for (uint32_t i = 0; i < nb; i++) {
-    const uint8_t* o = h.data() + (uint64_t)i * 64;
+  if (b1 <= b0) return 0;
+  if (!p->h_keys || b1 > p->emit_nblocks) {
+    if (err) *err = "fetch_block_keys: no prefetched chunk stats";
+    return -1;
+  }
+  for (size_t i = b0; i < b1; i++) {
+    const uint8_t* o = (const uint8_t*)p->h_keys + i * BLKSTAT_STRIDE;
     first_keys->emplace_back((const char*)o + 1, o[0]);
     last_keys->emplace_back((const char*)o + 33, o[32]);
   }
   return 0;
+}
+
+void GpuJob::block_stats(size_t b, uint64_t* mn, uint64_t* mx, uint64_t* tomb) {
+  const uint8_t* o = (const uint8_t*)p_->h_keys + b * BLKSTAT_STRIDE;
+  memcpy(mn, o + 64, 8);
+  memcpy(mx, o + 72, 8);
+  memcpy(tomb, o + 80, 8);
 }
 
 int GpuJob::gather_entries(uint64_t first, uint32_t count,

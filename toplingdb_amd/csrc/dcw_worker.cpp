@@ -370,6 +370,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     std::vector<SstIndexEntry> handles;
     std::vector<std::string> first_keys, last_keys;
     std::vector<uint32_t> block_counts;
+    uint64_t mn_seq = ~0ull, mx_seq = 0, n_tomb = 0; // file seq stats
     size_t cur = s;
     bool cut = false;
     uint64_t cut_entry = 0;
@@ -546,6 +547,11 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
         first_keys.push_back(cfirst[b]);
         last_keys.push_back(clast[b]);
         block_counts.push_back(blocks[b].count);
+        uint64_t bmn, bmx, bt;
+        job.block_stats(b, &bmn, &bmx, &bt);
+        if (bmn < mn_seq) mn_seq = bmn;
+        if (bmx > mx_seq) mx_seq = bmx;
+        n_tomb += bt;
       }
       cur = take ? blocks[take - 1].first + blocks[take - 1].count : cur;
       if (take < blocks.size()) break; // cut decided inside this chunk
@@ -579,14 +585,20 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       first_keys.push_back(kvs.front().first);
       last_keys.push_back(kvs.back().first);
       block_counts.push_back(partial_count);
+      for (auto& kv : kvs) {
+        uint64_t tag;
+        memcpy(&tag, kv.first.data() + kv.first.size() - 8, 8);
+        uint64_t seq = tag >> 8;
+        uint8_t vt = (uint8_t)tag;
+        if (seq < mn_seq) mn_seq = seq;
+        if (seq > mx_seq) mx_seq = seq;
+        if (vt == 0 || vt == 7) n_tomb++; // kTypeDeletion / kTypeSingleDeletion
+      }
     }
     if (cut) cur = cut_entry;
     uint64_t file_count = cur - s;
     if (handles.empty()) break; // nothing left
     uint64_t tw1 = now_usec();
-    uint64_t mn_seq = 0, mx_seq = 0, n_tomb = 0;
-    if (job.seq_minmax(file_first, file_count, &mn_seq, &mx_seq, &n_tomb, &err) != 0)
-      return fail(res, 25, err);
     // hand the whole per-file tail (separators, stats, meta tail, file
     // write) to a background thread — the GPU starts the next file now.
     char path[600];
