@@ -278,6 +278,19 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     return fail(res, 13, "unsupported compression (none/snappy only)");
 
   uint64_t t_start = now_usec();
+  // fine-grained wall attribution (DCW_PHASE_DEBUG=1 prints at job end)
+  struct WallProf {
+    const char* names[12] = {"load",  "stage", "decode", "merge",
+                             "dedup", "plan",  "emit",   "walkpack",
+                             "gather", "tailspawn", "join", "other"};
+    uint64_t us[12] = {0};
+    uint64_t t_last;
+    void mark(int slot, uint64_t now) {
+      us[slot] += now - t_last;
+      t_last = now;
+    }
+  } wp;
+  wp.t_last = t_start;
   std::string err;
 
   // ---- read + parse inputs (host I/O) ----
@@ -298,6 +311,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     in_bytes = staged->in_bytes;
   } else {
     if (load_inputs(d, &L, &err) != 0) return fail(res, 16, err);
+    wp.mark(0, now_usec());
     in_bytes = L.in_bytes;
   }
   res->t_read_usec = now_usec() - t0;
@@ -305,12 +319,16 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   // ---- GPU pipeline ----
   if (staged) {
     if (job.stage_adopt(staged->dev, &err) != 0) return fail(res, 17, err);
+    wp.mark(1, now_usec());
   } else {
     if (job.stage(L.gi, &err) != 0) return fail(res, 17, err);
   }
   if (job.decode(&err) != 0) return fail(res, 18, err);
+  wp.mark(2, now_usec());
   if (job.merge(&err) != 0) return fail(res, 19, err);
+  wp.mark(3, now_usec());
   if (job.dedup(d, &err) != 0) return fail(res, 20, err);
+  wp.mark(4, now_usec());
 
   size_t nsurv = job.num_survivors();
   const auto& klen = job.plan_klen();
@@ -323,6 +341,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   if (nsurv > 0 &&
       job.plan_all(base_for_plan, &plan_next, &plan_meta, &err) != 0)
     return fail(res, 28, err);
+  wp.mark(5, now_usec());
 
   // grandparent-aware file cutting (compaction_outputs.cc:231-352): per-
   // survivor boundary positions from the GPU + host FSM state
@@ -410,8 +429,10 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       plan_usec += now_usec() - tp0;
       if (blocks.empty()) break;
       std::vector<uint32_t> csizes;
+      wp.mark(11, now_usec());
       if (job.emit_blocks(blocks, o, &csizes, &err) != 0)
         return fail(res, 21, err);
+      wp.mark(6, now_usec());
       // walk csizes: find the cut (ShouldStopBefore semantics,
       // compaction_outputs.cc:231-352).  Without grandparents the file size
       // only changes at block flushes, so the walk is block-level and the
@@ -542,6 +563,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       std::vector<std::string> cfirst, clast;
       if (job.fetch_block_keys(0, take, &cfirst, &clast, &err) != 0)
         return fail(res, 23, err);
+      wp.mark(7, now_usec());
       for (size_t b = 0; b < take; b++) {
         handles.push_back({old + outoff[b], csizes[b]});
         first_keys.push_back(cfirst[b]);
@@ -559,6 +581,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     // partial block after a cut (1 entry for pure size cuts; up to a full
     // block's worth for grandparent-rule cuts)
     uint64_t file_first = s;
+    wp.mark(11, now_usec());
     if (cut && partial_count > 0) {
       std::vector<std::pair<std::string, std::string>> kvs;
       if (job.gather_entries(partial_first, partial_count, &kvs, &err) != 0)
@@ -595,6 +618,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
         if (vt == 0 || vt == 7) n_tomb++; // kTypeDeletion / kTypeSingleDeletion
       }
     }
+    wp.mark(8, now_usec());
     if (cut) cur = cut_entry;
     uint64_t file_count = cur - s;
     if (handles.empty()) break; // nothing left
@@ -705,15 +729,24 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       return 0;
     }));
     write_usec += now_usec() - tw1;
+    wp.mark(9, now_usec());
     s = cur;
   }
   {
     uint64_t tj2 = now_usec();
+    wp.mark(11, now_usec());
     for (auto& w : writers)
       if (w.get() != 0) return fail(res, 26, "output file tail/write failed");
     write_usec += now_usec() - tj2;
+    wp.mark(10, now_usec());
   }
   job.drain_d2h();
+  if (getenv("DCW_PHASE_DEBUG")) {
+    fprintf(stderr, "[phase]");
+    for (int i = 0; i < 12; i++)
+      fprintf(stderr, " %s=%.1fms", wp.names[i], wp.us[i] / 1000.0);
+    fprintf(stderr, " total=%.1fms\n", (now_usec() - t_start) / 1000.0);
+  }
 
   res->num_files = (uint32_t)out_files.size();
   res->files = (dcw_output_file*)malloc(sizeof(dcw_output_file) * out_files.size());
