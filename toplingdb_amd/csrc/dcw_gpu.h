@@ -102,7 +102,11 @@ class GpuJob {
   int gp_positions(const dcw_job_desc* d, std::vector<uint32_t>* pos,
                    std::vector<uint8_t>* nback, std::string* err);
   // min/max sequence + tombstone count over survivor range (post zeroing)
-  // per-block seq stats from the records prefetched by emit_blocks
+  // per-block records prefetched by emit_blocks: 96 B each =
+  // [klen_first u8 | first ikey (<=32) | klen_last u8 at 32 | last ikey |
+  //  minseq u64 at 64 | maxseq u64 at 72 | n_tombstones u64 at 80]
+  static constexpr size_t kBlkStatStride = 96;
+  const uint8_t* chunk_stats() const;
   void block_stats(size_t b, uint64_t* mn, uint64_t* mx, uint64_t* tomb);
   int seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* mx,
                  uint64_t* n_tombstones, std::string* err);

@@ -2261,6 +2261,11 @@ int GpuJob::fetch_block_keys(size_t b0, size_t b1,
   return 0;
 }
 
+const uint8_t* GpuJob::chunk_stats() const {
+  static_assert(GpuJob::kBlkStatStride == BLKSTAT_STRIDE, "stride mismatch");
+  return (const uint8_t*)p_->h_keys;
+}
+
 void GpuJob::block_stats(size_t b, uint64_t* mn, uint64_t* mx, uint64_t* tomb) {
   const uint8_t* o = (const uint8_t*)p_->h_keys + b * BLKSTAT_STRIDE;
   memcpy(mn, o + 64, 8);

@@ -189,6 +189,13 @@ struct RawBuf {
   }
 };
 
+// fixed-size block boundary key (ikey <= 24 B in the worker envelope);
+// avoids per-block heap strings on the main thread
+struct BKey {
+  uint8_t len = 0;
+  char b[32];
+};
+
 int fail(dcw_job_result* res, int code, const std::string& msg) {
   res->status = code;
   snprintf(res->error, sizeof(res->error), "%s", msg.c_str());
@@ -280,10 +287,11 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   uint64_t t_start = now_usec();
   // fine-grained wall attribution (DCW_PHASE_DEBUG=1 prints at job end)
   struct WallProf {
-    const char* names[12] = {"load",  "stage", "decode", "merge",
+    const char* names[13] = {"load",  "stage", "decode", "merge",
                              "dedup", "plan",  "emit",   "walkpack",
-                             "gather", "tailspawn", "join", "other"};
-    uint64_t us[12] = {0};
+                             "gather", "tailspawn", "join", "other",
+                             "chainwalk"};
+    uint64_t us[13] = {0};
     uint64_t t_last;
     void mark(int slot, uint64_t now) {
       us[slot] += now - t_last;
@@ -387,7 +395,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     std::vector<void*> pend;
     image.reserve(d->target_file_size + (d->target_file_size >> 2) + (2u << 20));
     std::vector<SstIndexEntry> handles;
-    std::vector<std::string> first_keys, last_keys;
+    std::vector<BKey> first_keys, last_keys;
     std::vector<uint32_t> block_counts;
     uint64_t mn_seq = ~0ull, mx_seq = 0, n_tomb = 0; // file seq stats
     size_t cur = s;
@@ -427,6 +435,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
         }
       }
       plan_usec += now_usec() - tp0;
+      wp.mark(12, now_usec());
       if (blocks.empty()) break;
       std::vector<uint32_t> csizes;
       wp.mark(11, now_usec());
@@ -560,21 +569,28 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
           }
         }
       }
-      std::vector<std::string> cfirst, clast;
-      if (job.fetch_block_keys(0, take, &cfirst, &clast, &err) != 0)
-        return fail(res, 23, err);
-      wp.mark(7, now_usec());
+      const uint8_t* stats = job.chunk_stats();
+      if (!stats) return fail(res, 23, "no prefetched chunk stats");
       for (size_t b = 0; b < take; b++) {
         handles.push_back({old + outoff[b], csizes[b]});
-        first_keys.push_back(cfirst[b]);
-        last_keys.push_back(clast[b]);
+        const uint8_t* r = stats + b * GpuJob::kBlkStatStride;
+        BKey fk, lk;
+        fk.len = r[0];
+        memcpy(fk.b, r + 1, fk.len);
+        lk.len = r[32];
+        memcpy(lk.b, r + 33, lk.len);
+        first_keys.push_back(fk);
+        last_keys.push_back(lk);
         block_counts.push_back(blocks[b].count);
         uint64_t bmn, bmx, bt;
-        job.block_stats(b, &bmn, &bmx, &bt);
+        memcpy(&bmn, r + 64, 8);
+        memcpy(&bmx, r + 72, 8);
+        memcpy(&bt, r + 80, 8);
         if (bmn < mn_seq) mn_seq = bmn;
         if (bmx > mx_seq) mx_seq = bmx;
         n_tomb += bt;
       }
+      wp.mark(7, now_usec());
       cur = take ? blocks[take - 1].first + blocks[take - 1].count : cur;
       if (take < blocks.size()) break; // cut decided inside this chunk
     }
@@ -605,8 +621,13 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       }
       image.append(pblk.data(), pblk.size());
       handles.push_back(h);
-      first_keys.push_back(kvs.front().first);
-      last_keys.push_back(kvs.back().first);
+      BKey fk, lk;
+      fk.len = (uint8_t)kvs.front().first.size();
+      memcpy(fk.b, kvs.front().first.data(), fk.len);
+      lk.len = (uint8_t)kvs.back().first.size();
+      memcpy(lk.b, kvs.back().first.data(), lk.len);
+      first_keys.push_back(fk);
+      last_keys.push_back(lk);
       block_counts.push_back(partial_count);
       for (auto& kv : kvs) {
         uint64_t tag;
@@ -644,7 +665,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       RawBuf image;
       std::vector<void*> pend; // D2H completion events for image regions
       std::vector<SstIndexEntry> handles;
-      std::vector<std::string> first_keys, last_keys;
+      std::vector<BKey> first_keys, last_keys;
       uint64_t file_first, file_count, mn_seq, mx_seq, n_tomb;
       std::string path;
     };
@@ -678,13 +699,12 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       std::vector<std::string> seps(nb);
       bool sep_key_plus_seq = false;
       for (size_t b = 0; b < nb; b++) {
-        std::string sep = tj->last_keys[b];
+        std::string sep(tj->last_keys[b].b, tj->last_keys[b].len);
         if (b + 1 < nb) {
-          shorten_separator(sep, (const uint8_t*)tj->first_keys[b + 1].data(),
-                            tj->first_keys[b + 1].size());
-          size_t su = sep.size() - 8, nu = tj->first_keys[b + 1].size() - 8;
-          if (su == nu &&
-              memcmp(sep.data(), tj->first_keys[b + 1].data(), su) == 0)
+          const BKey& nf = tj->first_keys[b + 1];
+          shorten_separator(sep, (const uint8_t*)nf.b, nf.len);
+          size_t su = sep.size() - 8, nu = (size_t)nf.len - 8;
+          if (su == nu && memcmp(sep.data(), nf.b, su) == 0)
             sep_key_plus_seq = true;
         }
         seps[b] = sep;
@@ -712,12 +732,12 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       snprintf(of.path, sizeof(of.path), "%s", tj->path.c_str());
       of.file_number = tj->o.orig_file_number;
       of.file_size = tj->image.len;
-      const std::string& smallest = tj->first_keys.front();
-      const std::string& largest = tj->last_keys.back();
-      of.smallest_len = (uint32_t)smallest.size();
-      memcpy(of.smallest_ikey, smallest.data(), std::min<size_t>(64, smallest.size()));
-      of.largest_len = (uint32_t)largest.size();
-      memcpy(of.largest_ikey, largest.data(), std::min<size_t>(64, largest.size()));
+      const BKey& smallest = tj->first_keys.front();
+      const BKey& largest = tj->last_keys.back();
+      of.smallest_len = smallest.len;
+      memcpy(of.smallest_ikey, smallest.b, smallest.len);
+      of.largest_len = largest.len;
+      memcpy(of.largest_ikey, largest.b, largest.len);
       of.smallest_seqno = tj->mn_seq == ~0ull ? 0 : tj->mn_seq;
       of.largest_seqno = tj->mx_seq;
       of.num_entries = tj->file_count;
@@ -743,7 +763,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   job.drain_d2h();
   if (getenv("DCW_PHASE_DEBUG")) {
     fprintf(stderr, "[phase]");
-    for (int i = 0; i < 12; i++)
+    for (int i = 0; i < 13; i++)
       fprintf(stderr, " %s=%.1fms", wp.names[i], wp.us[i] / 1000.0);
     fprintf(stderr, " total=%.1fms\n", (now_usec() - t_start) / 1000.0);
   }
