@@ -13,6 +13,8 @@
 #include <sys/stat.h>
 
 #include <cstdio>
+#include <fcntl.h>
+#include <unistd.h>
 #include <cstring>
 #include <future>
 #include <mutex>
@@ -722,11 +724,51 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       std::string tail =
           build_tail(tj->o, st, tj->handles, seps, !sep_key_plus_seq, tail_start);
       tj->image.append(tail.data(), tail.size());
-      FILE* f = fopen(tj->path.c_str(), "wb");
-      if (!f) return -1;
-      size_t w = fwrite(tj->image.p, 1, tj->image.len, f);
-      fclose(f);
-      if (w != tj->image.len) return -1;
+      // parallel segmented write: tmpfs writes are page-clear + memcpy
+      // bound per thread, and the last file's write sits on the job's
+      // critical path
+      int fd = open(tj->path.c_str(), O_WRONLY | O_CREAT | O_TRUNC, 0644);
+      if (fd < 0) return -1;
+      size_t len = tj->image.len;
+      if (ftruncate(fd, (off_t)len) != 0) {
+        close(fd);
+        return -1;
+      }
+      const size_t kSeg = 16u << 20;
+      size_t nseg = (len + kSeg - 1) / kSeg;
+      if (nseg > 4) nseg = 4;
+      size_t seg = (len + nseg - 1) / nseg;
+      std::vector<std::future<bool>> segw;
+      for (size_t si = 1; si < nseg; si++) {
+        size_t off = si * seg;
+        size_t cnt = off < len ? std::min(seg, len - off) : 0;
+        segw.emplace_back(std::async(std::launch::async, [fd, off, cnt, tj]() {
+          size_t done = 0;
+          while (done < cnt) {
+            ssize_t w = pwrite(fd, tj->image.p + off + done, cnt - done,
+                               (off_t)(off + done));
+            if (w <= 0) return false;
+            done += (size_t)w;
+          }
+          return true;
+        }));
+      }
+      bool ok = true;
+      {
+        size_t cnt = std::min(seg, len);
+        size_t done = 0;
+        while (done < cnt) {
+          ssize_t w = pwrite(fd, tj->image.p + done, cnt - done, (off_t)done);
+          if (w <= 0) {
+            ok = false;
+            break;
+          }
+          done += (size_t)w;
+        }
+      }
+      for (auto& f2 : segw) ok = f2.get() && ok;
+      close(fd);
+      if (!ok) return -1;
       dcw_output_file of;
       memset(&of, 0, sizeof(of));
       snprintf(of.path, sizeof(of.path), "%s", tj->path.c_str());
