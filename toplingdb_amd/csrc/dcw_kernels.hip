@@ -140,10 +140,11 @@ __global__ void k_verify_usize(const uint8_t* __restrict__ blob,
 // one wave per block: raw -> cooperative copy; snappy -> staged through LDS
 // (compressed in + decoded out both in LDS, lane 0 runs the serial decoder,
 // all lanes copy in/out); oversized blocks fall back to the direct path.
-#define DEC_MAX 4992 // 4992*4 waves = 20 KB LDS/WG -> 8 WGs (32 decoders)/CU
+#define DEC_MAX 4992 // 2*4992*4 waves = 39.9 KB LDS/WG -> 4 WGs (16 decoders)/CU
 struct DecLds {
-  uint8_t out[DEC_MAX]; // decoded output staged in LDS (byte-random writes);
-                        // compressed input is read from global (L1-resident)
+  uint8_t in[DEC_MAX]; // compressed input + decoded output both staged in
+  uint8_t out[DEC_MAX]; // LDS: the serial byte decoder is LDS-latency bound
+                        // (a global-input variant measured ~10% slower)
 };
 __device__ __forceinline__ void wave_lds_sync2() {
   __builtin_amdgcn_s_waitcnt(0);
@@ -174,9 +175,14 @@ __global__ __launch_bounds__(256) void k_decompress(
       }
       if (lane == 0)
         for (uint32_t t = n & ~15u; t < n; t++) dst[t] = src[t];
-    } else if (usize[i] <= DEC_MAX) {
+    } else if (n <= DEC_MAX && usize[i] <= DEC_MAX) {
+      for (uint32_t t = lane * 4; t < n; t += WAVE * 4) {
+        uint32_t chunk = n - t < 4 ? n - t : 4;
+        for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = src[t + x];
+      }
+      wave_lds_sync2();
       if (lane == 0) {
-        if (snappy_uncompress(src, n, L.out, usize[i]) != usize[i])
+        if (snappy_uncompress(L.in, n, L.out, usize[i]) != usize[i])
           set_err(err_flag, DE_SNAPPY);
       }
       wave_lds_sync2();
