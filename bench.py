@@ -59,6 +59,10 @@ def main():
     ap.add_argument("--cpu-baseline-runs", type=int, default=4,
                     help="bounded oracle sample (number of input runs)")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
+    ap.add_argument("--jobs-in-flight", type=int, default=2,
+                    help="concurrent compaction jobs per GPU (the production "
+                         "dcompact worker runs concurrent jobs per node, "
+                         "BASELINE.json configs[4]); 1 = strictly sequential")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -95,24 +99,40 @@ def main():
     log("rank %d: generated %d runs, %.1f MiB in %.1fs" %
         (rank, len(runs), in_bytes / 2**20, time.time() - t_gen))
 
-    def job(staged=0):
-        return dcw.make_job(runs, outd, compression=args.compression,
+    jif = max(1, args.jobs_in_flight)
+    slot_dirs = []
+    for i in range(jif):
+        d = os.path.join(work, "out%d" % i)
+        os.makedirs(d, exist_ok=True)
+        slot_dirs.append(d)
+
+    def job(slot, staged=0):
+        return dcw.make_job(runs, slot_dirs[slot], compression=args.compression,
                             target_file_size=64 << 20, bottommost_level=1,
                             staged_handle=staged)
 
-    handle = dcw.stage_inputs(job())
+    handle = dcw.stage_inputs(job(0))
     lib = dcw.lib()
     lib.dcw_kernel_stats_json.restype = ctypes.c_int32
     lib.dcw_kernel_stats_json.argtypes = [ctypes.c_char_p, ctypes.c_uint32]
 
-    def step():
-        res = dcw.execute(job(staged=handle))
+    def step(i):
+        # slot picks the output dir; concurrent jobs share the staged
+        # (read-only) inputs but nothing else
+        res = dcw.execute(job(i % jif, staged=handle))
         return res
+
+    from concurrent.futures import ThreadPoolExecutor
+    pool = ThreadPoolExecutor(max_workers=jif)
+
+    def run_steps(k):
+        futs = [pool.submit(step, i) for i in range(k)]
+        return [f.result() for f in futs]
 
     # warmup
     last = None
-    for _ in range(args.warmup):
-        last = step()
+    rs = run_steps(max(args.warmup, jif))
+    last = rs[-1]
     lib.dcw_kernel_stats_reset()
 
     def barrier():
@@ -122,8 +142,7 @@ def main():
 
     barrier()
     t0 = time.time()
-    for _ in range(args.steps):
-        last = step()
+    last = run_steps(args.steps)[-1]
     barrier()
     elapsed = time.time() - t0
     # MAX over ranks
@@ -198,6 +217,7 @@ def main():
             "compression": "snappy" if args.compression else "none",
             "target_file_size": 64 << 20,
             "output": "tmpfs (/dev/shm), D2H + file write inside the timed region",
+            "jobs_in_flight": jif,
         },
         "roofline": roofline,
         "cpu_baseline": cpu_baseline,

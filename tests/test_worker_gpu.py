@@ -231,6 +231,29 @@ def test_large_values_multi_chunk(tmp_path):
     assert len(rg["files"]) > 1
 
 
+def test_concurrent_jobs_bit_exact(tmp_path):
+    # two different jobs executing concurrently from two threads must each
+    # produce byte-identical output to their sequential oracle runs (the
+    # worker pools GpuJob instances; production runs concurrent jobs)
+    import concurrent.futures
+    runs_a = gen_runs(tmp_path, 4, 60000, seed0=11, compression=1)
+    runs_b = gen_runs(tmp_path, 4, 60000, seed0=77, compression=1)
+    dirs = {}
+    for tag in ("ga", "gb", "oa", "ob"):
+        d = tmp_path / tag
+        d.mkdir()
+        dirs[tag] = str(d)
+    kw = dict(compression=1, target_file_size=8 << 20)
+    with concurrent.futures.ThreadPoolExecutor(2) as ex:
+        fa = ex.submit(dcw.execute, dcw.make_job(runs_a, dirs["ga"], **kw))
+        fb = ex.submit(dcw.execute, dcw.make_job(runs_b, dirs["gb"], **kw))
+        ra, rb = fa.result(), fb.result()
+    oa = oracle.execute(oracle.make_job(runs_a, dirs["oa"], **kw))
+    ob = oracle.execute(oracle.make_job(runs_b, dirs["ob"], **kw))
+    assert_identical(ra, oa)
+    assert_identical(rb, ob)
+
+
 def test_grandparent_cuts_match_oracle(tmp_path):
     # grandparent-aware file cutting (ShouldStopBefore boundary rules,
     # compaction_outputs.cc:231-352): GPU worker vs oracle, bit-exact,

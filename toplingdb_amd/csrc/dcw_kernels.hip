@@ -1359,6 +1359,48 @@ struct GpuJob::Impl {
   int cur_outslot = 0;
   void* h_keys = nullptr; // pinned block-stats records (BLKSTAT_STRIDE each)
   uint64_t h_keys_cap = 0;
+  // pinned staging ring for small metadata H2D copies: stage into a pinned
+  // slot and hipMemcpyAsync on this job's own stream.  Keeps every copy
+  // ordered within the job without NULL-stream synchronization, which
+  // would serialize independent jobs sharing the device.  Slot reuse
+  // waits on the slot's event (ring depth 8 >> in-flight copies).
+  struct MetaSlot {
+    void* p = nullptr;
+    size_t cap = 0;
+    hipEvent_t ev = nullptr;
+    bool pending = false;
+  } meta[8];
+  int meta_cur = 0;
+  hipError_t h2d_meta(void* dst, const void* src, size_t n) {
+    if (n == 0) return hipSuccess;
+    MetaSlot& s = meta[meta_cur];
+    meta_cur = (meta_cur + 1) & 7;
+    if (!s.ev) {
+      hipError_t e = hipEventCreate(&s.ev);
+      if (e != hipSuccess) return e;
+    }
+    if (s.pending) {
+      hipError_t e = hipEventSynchronize(s.ev);
+      if (e != hipSuccess) return e;
+      s.pending = false;
+    }
+    if (s.cap < n) {
+      if (s.p) (void)hipHostFree(s.p);
+      s.p = nullptr;
+      s.cap = 0;
+      size_t c = n + n / 4 + 64;
+      hipError_t e = hipHostMalloc(&s.p, c);
+      if (e != hipSuccess) return e;
+      s.cap = c;
+    }
+    memcpy(s.p, src, n);
+    hipError_t e = hipMemcpyAsync(dst, s.p, n, hipMemcpyHostToDevice, stream);
+    if (e != hipSuccess) return e;
+    e = hipEventRecord(s.ev, stream);
+    if (e != hipSuccess) return e;
+    s.pending = true;
+    return hipSuccess;
+  }
   std::vector<KEv> kpending;
   void kbegin(const char* n, double bytes) {
     KEv e{n, bytes, nullptr, nullptr};
@@ -1499,8 +1541,8 @@ static double ms_between(hipEvent_t a, hipEvent_t b) {
 }
 
 GpuJob::GpuJob() : p_(new Impl) {
-  (void)hipStreamCreate(&p_->stream);
-  (void)hipStreamCreate(&p_->d2h_stream);
+  (void)hipStreamCreateWithFlags(&p_->stream, hipStreamNonBlocking);
+  (void)hipStreamCreateWithFlags(&p_->d2h_stream, hipStreamNonBlocking);
   for (auto& s : p_->outslots) {
     (void)hipEventCreate(&s.t0);
     (void)hipEventCreate(&s.done);
@@ -1588,6 +1630,11 @@ GpuJob::~GpuJob() {
     (void)hipEventDestroy(s.t0);
     (void)hipEventDestroy(s.done);
   }
+  for (auto& s : p->meta) {
+    if (s.pending) (void)hipEventSynchronize(s.ev);
+    if (s.p) (void)hipHostFree(s.p);
+    if (s.ev) (void)hipEventDestroy(s.ev);
+  }
   if (p->h_keys) (void)hipHostFree(p->h_keys);
   if (p->d2h_stream) (void)hipStreamDestroy(p->d2h_stream);
   if (p->stream) (void)hipStreamDestroy(p->stream);
@@ -1620,10 +1667,8 @@ int GpuJob::stage(const GpuInputs& in, std::string* err) {
   }
   HIPCHK(hipMalloc(&p->d_boff, sizeof(uint64_t) * p->n_blocks));
   HIPCHK(hipMalloc(&p->d_bsize, sizeof(uint32_t) * p->n_blocks));
-  HIPCHK(hipMemcpy(p->d_boff, boff.data(), sizeof(uint64_t) * p->n_blocks,
-                   hipMemcpyHostToDevice));
-  HIPCHK(hipMemcpy(p->d_bsize, bsize.data(), sizeof(uint32_t) * p->n_blocks,
-                   hipMemcpyHostToDevice));
+  HIPCHK(p->h2d_meta(p->d_boff, boff.data(), sizeof(uint64_t) * p->n_blocks));
+  HIPCHK(p->h2d_meta(p->d_bsize, bsize.data(), sizeof(uint32_t) * p->n_blocks));
   if (!p->d_crc) HIPCHK(hipMalloc(&p->d_crc, sizeof(Crc32cTables)));
   HIPCHK(hipMemcpyAsync(p->d_crc, &g_crc, sizeof(Crc32cTables),
                         hipMemcpyHostToDevice, p->stream));
@@ -1715,8 +1760,7 @@ int GpuJob::decode(std::string* err) {
   }
   p->ublob_size = acc;
   HIPCHK(p->ens((void**)&p->d_uoff, sizeof(uint64_t) * nb));
-  HIPCHK(hipMemcpy(p->d_uoff, uoff.data(), sizeof(uint64_t) * nb,
-                   hipMemcpyHostToDevice));
+  HIPCHK(p->h2d_meta(p->d_uoff, uoff.data(), sizeof(uint64_t) * nb));
   HIPCHK(p->ens((void**)&p->d_ublob, acc ? acc : 1));
   {
     std::lock_guard<std::mutex> lk(g_kmu);
@@ -1757,10 +1801,8 @@ int GpuJob::decode(std::string* err) {
   HIPCHK(p->ens((void**)&p->d_iv_local, sizeof(uint32_t) * niv));
   HIPCHK(p->ens((void**)&p->d_iv_cnt, sizeof(uint32_t) * niv));
   HIPCHK(p->ens((void**)&p->d_iv_base, sizeof(uint32_t) * niv));
-  HIPCHK(hipMemcpy(p->d_iv_block, iv_block.data(), sizeof(uint32_t) * niv,
-                   hipMemcpyHostToDevice));
-  HIPCHK(hipMemcpy(p->d_iv_local, iv_local.data(), sizeof(uint32_t) * niv,
-                   hipMemcpyHostToDevice));
+  HIPCHK(p->h2d_meta(p->d_iv_block, iv_block.data(), sizeof(uint32_t) * niv));
+  HIPCHK(p->h2d_meta(p->d_iv_local, iv_local.data(), sizeof(uint32_t) * niv));
   p->kbegin("count_entries", (double)p->ublob_size);
   hipLaunchKernelGGL(k_count_entries, dim3(grid_for(niv)), dim3(256), 0, p->stream,
                      p->d_ublob, p->d_uoff, p->d_usize, p->d_nrestarts,
@@ -1792,8 +1834,7 @@ int GpuJob::decode(std::string* err) {
                                                 : total_entries);
   }
   p->run_entry_begin.push_back(total_entries);
-  HIPCHK(hipMemcpy(p->d_iv_base, iv_base.data(), sizeof(uint32_t) * niv,
-                   hipMemcpyHostToDevice));
+  HIPCHK(p->h2d_meta(p->d_iv_base, iv_base.data(), sizeof(uint32_t) * niv));
   HIPCHK(p->ens((void**)&p->d_ent[0], sizeof(ulong4) * total_entries));
   HIPCHK(p->ens((void**)&p->d_ent[1], sizeof(ulong4) * total_entries));
   HIPCHK(p->ens((void**)&p->d_voff, sizeof(uint64_t) * total_entries));
@@ -1889,8 +1930,7 @@ static int scan_u8(GpuJob::Impl* p, const uint8_t* d_in, uint64_t n,
     bs[i] = (uint32_t)acc;
     acc += v;
   }
-  HIPCHK(hipMemcpy(d_bs, bs.data(), sizeof(uint32_t) * nblk,
-                   hipMemcpyHostToDevice));
+  HIPCHK(p->h2d_meta(d_bs, bs.data(), sizeof(uint32_t) * nblk));
   hipLaunchKernelGGL(k_scan_add_base, dim3((uint32_t)nblk), dim3(1024), 0,
                      p->stream, d_out, n, d_bs);
   // `bs` is a local: the async H2D must complete before it dies (ROCm may
@@ -1934,8 +1974,7 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   uint64_t* d_snaps = nullptr;
   if (d->num_snapshots) {
     HIPCHK(hipMalloc(&d_snaps, sizeof(uint64_t) * d->num_snapshots));
-    HIPCHK(hipMemcpy(d_snaps, d->snapshots, sizeof(uint64_t) * d->num_snapshots,
-                     hipMemcpyHostToDevice));
+    HIPCHK(p->h2d_meta(d_snaps, d->snapshots, sizeof(uint64_t) * d->num_snapshots));
   }
   P.snapshots = d_snaps;
   // levels below -> normkeys
@@ -1963,14 +2002,13 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
     HIPCHK(p->ens((void**)&p->d_lb_sm1, nb));
     HIPCHK(p->ens((void**)&p->d_lb_lg0, nb));
     HIPCHK(p->ens((void**)&p->d_lb_lg1, nb));
-    HIPCHK(hipMemcpy(p->d_lb_sm0, sm0.data(), nb, hipMemcpyHostToDevice));
-    HIPCHK(hipMemcpy(p->d_lb_sm1, sm1.data(), nb, hipMemcpyHostToDevice));
-    HIPCHK(hipMemcpy(p->d_lb_lg0, lg0.data(), nb, hipMemcpyHostToDevice));
-    HIPCHK(hipMemcpy(p->d_lb_lg1, lg1.data(), nb, hipMemcpyHostToDevice));
+    HIPCHK(p->h2d_meta(p->d_lb_sm0, sm0.data(), nb));
+    HIPCHK(p->h2d_meta(p->d_lb_sm1, sm1.data(), nb));
+    HIPCHK(p->h2d_meta(p->d_lb_lg0, lg0.data(), nb));
+    HIPCHK(p->h2d_meta(p->d_lb_lg1, lg1.data(), nb));
   }
   HIPCHK(p->ens((void**)&p->d_lb_beg, sizeof(uint32_t) * lbeg.size()));
-  HIPCHK(hipMemcpy(p->d_lb_beg, lbeg.data(), sizeof(uint32_t) * lbeg.size(),
-                   hipMemcpyHostToDevice));
+  HIPCHK(p->h2d_meta(p->d_lb_beg, lbeg.data(), sizeof(uint32_t) * lbeg.size()));
   P.lb_sm_k0 = p->d_lb_sm0;
   P.lb_sm_k1 = p->d_lb_sm1;
   P.lb_lg_k0 = p->d_lb_lg0;
@@ -2129,8 +2167,7 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   ENSURE(p->d_ebsize, p->eb_cap, sizeof(uint32_t) * nb * 2 + nb); // bsize+csum+btype
   p->d_ecsum = p->d_ebsize + nb;
   p->d_ebtype = (uint8_t*)(p->d_ecsum + nb);
-  HIPCHK(hipMemcpy(p->d_bds, bds.data(), sizeof(EmitBlockDesc) * nb,
-                   hipMemcpyHostToDevice));
+  HIPCHK(p->h2d_meta(p->d_bds, bds.data(), sizeof(EmitBlockDesc) * nb));
   p->kbegin("emit", 2.0 * (double)uout);
   hipLaunchKernelGGL(k_emit, dim3(nb < 4096 ? nb : 4096), dim3(256), 0, p->stream,
                      p->d_bds, nb, p->d_sk0, p->d_sk1, p->d_stag, p->d_svoff,
@@ -2167,7 +2204,12 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   // host needs no further GPU round trip for them
   HIPCHK(p->ens((void**)&p->d_scratch_keys, (uint64_t)nb * BLKSTAT_STRIDE));
   if (p->h_keys_cap < (uint64_t)nb * BLKSTAT_STRIDE) {
-    if (p->h_keys) (void)hipHostFree(p->h_keys);
+    for (auto& s : p->meta) {
+    if (s.pending) (void)hipEventSynchronize(s.ev);
+    if (s.p) (void)hipHostFree(s.p);
+    if (s.ev) (void)hipEventDestroy(s.ev);
+  }
+  if (p->h_keys) (void)hipHostFree(p->h_keys);
     p->h_keys_cap = (uint64_t)nb * BLKSTAT_STRIDE * 5 / 4;
     HIPCHK(hipHostMalloc(&p->h_keys, p->h_keys_cap));
   }
@@ -2202,8 +2244,7 @@ int GpuJob::pack_into(size_t b0, size_t b1, const std::vector<uint64_t>& outoff,
   uint64_t* d_outoff = (uint64_t*)p->d_outoff;
   // synchronous copy: NULL-stream ordering with the (blocking) pipeline
   // stream, no pageable-async lifetime/ordering hazards
-  HIPCHK(hipMemcpy(d_outoff, outoff.data(), sizeof(uint64_t) * nb,
-                   hipMemcpyHostToDevice));
+  HIPCHK(p->h2d_meta(d_outoff, outoff.data(), sizeof(uint64_t) * nb));
   Impl::OutSlot& slot = p->outslots[p->cur_outslot];
   p->cur_outslot ^= 1;
   if (slot.pending) { // slot reused two files later; transfer long done
@@ -2291,8 +2332,7 @@ int GpuJob::gather_entries(uint64_t first, uint32_t count,
   HIPCHK(p->ens(&p->d_scratch_gather, acc));
   uint64_t* d_recoff = (uint64_t*)p->d_scratch_recoff;
   uint8_t* d_out = (uint8_t*)p->d_scratch_gather;
-  HIPCHK(hipMemcpy(d_recoff, recoff.data(), sizeof(uint64_t) * count,
-                   hipMemcpyHostToDevice));
+  HIPCHK(p->h2d_meta(d_recoff, recoff.data(), sizeof(uint64_t) * count));
   hipLaunchKernelGGL(k_gather_range, dim3(grid_for(count)), dim3(256), 0,
                      p->stream, p->d_sk0, p->d_sk1, p->d_stag, p->d_svoff,
                      p->d_svlen, p->d_sklen, p->d_ublob, first, count, d_recoff,
@@ -2340,11 +2380,11 @@ int GpuJob::gp_positions(const dcw_job_desc* d, std::vector<uint32_t>* pos,
   HIPCHK(p->ens(&p->d_gp_tie, ng));
   HIPCHK(p->ens(&p->d_gp_pos, n * 4));
   HIPCHK(p->ens(&p->d_gp_nback, n));
-  HIPCHK(hipMemcpy(p->d_gp_sm0, sm0.data(), ng * 8, hipMemcpyHostToDevice));
-  HIPCHK(hipMemcpy(p->d_gp_sm1, sm1.data(), ng * 8, hipMemcpyHostToDevice));
-  HIPCHK(hipMemcpy(p->d_gp_lg0, lg0.data(), ng * 8, hipMemcpyHostToDevice));
-  HIPCHK(hipMemcpy(p->d_gp_lg1, lg1.data(), ng * 8, hipMemcpyHostToDevice));
-  HIPCHK(hipMemcpy(p->d_gp_tie, tie.data(), ng, hipMemcpyHostToDevice));
+  HIPCHK(p->h2d_meta(p->d_gp_sm0, sm0.data(), ng * 8));
+  HIPCHK(p->h2d_meta(p->d_gp_sm1, sm1.data(), ng * 8));
+  HIPCHK(p->h2d_meta(p->d_gp_lg0, lg0.data(), ng * 8));
+  HIPCHK(p->h2d_meta(p->d_gp_lg1, lg1.data(), ng * 8));
+  HIPCHK(p->h2d_meta(p->d_gp_tie, tie.data(), ng));
   hipLaunchKernelGGL(k_gp_positions, dim3(grid_for(n)), dim3(256), 0, p->stream,
                      p->d_sk0, p->d_sk1, n, (const uint64_t*)p->d_gp_sm0,
                      (const uint64_t*)p->d_gp_sm1, (const uint64_t*)p->d_gp_lg0,
@@ -2364,7 +2404,7 @@ int GpuJob::seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* m
   HIPCHK(p->ens(&p->d_scratch_mm, 24));
   unsigned long long* d = (unsigned long long*)p->d_scratch_mm;
   unsigned long long init[3] = {~0ull, 0, 0};
-  HIPCHK(hipMemcpy(d, init, 24, hipMemcpyHostToDevice));
+  HIPCHK(p->h2d_meta(d, init, 24));
   hipLaunchKernelGGL(k_seq_minmax, dim3(grid_for(count)), dim3(256), 0, p->stream,
                      p->d_stag, first, count, d, d + 1, d + 2);
   unsigned long long out[3];

@@ -31,6 +31,31 @@ namespace {
 
 std::mutex g_mu;
 std::mutex g_out_mu; // guards out_files slots filled by tail threads
+
+// Reusable GpuJob pool: device buffers are grow-only per job instance, so
+// pooling keeps the no-hipMalloc-per-job property while allowing several
+// jobs in flight on one device (the production dcompact worker runs
+// concurrent jobs per node; per-job streams are non-blocking so pipelines
+// interleave without cross-serialization).
+struct JobPool {
+  std::mutex mu;
+  std::vector<GpuJob*> free_jobs;
+  GpuJob* acquire() {
+    {
+      std::lock_guard<std::mutex> lk(mu);
+      if (!free_jobs.empty()) {
+        GpuJob* j = free_jobs.back();
+        free_jobs.pop_back();
+        return j;
+      }
+    }
+    return new GpuJob();
+  }
+  void put(GpuJob* j) {
+    std::lock_guard<std::mutex> lk(mu);
+    free_jobs.push_back(j);
+  }
+} g_jobs;
 bool g_inited = false;
 uint64_t g_next_stage_handle = 1;
 
@@ -275,7 +300,6 @@ void dcw_release_staged(uint64_t handle) {
 
 int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   memset(res, 0, sizeof(*res));
-  std::lock_guard<std::mutex> lk(g_mu); // one GPU job slot per process, round 1
   if (!g_inited)
     return fail(res, 10, "dcw_init not called or no gfx950 device (no CPU fallback)");
   if (d->struct_size != sizeof(dcw_job_desc))
@@ -305,16 +329,18 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
 
   // ---- read + parse inputs (host I/O) ----
   uint64_t t0 = now_usec();
-  // one cached GpuJob per process: device buffers are reused across jobs
-  // (grow-only), avoiding ~30 hipMalloc/hipFree pairs per job
-  static GpuJob* g_job = nullptr;
-  if (!g_job) g_job = new GpuJob();
-  g_job->reset();
-  GpuJob& job = *g_job;
+  struct PooledJob {
+    GpuJob* j;
+    PooledJob() : j(g_jobs.acquire()) {}
+    ~PooledJob() { g_jobs.put(j); }
+  } pj;
+  pj.j->reset();
+  GpuJob& job = *pj.j;
   uint64_t in_bytes = 0;
   LoadedInputs L;
   StagedJob* staged = nullptr;
   if (d->staged_handle) {
+    std::lock_guard<std::mutex> lk(g_mu);
     auto it = g_staged.find(d->staged_handle);
     if (it == g_staged.end()) return fail(res, 15, "bad staged handle");
     staged = it->second;
@@ -383,6 +409,15 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   std::vector<dcw_output_file> out_files;
   out_files.reserve(1024); // slots are written by tail threads; no realloc
   std::vector<std::future<int>> writers;
+  // join tail threads on EVERY exit path (they write into out_files and
+  // this stack frame's vectors; early error returns must not outlive them)
+  struct WriterJoin {
+    std::vector<std::future<int>>* w;
+    ~WriterJoin() {
+      for (auto& f : *w)
+        if (f.valid()) (void)f.get();
+    }
+  } wj{&writers};
   uint64_t total_out_bytes = 0, total_out_entries = 0;
   uint64_t plan_usec = 0, write_usec = 0;
 
