@@ -1373,6 +1373,8 @@ struct GpuJob::Impl {
   int meta_cur = 0;
   hipError_t h2d_meta(void* dst, const void* src, size_t n) {
     if (n == 0) return hipSuccess;
+    static const bool sync_meta = getenv("DCW_SYNC_META") != nullptr;
+    if (sync_meta) return hipMemcpy(dst, src, n, hipMemcpyHostToDevice);
     MetaSlot& s = meta[meta_cur];
     meta_cur = (meta_cur + 1) & 7;
     if (!s.ev) {
@@ -1541,8 +1543,10 @@ static double ms_between(hipEvent_t a, hipEvent_t b) {
 }
 
 GpuJob::GpuJob() : p_(new Impl) {
-  (void)hipStreamCreateWithFlags(&p_->stream, hipStreamNonBlocking);
-  (void)hipStreamCreateWithFlags(&p_->d2h_stream, hipStreamNonBlocking);
+  unsigned sf = getenv("DCW_BLOCKING_STREAMS") ? hipStreamDefault
+                                                : hipStreamNonBlocking;
+  (void)hipStreamCreateWithFlags(&p_->stream, sf);
+  (void)hipStreamCreateWithFlags(&p_->d2h_stream, sf);
   for (auto& s : p_->outslots) {
     (void)hipEventCreate(&s.t0);
     (void)hipEventCreate(&s.done);
