@@ -13,7 +13,9 @@
 #include <sys/stat.h>
 
 #include <cstdio>
+#include <execinfo.h>
 #include <fcntl.h>
+#include <csignal>
 #include <unistd.h>
 #include <cstring>
 #include <future>
@@ -238,7 +240,18 @@ extern "C" {
 
 const char* dcw_version(void) { return "toplingdb_amd dcompact worker r1 (gfx950)"; }
 
+static void dcw_segv_handler(int sig) {
+  void* frames[64];
+  int n = backtrace(frames, 64);
+  backtrace_symbols_fd(frames, n, 2);
+  _exit(139);
+}
+
 int32_t dcw_init(int32_t device_ordinal) {
+  if (getenv("DCW_SEGV_TRACE")) {
+    signal(SIGSEGV, dcw_segv_handler);
+    signal(SIGABRT, dcw_segv_handler);
+  }
   std::lock_guard<std::mutex> lk(g_mu);
   std::string err;
   if (gpu_init(device_ordinal, &err) != 0) {
