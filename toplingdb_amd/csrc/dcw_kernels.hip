@@ -1376,6 +1376,10 @@ struct GpuJob::Impl {
     static const bool sync_meta = getenv("DCW_SYNC_META") != nullptr;
     if (sync_meta) return hipMemcpy(dst, src, n, hipMemcpyHostToDevice);
     MetaSlot& s = meta[meta_cur];
+    static const bool dbg = getenv("DCW_META_DEBUG") != nullptr;
+    if (dbg)
+      fprintf(stderr, "[h2d_meta] slot=%d p=%p cap=%zu ev=%p pend=%d n=%zu dst=%p\n",
+              meta_cur, s.p, s.cap, (void*)s.ev, (int)s.pending, n, dst);
     meta_cur = (meta_cur + 1) & 7;
     if (!s.ev) {
       hipError_t e = hipEventCreate(&s.ev);
