@@ -1359,55 +1359,57 @@ struct GpuJob::Impl {
   int cur_outslot = 0;
   void* h_keys = nullptr; // pinned block-stats records (BLKSTAT_STRIDE each)
   uint64_t h_keys_cap = 0;
-  // pinned staging ring for small metadata H2D copies: stage into a pinned
-  // slot and hipMemcpyAsync on this job's own stream.  Keeps every copy
-  // ordered within the job without NULL-stream synchronization, which
-  // would serialize independent jobs sharing the device.  Slot reuse
-  // waits on the slot's event (ring depth 8 >> in-flight copies).
-  struct MetaSlot {
+  // Pinned metadata arena: every small H2D copy stages through a linear
+  // pinned allocation that is never reused within a job, then
+  // hipMemcpyAsync on this job's own stream.  Keeps every copy ordered
+  // within the job without NULL-stream synchronization (which would
+  // serialize independent jobs sharing the device) and without per-copy
+  // events.  arena_reset() is called from reset()/dtor when the stream
+  // is idle (job boundaries), so in-flight copies never see reuse.
+  struct MetaBlock {
     void* p = nullptr;
     size_t cap = 0;
-    hipEvent_t ev = nullptr;
-    bool pending = false;
-  } meta[8];
-  int meta_cur = 0;
+  };
+  std::vector<MetaBlock> meta_blocks;
+  size_t meta_cur_block = 0;
+  size_t meta_off = 0;
+  void arena_reset() {
+    // keep only the largest block to stay grow-only without hoarding
+    if (meta_blocks.size() > 1) {
+      size_t best = 0;
+      for (size_t i = 1; i < meta_blocks.size(); i++)
+        if (meta_blocks[i].cap > meta_blocks[best].cap) best = i;
+      for (size_t i = 0; i < meta_blocks.size(); i++)
+        if (i != best && meta_blocks[i].p) (void)hipHostFree(meta_blocks[i].p);
+      meta_blocks[0] = meta_blocks[best];
+      meta_blocks.resize(1);
+    }
+    meta_cur_block = 0;
+    meta_off = 0;
+  }
   hipError_t h2d_meta(void* dst, const void* src, size_t n) {
     if (n == 0) return hipSuccess;
     static const bool sync_meta = getenv("DCW_SYNC_META") != nullptr;
     if (sync_meta) return hipMemcpy(dst, src, n, hipMemcpyHostToDevice);
-    MetaSlot& s = meta[meta_cur];
-    static const bool dbg = getenv("DCW_META_DEBUG") != nullptr;
-    if (dbg)
-      fprintf(stderr, "[h2d_meta] slot=%d p=%p cap=%zu ev=%p pend=%d n=%zu dst=%p\n",
-              meta_cur, s.p, s.cap, (void*)s.ev, (int)s.pending, n, dst);
-    meta_cur = (meta_cur + 1) & 7;
-    if (!s.ev) {
-      hipError_t e = hipEventCreate(&s.ev);
-      if (e != hipSuccess) return e;
+    size_t need = (n + 63) & ~(size_t)63;
+    while (meta_cur_block < meta_blocks.size() &&
+           meta_off + need > meta_blocks[meta_cur_block].cap) {
+      meta_cur_block++;
+      meta_off = 0;
     }
-    if (s.pending) {
-      hipError_t e = hipEventSynchronize(s.ev);
+    if (meta_cur_block >= meta_blocks.size()) {
+      MetaBlock b;
+      b.cap = need > (1u << 20) ? need + need / 4 : (1u << 20);
+      hipError_t e = hipHostMalloc(&b.p, b.cap);
       if (e != hipSuccess) return e;
-      s.pending = false;
+      meta_blocks.push_back(b);
+      meta_off = 0;
     }
-    if (s.cap < n) {
-      if (s.p) (void)hipHostFree(s.p);
-      s.p = nullptr;
-      s.cap = 0;
-      size_t c = n + n / 4 + 64;
-      hipError_t e = hipHostMalloc(&s.p, c);
-      if (e != hipSuccess) return e;
-      s.cap = c;
-    }
-    memcpy(s.p, src, n);
-    hipError_t e = hipMemcpyAsync(dst, s.p, n, hipMemcpyHostToDevice, stream);
-    if (e != hipSuccess) return e;
-    e = hipEventRecord(s.ev, stream);
-    if (e != hipSuccess) return e;
-    s.pending = true;
-    return hipSuccess;
-  }
-  std::vector<KEv> kpending;
+    uint8_t* stagep = (uint8_t*)meta_blocks[meta_cur_block].p + meta_off;
+    meta_off += need;
+    memcpy(stagep, src, n);
+    return hipMemcpyAsync(dst, stagep, n, hipMemcpyHostToDevice, stream);
+  }  std::vector<KEv> kpending;
   void kbegin(const char* n, double bytes) {
     KEv e{n, bytes, nullptr, nullptr};
     (void)hipEventCreate(&e.a);
@@ -1575,6 +1577,9 @@ void GpuJob::drain_d2h() {
 // buffers belong to their StagedInput; owned ones are freed).
 void GpuJob::reset() {
   Impl* p = p_;
+  (void)hipStreamSynchronize(p->stream);
+  (void)hipStreamSynchronize(p->d2h_stream);
+  p->arena_reset();
   if (!p->borrowed_stage) {
     if (p->d_blob) (void)hipFree(p->d_blob);
     if (p->d_boff) (void)hipFree(p->d_boff);
@@ -1602,6 +1607,8 @@ void GpuJob::reset() {
 }
 GpuJob::~GpuJob() {
   Impl* p = p_;
+  if (p->stream) (void)hipStreamSynchronize(p->stream);
+  if (p->d2h_stream) (void)hipStreamSynchronize(p->d2h_stream);
   auto F = [](void* x) {
     if (x) (void)hipFree(x);
   };
@@ -1638,11 +1645,8 @@ GpuJob::~GpuJob() {
     (void)hipEventDestroy(s.t0);
     (void)hipEventDestroy(s.done);
   }
-  for (auto& s : p->meta) {
-    if (s.pending) (void)hipEventSynchronize(s.ev);
-    if (s.p) (void)hipHostFree(s.p);
-    if (s.ev) (void)hipEventDestroy(s.ev);
-  }
+  for (auto& b : p->meta_blocks)
+    if (b.p) (void)hipHostFree(b.p);
   if (p->h_keys) (void)hipHostFree(p->h_keys);
   if (p->d2h_stream) (void)hipStreamDestroy(p->d2h_stream);
   if (p->stream) (void)hipStreamDestroy(p->stream);
@@ -2212,11 +2216,8 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   // host needs no further GPU round trip for them
   HIPCHK(p->ens((void**)&p->d_scratch_keys, (uint64_t)nb * BLKSTAT_STRIDE));
   if (p->h_keys_cap < (uint64_t)nb * BLKSTAT_STRIDE) {
-    for (auto& s : p->meta) {
-    if (s.pending) (void)hipEventSynchronize(s.ev);
-    if (s.p) (void)hipHostFree(s.p);
-    if (s.ev) (void)hipEventDestroy(s.ev);
-  }
+    for (auto& b : p->meta_blocks)
+    if (b.p) (void)hipHostFree(b.p);
   if (p->h_keys) (void)hipHostFree(p->h_keys);
     p->h_keys_cap = (uint64_t)nb * BLKSTAT_STRIDE * 5 / 4;
     HIPCHK(hipHostMalloc(&p->h_keys, p->h_keys_cap));
