@@ -1370,9 +1370,18 @@ struct GpuJob::Impl {
     void* p = nullptr;
     size_t cap = 0;
   };
+  uint64_t canary1 = 0xdeadbeefcafef00dull;
   std::vector<MetaBlock> meta_blocks;
   size_t meta_cur_block = 0;
   size_t meta_off = 0;
+  uint64_t canary2 = 0x1122334455667788ull;
+  void check_canary(const char* where) {
+    if (canary1 != 0xdeadbeefcafef00dull || canary2 != 0x1122334455667788ull) {
+      fprintf(stderr, "[canary] DEAD at %s: c1=%016llx c2=%016llx\n", where,
+              (unsigned long long)canary1, (unsigned long long)canary2);
+      abort();
+    }
+  }
   void arena_reset() {
     // keep only the largest block to stay grow-only without hoarding
     if (meta_blocks.size() > 1) {
@@ -1388,6 +1397,7 @@ struct GpuJob::Impl {
     meta_off = 0;
   }
   hipError_t h2d_meta(void* dst, const void* src, size_t n) {
+    check_canary("h2d_meta");
     if (n == 0) return hipSuccess;
     static const bool sync_meta = getenv("DCW_SYNC_META") != nullptr;
     if (sync_meta) return hipMemcpy(dst, src, n, hipMemcpyHostToDevice);
@@ -1739,6 +1749,7 @@ int GpuJob::stage_release(StagedInput* s, std::string* err) {
 
 int GpuJob::decode(std::string* err) {
   Impl* p = p_;
+  p->check_canary("decode");
   hipEvent_t t0, t1;
   (void)hipEventCreate(&t0);
   (void)hipEventCreate(&t1);
@@ -1882,6 +1893,7 @@ int GpuJob::decode(std::string* err) {
 
 int GpuJob::merge(std::string* err) {
   Impl* p = p_;
+  p->check_canary("merge");
   hipEvent_t t0, t1;
   (void)hipEventCreate(&t0);
   (void)hipEventCreate(&t1);
@@ -2111,6 +2123,7 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
 int GpuJob::plan_all(const TableOpts& o, const uint32_t** next,
                      const uint32_t** meta, std::string* err) {
   Impl* p = p_;
+  p->check_canary("plan_all");
   uint64_t n = p->n_surv;
   // pinned host landing buffers (grow-only)
   if (p->h_plan_cap < n * 8) {
@@ -2151,6 +2164,7 @@ int GpuJob::plan_all(const TableOpts& o, const uint32_t** next,
 int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts& o,
                         std::vector<uint32_t>* comp_sizes, std::string* err) {
   Impl* p = p_;
+  p->check_canary("emit");
   hipEvent_t t0, t1;
   (void)hipEventCreate(&t0);
   (void)hipEventCreate(&t1);
@@ -2244,6 +2258,7 @@ int GpuJob::pack_into(size_t b0, size_t b1, const std::vector<uint64_t>& outoff,
                       uint8_t* host_dst, size_t total_bytes, void** done_event,
                       std::string* err) {
   Impl* p = p_;
+  p->check_canary("pack");
   uint32_t nb = (uint32_t)(b1 - b0);
   if (nb == 0) {
     *done_event = nullptr;
@@ -2328,6 +2343,7 @@ int GpuJob::gather_entries(uint64_t first, uint32_t count,
                            std::vector<std::pair<std::string, std::string>>* kvs,
                            std::string* err) {
   Impl* p = p_;
+  p->check_canary("gather");
   kvs->clear();
   if (!count) return 0;
   // record layout: [klen u8][key][vlen u32][value]
