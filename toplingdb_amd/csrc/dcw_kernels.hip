@@ -1417,6 +1417,12 @@ struct GpuJob::Impl {
     }
     uint8_t* stagep = (uint8_t*)meta_blocks[meta_cur_block].p + meta_off;
     meta_off += need;
+    static const bool dbg = getenv("DCW_META_DEBUG") != nullptr;
+    if (dbg)
+      fprintf(stderr,
+              "[h2d_meta] dst=%p src=%p n=%zu stagep=%p blk=%zu/%zu off=%zu cap=%zu\n",
+              dst, src, n, (void*)stagep, meta_cur_block, meta_blocks.size(),
+              meta_off, meta_blocks[meta_cur_block].cap);
     memcpy(stagep, src, n);
     return hipMemcpyAsync(dst, stagep, n, hipMemcpyHostToDevice, stream);
   }  std::vector<KEv> kpending;
@@ -2264,6 +2270,9 @@ int GpuJob::pack_into(size_t b0, size_t b1, const std::vector<uint64_t>& outoff,
     *done_event = nullptr;
     return 0;
   }
+  if (getenv("DCW_META_DEBUG"))
+    fprintf(stderr, "[pack] b0=%zu b1=%zu outoff.size=%zu data=%p total=%zu\n",
+            b0, b1, outoff.size(), (const void*)outoff.data(), total_bytes);
   ENSURE(p->d_outoff, p->outoff_cap, sizeof(uint64_t) * nb);
   uint64_t* d_outoff = (uint64_t*)p->d_outoff;
   // synchronous copy: NULL-stream ordering with the (blocking) pipeline
