@@ -1423,6 +1423,17 @@ struct GpuJob::Impl {
               "[h2d_meta] dst=%p src=%p n=%zu stagep=%p blk=%zu/%zu off=%zu cap=%zu\n",
               dst, src, n, (void*)stagep, meta_cur_block, meta_blocks.size(),
               meta_off, meta_blocks[meta_cur_block].cap);
+    if (dbg) { // probe both sides of the copy to attribute the fault
+      volatile uint8_t t0v = ((volatile uint8_t*)stagep)[0];
+      (void)t0v;
+      ((volatile uint8_t*)stagep)[0] = 0;
+      ((volatile uint8_t*)stagep)[n - 1] = 0;
+      fprintf(stderr, "[h2d_meta] stagep touch ok\n");
+      volatile uint8_t s0 = ((const volatile uint8_t*)src)[0];
+      volatile uint8_t s1 = ((const volatile uint8_t*)src)[n - 1];
+      (void)s0; (void)s1;
+      fprintf(stderr, "[h2d_meta] src touch ok\n");
+    }
     memcpy(stagep, src, n);
     return hipMemcpyAsync(dst, stagep, n, hipMemcpyHostToDevice, stream);
   }  std::vector<KEv> kpending;

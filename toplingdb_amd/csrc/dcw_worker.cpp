@@ -240,7 +240,12 @@ extern "C" {
 
 const char* dcw_version(void) { return "toplingdb_amd dcompact worker r1 (gfx950)"; }
 
-static void dcw_segv_handler(int sig) {
+static void dcw_segv_handler(int sig, siginfo_t* si, void*) {
+  char buf[128];
+  int m = snprintf(buf, sizeof(buf), "[segv] sig=%d fault_addr=%p\n", sig,
+                   si ? si->si_addr : nullptr);
+  ssize_t w = write(2, buf, m);
+  (void)w;
   void* frames[64];
   int n = backtrace(frames, 64);
   backtrace_symbols_fd(frames, n, 2);
@@ -249,8 +254,13 @@ static void dcw_segv_handler(int sig) {
 
 int32_t dcw_init(int32_t device_ordinal) {
   if (getenv("DCW_SEGV_TRACE")) {
-    signal(SIGSEGV, dcw_segv_handler);
-    signal(SIGABRT, dcw_segv_handler);
+    struct sigaction sa;
+    memset(&sa, 0, sizeof(sa));
+    sa.sa_sigaction = dcw_segv_handler;
+    sa.sa_flags = SA_SIGINFO;
+    sigaction(SIGSEGV, &sa, nullptr);
+    sigaction(SIGBUS, &sa, nullptr);
+    sigaction(SIGABRT, &sa, nullptr);
   }
   std::lock_guard<std::mutex> lk(g_mu);
   std::string err;
