@@ -2247,9 +2247,8 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   // host needs no further GPU round trip for them
   HIPCHK(p->ens((void**)&p->d_scratch_keys, (uint64_t)nb * BLKSTAT_STRIDE));
   if (p->h_keys_cap < (uint64_t)nb * BLKSTAT_STRIDE) {
-    for (auto& b : p->meta_blocks)
-    if (b.p) (void)hipHostFree(b.p);
-  if (p->h_keys) (void)hipHostFree(p->h_keys);
+    if (p->h_keys) (void)hipHostFree(p->h_keys);
+    p->h_keys = nullptr;
     p->h_keys_cap = (uint64_t)nb * BLKSTAT_STRIDE * 5 / 4;
     HIPCHK(hipHostMalloc(&p->h_keys, p->h_keys_cap));
   }
