@@ -1370,18 +1370,10 @@ struct GpuJob::Impl {
     void* p = nullptr;
     size_t cap = 0;
   };
-  uint64_t canary1 = 0xdeadbeefcafef00dull;
   std::vector<MetaBlock> meta_blocks;
   size_t meta_cur_block = 0;
   size_t meta_off = 0;
-  uint64_t canary2 = 0x1122334455667788ull;
-  void check_canary(const char* where) {
-    if (canary1 != 0xdeadbeefcafef00dull || canary2 != 0x1122334455667788ull) {
-      fprintf(stderr, "[canary] DEAD at %s: c1=%016llx c2=%016llx\n", where,
-              (unsigned long long)canary1, (unsigned long long)canary2);
-      abort();
-    }
-  }
+
   void arena_reset() {
     // keep only the largest block to stay grow-only without hoarding
     if (meta_blocks.size() > 1) {
@@ -1397,7 +1389,6 @@ struct GpuJob::Impl {
     meta_off = 0;
   }
   hipError_t h2d_meta(void* dst, const void* src, size_t n) {
-    check_canary("h2d_meta");
     if (n == 0) return hipSuccess;
     static const bool sync_meta = getenv("DCW_SYNC_META") != nullptr;
     if (sync_meta) return hipMemcpy(dst, src, n, hipMemcpyHostToDevice);
@@ -1417,23 +1408,6 @@ struct GpuJob::Impl {
     }
     uint8_t* stagep = (uint8_t*)meta_blocks[meta_cur_block].p + meta_off;
     meta_off += need;
-    static const bool dbg = getenv("DCW_META_DEBUG") != nullptr;
-    if (dbg)
-      fprintf(stderr,
-              "[h2d_meta] dst=%p src=%p n=%zu stagep=%p blk=%zu/%zu off=%zu cap=%zu\n",
-              dst, src, n, (void*)stagep, meta_cur_block, meta_blocks.size(),
-              meta_off, meta_blocks[meta_cur_block].cap);
-    if (dbg) { // probe both sides of the copy to attribute the fault
-      volatile uint8_t t0v = ((volatile uint8_t*)stagep)[0];
-      (void)t0v;
-      ((volatile uint8_t*)stagep)[0] = 0;
-      ((volatile uint8_t*)stagep)[n - 1] = 0;
-      fprintf(stderr, "[h2d_meta] stagep touch ok\n");
-      volatile uint8_t s0 = ((const volatile uint8_t*)src)[0];
-      volatile uint8_t s1 = ((const volatile uint8_t*)src)[n - 1];
-      (void)s0; (void)s1;
-      fprintf(stderr, "[h2d_meta] src touch ok\n");
-    }
     memcpy(stagep, src, n);
     return hipMemcpyAsync(dst, stagep, n, hipMemcpyHostToDevice, stream);
   }  std::vector<KEv> kpending;
@@ -1766,7 +1740,6 @@ int GpuJob::stage_release(StagedInput* s, std::string* err) {
 
 int GpuJob::decode(std::string* err) {
   Impl* p = p_;
-  p->check_canary("decode");
   hipEvent_t t0, t1;
   (void)hipEventCreate(&t0);
   (void)hipEventCreate(&t1);
@@ -1910,7 +1883,6 @@ int GpuJob::decode(std::string* err) {
 
 int GpuJob::merge(std::string* err) {
   Impl* p = p_;
-  p->check_canary("merge");
   hipEvent_t t0, t1;
   (void)hipEventCreate(&t0);
   (void)hipEventCreate(&t1);
@@ -2140,7 +2112,6 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
 int GpuJob::plan_all(const TableOpts& o, const uint32_t** next,
                      const uint32_t** meta, std::string* err) {
   Impl* p = p_;
-  p->check_canary("plan_all");
   uint64_t n = p->n_surv;
   // pinned host landing buffers (grow-only)
   if (p->h_plan_cap < n * 8) {
@@ -2181,7 +2152,6 @@ int GpuJob::plan_all(const TableOpts& o, const uint32_t** next,
 int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts& o,
                         std::vector<uint32_t>* comp_sizes, std::string* err) {
   Impl* p = p_;
-  p->check_canary("emit");
   hipEvent_t t0, t1;
   (void)hipEventCreate(&t0);
   (void)hipEventCreate(&t1);
@@ -2274,15 +2244,11 @@ int GpuJob::pack_into(size_t b0, size_t b1, const std::vector<uint64_t>& outoff,
                       uint8_t* host_dst, size_t total_bytes, void** done_event,
                       std::string* err) {
   Impl* p = p_;
-  p->check_canary("pack");
   uint32_t nb = (uint32_t)(b1 - b0);
   if (nb == 0) {
     *done_event = nullptr;
     return 0;
   }
-  if (getenv("DCW_META_DEBUG"))
-    fprintf(stderr, "[pack] b0=%zu b1=%zu outoff.size=%zu data=%p total=%zu\n",
-            b0, b1, outoff.size(), (const void*)outoff.data(), total_bytes);
   ENSURE(p->d_outoff, p->outoff_cap, sizeof(uint64_t) * nb);
   uint64_t* d_outoff = (uint64_t*)p->d_outoff;
   // synchronous copy: NULL-stream ordering with the (blocking) pipeline
@@ -2362,7 +2328,6 @@ int GpuJob::gather_entries(uint64_t first, uint32_t count,
                            std::vector<std::pair<std::string, std::string>>* kvs,
                            std::string* err) {
   Impl* p = p_;
-  p->check_canary("gather");
   kvs->clear();
   if (!count) return 0;
   // record layout: [klen u8][key][vlen u32][value]
