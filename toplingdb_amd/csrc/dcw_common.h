@@ -7,14 +7,23 @@
 // (citations at each function); the CPU oracle under /oracle is a separate
 // restatement used only to check this code.
 //
-// Snappy codec spec (self-pinned, SURVEY.md §8c — the reference pins no
-// compressed bytes): public snappy format; DCW encoder = greedy matcher,
-// hash table 1<<12 u16 slots over 4-byte windows,
-// hash = (load32_le(p) * 0x1e35a7bd) >> 20, candidate valid if within 65535
-// bytes and 4 bytes equal; extend forward; literals flushed before each
-// copy; copies emitted as: 1-byte-offset form when len in [4,11] and
-// offset < 2048, else 2-byte-offset chunks of <=64 bytes keeping a >=4 tail.
-// Identical in oracle/prims.c (C), here (host C++) and in the HIP kernels.
+// Snappy codec spec v4 (self-pinned, SURVEY.md §8c — the reference pins no
+// compressed bytes): public snappy format (any spec-compliant decoder reads
+// it); the DCW encoder is DETERMINISTIC AND PARALLEL by construction:
+//   segment size  seg = max(16, ceil(n/64)); segments [s*seg, min((s+1)*seg, n))
+//   hash table    1<<11 u32 slots; h = (load32_le(p) * 0x1e35a7bd) >> 21;
+//                 tab[h] = SMALLEST position p in [0, n-4] hashing to h
+//                 (first occurrence — order-independent, so a parallel
+//                 atomicMin build and a serial first-wins scan agree)
+//   per segment   greedy scan p from s0: candidate c = tab[h(p)] matches if
+//                 c < p and load32(c) == load32(p); extension capped at the
+//                 segment end (l <= s1 - p); on miss p += 1 (no skip-ahead —
+//                 a lane scans <= seg bytes); trailing bytes of the segment
+//                 are a literal.  Literals are never merged across segments.
+//   copies        1-byte-offset form when len in [4,11] and offset < 2048,
+//                 else 2-byte-offset chunks of <=64 bytes keeping a >=4 tail.
+// Identical in oracle/prims.c (C), here (host C++, serial restatement) and
+// in the HIP kernel (one wave per block, one lane per segment).
 #pragma once
 #include <stddef.h>
 #include <stdint.h>
@@ -364,35 +373,53 @@ DCW_HD uint8_t* snap_emit_copy(uint8_t* op, size_t offset, size_t len) {
   }
   return op;
 }
+DCW_HD size_t snap_segment_size(size_t n) {
+  size_t s = (n + 63) / 64;
+  return s < 16 ? 16 : s;
+}
+// Encode one segment [s0, s1) against a prebuilt first-occurrence table;
+// emits into op, returns the new op.  Shared verbatim by the serial host
+// restatement and the per-lane device path (spec v4).
+DCW_HD uint8_t* snap_encode_segment(const uint8_t* in, size_t n, size_t s0,
+                                    size_t s1, const uint32_t* tab,
+                                    uint8_t* op) {
+  (void)n;
+  size_t lit = s0, p = s0;
+  while (p + 4 <= s1) {
+    uint32_t w = load32(in + p);
+    uint32_t h = (w * kSnapHashMul) >> (32 - kSnapHashBits);
+    uint32_t c = tab[h];
+    if (c != 0xffffffffu && c < p && load32(in + c) == w) {
+      size_t l = 4;
+      while (p + l < s1 && in[c + l] == in[p + l]) l++;
+      op = snap_emit_literal(op, in + lit, p - lit);
+      op = snap_emit_copy(op, p - c, l);
+      p += l;
+      lit = p;
+    } else {
+      p++;
+    }
+  }
+  return snap_emit_literal(op, in + lit, s1 - lit);
+}
 // tab: caller-provided (1<<kSnapHashBits) u32 table initialized to
-// 0xffffffff (16 KiB; fits LDS on the device side).  Matches farther than
-// 65535 bytes back are rejected (2-byte copy offsets), like the oracle.
+// 0xffffffff (8 KiB; fits LDS on the device side).  Serial restatement of
+// spec v4: first-wins table build (== min position), then segments in
+// order.
 DCW_HD size_t snappy_compress_block(const uint8_t* in, size_t n, uint8_t* out,
                                     uint32_t* tab) {
   uint8_t* op = out;
   op += varint32_put(op, (uint32_t)n);
   if (n == 0) return (size_t)(op - out);
-  size_t lit_start = 0, p = 0;
-  uint32_t skip = 32; // spec v2: snappy-style skip acceleration
-  while (p + 4 <= n) {
-    uint32_t w = load32(in + p);
-    uint32_t h = (w * kSnapHashMul) >> (32 - kSnapHashBits);
-    uint32_t cand = tab[h];
-    tab[h] = (uint32_t)p;
-    if (cand != 0xffffffffu && p - cand < 65536 && load32(in + cand) == w) {
-      size_t mlen = 4;
-      while (p + mlen < n && in[cand + mlen] == in[p + mlen]) mlen++;
-      op = snap_emit_literal(op, in + lit_start, p - lit_start);
-      op = snap_emit_copy(op, p - cand, mlen);
-      p += mlen;
-      lit_start = p;
-      skip = 32;
-    } else {
-      p += skip >> 5;
-      skip++;
-    }
+  for (size_t p = 0; p + 4 <= n; p++) {
+    uint32_t h = (load32(in + p) * kSnapHashMul) >> (32 - kSnapHashBits);
+    if (tab[h] == 0xffffffffu) tab[h] = (uint32_t)p;
   }
-  op = snap_emit_literal(op, in + lit_start, n - lit_start);
+  size_t seg = snap_segment_size(n);
+  for (size_t s0 = 0; s0 < n; s0 += seg) {
+    size_t s1 = s0 + seg < n ? s0 + seg : n;
+    op = snap_encode_segment(in, n, s0, s1, tab, op);
+  }
   return (size_t)(op - out);
 }
 DCW_HD size_t snappy_uncompressed_len(const uint8_t* in, size_t n) {

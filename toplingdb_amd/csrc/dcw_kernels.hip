@@ -870,21 +870,18 @@ __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
   }
 }
 
-// snappy encode, one wave (64 threads) per 4 KiB-class block:
-// the block is staged into LDS by all lanes, the u16 hash table and the
-// output live in LDS too, lane 0 runs the serial matcher over LDS (the
-// greedy match chain is inherently sequential — the DETERMINISM of the
-// self-pinned codec spec is the point), all lanes copy the result out.
-// LDS per wave: in 5 KiB + tab 8 KiB + out 6.2 KiB ≈ 19.2 KiB -> with
-// 4 waves/WG ≈ 77 KiB/WG, 2 WGs/CU = 8 concurrent encoders per CU.
+// snappy encode, spec v4: one wave per 4 KiB-class block, WAVE-PARALLEL —
+// the spec's first-occurrence hash table is order-independent (min position
+// per slot), so all 64 lanes build it together with LDS atomicMin, and the
+// spec's fixed segmentation (max(16, ceil(n/64)) bytes) gives every lane an
+// independent greedy segment to encode.  Fragments are staged in per-lane
+// private buffers, laid out with a wave prefix sum (shfl), and copied to
+// the block's output slot.  The segment encoder itself is
+// dcw::snap_encode_segment — the SAME function the host C++ restatement
+// runs, so device/host byte-parity holds by construction.
 #define SNAP_MAX_UNC 5120 // block_size + slack; host guards this bound
 #define SNAP_MAX_OUT (32 + SNAP_MAX_UNC + SNAP_MAX_UNC / 6)
-struct SnapLds {
-  uint16_t tab[1u << kSnapHashBits];
-}; // 4 KiB per encoder: hash table only.  Input is read straight from
-// global (the 4 KiB block stays hot in L1/L2 for the single encoding
-// lane); output goes straight to global (write-only).  Keeping LDS per
-// encoder small lets ~9 workgroups (36 encoder waves) reside per CU.
+#define SNAP_FRAG_MAX 128 // worst-case encode of one <=80 B segment
 
 // wave-internal LDS ordering: drain DS ops + stop compiler reordering
 __device__ __forceinline__ void wave_lds_sync() {
@@ -892,148 +889,16 @@ __device__ __forceinline__ void wave_lds_sync() {
   __builtin_amdgcn_wave_barrier();
 }
 
-// u16-table variant of dcw::snappy_compress_block — byte-identical output
-// (positions < 65536 in a block; 0xffff = empty, position 0xffff cannot
-// occur since SNAP_MAX_UNC < 0xffff).  Match extension and literal copies
-// run 4 bytes at a time (same output; LDS 32-bit reads are
-// alignment-penalty-free, unlike _b64+).
-__device__ __forceinline__ uint8_t* snap_emit_literal4(uint8_t* op,
-                                                       const uint8_t* lit,
-                                                       size_t len) {
-  if (len == 0) return op;
-  size_t n = len - 1;
-  if (n < 60) {
-    *op++ = (uint8_t)(n << 2);
-  } else {
-    uint8_t tmp[4];
-    int count = 0;
-    size_t x = n;
-    while (x > 0) {
-      tmp[count++] = (uint8_t)(x & 0xff);
-      x >>= 8;
-    }
-    *op++ = (uint8_t)((59 + count) << 2);
-    for (int i = 0; i < count; i++) *op++ = tmp[i];
-  }
-  size_t t = 0;
-  for (; t + 4 <= len; t += 4) {
-    uint32_t v = load32(lit + t);
-    memcpy(op + t, &v, 4);
-  }
-  for (; t < len; t++) op[t] = lit[t];
-  return op + len;
-}
-// Exact-output encoder with a 2-position software pipeline: on the common
-// miss path two probe positions' loads and table reads overlap their LDS
-// latencies.  Semantics identical to the sequential spec: the second
-// probe's candidate accounts for the first probe's table insert
-// (cand2 = pz1 when the hashes collide), and speculation is discarded on a
-// match.
-__device__ static size_t snap_encode_lds(const uint8_t* in, uint32_t n,
-                                         uint8_t* out, uint16_t* tab) {
-  uint8_t* op = out;
-  op += varint32_put(op, n);
-  if (n == 0) return (size_t)(op - out);
-  uint32_t lit_start = 0, pz = 0;
-  uint32_t skip = 32; // spec v3: snappy-style skip acceleration
-  while (pz + 4 <= n) {
-    // 4-position speculative probe group.  The serial spec probes one
-    // position at a time: probe p (read tab, then insert p), on miss
-    // skip++ and advance by (skip>>5) computed BEFORE the increment.
-    // All four probe chains (in-load -> hash -> tab-load -> candidate
-    // compare-load) are issued as independent scalar chains so their LDS
-    // latencies overlap; serial semantics are reconstructed afterwards:
-    // a candidate read at p_j sees the in-group insertion of the latest
-    // p_i (i<j) hashing to the same slot, and insertions commit only up
-    // to the first matching position.
-    uint32_t p0 = pz;
-    uint32_t p1 = p0 + (skip >> 5);
-    uint32_t p2 = p1 + ((skip + 1) >> 5);
-    uint32_t p3 = p2 + ((skip + 2) >> 5);
-    int np = p3 + 4 <= n ? 4 : (p2 + 4 <= n ? 3 : (p1 + 4 <= n ? 2 : 1));
-    uint32_t w0 = load32(in + p0);
-    uint32_t w1 = np > 1 ? load32(in + p1) : 0;
-    uint32_t w2 = np > 2 ? load32(in + p2) : 0;
-    uint32_t w3 = np > 3 ? load32(in + p3) : 0;
-    uint32_t h0 = (w0 * kSnapHashMul) >> (32 - kSnapHashBits);
-    uint32_t h1 = (w1 * kSnapHashMul) >> (32 - kSnapHashBits);
-    uint32_t h2 = (w2 * kSnapHashMul) >> (32 - kSnapHashBits);
-    uint32_t h3 = (w3 * kSnapHashMul) >> (32 - kSnapHashBits);
-    uint32_t c0 = tab[h0];
-    uint32_t c1 = tab[h1];
-    uint32_t c2 = tab[h2];
-    uint32_t c3 = tab[h3];
-    // in-group overwrites (latest earlier position with the same slot wins)
-    if (h1 == h0) c1 = p0;
-    if (h2 == h1) c2 = p1; else if (h2 == h0) c2 = p0;
-    if (h3 == h2) c3 = p2; else if (h3 == h1) c3 = p1; else if (h3 == h0) c3 = p0;
-    // speculative candidate loads with sanitized offsets (0xffff = empty)
-    uint32_t a0 = load32(in + (c0 == 0xffffu ? 0 : c0));
-    uint32_t a1 = load32(in + (c1 == 0xffffu ? 0 : c1));
-    uint32_t a2 = load32(in + (c2 == 0xffffu ? 0 : c2));
-    uint32_t a3 = load32(in + (c3 == 0xffffu ? 0 : c3));
-    bool m0 = c0 != 0xffffu && c0 < p0 && a0 == w0;
-    bool m1 = np > 1 && c1 != 0xffffu && c1 < p1 && a1 == w1;
-    bool m2 = np > 2 && c2 != 0xffffu && c2 < p2 && a2 == w2;
-    bool m3 = np > 3 && c3 != 0xffffu && c3 < p3 && a3 == w3;
-    uint32_t mpos, mcand;
-    if (m0) {
-      tab[h0] = (uint16_t)p0;
-      mpos = p0; mcand = c0;
-    } else if (m1) {
-      tab[h0] = (uint16_t)p0; tab[h1] = (uint16_t)p1;
-      mpos = p1; mcand = c1;
-    } else if (m2) {
-      tab[h0] = (uint16_t)p0; tab[h1] = (uint16_t)p1; tab[h2] = (uint16_t)p2;
-      mpos = p2; mcand = c2;
-    } else if (m3) {
-      tab[h0] = (uint16_t)p0; tab[h1] = (uint16_t)p1; tab[h2] = (uint16_t)p2;
-      tab[h3] = (uint16_t)p3;
-      mpos = p3; mcand = c3;
-    } else {
-      tab[h0] = (uint16_t)p0;
-      if (np > 1) tab[h1] = (uint16_t)p1;
-      if (np > 2) tab[h2] = (uint16_t)p2;
-      if (np > 3) tab[h3] = (uint16_t)p3;
-      if (np < 4) break; // next probe would run past the block
-      pz = p3 + ((skip + 3) >> 5);
-      skip += 4;
-      continue;
-    }
-    // match at mpos against mcand
-    uint32_t mlen = 4;
-    while (mpos + mlen + 4 <= n) {
-      uint32_t a = load32(in + mcand + mlen);
-      uint32_t bz = load32(in + mpos + mlen);
-      uint32_t x = a ^ bz;
-      if (x) {
-        mlen += __builtin_ctz(x) >> 3;
-        goto done_ext;
-      }
-      mlen += 4;
-    }
-    while (mpos + mlen < n && in[mcand + mlen] == in[mpos + mlen]) mlen++;
-  done_ext:
-    op = snap_emit_literal4(op, in + lit_start, mpos - lit_start);
-    op = snap_emit_copy(op, mpos - mcand, mlen);
-    pz = mpos + mlen;
-    lit_start = pz;
-    skip = 32;
-  }
-  op = snap_emit_literal4(op, in + lit_start, n - lit_start);
-  return (size_t)(op - out);
-}
-
 __global__ __launch_bounds__(256) void k_compress(
     const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
     const uint8_t* __restrict__ ucblob, uint8_t* __restrict__ cblob,
     uint64_t ccap_per_block, uint32_t* __restrict__ bsize,
     uint8_t* __restrict__ btype, uint32_t* err_flag) {
-  __shared__ SnapLds lds[4]; // 4 encoders/WG = 16 KiB
+  __shared__ uint32_t tabs[4][1u << kSnapHashBits]; // 8 KiB per wave
   uint32_t wid = threadIdx.x / WAVE;  // wave within workgroup
   uint32_t lane = threadIdx.x % WAVE;
   uint32_t waves = blockDim.x / WAVE;
-  SnapLds& L = lds[wid];
+  uint32_t* tab = tabs[wid];
   for (uint32_t b = blockIdx.x * waves + wid; b < nblocks;
        b += gridDim.x * waves) {
     EmitBlockDesc d = bds[b];
@@ -1046,18 +911,46 @@ __global__ __launch_bounds__(256) void k_compress(
       continue;
     }
     const uint8_t* gin = ucblob + d.uout;
+    uint32_t n = d.unc_size;
     for (uint32_t t = lane; t < (1u << kSnapHashBits); t += WAVE)
-      L.tab[t] = 0xffffu;
+      tab[t] = 0xffffffffu;
     wave_lds_sync();
+    for (uint32_t p = lane; p + 4 <= n; p += WAVE) {
+      uint32_t h = (load32(gin + p) * kSnapHashMul) >> (32 - kSnapHashBits);
+      atomicMin(&tab[h], p);
+    }
+    wave_lds_sync();
+    uint32_t seg = (uint32_t)snap_segment_size(n);
+    uint32_t s0 = lane * seg;
+    uint8_t frag[SNAP_FRAG_MAX];
+    uint32_t fl = 0;
+    if (s0 < n) {
+      uint32_t s1 = s0 + seg < n ? s0 + seg : n;
+      uint8_t* e = snap_encode_segment(gin, n, s0, s1, tab, frag);
+      fl = (uint32_t)(e - frag);
+    }
+    // exclusive prefix of fragment sizes across the wave
+    uint32_t inc = fl;
+    for (int sh = 1; sh < WAVE; sh <<= 1) {
+      uint32_t v = __shfl_up(inc, sh);
+      if ((int)lane >= sh) inc += v;
+    }
+    uint32_t total = __shfl(inc, WAVE - 1);
+    uint32_t excl = inc - fl;
+    uint8_t* gout = cblob + (uint64_t)b * ccap_per_block;
+    uint8_t hdr[5];
+    uint32_t hl = varint32_put(hdr, n); // lane-uniform
+    if (lane == 0)
+      for (uint32_t t = 0; t < hl; t++) gout[t] = hdr[t];
+    for (uint32_t t = 0; t < fl; t++) gout[hl + excl + t] = frag[t];
     if (lane == 0) {
-      uint8_t* gout = cblob + (uint64_t)b * ccap_per_block;
-      uint32_t cn = (uint32_t)snap_encode_lds(gin, d.unc_size, gout, L.tab);
+      uint32_t cn = hl + total;
       // GoodCompressionRatio, default max_compressed_bytes_per_kb=896
-      if (cn <= (((uint64_t)896 * d.unc_size) >> 10)) {
+      if (cn <= (((uint64_t)896 * n) >> 10)) {
         bsize[b] = cn;
         btype[b] = 1;
       } else {
-        bsize[b] = d.unc_size;
+        bsize[b] = n;
         btype[b] = 0;
       }
     }
