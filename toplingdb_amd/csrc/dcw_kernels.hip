@@ -889,6 +889,62 @@ __device__ __forceinline__ void wave_lds_sync() {
   __builtin_amdgcn_wave_barrier();
 }
 
+// Device specialization of dcw::snap_encode_segment with identical output
+// bytes: literal copies run 4 bytes at a time and match extension compares
+// 4-byte words with ctz — only the data movement differs from the shared
+// host restatement, never the emitted stream.
+__device__ __forceinline__ uint8_t* snap_emit_literal4v(uint8_t* op,
+                                                        const uint8_t* lit,
+                                                        uint32_t len) {
+  if (len == 0) return op;
+  uint32_t n = len - 1;
+  if (n < 60) {
+    *op++ = (uint8_t)(n << 2);
+  } else { // len <= SNAP_FRAG_MAX < 256+1 -> single extra byte
+    *op++ = (uint8_t)(60 << 2);
+    *op++ = (uint8_t)n;
+  }
+  uint32_t t = 0;
+  for (; t + 4 <= len; t += 4) {
+    uint32_t v = load32(lit + t);
+    memcpy(op + t, &v, 4);
+  }
+  for (; t < len; t++) op[t] = lit[t];
+  return op + len;
+}
+__device__ __forceinline__ uint8_t* snap_encode_segment_dev(
+    const uint8_t* __restrict__ in, uint32_t s0, uint32_t s1,
+    const uint32_t* __restrict__ tab, uint8_t* op) {
+  uint32_t lit = s0, p = s0;
+  while (p + 4 <= s1) {
+    uint32_t w = load32(in + p);
+    uint32_t h = (w * kSnapHashMul) >> (32 - kSnapHashBits);
+    uint32_t c = tab[h];
+    if (c != 0xffffffffu && c < p && load32(in + c) == w) {
+      uint32_t l = 4;
+      while (p + l + 4 <= s1) {
+        uint32_t a = load32(in + c + l);
+        uint32_t bz = load32(in + p + l);
+        uint32_t x = a ^ bz;
+        if (x) {
+          l += __builtin_ctz(x) >> 3;
+          goto ext_done;
+        }
+        l += 4;
+      }
+      while (p + l < s1 && in[c + l] == in[p + l]) l++;
+    ext_done:
+      op = snap_emit_literal4v(op, in + lit, p - lit);
+      op = snap_emit_copy(op, p - c, l);
+      p += l;
+      lit = p;
+    } else {
+      p++;
+    }
+  }
+  return snap_emit_literal4v(op, in + lit, s1 - lit);
+}
+
 __global__ __launch_bounds__(256) void k_compress(
     const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
     const uint8_t* __restrict__ ucblob, uint8_t* __restrict__ cblob,
@@ -926,7 +982,7 @@ __global__ __launch_bounds__(256) void k_compress(
     uint32_t fl = 0;
     if (s0 < n) {
       uint32_t s1 = s0 + seg < n ? s0 + seg : n;
-      uint8_t* e = snap_encode_segment(gin, n, s0, s1, tab, frag);
+      uint8_t* e = snap_encode_segment_dev(gin, s0, s1, tab, frag);
       fl = (uint32_t)(e - frag);
     }
     // exclusive prefix of fragment sizes across the wave
