@@ -59,7 +59,7 @@ def main():
     ap.add_argument("--cpu-baseline-runs", type=int, default=4,
                     help="bounded oracle sample (number of input runs)")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
-    ap.add_argument("--jobs-in-flight", type=int, default=6,
+    ap.add_argument("--jobs-in-flight", type=int, default=10,
                     help="concurrent compaction jobs per GPU (the production "
                          "dcompact worker runs concurrent jobs per node, "
                          "BASELINE.json configs[4]); 1 = strictly sequential")
