@@ -137,18 +137,30 @@ __global__ void k_verify_usize(const uint8_t* __restrict__ blob,
   }
 }
 
-// one wave per block: raw -> cooperative copy; snappy -> staged through LDS
-// (compressed in + decoded out both in LDS, lane 0 runs the serial decoder,
-// all lanes copy in/out); oversized blocks fall back to the direct path.
-#define DEC_MAX 4992 // 2*4992*4 waves = 39.9 KB LDS/WG -> 4 WGs (16 decoders)/CU
+// one wave per block: raw -> cooperative copy; snappy -> PARALLEL decode:
+// lane 0 skims the op stream into LDS records (tag walk only — ~10x
+// cheaper than decoding), then all lanes apply literal ops concurrently
+// (every output byte is written exactly once: literals read input only,
+// so they commute), then copy ops are applied in stream order (a copy may
+// read bytes produced by literals or earlier copies) with the whole wave
+// moving words; self-overlapping copies (offset < length) fall back to a
+// serial byte loop for that op.  Works for ANY public-snappy stream, not
+// just our own spec-v4 output; op-count overflow falls back to the serial
+// decoder.
+#define DEC_MAX_OPS 1536
 struct DecLds {
-  uint8_t in[DEC_MAX]; // compressed input + decoded output both staged in
-  uint8_t out[DEC_MAX]; // LDS: the serial byte decoder is LDS-latency bound
-                        // (a global-input variant measured ~10% slower)
+  uint64_t ops[DEC_MAX_OPS]; // kind(1) | len(16) | out(16) | in(24)
+  uint32_t nops;
+  uint32_t nlit;
 };
 __device__ __forceinline__ void wave_lds_sync2() {
   __builtin_amdgcn_s_waitcnt(0);
   __builtin_amdgcn_wave_barrier();
+}
+// serial fallback (also the reference semantics for the parallel path)
+__device__ __forceinline__ bool dec_serial(const uint8_t* src, uint32_t n,
+                                           uint8_t* dst, uint32_t un) {
+  return snappy_uncompress(src, n, dst, un) == un;
 }
 __global__ __launch_bounds__(256) void k_decompress(
     const uint8_t* __restrict__ blob, const uint64_t* __restrict__ boff,
@@ -175,29 +187,116 @@ __global__ __launch_bounds__(256) void k_decompress(
       }
       if (lane == 0)
         for (uint32_t t = n & ~15u; t < n; t++) dst[t] = src[t];
-    } else if (n <= DEC_MAX && usize[i] <= DEC_MAX) {
-      for (uint32_t t = lane * 4; t < n; t += WAVE * 4) {
-        uint32_t chunk = n - t < 4 ? n - t : 4;
-        for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = src[t + x];
-      }
-      wave_lds_sync2();
-      if (lane == 0) {
-        if (snappy_uncompress(L.in, n, L.out, usize[i]) != usize[i])
-          set_err(err_flag, DE_SNAPPY);
-      }
-      wave_lds_sync2();
-      uint32_t un = usize[i];
-      for (uint32_t t = lane * 4; t < un; t += WAVE * 4) {
-        uint32_t chunk = un - t < 4 ? un - t : 4;
-        for (uint32_t x = 0; x < chunk; x++) dst[t + x] = L.out[t + x];
-      }
-      wave_lds_sync2();
-    } else {
-      if (lane == 0) {
-        if (snappy_uncompress(src, n, dst, usize[i]) != usize[i])
-          set_err(err_flag, DE_SNAPPY);
+      continue;
+    }
+    uint32_t un = usize[i];
+    // ---- phase 1: op skim (lane 0) ----
+    if (lane == 0) {
+      L.nops = 0;
+      L.nlit = 0;
+      uint32_t hdr;
+      int k = varint32_get(src, src + (n < 5 ? n : 5), &hdr);
+      if (k < 0 || hdr != un) {
+        L.nops = ~0u;
+      } else {
+        uint32_t ip = (uint32_t)k, op = 0, nops = 0, nlit = 0;
+        bool bad = false;
+        while (ip < n) {
+          uint8_t tag = src[ip++];
+          uint32_t len, inoff = ip, kind;
+          if ((tag & 3) == 0) {
+            len = (tag >> 2) + 1;
+            if (len > 60) {
+              uint32_t nb = len - 60;
+              if (ip + nb > n) { bad = true; break; }
+              len = 0;
+              for (uint32_t x = 0; x < nb; x++) len |= (uint32_t)src[ip + x] << (8 * x);
+              len += 1;
+              ip += nb;
+            }
+            inoff = ip;
+            if (ip + len > n) { bad = true; break; }
+            ip += len;
+            kind = 0;
+            nlit++;
+          } else {
+            uint32_t off;
+            if ((tag & 3) == 1) {
+              if (ip + 1 > n) { bad = true; break; }
+              len = ((tag >> 2) & 7) + 4;
+              off = ((uint32_t)(tag >> 5) << 8) | src[ip];
+              ip += 1;
+            } else if ((tag & 3) == 2) {
+              if (ip + 2 > n) { bad = true; break; }
+              len = (tag >> 2) + 1;
+              off = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8);
+              ip += 2;
+            } else {
+              bad = true; // 4-byte-offset form: our encoders never emit it
+              break;
+            }
+            if (off == 0 || off > op) { bad = true; break; }
+            inoff = off; // for copies, record the OFFSET instead
+            kind = 1;
+          }
+          if (op + len > un || nops >= DEC_MAX_OPS) { bad = true; break; }
+          L.ops[nops++] = (uint64_t)kind << 56 | (uint64_t)len << 40 |
+                          (uint64_t)op << 24 | inoff;
+          op += len;
+        }
+        if (bad || op != un) L.nops = ~0u;
+        else { L.nops = nops; L.nlit = nlit; }
       }
     }
+    wave_lds_sync2();
+    uint32_t nops = L.nops;
+    if (nops == ~0u) {
+      // overflow / unusual stream: serial reference decoder
+      if (lane == 0) {
+        if (!dec_serial(src, n, dst, un)) set_err(err_flag, DE_SNAPPY);
+      }
+      wave_lds_sync2();
+      continue;
+    }
+    // ---- phase 2: literals in parallel (disjoint outputs, input-only reads)
+    for (uint32_t o = lane; o < nops; o += WAVE) {
+      uint64_t r = L.ops[o];
+      if (r >> 56) continue;
+      uint32_t len = (uint32_t)(r >> 40) & 0xffff;
+      uint32_t out = (uint32_t)(r >> 24) & 0xffff;
+      uint32_t inp = (uint32_t)(r & 0xffffff);
+      uint32_t t = 0;
+      for (; t + 4 <= len; t += 4) {
+        uint32_t v = load32(src + inp + t);
+        memcpy(dst + out + t, &v, 4);
+      }
+      for (; t < len; t++) dst[out + t] = src[inp + t];
+    }
+    wave_lds_sync2();
+    // ---- phase 3: copies in stream order, wave-wide word moves ----
+    if (L.nlit != nops) {
+      for (uint32_t o = 0; o < nops; o++) {
+        uint64_t r = L.ops[o];
+        if (!(r >> 56)) continue;
+        uint32_t len = (uint32_t)(r >> 40) & 0xffff;
+        uint32_t out = (uint32_t)(r >> 24) & 0xffff;
+        uint32_t off = (uint32_t)(r & 0xffffff);
+        const uint8_t* s2 = dst + out - off;
+        uint8_t* d2 = dst + out;
+        if (off >= 4) { // 4-byte chunks never read their own output
+          for (uint32_t t = lane * 4; t < len; t += WAVE * 4) {
+            uint32_t chunk = len - t < 4 ? len - t : 4;
+            uint32_t v;
+            memcpy(&v, s2 + t, 4);
+            memcpy(d2 + t, &v, chunk == 4 ? 4u : chunk);
+          }
+        } else if (lane == 0) {
+          for (uint32_t t = 0; t < len; t++) d2[t] = s2[t];
+        }
+        wave_lds_sync2();
+      }
+    }
+    wave_lds_sync2();
   }
 }
 
