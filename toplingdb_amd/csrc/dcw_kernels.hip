@@ -283,15 +283,18 @@ __global__ __launch_bounds__(256) void k_decompress(
         uint32_t off = (uint32_t)(r & 0xffffff);
         const uint8_t* s2 = dst + out - off;
         uint8_t* d2 = dst + out;
-        if (off >= 4) { // 4-byte chunks never read their own output
+        if (off >= len) { // disjoint: word moves, no self-reference
           for (uint32_t t = lane * 4; t < len; t += WAVE * 4) {
             uint32_t chunk = len - t < 4 ? len - t : 4;
             uint32_t v;
             memcpy(&v, s2 + t, 4);
             memcpy(d2 + t, &v, chunk == 4 ? 4u : chunk);
           }
-        } else if (lane == 0) {
-          for (uint32_t t = 0; t < len; t++) d2[t] = s2[t];
+        } else {
+          // self-overlapping copy == replication of the off-byte window
+          // (out[j] = out[j-off] recursively); every read is from before
+          // the op's output, so lanes are independent
+          for (uint32_t t = lane; t < len; t += WAVE) d2[t] = s2[t % off];
         }
         wave_lds_sync2();
       }
