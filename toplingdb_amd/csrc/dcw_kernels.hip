@@ -149,18 +149,15 @@ __global__ void k_verify_usize(const uint8_t* __restrict__ blob,
 // decoder.
 #define DEC_MAX_OPS 1536
 struct DecLds {
-  uint64_t ops[DEC_MAX_OPS]; // kind(1) | len(16) | out(16) | in(24)
-  uint32_t nops;
+  // literal records grow from 0, copy records grow from the top (order
+  // preserved: copy k lives at ops[DEC_MAX_OPS-1-k])
+  uint64_t ops[DEC_MAX_OPS];
   uint32_t nlit;
+  uint32_t ncopy;
 };
 __device__ __forceinline__ void wave_lds_sync2() {
   __builtin_amdgcn_s_waitcnt(0);
   __builtin_amdgcn_wave_barrier();
-}
-// serial fallback (also the reference semantics for the parallel path)
-__device__ __forceinline__ bool dec_serial(const uint8_t* src, uint32_t n,
-                                           uint8_t* dst, uint32_t un) {
-  return snappy_uncompress(src, n, dst, un) == un;
 }
 __global__ __launch_bounds__(256) void k_decompress(
     const uint8_t* __restrict__ blob, const uint64_t* __restrict__ boff,
@@ -190,35 +187,42 @@ __global__ __launch_bounds__(256) void k_decompress(
       continue;
     }
     uint32_t un = usize[i];
-    // ---- phase 1: op skim (lane 0) ----
+    // ---- phase 1 (lane 0): skim ops into records.  Copies are grouped:
+    // a copy joins the current group iff its source bytes are fully
+    // decoded before the group runs (i.e. below the group's first output
+    // or covered by literals-only prefix); a self-overlapping copy always
+    // opens a group (its source is then the decoded prefix).  Bit 57
+    // marks a group head.  Our spec-v4 encoder matches against FIRST
+    // occurrences, so sources cluster early and groups are few.
     if (lane == 0) {
-      L.nops = 0;
       L.nlit = 0;
+      L.ncopy = 0;
       uint32_t hdr;
       int k = varint32_get(src, src + (n < 5 ? n : 5), &hdr);
       if (k < 0 || hdr != un) {
-        L.nops = ~0u;
+        L.nlit = ~0u;
       } else {
-        uint32_t ip = (uint32_t)k, op = 0, nops = 0, nlit = 0;
+        uint32_t ip = (uint32_t)k, op = 0, nlit = 0, ncopy = 0;
+        uint32_t group_start = 0; // first output byte of the current group
         bool bad = false;
         while (ip < n) {
           uint8_t tag = src[ip++];
-          uint32_t len, inoff = ip, kind;
+          uint32_t len, aux, kind, head = 0;
           if ((tag & 3) == 0) {
             len = (tag >> 2) + 1;
             if (len > 60) {
               uint32_t nb = len - 60;
               if (ip + nb > n) { bad = true; break; }
               len = 0;
-              for (uint32_t x = 0; x < nb; x++) len |= (uint32_t)src[ip + x] << (8 * x);
+              for (uint32_t x = 0; x < nb; x++)
+                len |= (uint32_t)src[ip + x] << (8 * x);
               len += 1;
               ip += nb;
             }
-            inoff = ip;
+            aux = ip; // input offset of the literal bytes
             if (ip + len > n) { bad = true; break; }
             ip += len;
             kind = 0;
-            nlit++;
           } else {
             uint32_t off;
             if ((tag & 3) == 1) {
@@ -232,36 +236,47 @@ __global__ __launch_bounds__(256) void k_decompress(
               off = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8);
               ip += 2;
             } else {
-              bad = true; // 4-byte-offset form: our encoders never emit it
+              bad = true; // 4-byte-offset form: never emitted by us
               break;
             }
             if (off == 0 || off > op) { bad = true; break; }
-            inoff = off; // for copies, record the OFFSET instead
+            aux = off;
             kind = 1;
+            // head a new group unless the source bytes are all decoded
+            // before this group runs (self-overlap source ends at op, so
+            // it always heads one)
+            uint32_t src_end = off >= len ? (op - off + len) : op;
+            if (ncopy == 0 || src_end > group_start) {
+              head = 1;
+              group_start = op;
+            }
           }
-          if (op + len > un || nops >= DEC_MAX_OPS) { bad = true; break; }
-          L.ops[nops++] = (uint64_t)kind << 56 | (uint64_t)len << 40 |
-                          (uint64_t)op << 24 | inoff;
+          if (op + len > un || nlit + ncopy >= DEC_MAX_OPS) { bad = true; break; }
+          uint64_t rec = (uint64_t)head << 57 | (uint64_t)kind << 56 |
+                         (uint64_t)len << 40 | (uint64_t)op << 24 | aux;
+          if (kind == 0)
+            L.ops[nlit++] = rec;
+          else
+            L.ops[DEC_MAX_OPS - 1 - ncopy++] = rec;
           op += len;
         }
-        if (bad || op != un) L.nops = ~0u;
-        else { L.nops = nops; L.nlit = nlit; }
+        if (bad || op != un) L.nlit = ~0u;
+        else { L.nlit = nlit; L.ncopy = ncopy; }
       }
     }
     wave_lds_sync2();
-    uint32_t nops = L.nops;
-    if (nops == ~0u) {
-      // overflow / unusual stream: serial reference decoder
+    uint32_t nlit = L.nlit;
+    if (nlit == ~0u) { // unusual stream / overflow: serial reference decode
       if (lane == 0) {
-        if (!dec_serial(src, n, dst, un)) set_err(err_flag, DE_SNAPPY);
+        if (snappy_uncompress(src, n, dst, un) != un)
+          set_err(err_flag, DE_SNAPPY);
       }
       wave_lds_sync2();
       continue;
     }
-    // ---- phase 2: literals in parallel (disjoint outputs, input-only reads)
-    for (uint32_t o = lane; o < nops; o += WAVE) {
+    // ---- phase 2: literals in parallel (disjoint outputs, input reads)
+    for (uint32_t o = lane; o < nlit; o += WAVE) {
       uint64_t r = L.ops[o];
-      if (r >> 56) continue;
       uint32_t len = (uint32_t)(r >> 40) & 0xffff;
       uint32_t out = (uint32_t)(r >> 24) & 0xffff;
       uint32_t inp = (uint32_t)(r & 0xffffff);
@@ -273,31 +288,43 @@ __global__ __launch_bounds__(256) void k_decompress(
       for (; t < len; t++) dst[out + t] = src[inp + t];
     }
     wave_lds_sync2();
-    // ---- phase 3: copies in stream order, wave-wide word moves ----
-    if (L.nlit != nops) {
-      for (uint32_t o = 0; o < nops; o++) {
-        uint64_t r = L.ops[o];
-        if (!(r >> 56)) continue;
+    // ---- phase 3: copy groups in order; within a group every copy's
+    // source is already decoded, so lanes apply whole copies in parallel
+    uint32_t ncopy = L.ncopy;
+    uint32_t c0 = 0;
+    while (c0 < ncopy) {
+      // group end: next head flag after c0 (wave ballot scan)
+      uint32_t cend = ncopy;
+      for (uint32_t base = c0 + 1; base < ncopy; base += WAVE) {
+        uint32_t idx = base + lane;
+        bool f = idx < ncopy && (L.ops[DEC_MAX_OPS - 1 - idx] >> 57 & 1);
+        uint64_t m = __ballot(f);
+        if (m) {
+          cend = base + __ffsll((unsigned long long)m) - 1;
+          break;
+        }
+      }
+      for (uint32_t c = c0 + lane; c < cend; c += WAVE) {
+        uint64_t r = L.ops[DEC_MAX_OPS - 1 - c];
         uint32_t len = (uint32_t)(r >> 40) & 0xffff;
         uint32_t out = (uint32_t)(r >> 24) & 0xffff;
         uint32_t off = (uint32_t)(r & 0xffffff);
         const uint8_t* s2 = dst + out - off;
         uint8_t* d2 = dst + out;
-        if (off >= len) { // disjoint: word moves, no self-reference
-          for (uint32_t t = lane * 4; t < len; t += WAVE * 4) {
-            uint32_t chunk = len - t < 4 ? len - t : 4;
+        if (off >= len) {
+          uint32_t t = 0;
+          for (; t + 4 <= len; t += 4) {
             uint32_t v;
             memcpy(&v, s2 + t, 4);
-            memcpy(d2 + t, &v, chunk == 4 ? 4u : chunk);
+            memcpy(d2 + t, &v, 4);
           }
+          for (; t < len; t++) d2[t] = s2[t];
         } else {
-          // self-overlapping copy == replication of the off-byte window
-          // (out[j] = out[j-off] recursively); every read is from before
-          // the op's output, so lanes are independent
-          for (uint32_t t = lane; t < len; t += WAVE) d2[t] = s2[t % off];
+          for (uint32_t t = 0; t < len; t++) d2[t] = s2[t % off];
         }
-        wave_lds_sync2();
       }
+      wave_lds_sync2();
+      c0 = cend;
     }
     wave_lds_sync2();
   }
