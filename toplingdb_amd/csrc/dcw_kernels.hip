@@ -147,11 +147,13 @@ __global__ void k_verify_usize(const uint8_t* __restrict__ blob,
 // serial byte loop for that op.  Works for ANY public-snappy stream, not
 // just our own spec-v4 output; op-count overflow falls back to the serial
 // decoder.
-#define DEC_MAX_OPS 1536
+#define DEC_MAX_OPS 1024
+#define DEC_IN_MAX 4992
 struct DecLds {
   // literal records grow from 0, copy records grow from the top (order
   // preserved: copy k lives at ops[DEC_MAX_OPS-1-k])
   uint64_t ops[DEC_MAX_OPS];
+  uint8_t in[DEC_IN_MAX]; // compressed bytes staged for the serial skim
   uint32_t nlit;
   uint32_t ncopy;
 };
@@ -187,6 +189,21 @@ __global__ __launch_bounds__(256) void k_decompress(
       continue;
     }
     uint32_t un = usize[i];
+    if (n > DEC_IN_MAX) { // oversized: serial reference decode
+      if (lane == 0) {
+        if (snappy_uncompress(src, n, dst, un) != un)
+          set_err(err_flag, DE_SNAPPY);
+      }
+      continue;
+    }
+    // stage compressed bytes: the skim is a serial byte walk and LDS
+    // loads are ~8x cheaper than global for a dependent chain
+    for (uint32_t t = lane * 4; t < n; t += WAVE * 4) {
+      uint32_t chunk = n - t < 4 ? n - t : 4;
+      for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = src[t + x];
+    }
+    wave_lds_sync2();
+    const uint8_t* lsrc = L.in;
     // ---- phase 1 (lane 0): skim ops into records.  Copies are grouped:
     // a copy joins the current group iff its source bytes are fully
     // decoded before the group runs (i.e. below the group's first output
@@ -198,7 +215,7 @@ __global__ __launch_bounds__(256) void k_decompress(
       L.nlit = 0;
       L.ncopy = 0;
       uint32_t hdr;
-      int k = varint32_get(src, src + (n < 5 ? n : 5), &hdr);
+      int k = varint32_get(lsrc, lsrc + (n < 5 ? n : 5), &hdr);
       if (k < 0 || hdr != un) {
         L.nlit = ~0u;
       } else {
@@ -206,7 +223,7 @@ __global__ __launch_bounds__(256) void k_decompress(
         uint32_t group_start = 0; // first output byte of the current group
         bool bad = false;
         while (ip < n) {
-          uint8_t tag = src[ip++];
+          uint8_t tag = lsrc[ip++];
           uint32_t len, aux, kind, head = 0;
           if ((tag & 3) == 0) {
             len = (tag >> 2) + 1;
@@ -215,7 +232,7 @@ __global__ __launch_bounds__(256) void k_decompress(
               if (ip + nb > n) { bad = true; break; }
               len = 0;
               for (uint32_t x = 0; x < nb; x++)
-                len |= (uint32_t)src[ip + x] << (8 * x);
+                len |= (uint32_t)lsrc[ip + x] << (8 * x);
               len += 1;
               ip += nb;
             }
@@ -228,12 +245,12 @@ __global__ __launch_bounds__(256) void k_decompress(
             if ((tag & 3) == 1) {
               if (ip + 1 > n) { bad = true; break; }
               len = ((tag >> 2) & 7) + 4;
-              off = ((uint32_t)(tag >> 5) << 8) | src[ip];
+              off = ((uint32_t)(tag >> 5) << 8) | lsrc[ip];
               ip += 1;
             } else if ((tag & 3) == 2) {
               if (ip + 2 > n) { bad = true; break; }
               len = (tag >> 2) + 1;
-              off = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8);
+              off = (uint32_t)lsrc[ip] | ((uint32_t)lsrc[ip + 1] << 8);
               ip += 2;
             } else {
               bad = true; // 4-byte-offset form: never emitted by us
@@ -282,10 +299,10 @@ __global__ __launch_bounds__(256) void k_decompress(
       uint32_t inp = (uint32_t)(r & 0xffffff);
       uint32_t t = 0;
       for (; t + 4 <= len; t += 4) {
-        uint32_t v = load32(src + inp + t);
+        uint32_t v = load32(lsrc + inp + t);
         memcpy(dst + out + t, &v, 4);
       }
-      for (; t < len; t++) dst[out + t] = src[inp + t];
+      for (; t < len; t++) dst[out + t] = lsrc[inp + t];
     }
     wave_lds_sync2();
     // ---- phase 3: copy groups in order; within a group every copy's
