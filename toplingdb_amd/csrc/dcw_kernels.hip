@@ -137,25 +137,14 @@ __global__ void k_verify_usize(const uint8_t* __restrict__ blob,
   }
 }
 
-// one wave per block: raw -> cooperative copy; snappy -> PARALLEL decode:
-// lane 0 skims the op stream into LDS records (tag walk only — ~10x
-// cheaper than decoding), then all lanes apply literal ops concurrently
-// (every output byte is written exactly once: literals read input only,
-// so they commute), then copy ops are applied in stream order (a copy may
-// read bytes produced by literals or earlier copies) with the whole wave
-// moving words; self-overlapping copies (offset < length) fall back to a
-// serial byte loop for that op.  Works for ANY public-snappy stream, not
-// just our own spec-v4 output; op-count overflow falls back to the serial
-// decoder.
-#define DEC_MAX_OPS 1024
-#define DEC_IN_MAX 4992
+// one wave per block: raw -> cooperative copy; snappy -> staged through LDS
+// (compressed in + decoded out both in LDS, lane 0 runs the serial decoder,
+// all lanes copy in/out); oversized blocks fall back to the direct path.
+#define DEC_MAX 4992 // 2*4992*4 waves = 39.9 KB LDS/WG -> 4 WGs (16 decoders)/CU
 struct DecLds {
-  // literal records grow from 0, copy records grow from the top (order
-  // preserved: copy k lives at ops[DEC_MAX_OPS-1-k])
-  uint64_t ops[DEC_MAX_OPS];
-  uint8_t in[DEC_IN_MAX]; // compressed bytes staged for the serial skim
-  uint32_t nlit;
-  uint32_t ncopy;
+  uint8_t in[DEC_MAX]; // compressed input + decoded output both staged in
+  uint8_t out[DEC_MAX]; // LDS: the serial byte decoder is LDS-latency bound
+                        // (a global-input variant measured ~10% slower)
 };
 __device__ __forceinline__ void wave_lds_sync2() {
   __builtin_amdgcn_s_waitcnt(0);
@@ -186,164 +175,29 @@ __global__ __launch_bounds__(256) void k_decompress(
       }
       if (lane == 0)
         for (uint32_t t = n & ~15u; t < n; t++) dst[t] = src[t];
-      continue;
-    }
-    uint32_t un = usize[i];
-    if (n > DEC_IN_MAX) { // oversized: serial reference decode
-      if (lane == 0) {
-        if (snappy_uncompress(src, n, dst, un) != un)
-          set_err(err_flag, DE_SNAPPY);
+    } else if (n <= DEC_MAX && usize[i] <= DEC_MAX) {
+      for (uint32_t t = lane * 4; t < n; t += WAVE * 4) {
+        uint32_t chunk = n - t < 4 ? n - t : 4;
+        for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = src[t + x];
       }
-      continue;
-    }
-    // stage compressed bytes: the skim is a serial byte walk and LDS
-    // loads are ~8x cheaper than global for a dependent chain
-    for (uint32_t t = lane * 4; t < n; t += WAVE * 4) {
-      uint32_t chunk = n - t < 4 ? n - t : 4;
-      for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = src[t + x];
-    }
-    wave_lds_sync2();
-    const uint8_t* lsrc = L.in;
-    // ---- phase 1 (lane 0): skim ops into records.  Copies are grouped:
-    // a copy joins the current group iff its source bytes are fully
-    // decoded before the group runs (i.e. below the group's first output
-    // or covered by literals-only prefix); a self-overlapping copy always
-    // opens a group (its source is then the decoded prefix).  Bit 57
-    // marks a group head.  Our spec-v4 encoder matches against FIRST
-    // occurrences, so sources cluster early and groups are few.
-    if (lane == 0) {
-      L.nlit = 0;
-      L.ncopy = 0;
-      uint32_t hdr;
-      int k = varint32_get(lsrc, lsrc + (n < 5 ? n : 5), &hdr);
-      if (k < 0 || hdr != un) {
-        L.nlit = ~0u;
-      } else {
-        uint32_t ip = (uint32_t)k, op = 0, nlit = 0, ncopy = 0;
-        uint32_t group_start = 0; // first output byte of the current group
-        bool bad = false;
-        while (ip < n) {
-          uint8_t tag = lsrc[ip++];
-          uint32_t len, aux, kind, head = 0;
-          if ((tag & 3) == 0) {
-            len = (tag >> 2) + 1;
-            if (len > 60) {
-              uint32_t nb = len - 60;
-              if (ip + nb > n) { bad = true; break; }
-              len = 0;
-              for (uint32_t x = 0; x < nb; x++)
-                len |= (uint32_t)lsrc[ip + x] << (8 * x);
-              len += 1;
-              ip += nb;
-            }
-            aux = ip; // input offset of the literal bytes
-            if (ip + len > n) { bad = true; break; }
-            ip += len;
-            kind = 0;
-          } else {
-            uint32_t off;
-            if ((tag & 3) == 1) {
-              if (ip + 1 > n) { bad = true; break; }
-              len = ((tag >> 2) & 7) + 4;
-              off = ((uint32_t)(tag >> 5) << 8) | lsrc[ip];
-              ip += 1;
-            } else if ((tag & 3) == 2) {
-              if (ip + 2 > n) { bad = true; break; }
-              len = (tag >> 2) + 1;
-              off = (uint32_t)lsrc[ip] | ((uint32_t)lsrc[ip + 1] << 8);
-              ip += 2;
-            } else {
-              bad = true; // 4-byte-offset form: never emitted by us
-              break;
-            }
-            if (off == 0 || off > op) { bad = true; break; }
-            aux = off;
-            kind = 1;
-            // head a new group unless the source bytes are all decoded
-            // before this group runs (self-overlap source ends at op, so
-            // it always heads one)
-            uint32_t src_end = off >= len ? (op - off + len) : op;
-            if (ncopy == 0 || src_end > group_start) {
-              head = 1;
-              group_start = op;
-            }
-          }
-          if (op + len > un || nlit + ncopy >= DEC_MAX_OPS) { bad = true; break; }
-          uint64_t rec = (uint64_t)head << 57 | (uint64_t)kind << 56 |
-                         (uint64_t)len << 40 | (uint64_t)op << 24 | aux;
-          if (kind == 0)
-            L.ops[nlit++] = rec;
-          else
-            L.ops[DEC_MAX_OPS - 1 - ncopy++] = rec;
-          op += len;
-        }
-        if (bad || op != un) L.nlit = ~0u;
-        else { L.nlit = nlit; L.ncopy = ncopy; }
-      }
-    }
-    wave_lds_sync2();
-    uint32_t nlit = L.nlit;
-    if (nlit == ~0u) { // unusual stream / overflow: serial reference decode
+      wave_lds_sync2();
       if (lane == 0) {
-        if (snappy_uncompress(src, n, dst, un) != un)
+        if (snappy_uncompress(L.in, n, L.out, usize[i]) != usize[i])
           set_err(err_flag, DE_SNAPPY);
       }
       wave_lds_sync2();
-      continue;
-    }
-    // ---- phase 2: literals in parallel (disjoint outputs, input reads)
-    for (uint32_t o = lane; o < nlit; o += WAVE) {
-      uint64_t r = L.ops[o];
-      uint32_t len = (uint32_t)(r >> 40) & 0xffff;
-      uint32_t out = (uint32_t)(r >> 24) & 0xffff;
-      uint32_t inp = (uint32_t)(r & 0xffffff);
-      uint32_t t = 0;
-      for (; t + 4 <= len; t += 4) {
-        uint32_t v = load32(lsrc + inp + t);
-        memcpy(dst + out + t, &v, 4);
-      }
-      for (; t < len; t++) dst[out + t] = lsrc[inp + t];
-    }
-    wave_lds_sync2();
-    // ---- phase 3: copy groups in order; within a group every copy's
-    // source is already decoded, so lanes apply whole copies in parallel
-    uint32_t ncopy = L.ncopy;
-    uint32_t c0 = 0;
-    while (c0 < ncopy) {
-      // group end: next head flag after c0 (wave ballot scan)
-      uint32_t cend = ncopy;
-      for (uint32_t base = c0 + 1; base < ncopy; base += WAVE) {
-        uint32_t idx = base + lane;
-        bool f = idx < ncopy && (L.ops[DEC_MAX_OPS - 1 - idx] >> 57 & 1);
-        uint64_t m = __ballot(f);
-        if (m) {
-          cend = base + __ffsll((unsigned long long)m) - 1;
-          break;
-        }
-      }
-      for (uint32_t c = c0 + lane; c < cend; c += WAVE) {
-        uint64_t r = L.ops[DEC_MAX_OPS - 1 - c];
-        uint32_t len = (uint32_t)(r >> 40) & 0xffff;
-        uint32_t out = (uint32_t)(r >> 24) & 0xffff;
-        uint32_t off = (uint32_t)(r & 0xffffff);
-        const uint8_t* s2 = dst + out - off;
-        uint8_t* d2 = dst + out;
-        if (off >= len) {
-          uint32_t t = 0;
-          for (; t + 4 <= len; t += 4) {
-            uint32_t v;
-            memcpy(&v, s2 + t, 4);
-            memcpy(d2 + t, &v, 4);
-          }
-          for (; t < len; t++) d2[t] = s2[t];
-        } else {
-          for (uint32_t t = 0; t < len; t++) d2[t] = s2[t % off];
-        }
+      uint32_t un = usize[i];
+      for (uint32_t t = lane * 4; t < un; t += WAVE * 4) {
+        uint32_t chunk = un - t < 4 ? un - t : 4;
+        for (uint32_t x = 0; x < chunk; x++) dst[t + x] = L.out[t + x];
       }
       wave_lds_sync2();
-      c0 = cend;
+    } else {
+      if (lane == 0) {
+        if (snappy_uncompress(src, n, dst, usize[i]) != usize[i])
+          set_err(err_flag, DE_SNAPPY);
+      }
     }
-    wave_lds_sync2();
   }
 }
 
