@@ -254,6 +254,22 @@ def test_concurrent_jobs_bit_exact(tmp_path):
     assert_identical(rb, ob)
 
 
+def test_corrupted_input_fails_loudly(tmp_path):
+    # flip one byte inside a data block: the worker must verify input
+    # block checksums and fail the job (block_fetcher.cc semantics), not
+    # emit output from corrupt data
+    runs = gen_runs(tmp_path, 1, 20000, compression=1)
+    p = runs[0][0]
+    blob = bytearray(open(p, "rb").read())
+    blob[len(blob) // 3] ^= 0xFF
+    open(p, "wb").write(bytes(blob))
+    out = tmp_path / "out"
+    out.mkdir()
+    jd = dcw.make_job(runs, str(out))
+    with pytest.raises(RuntimeError):
+        dcw.execute(jd)
+
+
 def test_grandparent_cuts_match_oracle(tmp_path):
     # grandparent-aware file cutting (ShouldStopBefore boundary rules,
     # compaction_outputs.cc:231-352): GPU worker vs oracle, bit-exact,
