@@ -23,11 +23,22 @@ def main(path, label=""):
     evc = cols(db, ev)
     # find the dispatch link column and the value column
     val = next(c for c in evc if "value" in c)
-    link = next((c for c in evc if "dispatch" in c), None)
+    link = next((c for c in evc if "dispatch" in c or "corr" in c or
+                 c.endswith("_id")), None)
+    if link is None:
+        print("rocpd_pmc_event columns:", evc)
+        for t in tables:
+            if "pmc" in t.lower() or "dispatch" in t.lower():
+                print(t, cols(db, t))
+        raise SystemExit("no link column")
     disp = next(t for t in tables if "kernel_dispatch" in t.lower()
                 and not t.endswith("_info"))
     dc = cols(db, disp)
-    did = next(c for c in dc if c in ("id", "dispatch_id"))
+    did = next((c for c in dc if c in ("id", "dispatch_id", "correlation_id")),
+               None)
+    if did is None:
+        print(disp, "columns:", dc)
+        raise SystemExit("no dispatch id column")
     kid = next(c for c in dc if "kernel" in c and "id" in c)
     sym = next(t for t in tables if "kernel_symbol" in t.lower())
     sc = cols(db, sym)
