@@ -998,7 +998,16 @@ __global__ __launch_bounds__(256) void k_compress(
     uint32_t hl = varint32_put(hdr, n); // lane-uniform
     if (lane == 0)
       for (uint32_t t = 0; t < hl; t++) gout[t] = hdr[t];
-    for (uint32_t t = 0; t < fl; t++) gout[hl + excl + t] = frag[t];
+    {
+      uint8_t* o2 = gout + hl + excl;
+      uint32_t t = 0;
+      for (; t + 4 <= fl; t += 4) { // misaligned dword stores are fine on gfx950
+        uint32_t v;
+        memcpy(&v, frag + t, 4);
+        memcpy(o2 + t, &v, 4);
+      }
+      for (; t < fl; t++) o2[t] = frag[t];
+    }
     if (lane == 0) {
       uint32_t cn = hl + total;
       // GoodCompressionRatio, default max_compressed_bytes_per_kb=896
