@@ -49,8 +49,8 @@ def gen_inputs(dcw, work_dir, rank, runs, entries, compression):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=5)
-    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--steps", type=int, default=20)  # >= 2x jobs-in-flight
+    ap.add_argument("--warmup", type=int, default=4)
     ap.add_argument("--runs", type=int, default=8)
     ap.add_argument("--entries-per-run", type=int, default=920_000,
                     help="~64 MiB snappy SST per run at 16B/100B")
