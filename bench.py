@@ -59,10 +59,12 @@ def main():
     ap.add_argument("--cpu-baseline-runs", type=int, default=4,
                     help="bounded oracle sample (number of input runs)")
     ap.add_argument("--skip-cpu-baseline", action="store_true")
-    ap.add_argument("--jobs-in-flight", type=int, default=10,
+    ap.add_argument("--jobs-in-flight", type=int, default=0,
                     help="concurrent compaction jobs per GPU (the production "
                          "dcompact worker runs concurrent jobs per node, "
-                         "BASELINE.json configs[4]); 1 = strictly sequential")
+                         "BASELINE.json configs[4]); 1 = strictly sequential; "
+                         "0 = auto: 10 at 1 rank, scaled down with ranks so "
+                         "per-node host threads and pinned memory stay sane")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -99,7 +101,10 @@ def main():
     log("rank %d: generated %d runs, %.1f MiB in %.1fs" %
         (rank, len(runs), in_bytes / 2**20, time.time() - t_gen))
 
-    jif = max(1, args.jobs_in_flight)
+    jif = args.jobs_in_flight
+    if jif <= 0:
+        jif = 10 if world == 1 else max(2, 12 // world)
+    jif = max(1, jif)
     slot_dirs = []
     for i in range(jif):
         d = os.path.join(work, "out%d" % i)
