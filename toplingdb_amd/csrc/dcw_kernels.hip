@@ -240,67 +240,45 @@ __device__ __forceinline__ int interval_bounds(
   return 0;
 }
 
-#define PARSE_STAGE_MAX 5120
-__global__ __launch_bounds__(256) void k_count_entries(
-    const uint8_t* __restrict__ ublob, const uint64_t* __restrict__ uoff,
-    const uint32_t* __restrict__ usize, const uint32_t* __restrict__ nrestarts,
-    const uint32_t* __restrict__ blk_iv0, uint32_t nblocks,
-    uint32_t* __restrict__ iv_cnt, uint32_t* err_flag) {
-  // one wave per block: stage the decoded block into LDS, then one lane
-  // per restart interval runs the serial varint walk against LDS (the
-  // walk is a dependent byte chain; LDS loads are ~8x cheaper than L2)
-  __shared__ uint8_t stage[4][PARSE_STAGE_MAX];
-  uint32_t waves_per_wg = blockDim.x / WAVE;
-  uint32_t wid = threadIdx.x / WAVE;
-  uint32_t wave = blockIdx.x * waves_per_wg + wid;
-  uint32_t lane = threadIdx.x % WAVE;
-  uint32_t stride = gridDim.x * waves_per_wg;
-  for (uint32_t bi = wave; bi < nblocks; bi += stride) {
-    const uint8_t* gsrc = ublob + uoff[bi];
-    uint32_t us = usize[bi];
-    const uint8_t* ublk = gsrc;
-    if (us <= PARSE_STAGE_MAX) {
-      for (uint32_t t = lane * 4; t < us; t += WAVE * 4) {
-        uint32_t chunk = us - t < 4 ? us - t : 4;
-        for (uint32_t x = 0; x < chunk; x++) stage[wid][t + x] = gsrc[t + x];
-      }
-      ublk = stage[wid];
+__global__ void k_count_entries(const uint8_t* __restrict__ ublob,
+                                const uint64_t* __restrict__ uoff,
+                                const uint32_t* __restrict__ usize,
+                                const uint32_t* __restrict__ nrestarts,
+                                const uint32_t* __restrict__ iv_block,
+                                const uint32_t* __restrict__ iv_local,
+                                uint32_t nintervals, uint32_t* __restrict__ iv_cnt,
+                                uint32_t* err_flag) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nintervals;
+       i += gridDim.x * blockDim.x) {
+    uint32_t b = iv_block[i];
+    const uint8_t* ublk = ublob + uoff[b];
+    uint32_t beg, end;
+    if (interval_bounds(ublk, usize[b], nrestarts[b], iv_local[i], &beg, &end) != 0) {
+      set_err(err_flag, DE_BLOCK_PARSE);
+      return;
     }
-    __builtin_amdgcn_s_waitcnt(0);
-    __builtin_amdgcn_wave_barrier();
-    uint32_t nr = nrestarts[bi];
-    uint32_t iv0 = blk_iv0[bi];
-    for (uint32_t j = lane; j < nr; j += WAVE) {
-      uint32_t beg, end;
-      if (interval_bounds(ublk, us, nr, j, &beg, &end) != 0) {
+    const uint8_t* p = ublk + beg;
+    const uint8_t* lim = ublk + end;
+    uint32_t n = 0;
+    while (p < lim) {
+      uint32_t shared, non_shared, vlen;
+      int a = varint32_get(p, lim, &shared);
+      if (a < 0) break;
+      p += a;
+      a = varint32_get(p, lim, &non_shared);
+      if (a < 0) break;
+      p += a;
+      a = varint32_get(p, lim, &vlen);
+      if (a < 0) break;
+      p += a;
+      p += non_shared + vlen;
+      if (p > lim) {
         set_err(err_flag, DE_BLOCK_PARSE);
-        continue;
+        return;
       }
-      const uint8_t* p = ublk + beg;
-      const uint8_t* lim = ublk + end;
-      uint32_t n = 0;
-      while (p < lim) {
-        uint32_t shared, non_shared, vlen;
-        int a = varint32_get(p, lim, &shared);
-        if (a < 0) break;
-        p += a;
-        a = varint32_get(p, lim, &non_shared);
-        if (a < 0) break;
-        p += a;
-        a = varint32_get(p, lim, &vlen);
-        if (a < 0) break;
-        p += a;
-        p += non_shared + vlen;
-        if (p > lim) {
-          set_err(err_flag, DE_BLOCK_PARSE);
-          break;
-        }
-        n++;
-      }
-      iv_cnt[iv0 + j] = n;
+      n++;
     }
-    __builtin_amdgcn_s_waitcnt(0);
-    __builtin_amdgcn_wave_barrier();
+    iv_cnt[i] = n;
   }
 }
 
@@ -322,124 +300,98 @@ __device__ __forceinline__ void key_set_byte(uint64_t& kb0, uint64_t& kb1,
   }
 }
 
-__global__ __launch_bounds__(256) void k_decode_entries(
+__global__ void k_decode_entries(
     const uint8_t* __restrict__ ublob, const uint64_t* __restrict__ uoff,
     const uint32_t* __restrict__ usize, const uint32_t* __restrict__ nrestarts,
-    const uint32_t* __restrict__ blk_iv0, const uint32_t* __restrict__ iv_base,
-    uint32_t nblocks, ulong4* __restrict__ ents, uint64_t* __restrict__ voff,
+    const uint32_t* __restrict__ iv_block, const uint32_t* __restrict__ iv_local,
+    const uint32_t* __restrict__ iv_base, uint32_t nintervals,
+    ulong4* __restrict__ ents, uint64_t* __restrict__ voff,
     uint32_t* __restrict__ vlen_out, uint8_t* __restrict__ klen_out,
     uint32_t* __restrict__ ukey_len_probe, uint32_t* err_flag) {
-  // one wave per block, one lane per restart interval, parsing from LDS
-  // (same staging rationale as k_count_entries above)
-  __shared__ uint8_t stage[4][PARSE_STAGE_MAX];
-  uint32_t waves_per_wg = blockDim.x / WAVE;
-  uint32_t wid = threadIdx.x / WAVE;
-  uint32_t wave = blockIdx.x * waves_per_wg + wid;
-  uint32_t lane = threadIdx.x % WAVE;
-  uint32_t stride = gridDim.x * waves_per_wg;
-  for (uint32_t bi = wave; bi < nblocks; bi += stride) {
-    const uint8_t* gsrc = ublob + uoff[bi];
-    uint32_t us = usize[bi];
-    uint64_t gbase = uoff[bi]; // voff values stay ublob-relative
-    const uint8_t* ublk = gsrc;
-    if (us <= PARSE_STAGE_MAX) {
-      for (uint32_t t = lane * 4; t < us; t += WAVE * 4) {
-        uint32_t chunk = us - t < 4 ? us - t : 4;
-        for (uint32_t x = 0; x < chunk; x++) stage[wid][t + x] = gsrc[t + x];
-      }
-      ublk = stage[wid];
-    }
-    __builtin_amdgcn_s_waitcnt(0);
-    __builtin_amdgcn_wave_barrier();
-    uint32_t nr = nrestarts[bi];
-    uint32_t iv0 = blk_iv0[bi];
-    for (uint32_t j = lane; j < nr; j += WAVE) {
-      uint32_t beg, end;
-      if (interval_bounds(ublk, us, nr, j, &beg, &end) != 0) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nintervals;
+       i += gridDim.x * blockDim.x) {
+    uint32_t b = iv_block[i];
+    const uint8_t* ublk = ublob + uoff[b];
+    uint32_t beg, end;
+    if (interval_bounds(ublk, usize[b], nrestarts[b], iv_local[i], &beg, &end) != 0)
+      return;
+    const uint8_t* p = ublk + beg;
+    const uint8_t* lim = ublk + end;
+    uint64_t kb0 = 0, kb1 = 0, ktail = 0; // key bytes in registers
+    uint32_t klen = 0;
+    uint32_t out = iv_base[i];
+    uint32_t probed_ulen = 0xffffffffu;
+    while (p < lim) {
+      uint32_t shared, non_shared, vl;
+      int a = varint32_get(p, lim, &shared);
+      if (a < 0) break;
+      p += a;
+      a = varint32_get(p, lim, &non_shared);
+      if (a < 0) break;
+      p += a;
+      a = varint32_get(p, lim, &vl);
+      if (a < 0) break;
+      p += a;
+      if (shared > klen || shared + non_shared > 24 ||
+          p + non_shared + vl > lim) {
         set_err(err_flag, DE_BLOCK_PARSE);
-        continue;
+        return;
       }
-      const uint8_t* p = ublk + beg;
-      const uint8_t* lim = ublk + end;
-      uint64_t kb0 = 0, kb1 = 0, ktail = 0; // key bytes in registers
-      uint32_t klen = 0;
-      uint32_t out = iv_base[iv0 + j];
-      uint32_t probed_ulen = 0xffffffffu;
-      while (p < lim) {
-        uint32_t shared, non_shared, vl;
-        int a = varint32_get(p, lim, &shared);
-        if (a < 0) break;
-        p += a;
-        a = varint32_get(p, lim, &non_shared);
-        if (a < 0) break;
-        p += a;
-        a = varint32_get(p, lim, &vl);
-        if (a < 0) break;
-        p += a;
-        if (shared > klen || shared + non_shared > 24 ||
-            p + non_shared + vl > lim) {
-          set_err(err_flag, DE_BLOCK_PARSE);
-          break;
-        }
-        for (uint32_t t = 0; t < non_shared; t++)
-          key_set_byte(kb0, kb1, ktail, shared + t, p[t]);
-        klen = shared + non_shared;
-        p += non_shared;
-        if (klen < 9) {
-          set_err(err_flag, DE_BLOCK_PARSE);
-          break;
-        }
-        uint32_t ulen = klen - 8;
-        if (ulen > 16) {
+      for (uint32_t t = 0; t < non_shared; t++)
+        key_set_byte(kb0, kb1, ktail, shared + t, p[t]);
+      klen = shared + non_shared;
+      p += non_shared;
+      if (klen < 9) {
+        set_err(err_flag, DE_BLOCK_PARSE);
+        return;
+      }
+      uint32_t ulen = klen - 8;
+      if (ulen > 16) {
+        set_err(err_flag, DE_UKEY_LEN);
+        return;
+      }
+      if (probed_ulen != ulen) {
+        uint32_t expect = atomicCAS(ukey_len_probe, 0xffffffffu, ulen);
+        if (expect != 0xffffffffu && expect != ulen) {
           set_err(err_flag, DE_UKEY_LEN);
-          break;
+          return;
         }
-        if (probed_ulen != ulen) {
-          uint32_t expect = atomicCAS(ukey_len_probe, 0xffffffffu, ulen);
-          if (expect != 0xffffffffu && expect != ulen) {
-            set_err(err_flag, DE_UKEY_LEN);
-            break;
-          }
-          probed_ulen = ulen;
-        }
-        // tag = 8 LE bytes starting at byte ulen of the key
-        uint64_t tag;
-        if (ulen == 16) {
-          tag = ktail;
-        } else if (ulen >= 8) {
-          uint32_t sh = 8 * (ulen - 8);
-          tag = sh ? ((kb1 >> sh) | (ktail << (64 - sh))) : kb1;
-        } else {
-          uint32_t sh = 8 * ulen;
-          tag = sh ? ((kb0 >> sh) | (kb1 << (64 - sh))) : kb0;
-        }
-        uint8_t vt = (uint8_t)tag;
-        if (!(vt == kTypeValue || vt == kTypeDeletion ||
-              vt == kTypeSingleDeletion)) {
-          set_err(err_flag, DE_TYPE);
-          break;
-        }
-        // normkey: big-endian words of the zero-padded user key
-        uint64_t u0, u1;
-        if (ulen >= 8) {
-          u0 = kb0;
-          uint32_t rem = ulen - 8; // bytes of user key in kb1
-          u1 = rem ? (kb1 & ((rem == 8) ? ~0ull : ((1ull << (8 * rem)) - 1))) : 0;
-        } else {
-          u0 = ulen ? (kb0 & ((1ull << (8 * ulen)) - 1)) : 0;
-          u1 = 0;
-        }
-        ents[out] = make_ulong4(__builtin_bswap64(u0), __builtin_bswap64(u1),
-                                ~tag, out);
-        voff[out] = gbase + (uint64_t)(p - ublk);
-        vlen_out[out] = vl;
-        klen_out[out] = (uint8_t)klen;
-        p += vl;
-        out++;
+        probed_ulen = ulen;
       }
+      // tag = 8 LE bytes starting at byte ulen of the key
+      uint64_t tag;
+      if (ulen == 16) {
+        tag = ktail;
+      } else if (ulen >= 8) {
+        uint32_t sh = 8 * (ulen - 8);
+        tag = sh ? ((kb1 >> sh) | (ktail << (64 - sh))) : kb1;
+      } else {
+        uint32_t sh = 8 * ulen;
+        tag = sh ? ((kb0 >> sh) | (kb1 << (64 - sh))) : kb0;
+      }
+      uint8_t vt = (uint8_t)tag;
+      if (!(vt == kTypeValue || vt == kTypeDeletion || vt == kTypeSingleDeletion)) {
+        set_err(err_flag, DE_TYPE);
+        return;
+      }
+      // normkey: big-endian words of the zero-padded user key
+      uint64_t u0, u1;
+      if (ulen >= 8) {
+        u0 = kb0;
+        uint32_t rem = ulen - 8; // bytes of user key in kb1
+        u1 = rem ? (kb1 & ((rem == 8) ? ~0ull : ((1ull << (8 * rem)) - 1))) : 0;
+      } else {
+        u0 = ulen ? (kb0 & ((1ull << (8 * ulen)) - 1)) : 0;
+        u1 = 0;
+      }
+      ents[out] = make_ulong4(__builtin_bswap64(u0), __builtin_bswap64(u1),
+                              ~tag, out);
+      voff[out] = uoff[b] + (uint64_t)(p - ublk);
+      vlen_out[out] = vl;
+      klen_out[out] = (uint8_t)klen;
+      p += vl;
+      out++;
     }
-    __builtin_amdgcn_s_waitcnt(0);
-    __builtin_amdgcn_wave_barrier();
   }
 }
 
@@ -1803,25 +1755,30 @@ int GpuJob::decode(std::string* err) {
     if (err) *err = "block decompress/parse failed, code " + std::to_string(err_host);
     return -1;
   }
-  // interval tables: intervals for block b are contiguous starting at
-  // blk_iv_base[b]; the parse kernels walk one wave per block
+  // interval tables
+  std::vector<uint32_t> iv_block, iv_local;
   std::vector<uint64_t> blk_iv_base(nb);
-  std::vector<uint32_t> blk_iv0(nb);
   uint64_t niv = 0;
   for (uint32_t b = 0; b < nb; b++) {
     blk_iv_base[b] = niv;
-    blk_iv0[b] = (uint32_t)niv;
+    for (uint32_t j = 0; j < nrestarts[b]; j++) {
+      iv_block.push_back(b);
+      iv_local.push_back(j);
+    }
     niv += nrestarts[b];
   }
   p->n_intervals = (uint32_t)niv;
-  HIPCHK(p->ens((void**)&p->d_iv_block, sizeof(uint32_t) * nb)); // blk_iv0
+  HIPCHK(p->ens((void**)&p->d_iv_block, sizeof(uint32_t) * niv));
+  HIPCHK(p->ens((void**)&p->d_iv_local, sizeof(uint32_t) * niv));
   HIPCHK(p->ens((void**)&p->d_iv_cnt, sizeof(uint32_t) * niv));
   HIPCHK(p->ens((void**)&p->d_iv_base, sizeof(uint32_t) * niv));
-  HIPCHK(p->h2d_meta(p->d_iv_block, blk_iv0.data(), sizeof(uint32_t) * nb));
+  HIPCHK(p->h2d_meta(p->d_iv_block, iv_block.data(), sizeof(uint32_t) * niv));
+  HIPCHK(p->h2d_meta(p->d_iv_local, iv_local.data(), sizeof(uint32_t) * niv));
   p->kbegin("count_entries", (double)p->ublob_size);
-  hipLaunchKernelGGL(k_count_entries, dim3(grid_for(nb * 4ull)), dim3(256), 0,
-                     p->stream, p->d_ublob, p->d_uoff, p->d_usize,
-                     p->d_nrestarts, p->d_iv_block, nb, p->d_iv_cnt, p->d_err);
+  hipLaunchKernelGGL(k_count_entries, dim3(grid_for(niv)), dim3(256), 0, p->stream,
+                     p->d_ublob, p->d_uoff, p->d_usize, p->d_nrestarts,
+                     p->d_iv_block, p->d_iv_local, (uint32_t)niv, p->d_iv_cnt,
+                     p->d_err);
   p->kend();
   std::vector<uint32_t> iv_cnt(niv);
   HIPCHK(hipMemcpyAsync(iv_cnt.data(), p->d_iv_cnt, sizeof(uint32_t) * niv,
@@ -1855,9 +1812,9 @@ int GpuJob::decode(std::string* err) {
   HIPCHK(p->ens((void**)&p->d_vlen, sizeof(uint32_t) * total_entries));
   HIPCHK(p->ens((void**)&p->d_klen, total_entries));
   p->kbegin("decode_entries", (double)p->ublob_size + 45.0 * total_entries);
-  hipLaunchKernelGGL(k_decode_entries, dim3(grid_for(nb * 4ull)), dim3(256), 0,
-                     p->stream, p->d_ublob, p->d_uoff, p->d_usize,
-                     p->d_nrestarts, p->d_iv_block, p->d_iv_base, nb,
+  hipLaunchKernelGGL(k_decode_entries, dim3(grid_for(niv)), dim3(256), 0,
+                     p->stream, p->d_ublob, p->d_uoff, p->d_usize, p->d_nrestarts,
+                     p->d_iv_block, p->d_iv_local, p->d_iv_base, (uint32_t)niv,
                      p->d_ent[0], p->d_voff, p->d_vlen, p->d_klen,
                      p->d_uklen_probe, p->d_err);
   p->kend();
