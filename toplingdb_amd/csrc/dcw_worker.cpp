@@ -18,7 +18,10 @@
 #include <csignal>
 #include <unistd.h>
 #include <cstring>
+#include <condition_variable>
+#include <deque>
 #include <future>
+#include <thread>
 #include <mutex>
 #include <string>
 #include <unordered_map>
@@ -39,6 +42,50 @@ std::mutex g_out_mu; // guards out_files slots filled by tail threads
 // jobs in flight on one device (the production dcompact worker runs
 // concurrent jobs per node; per-job streams are non-blocking so pipelines
 // interleave without cross-serialization).
+// Fixed worker pool for per-file tail jobs (separators, meta tail, file
+// write).  std::async spawns a fresh thread per task; at 10 jobs in
+// flight that is hundreds of transient threads contending with the
+// submission threads — a small fixed pool keeps host scheduling sane.
+// Tail jobs never wait on other tail jobs (their nested pwrite segments
+// use transient threads), so a bounded pool cannot deadlock.
+struct TailPool {
+  std::mutex mu;
+  std::condition_variable cv;
+  std::deque<std::packaged_task<int()>> q;
+  std::vector<std::thread> threads;
+  bool started = false;
+  void start_locked() {
+    unsigned n = std::thread::hardware_concurrency();
+    unsigned workers = n ? (n < 32 ? n : 32) : 8;
+    for (unsigned i = 0; i < workers; i++)
+      threads.emplace_back([this]() {
+        for (;;) {
+          std::packaged_task<int()> t;
+          {
+            std::unique_lock<std::mutex> lk(mu);
+            cv.wait(lk, [this]() { return !q.empty(); });
+            t = std::move(q.front());
+            q.pop_front();
+          }
+          t();
+        }
+      });
+    for (auto& t : threads) t.detach();
+    started = true;
+  }
+  std::future<int> submit(std::function<int()> fn) {
+    std::packaged_task<int()> task(std::move(fn));
+    std::future<int> f = task.get_future();
+    {
+      std::lock_guard<std::mutex> lk(mu);
+      if (!started) start_locked();
+      q.push_back(std::move(task));
+    }
+    cv.notify_one();
+    return f;
+  }
+} g_tails;
+
 struct JobPool {
   std::mutex mu;
   std::vector<GpuJob*> free_jobs;
@@ -749,9 +796,8 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     const uint32_t* vlen_p = vlen.data();
     uint64_t* out_bytes_p = &total_out_bytes;
     std::mutex* ob_mu = &g_out_mu;
-    writers.emplace_back(std::async(std::launch::async, [tj, klen_p, vlen_p,
-                                                         slot, &out_files,
-                                                         out_bytes_p, ob_mu]() -> int {
+    writers.emplace_back(g_tails.submit([tj, klen_p, vlen_p, slot, &out_files,
+                                         out_bytes_p, ob_mu]() -> int {
       // separators (FindShortestInternalKeySeparator between adjacent
       // blocks; last block keeps its last key — kShortenSeparators mode)
       for (void* e : tj->pend) GpuJob::wait_event(e); // image bytes complete
