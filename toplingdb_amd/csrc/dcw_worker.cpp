@@ -84,7 +84,14 @@ struct TailPool {
     cv.notify_one();
     return f;
   }
-} g_tails;
+};
+// intentionally leaked: destroying the pool's mutex/condvar at process
+// exit while detached workers wait on them hangs exit (pthread_cond
+// destruction blocks with waiters)
+TailPool& tails() {
+  static TailPool* p = new TailPool;
+  return *p;
+}
 
 struct JobPool {
   std::mutex mu;
@@ -796,7 +803,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     const uint32_t* vlen_p = vlen.data();
     uint64_t* out_bytes_p = &total_out_bytes;
     std::mutex* ob_mu = &g_out_mu;
-    writers.emplace_back(g_tails.submit([tj, klen_p, vlen_p, slot, &out_files,
+    writers.emplace_back(tails().submit([tj, klen_p, vlen_p, slot, &out_files,
                                          out_bytes_p, ob_mu]() -> int {
       // separators (FindShortestInternalKeySeparator between adjacent
       // blocks; last block keeps its last key — kShortenSeparators mode)
