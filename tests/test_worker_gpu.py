@@ -254,6 +254,22 @@ def test_concurrent_jobs_bit_exact(tmp_path):
     assert_identical(rb, ob)
 
 
+def test_nondefault_table_options(tmp_path):
+    # block_size / restart-interval / deviation variants flow through the
+    # plan FSM, emit kernel and index builder identically to the oracle
+    runs = gen_runs(tmp_path, 2, 40000, compression=1)
+    for bs, ri, dev in ((8192, 8, 10), (2048, 4, 25), (4096, 1, 0)):
+        og = tmp_path / ("g_%d_%d_%d" % (bs, ri, dev))
+        oo = tmp_path / ("o_%d_%d_%d" % (bs, ri, dev))
+        og.mkdir()
+        oo.mkdir()
+        kw = dict(compression=1, target_file_size=8 << 20, block_size=bs,
+                  block_restart_interval=ri, block_size_deviation=dev)
+        rg = dcw.execute(dcw.make_job(runs, str(og), **kw))
+        ro = oracle.execute(oracle.make_job(runs, str(oo), **kw))
+        assert_identical(rg, ro)
+
+
 def test_corrupted_input_fails_loudly(tmp_path):
     # flip one byte inside a data block: the worker must verify input
     # block checksums and fail the job (block_fetcher.cc semantics), not
