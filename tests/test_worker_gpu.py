@@ -258,10 +258,9 @@ def test_nondefault_table_options(tmp_path):
     # block_size / restart-interval / deviation variants flow through the
     # plan FSM, emit kernel and index builder identically to the oracle
     runs = gen_runs(tmp_path, 2, 40000, compression=1)
-    # snappy blocks are bounded by the GPU staging envelope (~5 KiB,
-    # DESIGN.md §7): 8 KiB blocks run uncompressed
     for bs, ri, dev, comp in ((8192, 8, 10, 0), (2048, 4, 25, 1),
-                              (4096, 1, 0, 1)):
+                              (4096, 1, 0, 1), (8192, 16, 10, 1),
+                              (16384, 16, 10, 1)):
         og = tmp_path / ("g_%d_%d_%d" % (bs, ri, dev))
         oo = tmp_path / ("o_%d_%d_%d" % (bs, ri, dev))
         og.mkdir()

@@ -794,7 +794,7 @@ struct EmitBlockDesc {
 // LDS offset table (<=512 entries per 4 KiB-class block), then all threads
 // encode their entries.  s_shared is the adjacent-survivor prefix from
 // k_shared_prefix (shared forced 0 at restart points).
-#define EMIT_MAX_ENTRIES 512
+#define EMIT_MAX_ENTRIES 1024
 __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
                        const uint64_t* __restrict__ s_k0,
                        const uint64_t* __restrict__ s_k1,
@@ -879,9 +879,9 @@ __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
 // the block's output slot.  The segment encoder itself is
 // dcw::snap_encode_segment — the SAME function the host C++ restatement
 // runs, so device/host byte-parity holds by construction.
-#define SNAP_MAX_UNC 5120 // block_size + slack; host guards this bound
+#define SNAP_MAX_UNC 16384 // block_size + slack; host guards this bound
 #define SNAP_MAX_OUT (32 + SNAP_MAX_UNC + SNAP_MAX_UNC / 6)
-#define SNAP_FRAG_MAX 128 // worst-case encode of one <=80 B segment
+#define SNAP_FRAG_MAX 304 // worst-case encode of one <=256 B segment
 
 // wave-internal LDS ordering: drain DS ops + stop compiler reordering
 __device__ __forceinline__ void wave_lds_sync() {
