@@ -190,6 +190,17 @@ typedef struct dcw_job_result {
 int32_t dcw_init(int32_t device_ordinal);
 void dcw_shutdown(void);
 
+/* Request cancellation of the job with this job_id (CompactionParams::
+ * job_id).  If the job is executing, it aborts at the next phase or
+ * output-chunk boundary and dcw_execute returns status DCW_CANCELLED; if
+ * no such job is running, the request is remembered and consumed by the
+ * next dcw_execute with that job_id.  Mirrors the executor's
+ * shutting-down checks (compaction_job.cc ShouldStopBefore/IsShuttingDown
+ * paths); the DB side treats the non-OK status like any worker failure
+ * (local fallback). */
+void dcw_cancel(int32_t job_id);
+#define DCW_CANCELLED 30
+
 /* Execute one compaction job (blocking).  Reentrant across threads after
  * dcw_init; one GPU job slot per call.  Mirrors
  * CompactionExecutor::Execute (compaction_executor.h:165-171). */

@@ -272,6 +272,25 @@ def test_nondefault_table_options(tmp_path):
         assert_identical(rg, ro)
 
 
+def test_cancel_pending_job(tmp_path):
+    # dcw_cancel before execute: the job aborts with DCW_CANCELLED (30)
+    # and produces no output files (include/dcw.h cancel contract)
+    runs = gen_runs(tmp_path, 1, 5000)
+    out = tmp_path / "out"
+    out.mkdir()
+    jd = dcw.make_job(runs, str(out))
+    jd.job_id = 7777
+    dcw.cancel(7777)
+    with pytest.raises(RuntimeError, match="cancel"):
+        dcw.execute(jd)
+    assert list(out.iterdir()) == []
+    # the request was consumed: the same job now runs normally
+    jd2 = dcw.make_job(runs, str(out))
+    jd2.job_id = 7777
+    r = dcw.execute(jd2)
+    assert r["out_entries"] == 5000
+
+
 def test_corrupted_input_fails_loudly(tmp_path):
     # flip one byte inside a data block: the worker must verify input
     # block checksums and fail the job (block_fetcher.cc semantics), not

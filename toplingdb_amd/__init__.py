@@ -133,6 +133,7 @@ def lib():
             C.c_char_p, C.c_uint64
         ]
         _lib.dcw_version.restype = C.c_char_p
+        _lib.dcw_cancel.argtypes = [C.c_int32]
     return _lib
 
 
@@ -144,6 +145,10 @@ def init(device=0):
 
 def shutdown():
     lib().dcw_shutdown()
+
+
+def cancel(job_id: int):
+    lib().dcw_cancel(job_id)
 
 
 def version() -> str:

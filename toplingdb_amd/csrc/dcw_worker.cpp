@@ -23,6 +23,7 @@
 #include <future>
 #include <thread>
 #include <mutex>
+#include <set>
 #include <string>
 #include <unordered_map>
 #include <vector>
@@ -120,6 +121,17 @@ struct StagedJob {
   uint64_t in_bytes = 0;
 };
 std::unordered_map<uint64_t, StagedJob*> g_staged;
+std::mutex g_cancel_mu;
+std::set<int32_t> g_cancelled;
+
+bool consume_cancel(int32_t job_id) {
+  std::lock_guard<std::mutex> lk(g_cancel_mu);
+  return g_cancelled.erase(job_id) != 0;
+}
+bool peek_cancel(int32_t job_id) {
+  std::lock_guard<std::mutex> lk(g_cancel_mu);
+  return g_cancelled.count(job_id) != 0;
+}
 
 uint64_t now_usec() {
   struct timespec ts;
@@ -335,6 +347,11 @@ void dcw_shutdown(void) {
   g_inited = false;
 }
 
+void dcw_cancel(int32_t job_id) {
+  std::lock_guard<std::mutex> lk(g_cancel_mu);
+  g_cancelled.insert(job_id);
+}
+
 void dcw_free_result(dcw_job_result* res) {
   free(res->files);
   res->files = nullptr;
@@ -436,6 +453,8 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   } else {
     if (job.stage(L.gi, &err) != 0) return fail(res, 17, err);
   }
+  if (consume_cancel(d->job_id))
+    return fail(res, 30 /*DCW_CANCELLED*/, "job cancelled");
   if (job.decode(&err) != 0) return fail(res, 18, err);
   wp.mark(2, now_usec());
   if (job.merge(&err) != 0) return fail(res, 19, err);
@@ -551,6 +570,10 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       plan_usec += now_usec() - tp0;
       wp.mark(12, now_usec());
       if (blocks.empty()) break;
+      if (peek_cancel(d->job_id)) {
+        consume_cancel(d->job_id);
+        return fail(res, 30 /*DCW_CANCELLED*/, "job cancelled");
+      }
       std::vector<uint32_t> csizes;
       wp.mark(11, now_usec());
       if (job.emit_blocks(blocks, o, &csizes, &err) != 0)
