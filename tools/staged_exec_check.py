@@ -1,4 +1,8 @@
-"""Minimal staged-execute repro with progress markers (crash bisection)."""
+"""Minimal staged-execute harness with progress markers.
+
+Runs gen -> stage_inputs -> two staged executes on cuda:0 with a marker
+printed after each step; useful for bisecting runtime issues on a GPU box
+(pairs with DCW_SEGV_TRACE=1 for a native backtrace on crash)."""
 import faulthandler
 import os
 import sys
