@@ -99,12 +99,13 @@ def ikey_sort(kvs):
     return sorted(kvs, key=lambda e: (e[0], -((e[1] << 8) | e[2])))
 
 
-def test_tombstones_and_snapshots_fuzz(tmp_path):
+@pytest.mark.parametrize("seed", [1234, 99, 424242])
+def test_tombstones_and_snapshots_fuzz(tmp_path, seed):
     # randomized Put/Delete/SingleDelete streams with snapshots; GPU FSM vs
     # oracle FSM must agree bit-for-bit.  SingleDelete contract: never mix
     # SD and non-SD writes for the same key (reference requirement); model
     # each key as either SD-managed or Put/Delete-managed.
-    rnd = random.Random(1234)
+    rnd = random.Random(seed)
     nkeys = 400
     sd_keys = {k for k in range(nkeys) if rnd.random() < 0.4}
     seq = 1
