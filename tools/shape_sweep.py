@@ -41,6 +41,76 @@ def run_case(tag, runs_n, entries, vlen, comp, tfs, snaps, bottom, seq0=1):
           (tag, len(rg["files"]), rg["out_entries"]), flush=True)
 
 
+def run_gp_case(tag, mcb):
+    import random
+    tmp = tempfile.mkdtemp(prefix="sweepgp_", dir="/dev/shm")
+    runs = []
+    for r in range(2):
+        p = os.path.join(tmp, "in%d.sst" % r)
+        dcw.gen_sst(p, seed=31 + r, num_entries=50000,
+                    seq_base=1 + r * 50000, compression=1)
+        runs.append([p])
+    rnd = random.Random(mcb & 0xFFFF)
+    gps = []
+    lo = b"\x00" * 16
+    for g in range(48):
+        hi = bytes([5 * g + rnd.randrange(1, 5)]) + bytes(
+            rnd.randrange(256) for _ in range(15))
+        if hi <= lo:
+            continue
+        gps.append((lo, hi, rnd.choice([1 << 20, 1 << 27, 1 << 31])))
+        lo = (hi + b"\x01")[:16]
+    og = os.path.join(tmp, "g")
+    oo = os.path.join(tmp, "o")
+    os.makedirs(og)
+    os.makedirs(oo)
+    kw = dict(compression=1, target_file_size=1 << 20,
+              max_compaction_bytes=mcb, grandparents=gps)
+    rg = dcw.execute(dcw.make_job(runs, og, **kw))
+    ro = oracle.execute(oracle.make_job(runs, oo, **kw))
+    assert len(rg["files"]) == len(ro["files"]), tag
+    for fg, fo in zip(rg["files"], ro["files"]):
+        with open(fg["path"], "rb") as a, open(fo["path"], "rb") as b:
+            assert a.read() == b.read(), (tag, fg["path"])
+    print("OK %-28s files=%d" % (tag, len(rg["files"])), flush=True)
+
+
+def run_concurrent_staged_case():
+    import concurrent.futures
+    tmp = tempfile.mkdtemp(prefix="sweepc_", dir="/dev/shm")
+    runs = []
+    for r in range(4):
+        p = os.path.join(tmp, "in%d.sst" % r)
+        dcw.gen_sst(p, seed=77 + r, num_entries=60000,
+                    seq_base=1 + r * 60000, compression=1)
+        runs.append([p])
+    jd0 = dcw.make_job(runs, tmp, compression=1)
+    h = dcw.stage_inputs(jd0)
+    dirs = []
+    for i in range(4):
+        d = os.path.join(tmp, "g%d" % i)
+        os.makedirs(d)
+        dirs.append(d)
+    kw = dict(compression=1, target_file_size=8 << 20)
+    with concurrent.futures.ThreadPoolExecutor(4) as ex:
+        futs = []
+        for i, d in enumerate(dirs):
+            jd = dcw.make_job(runs, d, staged_handle=h, **kw)
+            futs.append(ex.submit(dcw.execute, jd))
+        results = [f.result() for f in futs]
+    dcw.release_staged(h)
+    oo = os.path.join(tmp, "o")
+    os.makedirs(oo)
+    ro = oracle.execute(oracle.make_job(runs, oo, **kw))
+    for i, rg in enumerate(results):
+        assert len(rg["files"]) == len(ro["files"])
+        for fg, fo in zip(rg["files"], ro["files"]):
+            with open(fg["path"], "rb") as a, open(fo["path"], "rb") as b:
+                assert a.read() == b.read(), ("concurrent-staged", i, fg["path"])
+    print("OK %-28s 4 concurrent jobs, shared staged inputs" %
+          "concurrent-staged", flush=True)
+
+
 def main():
     dcw.init(0)
     cases = [
@@ -59,7 +129,10 @@ def main():
     ]
     for c in cases:
         run_case(*c)
-    print("SWEEP PASSED: %d shapes" % len(cases))
+    run_gp_case("grandparents-mcb-2g", 2 << 30)
+    run_gp_case("grandparents-mcb-256m", 256 << 20)
+    run_concurrent_staged_case()
+    print("SWEEP PASSED: %d shapes" % (len(cases) + 3))
     dcw.shutdown()
 
 
