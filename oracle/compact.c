@@ -635,6 +635,8 @@ static void out_open(outputs* o) {
   const dcw_job_desc* d = o->d;
   if (d->block_size) t.block_size = d->block_size;
   if (d->block_restart_interval) t.block_restart_interval = d->block_restart_interval;
+  if (d->index_block_restart_interval)
+    t.index_block_restart_interval = d->index_block_restart_interval;
   if (d->format_version) t.format_version = d->format_version;
   t.checksum_type = d->checksum_type;
   t.compression = d->compression;

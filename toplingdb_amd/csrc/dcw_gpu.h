@@ -73,11 +73,11 @@ class GpuJob {
   // Returns per-block final body sizes in comp_sizes.
   int emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts& o,
                   std::vector<uint32_t>* comp_sizes, std::string* err);
-  // GPU block planning: per-survivor next-block-start chain + packed
-  // (unc_size | nrestarts<<24) metadata, D2H into pinned host arrays owned
-  // by this job (valid until the next plan_all/reset).
-  int plan_all(const TableOpts& o, const uint32_t** next, const uint32_t** meta,
-               std::string* err);
+  // GPU block planning: per-survivor next-block-start chain + per-start
+  // uncompressed size (u32) and restart count (u16), D2H into pinned host
+  // arrays owned by this job (valid until the next plan_all/reset).
+  int plan_all(const TableOpts& o, const uint32_t** next, const uint32_t** unc,
+               const uint16_t** nrst, std::string* err);
   // Pack blocks [b0,b1) of the last emit_blocks call into one contiguous
   // [body|trailer]* image on device and D2H it straight into host_dst
   // (total_bytes = sum of (csize+5)); outoff[i] = image offset of block b0+i.

@@ -56,6 +56,7 @@ enum DevErr : uint32_t {
   DE_TYPE = 5,         // value type outside {Put, Delete, SingleDelete}
   DE_SD_CONTRACT = 6,  // SingleDelete + Delete mix (Corruption in reference)
   DE_COMPRESS_TYPE = 7,
+  DE_PLAN_WIDTH = 8,   // planned block exceeds the emit kernel's entry bound
 };
 
 static int g_device = -1;
@@ -794,7 +795,7 @@ struct EmitBlockDesc {
 // LDS offset table (<=512 entries per 4 KiB-class block), then all threads
 // encode their entries.  s_shared is the adjacent-survivor prefix from
 // k_shared_prefix (shared forced 0 at restart points).
-#define EMIT_MAX_ENTRIES 1024
+#define EMIT_MAX_ENTRIES 2048
 __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
                        const uint64_t* __restrict__ s_k0,
                        const uint64_t* __restrict__ s_k1,
@@ -1136,7 +1137,8 @@ __global__ void k_plan_next(const uint8_t* __restrict__ s_shared,
                             const uint32_t* __restrict__ s_vlen, uint64_t n,
                             uint32_t block_size, uint32_t restart_interval,
                             uint64_t dev_limit, uint32_t* __restrict__ next_out,
-                            uint32_t* __restrict__ meta_out) {
+                            uint32_t* __restrict__ unc_out,
+                            uint16_t* __restrict__ nr_out, uint32_t* err_flag) {
   for (uint64_t e = blockIdx.x * blockDim.x + threadIdx.x; e < n;
        e += (uint64_t)gridDim.x * blockDim.x) {
     uint64_t i = e;
@@ -1167,8 +1169,14 @@ __global__ void k_plan_next(const uint8_t* __restrict__ s_shared,
       i++;
     }
     next_out[e] = (uint32_t)i;
-    uint32_t unc = (uint32_t)(bytes + 4 * nrestarts + 4);
-    meta_out[e] = (unc & 0xffffffu) | (nrestarts << 24); // unc<=~4.3k, nr<=255
+    unc_out[e] = (uint32_t)(bytes + 4 * nrestarts + 4);
+    nr_out[e] = (uint16_t)nrestarts;
+    // unplannable shapes fail the JOB here (DB falls back local) instead of
+    // corrupting later: a block k_emit cannot hold, or a restart count that
+    // overflows u16 (blocks are <= block_size + one entry; EMIT_MAX_ENTRIES
+    // bounds restarts well under 64k for every emit-able block)
+    if (i - e > EMIT_MAX_ENTRIES || nrestarts > 0xffffu)
+      set_err(err_flag, DE_PLAN_WIDTH);
   }
 }
 
@@ -1468,7 +1476,7 @@ struct GpuJob::Impl {
   void *d_gp_sm0 = nullptr, *d_gp_sm1 = nullptr, *d_gp_lg0 = nullptr,
        *d_gp_lg1 = nullptr, *d_gp_tie = nullptr, *d_gp_pos = nullptr,
        *d_gp_nback = nullptr;
-  void *d_plan_next = nullptr, *d_plan_meta = nullptr;
+  void *d_plan_next = nullptr, *d_plan_meta = nullptr, *d_plan_nr = nullptr;
   void* h_plan = nullptr; // pinned host landing for next+meta
   size_t h_plan_cap = 0;
   bool h_plan_pageable = false;
@@ -1591,7 +1599,7 @@ GpuJob::~GpuJob() {
   F(p->d_scratch_mm);
   F(p->d_gp_sm0); F(p->d_gp_sm1); F(p->d_gp_lg0); F(p->d_gp_lg1);
   F(p->d_gp_tie); F(p->d_gp_pos); F(p->d_gp_nback);
-  F(p->d_plan_next); F(p->d_plan_meta);
+  F(p->d_plan_next); F(p->d_plan_meta); F(p->d_plan_nr);
   if (p->h_plan) {
     if (p->h_plan_pageable)
       free(p->h_plan);
@@ -2012,14 +2020,26 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
     if (d_snaps) (void)hipFree(d_snaps);
     return -1;
   }
+  bool lag_ran = false;
   for (uint64_t g = 0; g < ngroups; g++) {
     if (gflags[g] & GF_PRODUCED) {
       if (gflags[g] & GF_LAG_SENSITIVE) {
         hipLaunchKernelGGL(k_group_fsm, dim3(1), dim3(64), 0, p->stream, ents, n,
                            p->d_headidx, ngroups, P, p->d_survive, p->d_newtag,
                            p->d_clearv, (uint8_t*)nullptr, g, p->d_err);
+        lag_ran = true;
       }
       break;
+    }
+  }
+  if (lag_ran) { // a DE_SD_CONTRACT/DE_TYPE from the rerun must fail the job
+    HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost,
+                          p->stream));
+    HIPCHK(hipStreamSynchronize(p->stream));
+    if (err_host) {
+      if (err) *err = "dedup FSM (lag rerun) failed, code " + std::to_string(err_host);
+      if (d_snaps) (void)hipFree(d_snaps);
+      return -1;
     }
   }
   // survivor compaction
@@ -2068,14 +2088,15 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
 }
 
 int GpuJob::plan_all(const TableOpts& o, const uint32_t** next,
-                     const uint32_t** meta, std::string* err) {
+                     const uint32_t** unc, const uint16_t** nrst,
+                     std::string* err) {
   Impl* p = p_;
   uint64_t n = p->n_surv;
-  // pinned host landing buffers (grow-only)
-  if (p->h_plan_cap < n * 8) {
+  // pinned host landing buffers (grow-only): next u32 + unc u32 + nr u16
+  if (p->h_plan_cap < n * 10) {
     if (p->h_plan) (void)hipHostFree(p->h_plan);
     p->h_plan = nullptr;
-    size_t c = n * 8 + (n * 8) / 4 + 64;
+    size_t c = n * 10 + (n * 10) / 4 + 64;
     if (hipHostMalloc(&p->h_plan, c, hipHostMallocDefault) != hipSuccess) {
       p->h_plan = malloc(c); // pageable fallback, still correct
       p->h_plan_pageable = true;
@@ -2085,25 +2106,40 @@ int GpuJob::plan_all(const TableOpts& o, const uint32_t** next,
     p->h_plan_cap = c;
   }
   uint32_t* h_next = (uint32_t*)p->h_plan;
-  uint32_t* h_meta = h_next + n;
+  uint32_t* h_unc = h_next + n;
+  uint16_t* h_nr = (uint16_t*)(h_unc + n);
   HIPCHK(p->ens(&p->d_plan_next, n * 4 + 4));
   HIPCHK(p->ens(&p->d_plan_meta, n * 4 + 4));
+  HIPCHK(p->ens(&p->d_plan_nr, n * 2 + 4));
   uint64_t dev_limit =
       ((o.block_size * (100 - o.block_size_deviation)) + 99) / 100;
   p->kbegin("plan_next", 40.0 * n);
   hipLaunchKernelGGL(k_plan_next, dim3(grid_for(n)), dim3(256), 0, p->stream,
                      p->d_sshared, p->d_sklen, p->d_svlen, n, o.block_size,
                      o.block_restart_interval, dev_limit,
-                     (uint32_t*)p->d_plan_next, (uint32_t*)p->d_plan_meta);
+                     (uint32_t*)p->d_plan_next, (uint32_t*)p->d_plan_meta,
+                     (uint16_t*)p->d_plan_nr, p->d_err);
   p->kend();
   HIPCHK(hipMemcpyAsync(h_next, p->d_plan_next, n * 4, hipMemcpyDeviceToHost,
                         p->stream));
-  HIPCHK(hipMemcpyAsync(h_meta, p->d_plan_meta, n * 4, hipMemcpyDeviceToHost,
+  HIPCHK(hipMemcpyAsync(h_unc, p->d_plan_meta, n * 4, hipMemcpyDeviceToHost,
                         p->stream));
+  HIPCHK(hipMemcpyAsync(h_nr, p->d_plan_nr, n * 2, hipMemcpyDeviceToHost,
+                        p->stream));
+  uint32_t err_host = 0;
+  HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost, p->stream));
   HIPCHK(hipStreamSynchronize(p->stream));
   p->kresolve();
+  if (err_host) {
+    if (err) *err = "block plan failed, code " + std::to_string(err_host) +
+                    (err_host == DE_PLAN_WIDTH
+                         ? " (block shape exceeds the emit envelope)"
+                         : "");
+    return -1;
+  }
   *next = h_next;
-  *meta = h_meta;
+  *unc = h_unc;
+  *nrst = h_nr;
   return 0;
 }
 
@@ -2146,14 +2182,20 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
                      p->d_ucblob, o.block_restart_interval, p->d_err);
   p->kend();
   if (o.compression == 1) {
-    for (auto& pb : blocks)
+    uint32_t max_unc = 0;
+    for (auto& pb : blocks) {
       if (pb.unc_size > SNAP_MAX_UNC) {
         if (err)
           *err = "data block exceeds the GPU snappy staging bound (" +
                  std::to_string(pb.unc_size) + " B); job outside envelope";
         return -1;
       }
-    p->ccap_per_block = snappy_max_compressed(o.block_size + 1024);
+      if (pb.unc_size > max_unc) max_unc = pb.unc_size;
+    }
+    // per-block slot must hold the WORST-CASE encoder output of the largest
+    // planned block (a single large entry is always admitted past
+    // block_size), or k_compress would overrun into the next block's slot
+    p->ccap_per_block = (snappy_max_compressed(max_unc) + 63) & ~(uint64_t)63;
     ENSURE(p->d_cblob, p->cblob_cap, p->ccap_per_block * nb);
     p->kbegin("compress", 1.6 * (double)uout);
     uint32_t cgrid = (nb + 3) / 4;
@@ -2189,12 +2231,20 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   HIPCHK(hipMemcpyAsync(p->h_keys, p->d_scratch_keys,
                         (uint64_t)nb * BLKSTAT_STRIDE, hipMemcpyDeviceToHost,
                         p->stream));
+  uint32_t err_host = 0;
+  HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost, p->stream));
   (void)hipEventRecord(t1, p->stream);
   HIPCHK(hipStreamSynchronize(p->stream));
   ms_emit += ms_between(t0, t1);
   p->kresolve();
   (void)hipEventDestroy(t0);
   (void)hipEventDestroy(t1);
+  if (err_host) {
+    // an emit/compress error would otherwise be silently checksummed and
+    // written out as a corrupt block with status OK
+    if (err) *err = "block emit/compress failed, code " + std::to_string(err_host);
+    return -1;
+  }
   return 0;
 }
 

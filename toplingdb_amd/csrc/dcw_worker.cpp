@@ -199,6 +199,8 @@ TableOpts opts_from_desc(const dcw_job_desc* d) {
   TableOpts o;
   if (d->block_size) o.block_size = d->block_size;
   if (d->block_restart_interval) o.block_restart_interval = d->block_restart_interval;
+  if (d->index_block_restart_interval)
+    o.index_block_restart_interval = d->index_block_restart_interval;
   if (d->format_version) o.format_version = d->format_version;
   o.checksum_type = d->checksum_type;
   o.compression = d->compression;
@@ -403,6 +405,9 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     return fail(res, 12, "unsupported comparator (bytewise only)");
   if (d->compression > 1)
     return fail(res, 13, "unsupported compression (none/snappy only)");
+  if (d->checksum_type != 0 && d->checksum_type != 1 && d->checksum_type != 4)
+    return fail(res, 14,
+                "unsupported checksum_type (kNoChecksum/kCRC32c/kXXH3 only)");
 
   uint64_t t_start = now_usec();
   // fine-grained wall attribution (DCW_PHASE_DEBUG=1 prints at job end)
@@ -469,9 +474,10 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   // GPU block plan: per-survivor next-block-start chain (k_plan_next)
   TableOpts base_for_plan = opts_from_desc(d);
   const uint32_t* plan_next = nullptr;
-  const uint32_t* plan_meta = nullptr;
+  const uint32_t* plan_unc = nullptr;
+  const uint16_t* plan_nr = nullptr;
   if (nsurv > 0 &&
-      job.plan_all(base_for_plan, &plan_next, &plan_meta, &err) != 0)
+      job.plan_all(base_for_plan, &plan_next, &plan_unc, &plan_nr, &err) != 0)
     return fail(res, 28, err);
   wp.mark(5, now_usec());
 
@@ -556,12 +562,11 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
         size_t i = cur;
         while (i < nsurv && produced < min_unc) {
           uint32_t nx = plan_next[i];
-          uint32_t m = plan_meta[i];
           PlannedBlock pb;
           pb.first = (uint32_t)i;
           pb.count = nx - (uint32_t)i;
-          pb.unc_size = m & 0xffffffu;
-          pb.num_restarts = m >> 24;
+          pb.unc_size = plan_unc[i];
+          pb.num_restarts = plan_nr[i];
           blocks.push_back(pb);
           produced += pb.unc_size + kTrailerSize;
           i = nx;
@@ -967,6 +972,7 @@ int32_t dcw_gen_sst(const char* path, uint64_t seed, uint64_t num_entries,
                     uint32_t compression, uint32_t checksum_type,
                     uint64_t file_number, const char* db_id,
                     const char* db_session_id, uint64_t current_time) {
+  if (checksum_type != 0 && checksum_type != 1 && checksum_type != 4) return -1;
   TableOpts o;
   o.compression = compression;
   o.checksum_type = checksum_type;
