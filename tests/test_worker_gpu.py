@@ -308,6 +308,67 @@ def test_corrupted_input_fails_loudly(tmp_path):
         dcw.execute(jd)
 
 
+def test_oversize_incompressible_entry(tmp_path):
+    # regression (round-1 review): a single large entry is always admitted
+    # into a block past block_size; a poorly-compressible one used to overrun
+    # its fixed per-block compressed slot (silent device OOB write).  The
+    # slot is now sized from the chunk's largest planned block.
+    rnd = random.Random(31337)
+    kvs = []
+    for i in range(4000):
+        if i % 500 == 250:
+            v = bytes(rnd.getrandbits(8) for _ in range(12000))  # incompressible
+        else:
+            v = b"v" * 100
+        kvs.append((b"k%014d" % i, 1 + i, 1, v))
+    p = _write_kv_run(tmp_path, "big.sst", kvs, compression=0)
+    rg, ro = run_both(tmp_path, [[p]], compression=1, bottommost_level=1)
+    assert_identical(rg, ro)
+
+
+def test_dense_restarts_over_255(tmp_path):
+    # regression (round-1 review): >255 restarts per block used to wrap the
+    # packed 8-bit restart count in the GPU block plan (silent corruption).
+    # block_restart_interval=1 + empty values + 16 KiB blocks -> ~700
+    # restarts per block; bit-exact vs oracle.
+    runs = []
+    for r in range(2):
+        p = str(tmp_path / ("dr%d.sst" % r))
+        dcw.gen_sst(p, seed=101 + r, num_entries=30000, key_len=8, value_len=0,
+                    seq_base=1 + r * 30000)
+        runs.append([p])
+    rg, ro = run_both(tmp_path, runs, block_size=16384,
+                      block_restart_interval=1)
+    assert_identical(rg, ro)
+
+
+def test_rejects_unsupported_checksum_type(tmp_path):
+    # kXXHash(2)/kXXHash64(3) are declared in dcw.h but not implemented by
+    # the worker: the job must be REFUSED (DB falls back local), not written
+    # with zero checksums
+    runs = gen_runs(tmp_path, 1, 1000)
+    out = tmp_path / "out"
+    out.mkdir()
+    jd = dcw.make_job(runs, str(out), checksum_type=2)
+    with pytest.raises(RuntimeError, match="checksum"):
+        dcw.execute(jd)
+
+
+def test_index_restart_interval_honored(tmp_path):
+    # index_block_restart_interval flows into the index block for both
+    # worker and oracle (bit-compare covers the index bytes)
+    runs = gen_runs(tmp_path, 2, 20000)
+    rg, ro = run_both(tmp_path, runs, index_block_restart_interval=4)
+    assert_identical(rg, ro)
+    # and differs from the default-1 output (the field is not ignored)
+    og = tmp_path / "g1"
+    og.mkdir()
+    r1 = dcw.execute(dcw.make_job(runs, str(og)))
+    b_custom = open(rg["files"][0]["path"], "rb").read()
+    b_default = open(r1["files"][0]["path"], "rb").read()
+    assert b_custom != b_default
+
+
 def test_grandparent_cuts_match_oracle(tmp_path):
     # grandparent-aware file cutting (ShouldStopBefore boundary rules,
     # compaction_outputs.cc:231-352): GPU worker vs oracle, bit-exact,

@@ -241,6 +241,50 @@ __device__ __forceinline__ int interval_bounds(
   return 0;
 }
 
+// 16-byte register window over a sequential in-block scan.  The interval
+// walkers consume ~4 header bytes then skip ~100 value bytes; byte-granular
+// loads at that stride re-fetched each cache line ~2.6-5.5x (round-1 PMC,
+// profiles/pmc_hbm_traffic_r01_v4.txt).  Two u64 loads per 16 consumed
+// bytes at UNCHANGED thread parallelism cut the per-line touch count ~8x.
+// fill() may read up to 15 B past an interval/block end: d_ublob is
+// allocated with >=64 B of slack (Impl::ens), so the loads stay in bounds.
+struct ByteWin {
+  const uint8_t* base;
+  uint64_t w0, w1;
+  uint32_t off;
+  __device__ __forceinline__ void fill(uint32_t o) {
+    off = o;
+    memcpy(&w0, base + o, 8);
+    memcpy(&w1, base + o + 8, 8);
+  }
+  __device__ __forceinline__ uint8_t at(uint32_t pos) {
+    uint32_t d = pos - off;
+    if (d >= 16) {
+      fill(pos);
+      d = 0;
+    }
+    return d < 8 ? (uint8_t)(w0 >> (8 * d)) : (uint8_t)(w1 >> (8 * (d - 8)));
+  }
+};
+
+// varint32_get over the window; advances pos.  Same accept/reject set as
+// the pointer variant (dcw_common.h varint32_get).
+__device__ __forceinline__ bool win_varint32(ByteWin& W, uint32_t& pos,
+                                             uint32_t lim, uint32_t* v) {
+  uint32_t r = 0, sh = 0;
+  while (pos < lim && sh <= 28) {
+    uint8_t b = W.at(pos);
+    pos++;
+    r |= (uint32_t)(b & 0x7f) << sh;
+    if (!(b & 0x80)) {
+      *v = r;
+      return true;
+    }
+    sh += 7;
+  }
+  return false;
+}
+
 __global__ void k_count_entries(const uint8_t* __restrict__ ublob,
                                 const uint64_t* __restrict__ uoff,
                                 const uint32_t* __restrict__ usize,
@@ -258,22 +302,18 @@ __global__ void k_count_entries(const uint8_t* __restrict__ ublob,
       set_err(err_flag, DE_BLOCK_PARSE);
       return;
     }
-    const uint8_t* p = ublk + beg;
-    const uint8_t* lim = ublk + end;
+    ByteWin W;
+    W.base = ublk;
+    if (beg < end) W.fill(beg);
+    uint32_t pos = beg;
     uint32_t n = 0;
-    while (p < lim) {
+    while (pos < end) {
       uint32_t shared, non_shared, vlen;
-      int a = varint32_get(p, lim, &shared);
-      if (a < 0) break;
-      p += a;
-      a = varint32_get(p, lim, &non_shared);
-      if (a < 0) break;
-      p += a;
-      a = varint32_get(p, lim, &vlen);
-      if (a < 0) break;
-      p += a;
-      p += non_shared + vlen;
-      if (p > lim) {
+      if (!win_varint32(W, pos, end, &shared)) break;
+      if (!win_varint32(W, pos, end, &non_shared)) break;
+      if (!win_varint32(W, pos, end, &vlen)) break;
+      pos += non_shared + vlen;
+      if (pos > end) {
         set_err(err_flag, DE_BLOCK_PARSE);
         return;
       }
@@ -316,32 +356,28 @@ __global__ void k_decode_entries(
     uint32_t beg, end;
     if (interval_bounds(ublk, usize[b], nrestarts[b], iv_local[i], &beg, &end) != 0)
       return;
-    const uint8_t* p = ublk + beg;
-    const uint8_t* lim = ublk + end;
+    ByteWin W;
+    W.base = ublk;
+    if (beg < end) W.fill(beg);
+    uint32_t pos = beg;
     uint64_t kb0 = 0, kb1 = 0, ktail = 0; // key bytes in registers
     uint32_t klen = 0;
     uint32_t out = iv_base[i];
     uint32_t probed_ulen = 0xffffffffu;
-    while (p < lim) {
+    while (pos < end) {
       uint32_t shared, non_shared, vl;
-      int a = varint32_get(p, lim, &shared);
-      if (a < 0) break;
-      p += a;
-      a = varint32_get(p, lim, &non_shared);
-      if (a < 0) break;
-      p += a;
-      a = varint32_get(p, lim, &vl);
-      if (a < 0) break;
-      p += a;
+      if (!win_varint32(W, pos, end, &shared)) break;
+      if (!win_varint32(W, pos, end, &non_shared)) break;
+      if (!win_varint32(W, pos, end, &vl)) break;
       if (shared > klen || shared + non_shared > 24 ||
-          p + non_shared + vl > lim) {
+          (uint64_t)pos + non_shared + vl > end) {
         set_err(err_flag, DE_BLOCK_PARSE);
         return;
       }
       for (uint32_t t = 0; t < non_shared; t++)
-        key_set_byte(kb0, kb1, ktail, shared + t, p[t]);
+        key_set_byte(kb0, kb1, ktail, shared + t, W.at(pos + t));
       klen = shared + non_shared;
-      p += non_shared;
+      pos += non_shared;
       if (klen < 9) {
         set_err(err_flag, DE_BLOCK_PARSE);
         return;
@@ -387,10 +423,10 @@ __global__ void k_decode_entries(
       }
       ents[out] = make_ulong4(__builtin_bswap64(u0), __builtin_bswap64(u1),
                               ~tag, out);
-      voff[out] = uoff[b] + (uint64_t)(p - ublk);
+      voff[out] = uoff[b] + pos;
       vlen_out[out] = vl;
       klen_out[out] = (uint8_t)klen;
-      p += vl;
+      pos += vl;
       out++;
     }
   }
@@ -1400,10 +1436,14 @@ struct GpuJob::Impl {
     kpending.clear();
   }
   bool borrowed_stage = false; // staged buffers owned by a StagedInput
-  // staged input
+  // staged input: d_blob/d_boff/d_bsize are ALIASES of either this job's
+  // own grow-only slots (below) or a StagedInput's borrowed buffers
   uint8_t* d_blob = nullptr;
   uint64_t* d_boff = nullptr;
   uint32_t* d_bsize = nullptr;
+  uint8_t* d_blob_own = nullptr;
+  uint64_t* d_boff_own = nullptr;
+  uint32_t* d_bsize_own = nullptr;
   uint32_t n_blocks = 0;
   uint32_t checksum_type = 4;
   // decode
@@ -1547,11 +1587,8 @@ void GpuJob::reset() {
   (void)hipStreamSynchronize(p->stream);
   (void)hipStreamSynchronize(p->d2h_stream);
   p->arena_reset();
-  if (!p->borrowed_stage) {
-    if (p->d_blob) (void)hipFree(p->d_blob);
-    if (p->d_boff) (void)hipFree(p->d_boff);
-    if (p->d_bsize) (void)hipFree(p->d_bsize);
-  }
+  // own-slot staged buffers are grow-only (kept for the next job);
+  // borrowed ones belong to their StagedInput — either way just unalias
   p->d_blob = nullptr;
   p->d_boff = nullptr;
   p->d_bsize = nullptr;
@@ -1579,11 +1616,9 @@ GpuJob::~GpuJob() {
   auto F = [](void* x) {
     if (x) (void)hipFree(x);
   };
-  if (!p->borrowed_stage) {
-    F(p->d_blob);
-    F(p->d_boff);
-    F(p->d_bsize);
-  }
+  F(p->d_blob_own);
+  F(p->d_boff_own);
+  F(p->d_bsize_own);
   F(p->d_usize); F(p->d_btype_in);
   F(p->d_uoff); F(p->d_ublob); F(p->d_nrestarts); F(p->d_iv_block);
   F(p->d_iv_local); F(p->d_iv_cnt); F(p->d_iv_base); F(p->d_err);
@@ -1633,7 +1668,9 @@ int GpuJob::stage(const GpuInputs& in, std::string* err) {
   (void)hipEventCreate(&t0);
   (void)hipEventCreate(&t1);
   (void)hipEventRecord(t0, p->stream);
-  HIPCHK(hipMalloc(&p->d_blob, in.blob_size));
+  HIPCHK(p->ens((void**)&p->d_blob_own, in.blob_size));
+  p->d_blob = p->d_blob_own;
+  p->borrowed_stage = false;
   HIPCHK(hipMemcpyAsync(p->d_blob, in.blob, in.blob_size, hipMemcpyHostToDevice,
                         p->stream));
   p->n_blocks = (uint32_t)in.blocks.size();
@@ -1644,8 +1681,10 @@ int GpuJob::stage(const GpuInputs& in, std::string* err) {
     boff[i] = in.blocks[i].off;
     bsize[i] = in.blocks[i].size;
   }
-  HIPCHK(hipMalloc(&p->d_boff, sizeof(uint64_t) * p->n_blocks));
-  HIPCHK(hipMalloc(&p->d_bsize, sizeof(uint32_t) * p->n_blocks));
+  HIPCHK(p->ens((void**)&p->d_boff_own, sizeof(uint64_t) * p->n_blocks));
+  HIPCHK(p->ens((void**)&p->d_bsize_own, sizeof(uint32_t) * p->n_blocks));
+  p->d_boff = p->d_boff_own;
+  p->d_bsize = p->d_bsize_own;
   HIPCHK(p->h2d_meta(p->d_boff, boff.data(), sizeof(uint64_t) * p->n_blocks));
   HIPCHK(p->h2d_meta(p->d_bsize, bsize.data(), sizeof(uint32_t) * p->n_blocks));
   if (!p->d_crc) HIPCHK(hipMalloc(&p->d_crc, sizeof(Crc32cTables)));
@@ -1700,6 +1739,14 @@ int GpuJob::stage_release(StagedInput* s, std::string* err) {
   s->n_blocks = p->n_blocks;
   s->checksum_type = p->checksum_type;
   s->run_block_begin = run_blocks_;
+  // ownership moves to the StagedInput: drop the job's own-slot bookkeeping
+  // so ens() never frees or reuses the released buffers
+  p->caps[(void*)&p->d_blob_own] = 0;
+  p->caps[(void*)&p->d_boff_own] = 0;
+  p->caps[(void*)&p->d_bsize_own] = 0;
+  p->d_blob_own = nullptr;
+  p->d_boff_own = nullptr;
+  p->d_bsize_own = nullptr;
   p->borrowed_stage = true; // dtor must not free them now
   return 0;
 }

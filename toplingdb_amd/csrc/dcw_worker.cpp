@@ -139,61 +139,8 @@ uint64_t now_usec() {
   return (uint64_t)ts.tv_sec * 1000000 + ts.tv_nsec / 1000;
 }
 
-int read_file(const std::string& path, std::string* out, std::string* err) {
-  FILE* f = fopen(path.c_str(), "rb");
-  if (!f) {
-    *err = "cannot open " + path;
-    return -1;
-  }
-  fseek(f, 0, SEEK_END);
-  long sz = ftell(f);
-  fseek(f, 0, SEEK_SET);
-  out->resize((size_t)sz);
-  if (fread(&(*out)[0], 1, (size_t)sz, f) != (size_t)sz) {
-    fclose(f);
-    *err = "short read " + path;
-    return -1;
-  }
-  fclose(f);
-  return 0;
-}
-
-struct LoadedInputs {
-  std::string blob; // all files concatenated
-  GpuInputs gi;
-  uint64_t in_bytes = 0;
-};
-
-int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err) {
-  L->gi.run_block_begin.push_back(0);
-  uint32_t cstype = 0xffffffff;
-  for (uint32_t r = 0; r < d->num_runs; r++) {
-    for (uint32_t f = 0; f < d->runs[r].num_files; f++) {
-      std::string img;
-      if (read_file(d->runs[r].files[f], &img, err) != 0) return -1;
-      uint64_t base = L->blob.size();
-      ParsedSst ps = parse_sst((const uint8_t*)img.data(), img.size());
-      if (!ps.ok) {
-        *err = std::string(d->runs[r].files[f]) + ": " + ps.error;
-        return -1;
-      }
-      if (cstype == 0xffffffff) cstype = ps.checksum_type;
-      if (cstype != ps.checksum_type) {
-        *err = "mixed input checksum types unsupported";
-        return -1;
-      }
-      for (auto& h : ps.data_blocks)
-        L->gi.blocks.push_back({base + h.off, (uint32_t)h.size});
-      L->blob += img;
-      L->in_bytes += img.size();
-    }
-    L->gi.run_block_begin.push_back((uint32_t)L->gi.blocks.size());
-  }
-  L->gi.blob = (const uint8_t*)L->blob.data();
-  L->gi.blob_size = L->blob.size();
-  L->gi.checksum_type = cstype == 0xffffffff ? 4 : cstype;
-  return 0;
-}
+struct LoadedInputs; // fwd (RawBuf defined below the pin pool)
+int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err);
 
 TableOpts opts_from_desc(const dcw_job_desc* d) {
   TableOpts o;
@@ -226,14 +173,20 @@ struct PinPool {
   std::vector<std::pair<uint8_t*, size_t>> free_bufs;
   uint8_t* acquire(size_t n, size_t* cap) {
     {
+      // best fit: input blobs (~600 MB) and output images (~70 MB) share
+      // this pool; first-fit would hand a blob-sized buffer to an image
+      // and force a fresh pin for the next blob
       std::lock_guard<std::mutex> lk(mu);
-      for (size_t i = 0; i < free_bufs.size(); i++) {
-        if (free_bufs[i].second >= n) {
-          auto b = free_bufs[i];
-          free_bufs.erase(free_bufs.begin() + i);
-          *cap = b.second;
-          return b.first;
-        }
+      size_t best = free_bufs.size();
+      for (size_t i = 0; i < free_bufs.size(); i++)
+        if (free_bufs[i].second >= n &&
+            (best == free_bufs.size() || free_bufs[i].second < free_bufs[best].second))
+          best = i;
+      if (best < free_bufs.size()) {
+        auto b = free_bufs[best];
+        free_bufs.erase(free_bufs.begin() + best);
+        *cap = b.second;
+        return b.first;
       }
     }
     size_t c = n + n / 8 + (1 << 20);
@@ -285,6 +238,74 @@ struct RawBuf {
     if (p) g_pin_pool.release(p, cap);
   }
 };
+
+// Input images live in PINNED memory (RawBuf over the pin pool): the
+// per-job H2D of ~600 MB runs at PCIe rate and overlaps other jobs'
+// kernels on this job's own stream.  A real dcompact worker pays this
+// read+upload for every job (the reference hot loop reads its inputs,
+// table/block_fetcher.cc:242), so the bench times it by default.
+struct LoadedInputs {
+  RawBuf blob; // all files concatenated (pinned)
+  GpuInputs gi;
+  uint64_t in_bytes = 0;
+};
+
+int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err) {
+  // pre-size the pinned blob so reads land in place (no regrow memcpy)
+  uint64_t total = 0;
+  for (uint32_t r = 0; r < d->num_runs; r++)
+    for (uint32_t f = 0; f < d->runs[r].num_files; f++) {
+      struct stat st;
+      if (stat(d->runs[r].files[f], &st) != 0) {
+        *err = std::string("cannot stat ") + d->runs[r].files[f];
+        return -1;
+      }
+      total += (uint64_t)st.st_size;
+    }
+  L->blob.reserve(total);
+  L->gi.run_block_begin.push_back(0);
+  uint32_t cstype = 0xffffffff;
+  for (uint32_t r = 0; r < d->num_runs; r++) {
+    for (uint32_t f = 0; f < d->runs[r].num_files; f++) {
+      const char* path = d->runs[r].files[f];
+      FILE* fp = fopen(path, "rb");
+      if (!fp) {
+        *err = std::string("cannot open ") + path;
+        return -1;
+      }
+      fseek(fp, 0, SEEK_END);
+      uint64_t sz = (uint64_t)ftell(fp);
+      fseek(fp, 0, SEEK_SET);
+      uint64_t base = L->blob.len;
+      L->blob.reserve(base + sz);
+      if (fread(L->blob.p + base, 1, sz, fp) != sz) {
+        fclose(fp);
+        *err = std::string("short read ") + path;
+        return -1;
+      }
+      fclose(fp);
+      L->blob.len = base + sz;
+      ParsedSst ps = parse_sst(L->blob.p + base, sz);
+      if (!ps.ok) {
+        *err = std::string(path) + ": " + ps.error;
+        return -1;
+      }
+      if (cstype == 0xffffffff) cstype = ps.checksum_type;
+      if (cstype != ps.checksum_type) {
+        *err = "mixed input checksum types unsupported";
+        return -1;
+      }
+      for (auto& h : ps.data_blocks)
+        L->gi.blocks.push_back({base + h.off, (uint32_t)h.size});
+      L->in_bytes += sz;
+    }
+    L->gi.run_block_begin.push_back((uint32_t)L->gi.blocks.size());
+  }
+  L->gi.blob = L->blob.p;
+  L->gi.blob_size = L->blob.len;
+  L->gi.checksum_type = cstype == 0xffffffff ? 4 : cstype;
+  return 0;
+}
 
 // fixed-size block boundary key (ikey <= 24 B in the worker envelope);
 // avoids per-block heap strings on the main thread
