@@ -13,6 +13,7 @@ class valvec32 : public std::vector<T> {
   using std::vector<T>::vector;
   void reserve_aligned(size_t /*align*/, size_t cap) { this->reserve(cap); }
   void risk_set_size(size_t n) { this->resize(n); }
+  void erase_all() { this->clear(); }
 };
 
 } // namespace terark
