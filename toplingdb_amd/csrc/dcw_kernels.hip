@@ -202,6 +202,92 @@ __global__ __launch_bounds__(256) void k_decompress(
   }
 }
 
+// Decompress variant 1: LDS stages the DECODED side only; the serial
+// decoder reads compressed input straight from global (tag bytes are ~1
+// per ~35 output bytes; literal/copy payloads already move as words in
+// snappy_uncompress).  Halving the LDS footprint doubles resident
+// decoders: 4 waves/WG x 8 WGs/CU = 32 serial decoders per CU.
+struct DecLdsOut {
+  uint8_t out[DEC_MAX];
+};
+__global__ __launch_bounds__(256) void k_decompress_v1(
+    const uint8_t* __restrict__ blob, const uint64_t* __restrict__ boff,
+    const uint32_t* __restrict__ bsize, const uint8_t* __restrict__ btype,
+    const uint64_t* __restrict__ uoff, const uint32_t* __restrict__ usize,
+    uint32_t nblocks, uint8_t* __restrict__ ublob, uint32_t* err_flag) {
+  __shared__ DecLdsOut lds[4];
+  uint32_t waves_per_wg = blockDim.x / WAVE;
+  uint32_t wid = threadIdx.x / WAVE;
+  uint32_t wave = blockIdx.x * waves_per_wg + wid;
+  uint32_t lane = threadIdx.x % WAVE;
+  uint32_t stride = gridDim.x * waves_per_wg;
+  DecLdsOut& L = lds[wid];
+  for (uint32_t i = wave; i < nblocks; i += stride) {
+    const uint8_t* src = blob + boff[i];
+    uint8_t* dst = ublob + uoff[i];
+    uint32_t n = bsize[i];
+    if (btype[i] == 0) {
+      uint32_t pos = lane * 16;
+      for (; pos + 16 <= n; pos += WAVE * 16) {
+        ulong2 v;
+        memcpy(&v, src + pos, 16);
+        memcpy(dst + pos, &v, 16);
+      }
+      if (lane == 0)
+        for (uint32_t t = n & ~15u; t < n; t++) dst[t] = src[t];
+    } else if (usize[i] <= DEC_MAX) {
+      if (lane == 0) {
+        if (snappy_uncompress(src, n, L.out, usize[i]) != usize[i])
+          set_err(err_flag, DE_SNAPPY);
+      }
+      wave_lds_sync2();
+      uint32_t un = usize[i];
+      for (uint32_t t = lane * 4; t < un; t += WAVE * 4) {
+        uint32_t chunk = un - t < 4 ? un - t : 4;
+        for (uint32_t x = 0; x < chunk; x++) dst[t + x] = L.out[t + x];
+      }
+      wave_lds_sync2();
+    } else {
+      if (lane == 0) {
+        if (snappy_uncompress(src, n, dst, usize[i]) != usize[i])
+          set_err(err_flag, DE_SNAPPY);
+      }
+    }
+  }
+}
+
+// Decompress variant 2: no LDS at all — the serial decoder reads global
+// and writes global; back-references read recently-written (L1-hot)
+// output.  Occupancy is register-bound only.
+__global__ __launch_bounds__(256) void k_decompress_v2(
+    const uint8_t* __restrict__ blob, const uint64_t* __restrict__ boff,
+    const uint32_t* __restrict__ bsize, const uint8_t* __restrict__ btype,
+    const uint64_t* __restrict__ uoff, const uint32_t* __restrict__ usize,
+    uint32_t nblocks, uint8_t* __restrict__ ublob, uint32_t* err_flag) {
+  uint32_t waves_per_wg = blockDim.x / WAVE;
+  uint32_t wave = blockIdx.x * waves_per_wg + threadIdx.x / WAVE;
+  uint32_t lane = threadIdx.x % WAVE;
+  uint32_t stride = gridDim.x * waves_per_wg;
+  for (uint32_t i = wave; i < nblocks; i += stride) {
+    const uint8_t* src = blob + boff[i];
+    uint8_t* dst = ublob + uoff[i];
+    uint32_t n = bsize[i];
+    if (btype[i] == 0) {
+      uint32_t pos = lane * 16;
+      for (; pos + 16 <= n; pos += WAVE * 16) {
+        ulong2 v;
+        memcpy(&v, src + pos, 16);
+        memcpy(dst + pos, &v, 16);
+      }
+      if (lane == 0)
+        for (uint32_t t = n & ~15u; t < n; t++) dst[t] = src[t];
+    } else if (lane == 0) {
+      if (snappy_uncompress(src, n, dst, usize[i]) != usize[i])
+        set_err(err_flag, DE_SNAPPY);
+    }
+  }
+}
+
 __global__ void k_num_restarts(const uint8_t* __restrict__ ublob,
                                const uint64_t* __restrict__ uoff,
                                const uint32_t* __restrict__ usize, uint32_t nblocks,
@@ -1060,6 +1146,142 @@ __global__ __launch_bounds__(256) void k_compress(
   }
 }
 
+// ---- two-pass compress variants ----
+// Pass 1 measures each lane's fragment length (no stores), a shfl prefix
+// sum places the fragments, pass 2 re-runs the SAME segment encoder
+// writing directly to the block's output slot.  Removes the per-lane
+// private fragment buffer entirely: no scratch write+read traffic, no
+// 304 B/lane register/scratch footprint, and incompressible blocks skip
+// the output writes (the ratio check runs before pass 2).
+__device__ __forceinline__ uint32_t snap_lit_len(uint32_t n) {
+  // mirrors snap_emit_literal4v: 1-byte header below 60, else 2 (len<=256)
+  return n == 0 ? 0 : (n <= 60 ? 1 + n : 2 + n);
+}
+__device__ __forceinline__ uint32_t snap_copy_len(uint32_t offset, uint32_t len) {
+  // mirrors snap_emit_copy's chunking exactly
+  uint32_t c = 0;
+  while (len > 0) {
+    if (len >= 4 && len <= 11 && offset < 2048) return c + 2;
+    uint32_t chunk = len > 64 ? 64 : len;
+    if (len - chunk > 0 && len - chunk < 4) chunk = len - 4;
+    c += 3;
+    len -= chunk;
+  }
+  return c;
+}
+__device__ __forceinline__ uint32_t snap_measure_segment_dev(
+    const uint8_t* __restrict__ in, uint32_t s0, uint32_t s1,
+    const uint32_t* __restrict__ tab) {
+  uint32_t lit = s0, p = s0, out = 0;
+  while (p + 4 <= s1) {
+    uint32_t w = load32(in + p);
+    uint32_t h = (w * kSnapHashMul) >> (32 - kSnapHashBits);
+    uint32_t c = tab[h];
+    if (c != 0xffffffffu && c < p && load32(in + c) == w) {
+      uint32_t l = 4;
+      while (p + l + 4 <= s1) {
+        uint32_t a = load32(in + c + l);
+        uint32_t bz = load32(in + p + l);
+        uint32_t x = a ^ bz;
+        if (x) {
+          l += __builtin_ctz(x) >> 3;
+          goto mext_done;
+        }
+        l += 4;
+      }
+      while (p + l < s1 && in[c + l] == in[p + l]) l++;
+    mext_done:
+      out += snap_lit_len(p - lit) + snap_copy_len(p - c, l);
+      p += l;
+      lit = p;
+    } else {
+      p++;
+    }
+  }
+  return out + snap_lit_len(s1 - lit);
+}
+
+#define SNAP_LDSIN_MAX 5376 // LDS-staged input bound (default 4 KiB blocks)
+template <int LDSIN>
+__global__ __launch_bounds__(256) void k_compress_2p(
+    const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
+    const uint8_t* __restrict__ ucblob, uint8_t* __restrict__ cblob,
+    uint64_t ccap_per_block, uint32_t* __restrict__ bsize,
+    uint8_t* __restrict__ btype, uint32_t* err_flag) {
+  __shared__ uint32_t tabs[4][1u << kSnapHashBits]; // 8 KiB per wave
+  __shared__ uint8_t ins[LDSIN ? 4 : 1][LDSIN ? SNAP_LDSIN_MAX : 4];
+  uint32_t wid = threadIdx.x / WAVE;
+  uint32_t lane = threadIdx.x % WAVE;
+  uint32_t waves = blockDim.x / WAVE;
+  uint32_t* tab = tabs[wid];
+  for (uint32_t b = blockIdx.x * waves + wid; b < nblocks;
+       b += gridDim.x * waves) {
+    EmitBlockDesc d = bds[b];
+    if (d.unc_size > SNAP_MAX_UNC) {
+      if (lane == 0) {
+        set_err(err_flag, DE_BLOCK_PARSE);
+        bsize[b] = d.unc_size;
+        btype[b] = 0;
+      }
+      continue;
+    }
+    const uint8_t* gin = ucblob + d.uout;
+    uint32_t n = d.unc_size;
+    const uint8_t* in = gin;
+    if (LDSIN && n <= SNAP_LDSIN_MAX) {
+      uint8_t* li = ins[wid];
+      for (uint32_t t = lane * 16; t < n; t += WAVE * 16) {
+        uint32_t chunk = n - t < 16 ? n - t : 16;
+        if (chunk == 16) {
+          ulong2 v;
+          memcpy(&v, gin + t, 16);
+          memcpy(li + t, &v, 16);
+        } else {
+          for (uint32_t x = 0; x < chunk; x++) li[t + x] = gin[t + x];
+        }
+      }
+      in = li;
+    }
+    for (uint32_t t = lane; t < (1u << kSnapHashBits); t += WAVE)
+      tab[t] = 0xffffffffu;
+    wave_lds_sync();
+    for (uint32_t p = lane; p + 4 <= n; p += WAVE) {
+      uint32_t h = (load32(in + p) * kSnapHashMul) >> (32 - kSnapHashBits);
+      atomicMin(&tab[h], p);
+    }
+    wave_lds_sync();
+    uint32_t seg = (uint32_t)snap_segment_size(n);
+    uint32_t s0 = lane * seg;
+    uint32_t s1 = s0 + seg < n ? s0 + seg : n;
+    uint32_t fl = s0 < n ? snap_measure_segment_dev(in, s0, s1, tab) : 0;
+    uint32_t inc = fl;
+    for (int sh = 1; sh < WAVE; sh <<= 1) {
+      uint32_t v = __shfl_up(inc, sh);
+      if ((int)lane >= sh) inc += v;
+    }
+    uint32_t total = __shfl(inc, WAVE - 1);
+    uint32_t excl = inc - fl;
+    uint8_t hdr[5];
+    uint32_t hl = varint32_put(hdr, n); // lane-uniform
+    uint32_t cn = hl + total;
+    // GoodCompressionRatio check BEFORE emitting: incompressible blocks
+    // write nothing (pack copies the raw block)
+    if (cn <= (((uint64_t)896 * n) >> 10)) {
+      uint8_t* gout = cblob + (uint64_t)b * ccap_per_block;
+      if (lane == 0) {
+        for (uint32_t t = 0; t < hl; t++) gout[t] = hdr[t];
+        bsize[b] = cn;
+        btype[b] = 1;
+      }
+      if (s0 < n) (void)snap_encode_segment_dev(in, s0, s1, tab, gout + hl + excl);
+    } else if (lane == 0) {
+      bsize[b] = n;
+      btype[b] = 0;
+    }
+    wave_lds_sync();
+  }
+}
+
 __global__ void k_checksum(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
                            const uint8_t* __restrict__ ucblob,
                            const uint8_t* __restrict__ cblob, uint64_t ccap,
@@ -1792,10 +2014,24 @@ int GpuJob::decode(std::string* err) {
     std::lock_guard<std::mutex> lk(g_kmu);
     kstats()["verify_checksum"].alg_bytes += in_block_bytes;
   }
+  static const int decomp_v =
+      getenv("DCW_DECOMP_V") ? atoi(getenv("DCW_DECOMP_V")) : 0;
   p->kbegin("decompress", in_block_bytes + (double)acc);
-  hipLaunchKernelGGL(k_decompress, dim3(grid_for(nb * 4ull)), dim3(256), 0,
-                     p->stream, p->d_blob, p->d_boff, p->d_bsize, p->d_btype_in,
-                     p->d_uoff, p->d_usize, nb, p->d_ublob, p->d_err);
+  if (decomp_v == 1)
+    hipLaunchKernelGGL(k_decompress_v1, dim3(grid_for(nb * 4ull)), dim3(256), 0,
+                       p->stream, p->d_blob, p->d_boff, p->d_bsize,
+                       p->d_btype_in, p->d_uoff, p->d_usize, nb, p->d_ublob,
+                       p->d_err);
+  else if (decomp_v == 2)
+    hipLaunchKernelGGL(k_decompress_v2, dim3(grid_for(nb * 4ull)), dim3(256), 0,
+                       p->stream, p->d_blob, p->d_boff, p->d_bsize,
+                       p->d_btype_in, p->d_uoff, p->d_usize, nb, p->d_ublob,
+                       p->d_err);
+  else
+    hipLaunchKernelGGL(k_decompress, dim3(grid_for(nb * 4ull)), dim3(256), 0,
+                       p->stream, p->d_blob, p->d_boff, p->d_bsize,
+                       p->d_btype_in, p->d_uoff, p->d_usize, nb, p->d_ublob,
+                       p->d_err);
   p->kend();
   HIPCHK(p->ens((void**)&p->d_nrestarts, sizeof(uint32_t) * nb));
   hipLaunchKernelGGL(k_num_restarts, dim3(grid_for(nb)), dim3(256), 0, p->stream,
@@ -2244,11 +2480,25 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
     // block_size), or k_compress would overrun into the next block's slot
     p->ccap_per_block = (snappy_max_compressed(max_unc) + 63) & ~(uint64_t)63;
     ENSURE(p->d_cblob, p->cblob_cap, p->ccap_per_block * nb);
+    static const int comp_v =
+        getenv("DCW_COMPRESS_V") ? atoi(getenv("DCW_COMPRESS_V")) : 0;
     p->kbegin("compress", 1.6 * (double)uout);
     uint32_t cgrid = (nb + 3) / 4;
-    hipLaunchKernelGGL(k_compress, dim3(cgrid < 4096 ? cgrid : 4096), dim3(256),
-                       0, p->stream, p->d_bds, nb, p->d_ucblob, p->d_cblob,
-                       p->ccap_per_block, p->d_ebsize, p->d_ebtype, p->d_err);
+    if (comp_v == 2)
+      hipLaunchKernelGGL(k_compress_2p<1>, dim3(cgrid < 4096 ? cgrid : 4096),
+                         dim3(256), 0, p->stream, p->d_bds, nb, p->d_ucblob,
+                         p->d_cblob, p->ccap_per_block, p->d_ebsize,
+                         p->d_ebtype, p->d_err);
+    else if (comp_v == 3)
+      hipLaunchKernelGGL(k_compress_2p<0>, dim3(cgrid < 4096 ? cgrid : 4096),
+                         dim3(256), 0, p->stream, p->d_bds, nb, p->d_ucblob,
+                         p->d_cblob, p->ccap_per_block, p->d_ebsize,
+                         p->d_ebtype, p->d_err);
+    else
+      hipLaunchKernelGGL(k_compress, dim3(cgrid < 4096 ? cgrid : 4096),
+                         dim3(256), 0, p->stream, p->d_bds, nb, p->d_ucblob,
+                         p->d_cblob, p->ccap_per_block, p->d_ebsize,
+                         p->d_ebtype, p->d_err);
     p->kend();
   } else {
     hipLaunchKernelGGL(k_sizes_nocomp, dim3(grid_for(nb)), dim3(256), 0,
