@@ -141,6 +141,13 @@ def main():
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--skip-hbm-resident", action="store_true",
                     help="skip the secondary staged-inputs region")
+    ap.add_argument("--job-mix", choices=["uniform", "mixed"], default="uniform",
+                    help="mixed = BASELINE configs[4] job mix (alternating "
+                         "L0→L1 4-way and L1→L2 8-way jobs) through the "
+                         "node job queue; uniform = the headline configs[2] "
+                         "workload (still queue-dispatched at N>1)")
+    ap.add_argument("--mix-jobs", type=int, default=0,
+                    help="job count for --job-mix mixed (0 = 8 per rank)")
     ap.add_argument("--jobs-in-flight", type=int, default=0,
                     help="concurrent compaction jobs per GPU (the production "
                          "dcompact worker runs concurrent jobs per node, "
@@ -214,6 +221,64 @@ def main():
 
     from concurrent.futures import ThreadPoolExecutor
     pool = ThreadPoolExecutor(max_workers=jif)
+
+    # ---- node job queue (toplingdb_amd/dcompact_queue.py): metadata
+    # broadcast from rank 0 (the only collective), deterministic LPT
+    # assignment, per-rank concurrent execution ----
+    from toplingdb_amd import dcompact_queue as dq
+    coord = dq.QueueCoordinator(dist, rank, world)
+
+    def rank_input_paths(r):
+        w = os.path.join(args.workdir, "r%d" % r)
+        return [[os.path.join(w, "in_r%d_%d.sst" % (r, i))]
+                for i in range(args.runs)]
+
+    my_queue_jobs = None
+    queue_total_bytes = None
+    mix_jobs_meta = None
+    if args.job_mix == "mixed":
+        # BASELINE configs[4] mix through the queue: generation sharded
+        # round-robin over ranks, then broadcast + LPT by input bytes
+        n_mix = args.mix_jobs if args.mix_jobs > 0 else 8 * world
+        mixdir = os.path.join(args.workdir, "mix")
+        os.makedirs(mixdir, exist_ok=True)
+        all_mix = dq.build_job_mix(mixdir, dcw.gen_sst, n_jobs=n_mix,
+                                   world=world)
+        for j in all_mix:
+            if j["job_id"] % world == rank:
+                dq.gen_job_inputs(j, dcw.gen_sst,
+                                  compression=args.compression)
+        if dist:
+            dist.barrier()
+        mix_jobs_meta = coord.broadcast_jobs(all_mix if rank == 0 else [])
+        weights = [sum(os.path.getsize(p) for p in j["paths"])
+                   for j in mix_jobs_meta]
+        my_queue_jobs, _ = coord.my_jobs(mix_jobs_meta, weights)
+        queue_total_bytes = sum(weights)
+    elif world > 1:
+        # headline workload, queue-dispatched: steps*world equal jobs, job
+        # i over rank (i % world)'s input set; any rank may execute any set
+        meta = [{"job_id": 5000 + i, "input_rank": i % world}
+                for i in range(args.steps * world)]
+        meta = coord.broadcast_jobs(meta if rank == 0 else [])
+        my_queue_jobs, _ = coord.my_jobs(meta, [1] * len(meta))
+        queue_total_bytes = in_bytes * len(meta)
+
+    def exec_queue_job(jm, slot):
+        if "paths" in jm:  # mixed-mode job
+            return dcw.execute(dcw.make_job(
+                [[p] for p in jm["paths"]], slot_dirs[slot],
+                compression=args.compression, target_file_size=64 << 20,
+                bottommost_level=jm["bottommost"]))
+        return dcw.execute(dcw.make_job(
+            rank_input_paths(jm["input_rank"]), slot_dirs[slot],
+            compression=args.compression, target_file_size=64 << 20,
+            bottommost_level=1))
+
+    def run_queue():
+        futs = [pool.submit(exec_queue_job, jm, i % jif)
+                for i, jm in enumerate(my_queue_jobs)]
+        return [f.result() for f in futs]
 
     def run_steps(k, staged=0):
         futs = [pool.submit(lambda i: dcw.execute(job(i % jif, staged=staged)), i)
