@@ -366,9 +366,17 @@ DCW_HD uint8_t* snap_emit_copy(uint8_t* op, size_t offset, size_t len) {
     }
     size_t chunk = len > 64 ? 64 : len;
     if (len - chunk > 0 && len - chunk < 4) chunk = len - 4;
-    *op++ = (uint8_t)(2 | ((chunk - 1) << 2));
-    *op++ = (uint8_t)(offset & 0xff);
-    *op++ = (uint8_t)(offset >> 8);
+    if (offset < 65536) {
+      *op++ = (uint8_t)(2 | ((chunk - 1) << 2));
+      *op++ = (uint8_t)(offset & 0xff);
+      *op++ = (uint8_t)(offset >> 8);
+    } else { // 4-byte-offset form (dictionary matches reach past 64 KiB)
+      *op++ = (uint8_t)(3 | ((chunk - 1) << 2));
+      *op++ = (uint8_t)(offset & 0xff);
+      *op++ = (uint8_t)((offset >> 8) & 0xff);
+      *op++ = (uint8_t)((offset >> 16) & 0xff);
+      *op++ = (uint8_t)((offset >> 24) & 0xff);
+    }
     len -= chunk;
   }
   return op;
