@@ -503,6 +503,140 @@ DCW_HD size_t snappy_uncompress(const uint8_t* in, size_t n, uint8_t* out,
   return op == oend ? ulen : 0;
 }
 
+// ---------- dictionary snappy (DcwZipTable value blocks) ----------
+// "DZT dict codec v1" (self-pinned, like the base codec): virtual stream =
+// dict[0,D) || block[D, D+n).  Hash table = first occurrence over dict
+// positions [0, D-4] (precomputed once per file) merged with block
+// positions [D, D+n-4] under the min-position rule (dict always wins;
+// block first-occurrence fills empty slots — order-independent, so the
+// serial first-wins scan and a parallel atomicMin build agree).  Segments
+// = spec-v4 segmentation of the BLOCK bytes; copies may reach into the
+// dict (4-byte-offset copy form for offsets >= 64 KiB); literals come
+// from block bytes only.  Encoded stream = varint(n) + segment streams.
+DCW_HD uint8_t snap_vbyte(const uint8_t* dict, uint32_t D, const uint8_t* in,
+                          uint32_t pos) {
+  return pos < D ? dict[pos] : in[pos - D];
+}
+DCW_HD uint32_t snap_vload32(const uint8_t* dict, uint32_t D,
+                             const uint8_t* in, uint32_t pos) {
+  if (pos >= D) return load32(in + pos - D);
+  if (pos + 4 <= D) return load32(dict + pos);
+  uint8_t b[4];
+  for (int i = 0; i < 4; i++) b[i] = snap_vbyte(dict, D, in, pos + i);
+  uint32_t v;
+  memcpy(&v, b, 4);
+  return v;
+}
+// first-occurrence table over the dict alone (built once per file)
+DCW_HD void snap_dict_table(const uint8_t* dict, uint32_t D,
+                            uint32_t* __restrict__ tab) {
+  for (uint32_t t = 0; t < (1u << kSnapHashBits); t++) tab[t] = 0xffffffffu;
+  for (uint32_t p = 0; p + 4 <= D; p++) {
+    uint32_t h = (load32(dict + p) * kSnapHashMul) >> (32 - kSnapHashBits);
+    if (tab[h] == 0xffffffffu) tab[h] = p;
+  }
+}
+// encode one block segment; s0/s1 are VIRTUAL positions (>= D)
+DCW_HD uint8_t* snap_encode_segment_dict(const uint8_t* dict, uint32_t D,
+                                         const uint8_t* in, uint32_t s0,
+                                         uint32_t s1, const uint32_t* tab,
+                                         uint8_t* op) {
+  uint32_t lit = s0, p = s0;
+  while (p + 4 <= s1) {
+    uint32_t w = load32(in + (p - D));
+    uint32_t h = (w * kSnapHashMul) >> (32 - kSnapHashBits);
+    uint32_t c = tab[h];
+    if (c != 0xffffffffu && c < p && snap_vload32(dict, D, in, c) == w) {
+      uint32_t l = 4;
+      while (p + l < s1 && snap_vbyte(dict, D, in, c + l) == in[p + l - D]) l++;
+      op = snap_emit_literal(op, in + (lit - D), p - lit);
+      op = snap_emit_copy(op, p - c, l);
+      p += l;
+      lit = p;
+    } else {
+      p++;
+    }
+  }
+  return snap_emit_literal(op, in + (lit - D), s1 - lit);
+}
+// serial whole-block dict compress (oracle/host restatement path)
+DCW_HD size_t snappy_compress_block_dict(const uint8_t* dict, uint32_t D,
+                                         const uint32_t* dict_tab,
+                                         const uint8_t* in, size_t n,
+                                         uint8_t* out, uint32_t* tab) {
+  uint8_t* op = out;
+  op += varint32_put(op, (uint32_t)n);
+  if (n == 0) return (size_t)(op - out);
+  for (uint32_t t = 0; t < (1u << kSnapHashBits); t++) tab[t] = dict_tab[t];
+  for (uint32_t p = 0; p + 4 <= n; p++) {
+    uint32_t h = (load32(in + p) * kSnapHashMul) >> (32 - kSnapHashBits);
+    if (tab[h] == 0xffffffffu) tab[h] = D + (uint32_t)p; // dict pos always wins
+  }
+  size_t seg = snap_segment_size(n);
+  for (size_t s0 = 0; s0 < n; s0 += seg) {
+    size_t s1 = s0 + seg < n ? s0 + seg : n;
+    op = snap_encode_segment_dict(dict, D, in, D + (uint32_t)s0,
+                                  D + (uint32_t)s1, tab, op);
+  }
+  return (size_t)(op - out);
+}
+// decode with dictionary context; returns ulen on success, 0 on corruption
+DCW_HD size_t snappy_uncompress_dict(const uint8_t* dict, size_t D,
+                                     const uint8_t* in, size_t n,
+                                     uint8_t* out, size_t cap) {
+  uint32_t ulen;
+  int k = varint32_get(in, in + n, &ulen);
+  if (k < 0 || ulen > cap) return 0;
+  const uint8_t* ip = in + k;
+  const uint8_t* iend = in + n;
+  uint8_t* op = out;
+  uint8_t* oend = out + ulen;
+  while (ip < iend) {
+    uint8_t tag = *ip++;
+    if ((tag & 3) == 0) {
+      size_t len = (size_t)(tag >> 2) + 1;
+      if (len > 60) {
+        int nb = (int)len - 60;
+        if (ip + nb > iend) return 0;
+        len = 0;
+        for (int i = 0; i < nb; i++) len |= (size_t)ip[i] << (8 * i);
+        len += 1;
+        ip += nb;
+      }
+      if (ip + len > iend || op + len > oend) return 0;
+      for (size_t i = 0; i < len; i++) op[i] = ip[i];
+      ip += len;
+      op += len;
+    } else {
+      size_t len, offset;
+      if ((tag & 3) == 1) {
+        len = ((tag >> 2) & 7) + 4;
+        if (ip >= iend) return 0;
+        offset = ((size_t)(tag >> 5) << 8) | *ip++;
+      } else if ((tag & 3) == 2) {
+        len = (size_t)(tag >> 2) + 1;
+        if (ip + 2 > iend) return 0;
+        offset = (size_t)ip[0] | ((size_t)ip[1] << 8);
+        ip += 2;
+      } else {
+        len = (size_t)(tag >> 2) + 1;
+        if (ip + 4 > iend) return 0;
+        offset = (size_t)load32(ip);
+        ip += 4;
+      }
+      size_t produced = (size_t)(op - out);
+      if (offset == 0 || offset > produced + D || op + len > oend) return 0;
+      for (size_t i = 0; i < len; i++) {
+        op[i] = (produced + i >= offset)
+                    ? out[produced + i - offset]
+                    : dict[D - offset + produced + i];
+      }
+      op += len;
+    }
+  }
+  return op == oend ? (size_t)ulen : 0;
+}
+
 // ---------- internal keys ----------
 // Normalized 24-byte sort key for uniform-length user keys (<=16 B):
 // k0,k1 = big-endian words of the zero-padded user key, k2 = ~tag.

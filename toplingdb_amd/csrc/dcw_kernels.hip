@@ -1201,6 +1201,103 @@ __device__ __forceinline__ uint32_t snap_measure_segment_dev(
   return out + snap_lit_len(s1 - lit);
 }
 
+// Single-pass wave-parallel compress with the block staged in LDS: the
+// encoder's per-iteration dependent chain (load -> hash -> table probe ->
+// candidate compare) runs at LDS latency instead of vmem latency.  PMC
+// shows k_compress moves only ~0.4 GB/job — it is latency-bound, so the
+// chain length is the lever, not traffic (profiles/pmc_per_launch.json).
+__global__ __launch_bounds__(256) void k_compress_ldsin(
+    const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
+    const uint8_t* __restrict__ ucblob, uint8_t* __restrict__ cblob,
+    uint64_t ccap_per_block, uint32_t* __restrict__ bsize,
+    uint8_t* __restrict__ btype, uint32_t* err_flag) {
+  __shared__ uint32_t tabs[4][1u << kSnapHashBits]; // 8 KiB per wave
+  __shared__ uint8_t ins[4][5376];
+  uint32_t wid = threadIdx.x / WAVE;
+  uint32_t lane = threadIdx.x % WAVE;
+  uint32_t waves = blockDim.x / WAVE;
+  uint32_t* tab = tabs[wid];
+  for (uint32_t b = blockIdx.x * waves + wid; b < nblocks;
+       b += gridDim.x * waves) {
+    EmitBlockDesc d = bds[b];
+    if (d.unc_size > SNAP_MAX_UNC) {
+      if (lane == 0) {
+        set_err(err_flag, DE_BLOCK_PARSE);
+        bsize[b] = d.unc_size;
+        btype[b] = 0;
+      }
+      continue;
+    }
+    const uint8_t* gin = ucblob + d.uout;
+    uint32_t n = d.unc_size;
+    const uint8_t* in = gin;
+    if (n <= sizeof(ins[0])) {
+      uint8_t* li = ins[wid];
+      for (uint32_t t = lane * 16; t < n; t += WAVE * 16) {
+        uint32_t chunk = n - t < 16 ? n - t : 16;
+        if (chunk == 16) {
+          ulong2 v;
+          memcpy(&v, gin + t, 16);
+          memcpy(li + t, &v, 16);
+        } else {
+          for (uint32_t x = 0; x < chunk; x++) li[t + x] = gin[t + x];
+        }
+      }
+      in = li;
+    }
+    for (uint32_t t = lane; t < (1u << kSnapHashBits); t += WAVE)
+      tab[t] = 0xffffffffu;
+    wave_lds_sync();
+    for (uint32_t p = lane; p + 4 <= n; p += WAVE) {
+      uint32_t h = (load32(in + p) * kSnapHashMul) >> (32 - kSnapHashBits);
+      atomicMin(&tab[h], p);
+    }
+    wave_lds_sync();
+    uint32_t seg = (uint32_t)snap_segment_size(n);
+    uint32_t s0 = lane * seg;
+    uint8_t frag[SNAP_FRAG_MAX];
+    uint32_t fl = 0;
+    if (s0 < n) {
+      uint32_t s1 = s0 + seg < n ? s0 + seg : n;
+      uint8_t* e = snap_encode_segment_dev(in, s0, s1, tab, frag);
+      fl = (uint32_t)(e - frag);
+    }
+    uint32_t inc = fl;
+    for (int sh = 1; sh < WAVE; sh <<= 1) {
+      uint32_t v = __shfl_up(inc, sh);
+      if ((int)lane >= sh) inc += v;
+    }
+    uint32_t total = __shfl(inc, WAVE - 1);
+    uint32_t excl = inc - fl;
+    uint8_t* gout = cblob + (uint64_t)b * ccap_per_block;
+    uint8_t hdr[5];
+    uint32_t hl = varint32_put(hdr, n);
+    if (lane == 0)
+      for (uint32_t t = 0; t < hl; t++) gout[t] = hdr[t];
+    {
+      uint8_t* o2 = gout + hl + excl;
+      uint32_t t = 0;
+      for (; t + 4 <= fl; t += 4) {
+        uint32_t v;
+        memcpy(&v, frag + t, 4);
+        memcpy(o2 + t, &v, 4);
+      }
+      for (; t < fl; t++) o2[t] = frag[t];
+    }
+    if (lane == 0) {
+      uint32_t cn = hl + total;
+      if (cn <= (((uint64_t)896 * n) >> 10)) {
+        bsize[b] = cn;
+        btype[b] = 1;
+      } else {
+        bsize[b] = n;
+        btype[b] = 0;
+      }
+    }
+    wave_lds_sync();
+  }
+}
+
 #define SNAP_LDSIN_MAX 5376 // LDS-staged input bound (default 4 KiB blocks)
 template <int LDSIN>
 __global__ __launch_bounds__(256) void k_compress_2p(
@@ -2484,7 +2581,12 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
         getenv("DCW_COMPRESS_V") ? atoi(getenv("DCW_COMPRESS_V")) : 0;
     p->kbegin("compress", 1.6 * (double)uout);
     uint32_t cgrid = (nb + 3) / 4;
-    if (comp_v == 2)
+    if (comp_v == 1)
+      hipLaunchKernelGGL(k_compress_ldsin, dim3(cgrid < 4096 ? cgrid : 4096),
+                         dim3(256), 0, p->stream, p->d_bds, nb, p->d_ucblob,
+                         p->d_cblob, p->ccap_per_block, p->d_ebsize,
+                         p->d_ebtype, p->d_err);
+    else if (comp_v == 2)
       hipLaunchKernelGGL(k_compress_2p<1>, dim3(cgrid < 4096 ? cgrid : 4096),
                          dim3(256), 0, p->stream, p->d_bds, nb, p->d_ucblob,
                          p->d_cblob, p->ccap_per_block, p->d_ebsize,
