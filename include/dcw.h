@@ -140,6 +140,15 @@ typedef struct dcw_job_desc {
 
   const char* comparator_name; /* must be "leveldb.BytewiseComparator" */
 
+  /* Output table format, the TableFactory seam (include/rocksdb/table.h:
+   * 844-934; CompactionParams::table_factory ObjectRpcParam).
+   * 0 = BlockBasedTable format_version=5 (default).
+   * 1 = DcwZipTable "DZT1": the searchable-compressed SST of BASELINE.json
+   *     configs[3] (the reference's ToplingZipTable is absent/private,
+   *     README.md:53 — own design, parity self-pinned; layout documented
+   *     in toplingdb_amd/csrc/dcw_dzt.h). */
+  uint32_t output_table_factory;
+
   /* bench hook: when nonzero the worker keeps input SST images and the
    * device staging for this handle alive across calls (see dcw_stage_*) */
   uint64_t staged_handle;

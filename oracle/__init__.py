@@ -72,6 +72,7 @@ class JobDesc(C.Structure):
         ("level_compaction_dynamic_file_size", C.c_uint32),
         ("block_size_deviation", C.c_uint64),
         ("comparator_name", C.c_char_p),
+        ("output_table_factory", C.c_uint32),
         ("staged_handle", C.c_uint64),
     ]
 
@@ -171,6 +172,72 @@ def ikey_compare(a: bytes, b: bytes) -> int:
     return _lib.orc_ikey_compare(a, len(a), b, len(b))
 
 
+# ---- dictionary snappy ("DZT dict codec v1") ----
+def snap_dict_table(dictionary: bytes):
+    tab = (C.c_uint32 * 2048)()
+    _lib.orc_snap_dict_table(dictionary, len(dictionary), tab)
+    return tab
+
+
+def snappy_compress_dict(dictionary: bytes, data: bytes) -> bytes:
+    tab = snap_dict_table(dictionary)
+    out = C.create_string_buffer(_lib.orc_snappy_max_compressed(len(data)) + 16)
+    n = _lib.orc_snappy_compress_dict(dictionary, len(dictionary), tab, data,
+                                      len(data), out)
+    return out.raw[:n]
+
+
+def snappy_uncompress_dict(dictionary: bytes, data: bytes) -> bytes:
+    ulen = _lib.orc_snappy_uncompressed_len(data, len(data))
+    if ulen == C.c_size_t(-1).value:
+        raise ValueError("bad snappy preamble")
+    out = C.create_string_buffer(max(ulen, 1))
+    n = _lib.orc_snappy_uncompress_dict(dictionary, len(dictionary), data,
+                                        len(data), out, ulen)
+    if n != ulen:
+        raise ValueError("dict snappy corruption")
+    return out.raw[:ulen]
+
+
+# ---- DcwZipTable reader (searchability verification) ----
+def dzt_read(data: bytes):
+    """Full scan of a DZT1 file -> list of (internal_key, value)."""
+    r = _lib.orc_dzt_open(data, len(data))
+    if not r:
+        raise ValueError("DZT open failed")
+    out = []
+
+    @_DZTCB
+    def cb(_arg, k, klen, v, vlen):
+        out.append((C.string_at(k, klen), C.string_at(v, vlen)))
+        return 0
+
+    rc = _lib.orc_dzt_iterate(r, cb, None)
+    _lib.orc_dzt_close(r)
+    if rc != 0:
+        raise ValueError("DZT iterate failed")
+    return out
+
+
+def dzt_get(data: bytes, user_key: bytes):
+    """Point lookup -> (value, tag) or None (the searchable property)."""
+    r = _lib.orc_dzt_open(data, len(data))
+    if not r:
+        raise ValueError("DZT open failed")
+    buf = Buf()
+    tag = C.c_uint64()
+    rc = _lib.orc_dzt_get(r, user_key, len(user_key), C.byref(buf),
+                          C.byref(tag))
+    out = None
+    if rc == 0:
+        out = (C.string_at(buf.data, buf.size), tag.value)
+        _lib.orc_buf_free(C.byref(buf))
+    _lib.orc_dzt_close(r)
+    if rc < 0:
+        raise ValueError("DZT corruption")
+    return out
+
+
 # ---- table builder (for golden tests) ----
 class TableOpts(C.Structure):
     _fields_ = [
@@ -207,6 +274,27 @@ _lib.orc_table_builder_file_size.argtypes = [C.c_void_p]
 _lib.orc_table_builder_finish.restype = C.c_int
 _lib.orc_table_builder_finish.argtypes = [C.c_void_p, C.POINTER(Buf)]
 _lib.orc_table_builder_delete.argtypes = [C.c_void_p]
+_lib.orc_snap_dict_table.argtypes = [C.c_char_p, C.c_uint32,
+                                     C.POINTER(C.c_uint32)]
+_lib.orc_snappy_compress_dict.restype = C.c_size_t
+_lib.orc_snappy_compress_dict.argtypes = [C.c_char_p, C.c_uint32,
+                                          C.POINTER(C.c_uint32), C.c_char_p,
+                                          C.c_size_t, C.c_void_p]
+_lib.orc_snappy_uncompress_dict.restype = C.c_size_t
+_lib.orc_snappy_uncompress_dict.argtypes = [C.c_char_p, C.c_size_t, C.c_char_p,
+                                            C.c_size_t, C.c_void_p, C.c_size_t]
+_DZTCB = C.CFUNCTYPE(C.c_int, C.c_void_p, C.POINTER(C.c_uint8), C.c_uint32,
+                     C.POINTER(C.c_uint8), C.c_uint32)
+_lib.orc_dzt_open.restype = C.c_void_p
+_lib.orc_dzt_open.argtypes = [C.c_char_p, C.c_size_t]
+_lib.orc_dzt_close.argtypes = [C.c_void_p]
+_lib.orc_dzt_num_entries.restype = C.c_uint64
+_lib.orc_dzt_num_entries.argtypes = [C.c_void_p]
+_lib.orc_dzt_iterate.restype = C.c_int
+_lib.orc_dzt_iterate.argtypes = [C.c_void_p, _DZTCB, C.c_void_p]
+_lib.orc_dzt_get.restype = C.c_int
+_lib.orc_dzt_get.argtypes = [C.c_void_p, C.c_char_p, C.c_uint32,
+                             C.POINTER(Buf), C.POINTER(C.c_uint64)]
 _lib.orc_buf_free.argtypes = [C.POINTER(Buf)]
 
 
@@ -362,6 +450,7 @@ def make_job(runs, output_dir, **kw) -> JobDesc:
         "level_compaction_dynamic_file_size", 1)
     d.block_size_deviation = kw.pop("block_size_deviation", 10)
     d.comparator_name = b"leveldb.BytewiseComparator"
+    d.output_table_factory = kw.pop("output_table_factory", 0)
     for k, v in kw.items():
         setattr(d, k, v)
     return d

@@ -520,6 +520,12 @@ static void ci_next(citer* c) {
 typedef struct outputs {
   const dcw_job_desc* d;
   orc_table_builder* builder;
+  /* DcwZipTable output mode (d->output_table_factory == 1, dzt.c spec):
+   * the cut rule runs on UNCOMPRESSED bytes at value-block boundaries */
+  orc_dzt_builder* zb;
+  uint64_t dzt_unc;       /* key-record + value bytes so far (this file) */
+  uint64_t dzt_vb_ulen;   /* open value block's uncompressed bytes */
+  uint32_t dzt_vb_count;  /* open value block's entry count */
   uint64_t current_output_file_size;
   /* grandparent state (compaction_outputs.cc:121-230) */
   size_t grandparent_index;
@@ -629,30 +635,38 @@ static int out_should_stop_before(outputs* o, const uint8_t* ikey, size_t iklen)
   return 0;
 }
 
-static void out_open(outputs* o) {
-  orc_table_opts t;
-  orc_table_opts_default(&t);
+static void fill_topts(outputs* o, orc_table_opts* t) {
+  orc_table_opts_default(t);
   const dcw_job_desc* d = o->d;
-  if (d->block_size) t.block_size = d->block_size;
-  if (d->block_restart_interval) t.block_restart_interval = d->block_restart_interval;
+  if (d->block_size) t->block_size = d->block_size;
+  if (d->block_restart_interval) t->block_restart_interval = d->block_restart_interval;
   if (d->index_block_restart_interval)
-    t.index_block_restart_interval = d->index_block_restart_interval;
-  if (d->format_version) t.format_version = d->format_version;
-  t.checksum_type = d->checksum_type;
-  t.compression = d->compression;
-  if (d->block_size_deviation) t.block_size_deviation = d->block_size_deviation;
-  t.db_id = d->db_id;
-  t.db_session_id = d->db_session_id;
-  t.db_host_id = d->db_host_id;
-  t.cf_name = d->cf_name;
-  t.cf_id = d->cf_id;
+    t->index_block_restart_interval = d->index_block_restart_interval;
+  if (d->format_version) t->format_version = d->format_version;
+  t->checksum_type = d->checksum_type;
+  t->compression = d->compression;
+  if (d->block_size_deviation) t->block_size_deviation = d->block_size_deviation;
+  t->db_id = d->db_id;
+  t->db_session_id = d->db_session_id;
+  t->db_host_id = d->db_host_id;
+  t->cf_name = d->cf_name;
+  t->cf_id = d->cf_id;
+  t->orig_file_number = o->file_number;
+  t->creation_time = d->oldest_ancester_time ? d->oldest_ancester_time : d->current_time;
+  t->file_creation_time = d->current_time;
+  t->oldest_key_time = 0;
+  t->level_at_creation = d->output_level;
+}
+
+static void out_open(outputs* o) {
   o->file_number = o->next_file_number++;
-  t.orig_file_number = o->file_number;
-  t.creation_time = d->oldest_ancester_time ? d->oldest_ancester_time : d->current_time;
-  t.file_creation_time = d->current_time;
-  t.oldest_key_time = 0;
-  t.level_at_creation = d->output_level;
-  o->builder = orc_table_builder_new(&t);
+  orc_table_opts t;
+  fill_topts(o, &t);
+  if (o->d->output_table_factory != 1) o->builder = orc_table_builder_new(&t);
+  o->zb = NULL; /* DZT builder created in out_add once ukey_len is known */
+  o->dzt_unc = 0;
+  o->dzt_vb_ulen = 0;
+  o->dzt_vb_count = 0;
   o->smallest_len = o->largest_len = 0;
   o->smallest_seqno = KMAXSEQ;
   o->largest_seqno = 0;
@@ -661,9 +675,15 @@ static void out_open(outputs* o) {
 
 static int out_close(outputs* o) {
   orc_buf file = {0};
-  orc_table_builder_finish(o->builder, &file);
-  orc_table_builder_delete(o->builder);
-  o->builder = NULL;
+  if (o->zb) {
+    orc_dzt_builder_finish(o->zb, &file);
+    orc_dzt_builder_delete(o->zb);
+    o->zb = NULL;
+  } else {
+    orc_table_builder_finish(o->builder, &file);
+    orc_table_builder_delete(o->builder);
+    o->builder = NULL;
+  }
   char path[600];
   snprintf(path, sizeof(path), "%s/%06" PRIu64 ".sst", o->d->output_dir,
            o->file_number);
@@ -702,6 +722,49 @@ static int out_close(outputs* o) {
 /* AddToOutput (compaction_outputs.cc:356-420) */
 static int out_add(outputs* o, const uint8_t* key, size_t klen, const uint8_t* val,
                    size_t vlen) {
+  if (o->d->output_table_factory == 1) {
+    /* DcwZipTable output (dzt.c spec): cut when this entry would start a
+     * new value block and the file's uncompressed bytes reached target */
+    int vb_closes = o->zb && (o->dzt_vb_count >= 256 ||
+                              (o->dzt_vb_count > 0 &&
+                               o->dzt_vb_ulen + vlen > 16384));
+    if (o->zb && vb_closes && o->dzt_unc >= o->d->target_file_size) {
+      if (out_close(o) != 0) return -1;
+      vb_closes = 0;
+    }
+    if (o->zb == NULL && o->builder == NULL) out_open(o);
+    if (o->zb == NULL) o->zb = ({
+          orc_table_opts t;
+          fill_topts(o, &t);
+          orc_dzt_builder_new(&t, (uint32_t)(klen - 8));
+        });
+    if (vb_closes) {
+      o->dzt_vb_count = 0;
+      o->dzt_vb_ulen = 0;
+    }
+    o->dzt_vb_count++;
+    o->dzt_vb_ulen += vlen;
+    int krec = orc_dzt_builder_add(o->zb, key, klen, val, vlen);
+    if (krec < 0) {
+      snprintf(o->err, sizeof(o->err), "DZT: non-uniform user key length");
+      return -1;
+    }
+    o->dzt_unc += (uint64_t)krec + vlen;
+    o->current_output_file_size = o->dzt_unc;
+    uint64_t tag;
+    memcpy(&tag, key + klen - 8, 8);
+    uint64_t seq = tag >> 8;
+    if (o->file_entries == 0) {
+      memcpy(o->smallest, key, klen);
+      o->smallest_len = klen;
+    }
+    memcpy(o->largest, key, klen);
+    o->largest_len = klen;
+    if (seq < o->smallest_seqno) o->smallest_seqno = seq;
+    if (seq > o->largest_seqno) o->largest_seqno = seq;
+    o->file_entries++;
+    return 0;
+  }
   if (out_should_stop_before(o, key, klen) && o->builder != NULL) {
     if (out_close(o) != 0) return -1;
     o->grandparent_overlapped_bytes = gp_cur_overlap(o, key, klen);
@@ -786,7 +849,7 @@ int32_t orc_execute(const dcw_job_desc* d, dcw_job_result* res) {
     free(out.files);
     return 5;
   }
-  if (out.builder) {
+  if (out.builder || out.zb) {
     if (out_close(&out) != 0) {
       snprintf(res->error, sizeof(res->error), "%s", out.err);
       res->status = 4;

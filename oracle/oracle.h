@@ -115,6 +115,37 @@ typedef int (*orc_kv_cb)(void* arg, const uint8_t* ikey, size_t klen,
                          const uint8_t* val, size_t vlen);
 int orc_table_iterate(orc_table_reader* r, orc_kv_cb cb, void* arg);
 
+/* ---------- dictionary snappy ("DZT dict codec v1", dzt.c spec) ---------- */
+void orc_snap_dict_table(const uint8_t* dict, uint32_t D, uint32_t* tab /*2048*/);
+size_t orc_snappy_compress_dict(const uint8_t* dict, uint32_t D,
+                                const uint32_t* dict_tab, const uint8_t* in,
+                                size_t n, uint8_t* out);
+size_t orc_snappy_uncompress_dict(const uint8_t* dict, size_t D,
+                                  const uint8_t* in, size_t n, uint8_t* out,
+                                  size_t cap);
+
+/* ---------- DcwZipTable "DZT1" (searchable-compressed SST; dzt.c) ------- */
+typedef struct orc_dzt_builder orc_dzt_builder;
+orc_dzt_builder* orc_dzt_builder_new(const orc_table_opts* o, uint32_t ukey_len);
+/* returns the entry's key-record size (cut-rule bytes) or -1 */
+int orc_dzt_builder_add(orc_dzt_builder* b, const uint8_t* ikey, size_t klen,
+                        const uint8_t* value, size_t vlen);
+uint64_t orc_dzt_builder_num_entries(const orc_dzt_builder* b);
+int orc_dzt_builder_finish(orc_dzt_builder* b, orc_buf* out);
+void orc_dzt_builder_delete(orc_dzt_builder* b);
+
+typedef struct orc_dzt_reader orc_dzt_reader;
+orc_dzt_reader* orc_dzt_open(const uint8_t* data, size_t size);
+void orc_dzt_close(orc_dzt_reader* r);
+uint64_t orc_dzt_num_entries(const orc_dzt_reader* r);
+/* 0 = found (value+tag filled), 1 = not found, -1 = corruption */
+int orc_dzt_get(orc_dzt_reader* r, const uint8_t* uk, uint32_t uklen,
+                orc_buf* value, uint64_t* tag_out);
+int orc_dzt_iterate(orc_dzt_reader* r,
+                    int (*cb)(void* ctx, const uint8_t* ikey, uint32_t klen,
+                              const uint8_t* val, uint32_t vlen),
+                    void* ctx);
+
 /* ---------- the worker oracle ---------- */
 
 /* Execute a compaction job on the CPU with reference semantics; writes

@@ -323,9 +323,17 @@ static uint8_t* sn_emit_copy(uint8_t* op, size_t offset, size_t len) {
     }
     size_t chunk = len > 64 ? 64 : len;
     if (len - chunk > 0 && len - chunk < 4) chunk = len - 4; /* keep tail >= 4 */
-    *op++ = (uint8_t)(2 | ((chunk - 1) << 2));
-    *op++ = (uint8_t)(offset & 0xff);
-    *op++ = (uint8_t)(offset >> 8);
+    if (offset < 65536) {
+      *op++ = (uint8_t)(2 | ((chunk - 1) << 2));
+      *op++ = (uint8_t)(offset & 0xff);
+      *op++ = (uint8_t)(offset >> 8);
+    } else { /* 4-byte-offset form: dictionary matches past 64 KiB */
+      *op++ = (uint8_t)(3 | ((chunk - 1) << 2));
+      *op++ = (uint8_t)(offset & 0xff);
+      *op++ = (uint8_t)((offset >> 8) & 0xff);
+      *op++ = (uint8_t)((offset >> 16) & 0xff);
+      *op++ = (uint8_t)((offset >> 24) & 0xff);
+    }
     len -= chunk;
   }
   return op;
@@ -421,6 +429,128 @@ size_t orc_snappy_uncompress(const uint8_t* in, size_t n, uint8_t* out, size_t c
       if (offset == 0 || (size_t)(op - out) < offset || op + len > oend) return 0;
       const uint8_t* src = op - offset;
       for (size_t i = 0; i < len; i++) op[i] = src[i]; /* overlap-safe fwd copy */
+      op += len;
+    }
+  }
+  return op == oend ? ulen : 0;
+}
+
+/* ---------------- dictionary snappy ("DZT dict codec v1") ----------------
+ * Self-pinned like the base codec; full spec at dcw_common.h (virtual
+ * stream dict||block, min-position merged hash table, spec-v4 segments,
+ * 4-byte-offset copies past 64 KiB).  This is the oracle's independent
+ * restatement; parity tests bit-compare it against the HIP encoder. */
+static uint8_t sn_vbyte(const uint8_t* dict, uint32_t D, const uint8_t* in,
+                        uint32_t pos) {
+  return pos < D ? dict[pos] : in[pos - D];
+}
+static uint32_t sn_vload32(const uint8_t* dict, uint32_t D, const uint8_t* in,
+                           uint32_t pos) {
+  if (pos >= D) return rd32(in + pos - D);
+  if (pos + 4 <= D) return rd32(dict + pos);
+  uint8_t b[4];
+  for (int i = 0; i < 4; i++) b[i] = sn_vbyte(dict, D, in, pos + i);
+  uint32_t v;
+  memcpy(&v, b, 4);
+  return v;
+}
+void orc_snap_dict_table(const uint8_t* dict, uint32_t D, uint32_t* tab) {
+  enum { HBITS = 11 };
+  static const uint32_t HMUL = 0x1e35a7bdu;
+  memset(tab, 0xff, sizeof(uint32_t) << HBITS);
+  for (uint32_t p = 0; p + 4 <= D; p++) {
+    uint32_t h = (rd32(dict + p) * HMUL) >> (32 - HBITS);
+    if (tab[h] == 0xffffffffu) tab[h] = p;
+  }
+}
+size_t orc_snappy_compress_dict(const uint8_t* dict, uint32_t D,
+                                const uint32_t* dict_tab, const uint8_t* in,
+                                size_t n, uint8_t* out) {
+  enum { HBITS = 11 };
+  static const uint32_t HMUL = 0x1e35a7bdu;
+  uint8_t* op = out;
+  op += orc_varint32_put(op, (uint32_t)n);
+  if (n == 0) return (size_t)(op - out);
+  uint32_t* tab = (uint32_t*)malloc(sizeof(uint32_t) << HBITS);
+  memcpy(tab, dict_tab, sizeof(uint32_t) << HBITS);
+  for (size_t p = 0; p + 4 <= n; p++) {
+    uint32_t h = (rd32(in + p) * HMUL) >> (32 - HBITS);
+    if (tab[h] == 0xffffffffu) tab[h] = D + (uint32_t)p; /* dict pos wins */
+  }
+  size_t seg = (n + 63) / 64;
+  if (seg < 16) seg = 16;
+  for (size_t s0 = 0; s0 < n; s0 += seg) {
+    size_t s1 = s0 + seg < n ? s0 + seg : n;
+    uint32_t lit = D + (uint32_t)s0, p = D + (uint32_t)s0,
+             vs1 = D + (uint32_t)s1;
+    while (p + 4 <= vs1) {
+      uint32_t w = rd32(in + (p - D));
+      uint32_t h = (w * HMUL) >> (32 - HBITS);
+      uint32_t c = tab[h];
+      if (c != 0xffffffffu && c < p && sn_vload32(dict, D, in, c) == w) {
+        uint32_t l = 4;
+        while (p + l < vs1 && sn_vbyte(dict, D, in, c + l) == in[p + l - D]) l++;
+        op = sn_emit_literal(op, in + (lit - D), p - lit);
+        op = sn_emit_copy(op, p - c, l);
+        p += l;
+        lit = p;
+      } else {
+        p++;
+      }
+    }
+    op = sn_emit_literal(op, in + (lit - D), vs1 - lit);
+  }
+  free(tab);
+  return (size_t)(op - out);
+}
+size_t orc_snappy_uncompress_dict(const uint8_t* dict, size_t D,
+                                  const uint8_t* in, size_t n, uint8_t* out,
+                                  size_t cap) {
+  uint32_t ulen;
+  int k = orc_varint32_get(in, in + n, &ulen);
+  if (k < 0 || ulen > cap) return 0;
+  const uint8_t* ip = in + k;
+  const uint8_t* iend = in + n;
+  uint8_t* op = out;
+  uint8_t* oend = out + ulen;
+  while (ip < iend) {
+    uint8_t tag = *ip++;
+    if ((tag & 3) == 0) {
+      size_t len = (tag >> 2) + 1;
+      if (len > 60) {
+        int nb = (int)len - 60;
+        if (ip + nb > iend) return 0;
+        len = 0;
+        for (int i = 0; i < nb; i++) len |= (size_t)ip[i] << (8 * i);
+        len += 1;
+        ip += nb;
+      }
+      if (ip + len > iend || op + len > oend) return 0;
+      memcpy(op, ip, len);
+      ip += len;
+      op += len;
+    } else {
+      size_t len, offset;
+      if ((tag & 3) == 1) {
+        len = ((tag >> 2) & 7) + 4;
+        if (ip >= iend) return 0;
+        offset = ((size_t)(tag >> 5) << 8) | *ip++;
+      } else if ((tag & 3) == 2) {
+        len = (tag >> 2) + 1;
+        if (ip + 2 > iend) return 0;
+        offset = (size_t)ip[0] | ((size_t)ip[1] << 8);
+        ip += 2;
+      } else {
+        len = (tag >> 2) + 1;
+        if (ip + 4 > iend) return 0;
+        offset = rd32(ip);
+        ip += 4;
+      }
+      size_t produced = (size_t)(op - out);
+      if (offset == 0 || offset > produced + D || op + len > oend) return 0;
+      for (size_t i = 0; i < len; i++)
+        op[i] = (produced + i >= offset) ? out[produced + i - offset]
+                                         : dict[D - offset + produced + i];
       op += len;
     }
   }

@@ -65,6 +65,7 @@ class JobDesc(C.Structure):
         ("level_compaction_dynamic_file_size", C.c_uint32),
         ("block_size_deviation", C.c_uint64),
         ("comparator_name", C.c_char_p),
+        ("output_table_factory", C.c_uint32),
         ("staged_handle", C.c_uint64),
     ]
 
@@ -247,6 +248,7 @@ def make_job(runs, output_dir, **kw) -> JobDesc:
     d.level_compaction_dynamic_file_size = 1
     d.block_size_deviation = kw.pop("block_size_deviation", 10)
     d.comparator_name = b"leveldb.BytewiseComparator"
+    d.output_table_factory = kw.pop("output_table_factory", 0)
     d.staged_handle = kw.pop("staged_handle", 0)
     for k, v in kw.items():
         setattr(d, k, v)
