@@ -141,6 +141,9 @@ def main():
     ap.add_argument("--skip-cpu-baseline", action="store_true")
     ap.add_argument("--skip-hbm-resident", action="store_true",
                     help="skip the secondary staged-inputs region")
+    ap.add_argument("--table-factory", choices=["bbt", "dzt"], default="bbt",
+                    help="dzt = DcwZipTable searchable-compressed output "
+                         "(BASELINE configs[3]); bbt = BlockBasedTable")
     ap.add_argument("--job-mix", choices=["uniform", "mixed"], default="uniform",
                     help="mixed = BASELINE configs[4] job mix (alternating "
                          "L0→L1 4-way and L1→L2 8-way jobs) through the "
@@ -214,10 +217,12 @@ def main():
     lib.dcw_kernel_stats_json.restype = ctypes.c_int32
     lib.dcw_kernel_stats_json.argtypes = [ctypes.c_char_p, ctypes.c_uint32]
 
+    otf = 1 if args.table_factory == "dzt" else 0
+
     def job(slot, staged=0):
         return dcw.make_job(runs, slot_dirs[slot], compression=args.compression,
                             target_file_size=64 << 20, bottommost_level=1,
-                            staged_handle=staged)
+                            staged_handle=staged, output_table_factory=otf)
 
     from concurrent.futures import ThreadPoolExecutor
     pool = ThreadPoolExecutor(max_workers=jif)
@@ -361,8 +366,11 @@ def main():
         "dtype": "u8",
         "data": "synthetic",
         "config": {
-            "workload": "8-way L1→L2 merge, 8×~64MiB snappy SSTs, "
-                        "16B key / 100B value (BASELINE.json configs[2])",
+            "workload": ("DcwZipTable build: searchable-compressed output "
+                         "(BASELINE.json configs[3])" if otf else
+                         "8-way L1→L2 merge, 8×~64MiB snappy SSTs, "
+                         "16B key / 100B value (BASELINE.json configs[2])"),
+            "output_table_factory": "dzt" if otf else "bbt",
             "runs": args.runs,
             "entries_per_run": args.entries_per_run,
             "input_bytes_per_job": in_bytes,

@@ -1,0 +1,111 @@
+"""GPU DcwZipTable build-path parity (BASELINE configs[3]): whole output
+files must be BIT-IDENTICAL to the CPU oracle's DZT builder on the same
+job descriptors, and the outputs must be searchable (point lookups through
+the oracle reader)."""
+import os
+import random
+
+import pytest
+
+import oracle
+import toplingdb_amd as dcw
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module", autouse=True)
+def gpu():
+    dcw.init(0)
+    yield
+    dcw.shutdown()
+
+
+def gen_runs(tmp_path, n_runs, entries, seed0=1, compression=1, value_len=100):
+    runs = []
+    for r in range(n_runs):
+        p = str(tmp_path / ("dz_in_%d.sst" % r))
+        dcw.gen_sst(p, seed=seed0 + r, num_entries=entries, value_len=value_len,
+                    seq_base=1 + r * entries, compression=compression)
+        runs.append([p])
+    return runs
+
+
+def run_both(tmp_path, runs, **kw):
+    og = tmp_path / "gz"
+    oo = tmp_path / "oz"
+    og.mkdir()
+    oo.mkdir()
+    rg = dcw.execute(dcw.make_job(runs, str(og), output_table_factory=1, **kw))
+    ro = oracle.execute(oracle.make_job(runs, str(oo), output_table_factory=1,
+                                        **kw))
+    return rg, ro
+
+
+def assert_identical(rg, ro):
+    assert rg["out_entries"] == ro["out_entries"]
+    assert len(rg["files"]) == len(ro["files"])
+    for fg, fo in zip(rg["files"], ro["files"]):
+        assert fg["file_number"] == fo["file_number"]
+        da = open(fg["path"], "rb").read()
+        db = open(fo["path"], "rb").read()
+        assert da == db, "GPU DZT %s differs from oracle (%d vs %d bytes)" % (
+            fg["path"], len(da), len(db))
+        assert fg["smallest"] == fo["smallest"]
+        assert fg["largest"] == fo["largest"]
+        assert fg["smallest_seqno"] == fo["smallest_seqno"]
+        assert fg["largest_seqno"] == fo["largest_seqno"]
+
+
+def test_dzt_gpu_bit_exact_single_file(tmp_path):
+    runs = gen_runs(tmp_path, 2, 30000)
+    rg, ro = run_both(tmp_path, runs, compression=1, bottommost_level=1)
+    assert_identical(rg, ro)
+    assert len(rg["files"]) == 1
+
+
+def test_dzt_gpu_bit_exact_multi_file(tmp_path):
+    runs = gen_runs(tmp_path, 4, 60000)
+    rg, ro = run_both(tmp_path, runs, compression=1, bottommost_level=1,
+                      target_file_size=4 << 20)
+    assert_identical(rg, ro)
+    assert len(rg["files"]) > 1
+
+
+def test_dzt_gpu_uncompressed(tmp_path):
+    runs = gen_runs(tmp_path, 2, 20000)
+    rg, ro = run_both(tmp_path, runs, compression=0, bottommost_level=1)
+    assert_identical(rg, ro)
+
+
+def test_dzt_gpu_large_values(tmp_path):
+    # values near/above the value-block bound: single-value blocks + raw
+    # oversize fallback
+    runs = []
+    for r in range(2):
+        p = str(tmp_path / ("lv%d.sst" % r))
+        dcw.gen_sst(p, seed=5 + r, num_entries=3000, value_len=9000,
+                    seq_base=1 + r * 3000, compression=1)
+        runs.append([p])
+    rg, ro = run_both(tmp_path, runs, compression=1, bottommost_level=1)
+    assert_identical(rg, ro)
+
+
+def test_dzt_gpu_searchable(tmp_path):
+    runs = gen_runs(tmp_path, 2, 25000)
+    rg, _ = run_both(tmp_path, runs, compression=1, bottommost_level=1)
+    data = open(rg["files"][0]["path"], "rb").read()
+    stream = oracle.dzt_read(data)
+    assert len(stream) == rg["out_entries"]
+    rnd = random.Random(2)
+    for ik, v in rnd.sample(stream, 40):
+        got = oracle.dzt_get(data, ik[:-8])
+        assert got is not None and got[0] == v
+    assert oracle.dzt_get(data, b"\xff" * 16) is None
+
+
+def test_dzt_gpu_tombstones(tmp_path):
+    # deletes + snapshots flow through the same FSM before the DZT build
+    runs = gen_runs(tmp_path, 3, 20000, seed0=77)
+    rg, ro = run_both(tmp_path, runs, compression=1, bottommost_level=0,
+                      snapshots=[15000], earliest_write_conflict_snapshot=15000)
+    assert_identical(rg, ro)

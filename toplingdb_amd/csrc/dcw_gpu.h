@@ -111,6 +111,42 @@ class GpuJob {
   int seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* mx,
                  uint64_t* n_tombstones, std::string* err);
 
+  // ---- DcwZipTable ("DZT1") output path (BASELINE configs[3]);
+  //      format spec: oracle/dzt.c header comment ----
+  struct DztVBlock {
+    uint32_t first;   // survivor index of the block's first entry (absolute)
+    uint32_t count;
+    uint32_t ulen;    // uncompressed value bytes
+    uint64_t stage_off; // offset in the staging blob
+  };
+  struct DztKBlock {
+    uint32_t first;   // absolute survivor index
+    uint32_t count;   // <= 64
+    uint64_t koff;    // offset in the key area
+  };
+  // value bytes of sampled survivors (dict build): out[i*256..] gets the
+  // first min(vlen,256) bytes of survivor idx[i]
+  int dzt_sample(const std::vector<uint32_t>& idx, uint8_t* out,
+                 std::string* err);
+  // gather + dict-compress + checksum all value blocks of one file;
+  // voff_entry[i] = in-block offset of survivor (vb.first..) values
+  int dzt_values(const std::vector<DztVBlock>& vbs,
+                 const std::vector<uint32_t>& voff_entry, uint64_t ent_base,
+                 const uint8_t* dict, uint32_t dict_size, const TableOpts& o,
+                 std::vector<uint32_t>* csize, std::vector<uint8_t>* btype,
+                 std::vector<uint32_t>* csum, std::string* err);
+  // pack value bodies + [btype u8][csum u32] trailers into host_dst
+  int dzt_pack_values(const std::vector<DztVBlock>& vbs,
+                      const std::vector<uint64_t>& outoff,
+                      uint64_t total_bytes, uint8_t* host_dst,
+                      std::string* err);
+  // emit the key area (records per oracle/dzt.c) + per-key-block first
+  // internal keys (stride ukey_len+8) straight into host buffers
+  int dzt_keyarea(const std::vector<DztKBlock>& kbs,
+                  const std::vector<uint32_t>& voff_entry, uint64_t ent_base,
+                  uint64_t key_area_size, uint8_t* host_keyarea,
+                  uint8_t* host_first_ikeys, std::string* err);
+
   double ms_decode = 0, ms_merge = 0, ms_dedup = 0, ms_emit = 0, ms_h2d = 0,
          ms_d2h = 0;
   // set by decode(): uniform user key length of the job's entries

@@ -1836,6 +1836,14 @@ struct GpuJob::Impl {
        *d_gp_lg1 = nullptr, *d_gp_tie = nullptr, *d_gp_pos = nullptr,
        *d_gp_nback = nullptr;
   void *d_plan_next = nullptr, *d_plan_meta = nullptr, *d_plan_nr = nullptr;
+  // DZT output path slots (grow-only)
+  void *d_dzt_idx = nullptr, *d_dzt_sample = nullptr, *d_dzt_vbs = nullptr,
+       *d_dzt_voff = nullptr, *d_dzt_vstage = nullptr, *d_dzt_dict = nullptr,
+       *d_dzt_dict_tab = nullptr, *d_dzt_cblob = nullptr,
+       *d_dzt_bsize = nullptr, *d_dzt_btype = nullptr, *d_dzt_csum = nullptr,
+       *d_dzt_kbs = nullptr, *d_dzt_keyarea = nullptr, *d_dzt_kidx = nullptr,
+       *d_dzt_img = nullptr;
+  uint64_t dzt_ccap = 0;
   void* h_plan = nullptr; // pinned host landing for next+meta
   size_t h_plan_cap = 0;
   bool h_plan_pageable = false;
@@ -1954,6 +1962,10 @@ GpuJob::~GpuJob() {
   F(p->d_gp_sm0); F(p->d_gp_sm1); F(p->d_gp_lg0); F(p->d_gp_lg1);
   F(p->d_gp_tie); F(p->d_gp_pos); F(p->d_gp_nback);
   F(p->d_plan_next); F(p->d_plan_meta); F(p->d_plan_nr);
+  F(p->d_dzt_idx); F(p->d_dzt_sample); F(p->d_dzt_vbs); F(p->d_dzt_voff);
+  F(p->d_dzt_vstage); F(p->d_dzt_dict); F(p->d_dzt_dict_tab);
+  F(p->d_dzt_cblob); F(p->d_dzt_bsize); F(p->d_dzt_btype); F(p->d_dzt_csum);
+  F(p->d_dzt_kbs); F(p->d_dzt_keyarea); F(p->d_dzt_kidx); F(p->d_dzt_img);
   if (p->h_plan) {
     if (p->h_plan_pageable)
       free(p->h_plan);
@@ -2829,6 +2841,431 @@ int GpuJob::seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* m
   *mn = out[0];
   *mx = out[1];
   *n_tombstones = out[2];
+  return 0;
+}
+
+// ------------------------------------------------------------------
+// DcwZipTable ("DZT1") build kernels — the searchable-compressed SST of
+// BASELINE configs[3].  Format + parity: oracle/dzt.c header comment.
+// ------------------------------------------------------------------
+#define DZTK 64
+// value bytes of sampled survivors -> fixed 256 B stride (dict build)
+__global__ void k_dzt_sample(const uint32_t* __restrict__ idx, uint32_t n,
+                             const uint64_t* __restrict__ s_voff,
+                             const uint32_t* __restrict__ s_vlen,
+                             const uint8_t* __restrict__ ublob,
+                             uint8_t* __restrict__ out) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += gridDim.x * blockDim.x) {
+    uint32_t e = idx[i];
+    uint32_t take = s_vlen[e] < 256 ? s_vlen[e] : 256;
+    const uint8_t* src = ublob + s_voff[e];
+    uint8_t* dst = out + (uint64_t)i * 256;
+    for (uint32_t t = 0; t < take; t++) dst[t] = src[t];
+  }
+}
+
+// first-occurrence hash table over the dict (min-position; matches the
+// oracle's serial first-wins scan)
+__global__ void k_dict_tab(const uint8_t* __restrict__ dict, uint32_t D,
+                           uint32_t* __restrict__ tab) {
+  for (uint32_t p = blockIdx.x * blockDim.x + threadIdx.x; p + 4 <= D;
+       p += gridDim.x * blockDim.x) {
+    uint32_t h = (load32(dict + p) * kSnapHashMul) >> (32 - kSnapHashBits);
+    atomicMin(&tab[h], p);
+  }
+}
+
+// gather value bytes of each value block into the contiguous staging blob
+__global__ void k_dzt_gather(const GpuJob::DztVBlock* __restrict__ vbs,
+                             uint32_t nvb,
+                             const uint32_t* __restrict__ voff_entry,
+                             uint64_t ent_base,
+                             const uint64_t* __restrict__ s_voff,
+                             const uint32_t* __restrict__ s_vlen,
+                             const uint8_t* __restrict__ ublob,
+                             uint8_t* __restrict__ vstage) {
+  for (uint32_t b = blockIdx.x; b < nvb; b += gridDim.x) {
+    GpuJob::DztVBlock v = vbs[b];
+    for (uint32_t li = threadIdx.x; li < v.count; li += blockDim.x) {
+      uint64_t e = v.first + li;
+      uint8_t* dst = vstage + v.stage_off + voff_entry[e - ent_base];
+      const uint8_t* src = ublob + s_voff[e];
+      uint32_t vl = s_vlen[e];
+      uint32_t t = 0;
+      for (; t + 4 <= vl; t += 4) {
+        uint32_t w;
+        memcpy(&w, src + t, 4);
+        memcpy(dst + t, &w, 4);
+      }
+      for (; t < vl; t++) dst[t] = src[t];
+    }
+  }
+}
+
+// dict-snappy per value block, wave-parallel (spec v4 segmentation; the
+// segment encoder is the SAME DCW_HD function the oracle restates)
+__global__ __launch_bounds__(256) void k_dzt_compress(
+    const GpuJob::DztVBlock* __restrict__ vbs, uint32_t nvb,
+    const uint8_t* __restrict__ vstage, const uint8_t* __restrict__ dict,
+    uint32_t D, const uint32_t* __restrict__ dict_tab,
+    uint8_t* __restrict__ cblob, uint64_t ccap,
+    uint32_t* __restrict__ bsize, uint8_t* __restrict__ btype,
+    uint32_t* err_flag) {
+  __shared__ uint32_t tabs[4][1u << kSnapHashBits];
+  uint32_t wid = threadIdx.x / WAVE;
+  uint32_t lane = threadIdx.x % WAVE;
+  uint32_t waves = blockDim.x / WAVE;
+  uint32_t* tab = tabs[wid];
+  for (uint32_t b = blockIdx.x * waves + wid; b < nvb; b += gridDim.x * waves) {
+    GpuJob::DztVBlock v = vbs[b];
+    uint32_t n = v.ulen;
+    if (n > SNAP_MAX_UNC) { // oversize single-value block stays raw
+      if (lane == 0) {
+        bsize[b] = n;
+        btype[b] = 0;
+      }
+      continue;
+    }
+    const uint8_t* gin = vstage + v.stage_off;
+    for (uint32_t t = lane; t < (1u << kSnapHashBits); t += WAVE)
+      tab[t] = dict_tab[t];
+    wave_lds_sync();
+    for (uint32_t p = lane; p + 4 <= n; p += WAVE) {
+      uint32_t h = (load32(gin + p) * kSnapHashMul) >> (32 - kSnapHashBits);
+      atomicMin(&tab[h], D + p); // dict positions (< D) always win
+    }
+    wave_lds_sync();
+    uint32_t seg = (uint32_t)snap_segment_size(n);
+    uint32_t s0 = lane * seg;
+    uint8_t frag[SNAP_FRAG_MAX];
+    uint32_t fl = 0;
+    if (s0 < n) {
+      uint32_t s1 = s0 + seg < n ? s0 + seg : n;
+      uint8_t* e = snap_encode_segment_dict(dict, D, gin, D + s0, D + s1, tab,
+                                            frag);
+      fl = (uint32_t)(e - frag);
+      if (fl > SNAP_FRAG_MAX) set_err(err_flag, DE_SNAPPY);
+    }
+    uint32_t inc = fl;
+    for (int sh = 1; sh < WAVE; sh <<= 1) {
+      uint32_t x = __shfl_up(inc, sh);
+      if ((int)lane >= sh) inc += x;
+    }
+    uint32_t total = __shfl(inc, WAVE - 1);
+    uint32_t excl = inc - fl;
+    uint8_t* gout = cblob + (uint64_t)b * ccap;
+    uint8_t hdr[5];
+    uint32_t hl = varint32_put(hdr, n);
+    if (lane == 0)
+      for (uint32_t t = 0; t < hl; t++) gout[t] = hdr[t];
+    {
+      uint8_t* o2 = gout + hl + excl;
+      uint32_t t = 0;
+      for (; t + 4 <= fl; t += 4) {
+        uint32_t w;
+        memcpy(&w, frag + t, 4);
+        memcpy(o2 + t, &w, 4);
+      }
+      for (; t < fl; t++) o2[t] = frag[t];
+    }
+    if (lane == 0) {
+      uint32_t cn = hl + total;
+      if (cn <= (((uint64_t)896 * n) >> 10)) {
+        bsize[b] = cn;
+        btype[b] = 2; // dict-snappy
+      } else {
+        bsize[b] = n;
+        btype[b] = 0;
+      }
+    }
+    wave_lds_sync();
+  }
+}
+
+__global__ void k_dzt_checksum(const GpuJob::DztVBlock* __restrict__ vbs,
+                               uint32_t nvb, const uint8_t* __restrict__ vstage,
+                               const uint8_t* __restrict__ cblob, uint64_t ccap,
+                               const uint32_t* __restrict__ bsize,
+                               const uint8_t* __restrict__ btype,
+                               uint32_t checksum_type,
+                               const Crc32cTables* __restrict__ crc_tt,
+                               uint32_t* __restrict__ csum) {
+  for (uint32_t b = blockIdx.x * blockDim.x + threadIdx.x; b < nvb;
+       b += gridDim.x * blockDim.x) {
+    const uint8_t* body =
+        btype[b] == 2 ? cblob + (uint64_t)b * ccap : vstage + vbs[b].stage_off;
+    csum[b] = block_checksum(checksum_type, crc_tt, body, bsize[b], btype[b]);
+  }
+}
+
+__global__ void k_dzt_pack(const GpuJob::DztVBlock* __restrict__ vbs,
+                           uint32_t nvb, const uint8_t* __restrict__ vstage,
+                           const uint8_t* __restrict__ cblob, uint64_t ccap,
+                           const uint32_t* __restrict__ bsize,
+                           const uint8_t* __restrict__ btype,
+                           const uint32_t* __restrict__ csum,
+                           const uint64_t* __restrict__ outoff,
+                           uint8_t* __restrict__ out) {
+  uint32_t waves_per_wg = blockDim.x / WAVE;
+  uint32_t wave = blockIdx.x * waves_per_wg + threadIdx.x / WAVE;
+  uint32_t lane = threadIdx.x % WAVE;
+  for (uint32_t b = wave; b < nvb; b += gridDim.x * waves_per_wg) {
+    const uint8_t* body =
+        btype[b] == 2 ? cblob + (uint64_t)b * ccap : vstage + vbs[b].stage_off;
+    uint8_t* dst = out + outoff[b];
+    uint32_t n = bsize[b];
+    for (uint32_t pos = lane * 16; pos < n; pos += WAVE * 16) {
+      uint32_t chunk = n - pos < 16 ? n - pos : 16;
+      for (uint32_t t = 0; t < chunk; t++) dst[pos + t] = body[pos + t];
+    }
+    if (lane == 0) {
+      dst[n] = btype[b];
+      uint32_t cs = csum[b];
+      memcpy(dst + n + 1, &cs, 4);
+    }
+  }
+}
+
+// key area: one workgroup per key block (record layout per oracle/dzt.c);
+// thread 0 also writes the block's first internal key (key index input)
+__global__ void k_dzt_keys(const GpuJob::DztKBlock* __restrict__ kbs,
+                           uint32_t nkb, const uint64_t* __restrict__ s_k0,
+                           const uint64_t* __restrict__ s_k1,
+                           const uint64_t* __restrict__ s_tag,
+                           const uint8_t* __restrict__ s_klen,
+                           const uint8_t* __restrict__ s_sshared,
+                           const uint32_t* __restrict__ voff_entry,
+                           uint64_t ent_base,
+                           const uint32_t* __restrict__ s_vlen,
+                           uint32_t ukey_len, uint8_t* __restrict__ keyarea,
+                           uint8_t* __restrict__ first_ikeys,
+                           uint32_t ik_stride, uint32_t* err_flag) {
+  __shared__ uint32_t offs[DZTK];
+  __shared__ uint32_t esz[DZTK];
+  for (uint32_t kb = blockIdx.x; kb < nkb; kb += gridDim.x) {
+    GpuJob::DztKBlock K = kbs[kb];
+    if (K.count > DZTK) {
+      if (threadIdx.x == 0) set_err(err_flag, DE_BLOCK_PARSE);
+      __syncthreads();
+      continue;
+    }
+    for (uint32_t li = threadIdx.x; li < K.count; li += blockDim.x) {
+      uint64_t e = K.first + li;
+      uint32_t sh = li == 0 ? 0 : s_sshared[e];
+      if (sh > ukey_len) sh = ukey_len;
+      uint32_t ns = ukey_len - sh;
+      esz[li] = varint_len(sh) + varint_len(ns) + ns + 8 + 4 + 4;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      uint32_t off = 0;
+      for (uint32_t li = 0; li < K.count; li++) {
+        offs[li] = off;
+        off += esz[li];
+      }
+    }
+    __syncthreads();
+    for (uint32_t li = threadIdx.x; li < K.count; li += blockDim.x) {
+      uint64_t e = K.first + li;
+      uint32_t sh = li == 0 ? 0 : s_sshared[e];
+      if (sh > ukey_len) sh = ukey_len;
+      uint32_t ns = ukey_len - sh;
+      uint8_t key[24];
+      build_ikey(s_k0[e], s_k1[e], s_tag[e], s_klen[e], key);
+      uint8_t* p = keyarea + K.koff + offs[li];
+      p += varint32_put(p, sh);
+      p += varint32_put(p, ns);
+      for (uint32_t t = 0; t < ns; t++) p[t] = key[sh + t];
+      p += ns;
+      memcpy(p, key + ukey_len, 8); // tag
+      p += 8;
+      uint32_t vo = voff_entry[e - ent_base];
+      memcpy(p, &vo, 4);
+      p += 4;
+      uint32_t vl = s_vlen[e];
+      memcpy(p, &vl, 4);
+    }
+    if (threadIdx.x == 0) {
+      uint8_t key[24];
+      build_ikey(s_k0[K.first], s_k1[K.first], s_tag[K.first], s_klen[K.first],
+                 key);
+      memcpy(first_ikeys + (uint64_t)kb * ik_stride, key, ik_stride);
+    }
+    __syncthreads();
+  }
+}
+
+int GpuJob::dzt_sample(const std::vector<uint32_t>& idx, uint8_t* out,
+                       std::string* err) {
+  Impl* p = p_;
+  uint32_t n = (uint32_t)idx.size();
+  if (!n) return 0;
+  HIPCHK(p->ens(&p->d_dzt_idx, n * 4));
+  HIPCHK(p->ens(&p->d_dzt_sample, (uint64_t)n * 256));
+  HIPCHK(p->h2d_meta(p->d_dzt_idx, idx.data(), n * 4));
+  hipLaunchKernelGGL(k_dzt_sample, dim3(grid_for(n)), dim3(256), 0, p->stream,
+                     (const uint32_t*)p->d_dzt_idx, n, p->d_svoff, p->d_svlen,
+                     p->d_ublob, (uint8_t*)p->d_dzt_sample);
+  HIPCHK(hipMemcpyAsync(out, p->d_dzt_sample, (uint64_t)n * 256,
+                        hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  return 0;
+}
+
+int GpuJob::dzt_values(const std::vector<DztVBlock>& vbs,
+                       const std::vector<uint32_t>& voff_entry,
+                       uint64_t ent_base, const uint8_t* dict,
+                       uint32_t dict_size, const TableOpts& o,
+                       std::vector<uint32_t>* csize, std::vector<uint8_t>* btype,
+                       std::vector<uint32_t>* csum, std::string* err) {
+  Impl* p = p_;
+  uint32_t nvb = (uint32_t)vbs.size();
+  if (!nvb) {
+    csize->clear();
+    btype->clear();
+    csum->clear();
+    return 0;
+  }
+  uint64_t stage_total = vbs.back().stage_off + vbs.back().ulen;
+  HIPCHK(p->ens(&p->d_dzt_vbs, sizeof(DztVBlock) * nvb));
+  HIPCHK(p->ens(&p->d_dzt_voff, voff_entry.size() * 4));
+  HIPCHK(p->ens(&p->d_dzt_vstage, stage_total));
+  HIPCHK(p->h2d_meta(p->d_dzt_vbs, vbs.data(), sizeof(DztVBlock) * nvb));
+  HIPCHK(p->h2d_meta(p->d_dzt_voff, voff_entry.data(), voff_entry.size() * 4));
+  p->kbegin("dzt_gather", 2.0 * (double)stage_total);
+  hipLaunchKernelGGL(k_dzt_gather, dim3(grid_for(nvb * 64ull)), dim3(256), 0,
+                     p->stream, (const DztVBlock*)p->d_dzt_vbs, nvb,
+                     (const uint32_t*)p->d_dzt_voff, ent_base, p->d_svoff,
+                     p->d_svlen, p->d_ublob, (uint8_t*)p->d_dzt_vstage);
+  p->kend();
+  p->dzt_ccap = (snappy_max_compressed(SNAP_MAX_UNC) + 63) & ~(uint64_t)63;
+  HIPCHK(p->ens(&p->d_dzt_bsize, nvb * 4));
+  HIPCHK(p->ens(&p->d_dzt_btype, nvb));
+  HIPCHK(p->ens(&p->d_dzt_csum, nvb * 4));
+  if (o.compression != 0) {
+    HIPCHK(p->ens(&p->d_dzt_dict, dict_size ? dict_size : 1));
+    HIPCHK(p->ens(&p->d_dzt_dict_tab, sizeof(uint32_t) << kSnapHashBits));
+    if (dict_size)
+      HIPCHK(p->h2d_meta(p->d_dzt_dict, dict, dict_size));
+    HIPCHK(hipMemsetAsync(p->d_dzt_dict_tab, 0xff,
+                          sizeof(uint32_t) << kSnapHashBits, p->stream));
+    if (dict_size >= 4)
+      hipLaunchKernelGGL(k_dict_tab, dim3(grid_for(dict_size)), dim3(256), 0,
+                         p->stream, (const uint8_t*)p->d_dzt_dict, dict_size,
+                         (uint32_t*)p->d_dzt_dict_tab);
+    HIPCHK(p->ens(&p->d_dzt_cblob, p->dzt_ccap * nvb));
+    p->kbegin("dzt_compress", 1.6 * (double)stage_total);
+    uint32_t cgrid = (nvb + 3) / 4;
+    hipLaunchKernelGGL(k_dzt_compress, dim3(cgrid < 4096 ? cgrid : 4096),
+                       dim3(256), 0, p->stream, (const DztVBlock*)p->d_dzt_vbs,
+                       nvb, (const uint8_t*)p->d_dzt_vstage,
+                       (const uint8_t*)p->d_dzt_dict, dict_size,
+                       (const uint32_t*)p->d_dzt_dict_tab,
+                       (uint8_t*)p->d_dzt_cblob, p->dzt_ccap,
+                       (uint32_t*)p->d_dzt_bsize, (uint8_t*)p->d_dzt_btype,
+                       p->d_err);
+    p->kend();
+  } else {
+    // raw blocks: bsize = ulen, btype = 0 (host fills, tiny)
+    std::vector<uint32_t> bs(nvb);
+    std::vector<uint8_t> bt(nvb, 0);
+    for (uint32_t b = 0; b < nvb; b++) bs[b] = vbs[b].ulen;
+    HIPCHK(p->h2d_meta(p->d_dzt_bsize, bs.data(), nvb * 4));
+    HIPCHK(p->h2d_meta(p->d_dzt_btype, bt.data(), nvb));
+    HIPCHK(p->ens(&p->d_dzt_cblob, 64)); // unused
+  }
+  p->kbegin("dzt_checksum", (double)stage_total);
+  hipLaunchKernelGGL(k_dzt_checksum, dim3(grid_for(nvb)), dim3(256), 0,
+                     p->stream, (const DztVBlock*)p->d_dzt_vbs, nvb,
+                     (const uint8_t*)p->d_dzt_vstage,
+                     (const uint8_t*)p->d_dzt_cblob, p->dzt_ccap,
+                     (const uint32_t*)p->d_dzt_bsize,
+                     (const uint8_t*)p->d_dzt_btype, o.checksum_type, p->d_crc,
+                     (uint32_t*)p->d_dzt_csum);
+  p->kend();
+  csize->resize(nvb);
+  btype->resize(nvb);
+  csum->resize(nvb);
+  HIPCHK(hipMemcpyAsync(csize->data(), p->d_dzt_bsize, nvb * 4,
+                        hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipMemcpyAsync(btype->data(), p->d_dzt_btype, nvb,
+                        hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipMemcpyAsync(csum->data(), p->d_dzt_csum, nvb * 4,
+                        hipMemcpyDeviceToHost, p->stream));
+  uint32_t err_host = 0;
+  HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  p->kresolve();
+  if (err_host) {
+    if (err) *err = "DZT value compress failed, code " + std::to_string(err_host);
+    return -1;
+  }
+  return 0;
+}
+
+int GpuJob::dzt_pack_values(const std::vector<DztVBlock>& vbs,
+                            const std::vector<uint64_t>& outoff,
+                            uint64_t total_bytes, uint8_t* host_dst,
+                            std::string* err) {
+  Impl* p = p_;
+  uint32_t nvb = (uint32_t)vbs.size();
+  if (!nvb) return 0;
+  HIPCHK(p->ens(&p->d_outoff, sizeof(uint64_t) * nvb));
+  HIPCHK(p->h2d_meta(p->d_outoff, outoff.data(), sizeof(uint64_t) * nvb));
+  HIPCHK(p->ens(&p->d_dzt_img, total_bytes));
+  p->kbegin("dzt_pack", 2.0 * (double)total_bytes);
+  hipLaunchKernelGGL(k_dzt_pack, dim3(grid_for(nvb * 4ull)), dim3(256), 0,
+                     p->stream, (const DztVBlock*)p->d_dzt_vbs, nvb,
+                     (const uint8_t*)p->d_dzt_vstage,
+                     (const uint8_t*)p->d_dzt_cblob, p->dzt_ccap,
+                     (const uint32_t*)p->d_dzt_bsize,
+                     (const uint8_t*)p->d_dzt_btype,
+                     (const uint32_t*)p->d_dzt_csum,
+                     (const uint64_t*)p->d_outoff, (uint8_t*)p->d_dzt_img);
+  p->kend();
+  HIPCHK(hipMemcpyAsync(host_dst, p->d_dzt_img, total_bytes,
+                        hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  p->kresolve();
+  return 0;
+}
+
+int GpuJob::dzt_keyarea(const std::vector<DztKBlock>& kbs,
+                        const std::vector<uint32_t>& voff_entry,
+                        uint64_t ent_base, uint64_t key_area_size,
+                        uint8_t* host_keyarea, uint8_t* host_first_ikeys,
+                        std::string* err) {
+  Impl* p = p_;
+  uint32_t nkb = (uint32_t)kbs.size();
+  if (!nkb) return 0;
+  uint32_t ik = ukey_len + 8;
+  HIPCHK(p->ens(&p->d_dzt_kbs, sizeof(DztKBlock) * nkb));
+  HIPCHK(p->ens(&p->d_dzt_keyarea, key_area_size));
+  HIPCHK(p->ens(&p->d_dzt_kidx, (uint64_t)nkb * ik));
+  HIPCHK(p->h2d_meta(p->d_dzt_kbs, kbs.data(), sizeof(DztKBlock) * nkb));
+  // voff_entry already resident from dzt_values (same file)
+  (void)voff_entry;
+  p->kbegin("dzt_keys", 24.0 * (double)nkb * DZTK);
+  hipLaunchKernelGGL(k_dzt_keys, dim3(grid_for(nkb * 64ull)), dim3(64), 0,
+                     p->stream, (const DztKBlock*)p->d_dzt_kbs, nkb, p->d_sk0,
+                     p->d_sk1, p->d_stag, p->d_sklen, p->d_sshared,
+                     (const uint32_t*)p->d_dzt_voff, ent_base, p->d_svlen,
+                     ukey_len, (uint8_t*)p->d_dzt_keyarea,
+                     (uint8_t*)p->d_dzt_kidx, ik, p->d_err);
+  p->kend();
+  HIPCHK(hipMemcpyAsync(host_keyarea, p->d_dzt_keyarea, key_area_size,
+                        hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipMemcpyAsync(host_first_ikeys, p->d_dzt_kidx, (uint64_t)nkb * ik,
+                        hipMemcpyDeviceToHost, p->stream));
+  uint32_t err_host = 0;
+  HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  p->kresolve();
+  if (err_host) {
+    if (err) *err = "DZT key emit failed, code " + std::to_string(err_host);
+    return -1;
+  }
   return 0;
 }
 

@@ -361,6 +361,266 @@ int fail(dcw_job_result* res, int code, const std::string& msg) {
   return code;
 }
 
+// ---- DcwZipTable ("DZT1") output path: plan + build + write ----
+// Mirrors the oracle's DZT out_add/out_close semantics exactly
+// (oracle/dzt.c format spec; oracle/compact.c DZT branch): value-block
+// greedy grouping (<=256 values, <=16 KiB), the uncompressed-bytes file
+// cut rule, per-file dict sampling, key blocks of 64.
+enum : uint32_t {
+  kDztKB = 64,
+  kDztVbMax = 256,
+  kDztVbUlenMax = 16384,
+  kDztDictMax = 49152,
+  kDztDictSampleBytes = 256,
+};
+
+int dzt_finish(const dcw_job_desc* d, dcw_job_result* res, GpuJob& job,
+               uint64_t in_bytes, uint64_t t_start) {
+  std::string err;
+  size_t nsurv = job.num_survivors();
+  const auto& klen = job.plan_klen();
+  const auto& vlen = job.plan_vlen();
+  const auto& shared = job.plan_shared();
+  (void)klen;
+  const uint32_t U = job.ukey_len;
+  const uint32_t IK = U + 8;
+  TableOpts base = opts_from_desc(d);
+  uint64_t next_file_number = d->next_file_number;
+  std::vector<dcw_output_file> out_files;
+  uint64_t total_out_bytes = 0, total_out_entries = 0;
+
+  struct FilePlan {
+    uint64_t first = 0, count = 0;
+    std::vector<GpuJob::DztVBlock> vbs;
+    std::vector<GpuJob::DztKBlock> kbs;
+    std::vector<uint32_t> voff; // per entry (file-local order)
+    uint64_t key_area_size = 0;
+    uint64_t raw_value = 0;
+  };
+  std::vector<FilePlan> files;
+  {
+    uint64_t i = 0;
+    while (i < nsurv) {
+      FilePlan fp;
+      fp.first = i;
+      uint64_t unc = 0, vb_stage = 0, vb_first = i, vb_ulen = 0;
+      uint32_t vb_count = 0;
+      while (i < nsurv) {
+        uint32_t vl = vlen[i];
+        bool vb_closes =
+            vb_count >= kDztVbMax ||
+            (vb_count > 0 && vb_ulen + vl > kDztVbUlenMax);
+        if (vb_closes && fp.count > 0 && unc >= d->target_file_size)
+          break; // file cut before entry i
+        if (vb_closes) {
+          fp.vbs.push_back({(uint32_t)vb_first, vb_count, (uint32_t)vb_ulen,
+                            vb_stage});
+          vb_stage += vb_ulen;
+          vb_first = i;
+          vb_count = 0;
+          vb_ulen = 0;
+        }
+        uint64_t fl = i - fp.first;
+        if (fl % kDztKB == 0)
+          fp.kbs.push_back({(uint32_t)i, 0, fp.key_area_size});
+        uint32_t sh = (fl % kDztKB == 0)
+                          ? 0
+                          : (shared[i] < U ? shared[i] : U);
+        uint32_t ns = U - sh;
+        uint32_t krec = varint_len(sh) + varint_len(ns) + ns + 8 + 4 + 4;
+        fp.kbs.back().count++;
+        fp.key_area_size += krec;
+        fp.voff.push_back((uint32_t)vb_ulen);
+        vb_count++;
+        vb_ulen += vl;
+        unc += krec + vl;
+        fp.raw_value += vl;
+        fp.count++;
+        i++;
+      }
+      if (vb_count)
+        fp.vbs.push_back({(uint32_t)vb_first, vb_count, (uint32_t)vb_ulen,
+                          vb_stage});
+      if (fp.count) files.push_back(std::move(fp));
+    }
+  }
+
+  auto put32 = [](std::string& s, uint32_t v) { s.append((const char*)&v, 4); };
+  auto put64 = [](std::string& s, uint64_t v) { s.append((const char*)&v, 8); };
+
+  for (auto& fp : files) {
+    TableOpts o = base;
+    o.orig_file_number = next_file_number++;
+    // ---- dict (sampling rule identical to oracle/dzt.c) ----
+    std::vector<uint32_t> sidx;
+    uint64_t stride = fp.count / 256;
+    if (!stride) stride = 1;
+    for (uint64_t j = 0; j < fp.count; j += stride)
+      sidx.push_back((uint32_t)(fp.first + j));
+    std::vector<uint8_t> samp(sidx.size() * 256);
+    if (job.dzt_sample(sidx, samp.data(), &err) != 0)
+      return fail(res, 31, err);
+    std::string dict;
+    for (size_t si = 0; si < sidx.size() && dict.size() < kDztDictMax; si++) {
+      uint32_t take = vlen[sidx[si]] < kDztDictSampleBytes
+                          ? vlen[sidx[si]]
+                          : kDztDictSampleBytes;
+      if (dict.size() + take > kDztDictMax)
+        take = (uint32_t)(kDztDictMax - dict.size());
+      dict.append((const char*)samp.data() + si * 256, take);
+    }
+    // ---- value blocks on the GPU ----
+    std::vector<uint32_t> csize, csum;
+    std::vector<uint8_t> btype;
+    if (job.dzt_values(fp.vbs, fp.voff, fp.first, (const uint8_t*)dict.data(),
+                       (uint32_t)dict.size(), o, &csize, &btype, &csum,
+                       &err) != 0)
+      return fail(res, 32, err);
+    std::vector<uint64_t> outoff(fp.vbs.size());
+    uint64_t vtotal = 0;
+    for (size_t b = 0; b < fp.vbs.size(); b++) {
+      outoff[b] = vtotal;
+      vtotal += csize[b] + 5;
+    }
+    // ---- layout ----
+    uint64_t dict_off = fp.key_area_size;
+    uint64_t value_off = dict_off + dict.size();
+    uint64_t kindex_off = value_off + vtotal;
+    size_t nkb = fp.kbs.size();
+    uint64_t kindex_size = (uint64_t)nkb * (IK + 20);
+    uint64_t vindex_off = kindex_off + kindex_size;
+    uint64_t vindex_size = (uint64_t)fp.vbs.size() * 24;
+    uint64_t props_off = vindex_off + vindex_size;
+    // ---- GPU emits straight into the host image ----
+    RawBuf image;
+    std::vector<uint8_t> first_ikeys((uint64_t)nkb * IK);
+    image.resize_uninit(props_off); // props+footer appended after
+    if (job.dzt_keyarea(fp.kbs, fp.voff, fp.first, fp.key_area_size, image.p,
+                        first_ikeys.data(), &err) != 0)
+      return fail(res, 33, err);
+    memcpy(image.p + dict_off, dict.data(), dict.size());
+    if (job.dzt_pack_values(fp.vbs, outoff, vtotal, image.p + value_off,
+                            &err) != 0)
+      return fail(res, 34, err);
+    // ---- key index ----
+    std::string kindex;
+    kindex.reserve(kindex_size);
+    for (size_t kb = 0; kb < nkb; kb++) {
+      kindex.append((const char*)first_ikeys.data() + kb * IK, IK);
+      put64(kindex, fp.kbs[kb].koff);
+      uint64_t kend = kb + 1 < nkb ? fp.kbs[kb + 1].koff : fp.key_area_size;
+      put32(kindex, (uint32_t)(kend - fp.kbs[kb].koff));
+      put64(kindex, fp.kbs[kb].first - fp.first);
+    }
+    memcpy(image.p + kindex_off, kindex.data(), kindex.size());
+    // ---- value index ----
+    std::string vindex;
+    vindex.reserve(vindex_size);
+    for (size_t b = 0; b < fp.vbs.size(); b++) {
+      put64(vindex, outoff[b]);
+      put32(vindex, csize[b]);
+      put32(vindex, fp.vbs[b].ulen);
+      put64(vindex, fp.vbs[b].first - fp.first);
+    }
+    memcpy(image.p + vindex_off, vindex.data(), vindex.size());
+    // ---- props + footer (oracle/dzt.c layout) ----
+    std::string props;
+    put64(props, fp.count);
+    put64(props, nkb);
+    put64(props, fp.vbs.size());
+    put64(props, dict.size());
+    put64(props, fp.count * (uint64_t)IK);
+    put64(props, fp.raw_value);
+    put64(props, o.orig_file_number);
+    put64(props, o.creation_time);
+    put64(props, o.file_creation_time);
+    put32(props, U);
+    put32(props, o.checksum_type);
+    put32(props, o.cf_id);
+    int32_t lvl = o.level_at_creation;
+    props.append((const char*)&lvl, 4);
+    for (const std::string* sp : {&o.db_id, &o.db_session_id, &o.db_host_id,
+                                  &o.cf_name}) {
+      uint8_t tmp[10];
+      int m = varint32_put(tmp, (uint32_t)sp->size());
+      props.append((const char*)tmp, m);
+      props.append(*sp);
+    }
+    std::string footer;
+    uint32_t ct = o.checksum_type;
+    put64(footer, dict_off);
+    put64(footer, value_off);
+    put64(footer, kindex_off);
+    put64(footer, vindex_off);
+    put64(footer, props_off);
+    put64(footer, props.size());
+    put32(footer, block_checksum(ct, &g_crc, image.p, dict_off, 0));
+    put32(footer, block_checksum(ct, &g_crc, (const uint8_t*)dict.data(),
+                                 dict.size(), 0));
+    put32(footer, block_checksum(ct, &g_crc, (const uint8_t*)kindex.data(),
+                                 kindex.size(), 0));
+    put32(footer, block_checksum(ct, &g_crc, (const uint8_t*)vindex.data(),
+                                 vindex.size(), 0));
+    put32(footer, block_checksum(ct, &g_crc, (const uint8_t*)props.data(),
+                                 props.size(), 0));
+    put32(footer, ct);
+    put32(footer, 1);  // version
+    put32(footer, 0);  // pad
+    put64(footer, 0);  // reserved
+    put64(footer, 0x313050495A574344ull); // "DCWZIP01"
+    image.append(props.data(), props.size());
+    image.append(footer.data(), footer.size());
+    // ---- write + meta ----
+    char path[600];
+    snprintf(path, sizeof(path), "%s/%06" PRIu64 ".sst", d->output_dir,
+             o.orig_file_number);
+    FILE* f = fopen(path, "wb");
+    if (!f) return fail(res, 35, std::string("cannot write ") + path);
+    if (fwrite(image.p, 1, image.len, f) != image.len) {
+      fclose(f);
+      return fail(res, 35, std::string("short write ") + path);
+    }
+    fclose(f);
+    dcw_output_file of;
+    memset(&of, 0, sizeof(of));
+    snprintf(of.path, sizeof(of.path), "%s", path);
+    of.file_number = o.orig_file_number;
+    of.file_size = image.len;
+    memcpy(of.smallest_ikey, first_ikeys.data(), IK);
+    of.smallest_len = IK;
+    std::vector<std::pair<std::string, std::string>> lastkv;
+    if (job.gather_entries(fp.first + fp.count - 1, 1, &lastkv, &err) != 0)
+      return fail(res, 36, err);
+    memcpy(of.largest_ikey, lastkv[0].first.data(), IK);
+    of.largest_len = IK;
+    uint64_t mn, mx, tomb;
+    if (job.seq_minmax(fp.first, fp.count, &mn, &mx, &tomb, &err) != 0)
+      return fail(res, 36, err);
+    of.smallest_seqno = mn == ~0ull ? 0 : mn;
+    of.largest_seqno = mx;
+    of.num_entries = fp.count;
+    out_files.push_back(of);
+    total_out_bytes += image.len;
+    total_out_entries += fp.count;
+  }
+
+  res->num_files = (uint32_t)out_files.size();
+  res->files =
+      (dcw_output_file*)malloc(sizeof(dcw_output_file) * (out_files.size() + 1));
+  memcpy(res->files, out_files.data(),
+         sizeof(dcw_output_file) * out_files.size());
+  res->in_bytes = in_bytes;
+  res->out_bytes = total_out_bytes;
+  res->in_entries = job.num_input_entries();
+  res->out_entries = total_out_entries;
+  res->t_h2d_usec = (uint64_t)(job.ms_h2d * 1000);
+  res->t_gpu_usec = (uint64_t)((job.ms_decode + job.ms_merge + job.ms_dedup +
+                                job.ms_emit) * 1000);
+  res->work_time_usec = now_usec() - t_start;
+  res->status = 0;
+  return 0;
+}
+
 } // namespace
 } // namespace dcw
 
@@ -470,6 +730,8 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   if (d->checksum_type != 0 && d->checksum_type != 1 && d->checksum_type != 4)
     return fail(res, 14,
                 "unsupported checksum_type (kNoChecksum/kCRC32c/kXXH3 only)");
+  if (d->output_table_factory > 1)
+    return fail(res, 14, "unsupported output_table_factory");
 
   uint64_t t_start = now_usec();
   // fine-grained wall attribution (DCW_PHASE_DEBUG=1 prints at job end)
@@ -528,6 +790,16 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   wp.mark(3, now_usec());
   if (job.dedup(d, &err) != 0) return fail(res, 20, err);
   wp.mark(4, now_usec());
+
+  if (d->output_table_factory == 1) {
+    // DcwZipTable output (BASELINE configs[3]): same decode/merge/dedup
+    // pipeline, different build path
+    int rc = dzt_finish(d, res, job, in_bytes, t_start);
+    if (rc == 0 && getenv("DCW_PHASE_DEBUG"))
+      fprintf(stderr, "[phase] dzt total=%.1fms\n",
+              (now_usec() - t_start) / 1000.0);
+    return rc;
+  }
 
   size_t nsurv = job.num_survivors();
   const auto& klen = job.plan_klen();
