@@ -274,6 +274,8 @@ _lib.orc_table_builder_file_size.argtypes = [C.c_void_p]
 _lib.orc_table_builder_finish.restype = C.c_int
 _lib.orc_table_builder_finish.argtypes = [C.c_void_p, C.POINTER(Buf)]
 _lib.orc_table_builder_delete.argtypes = [C.c_void_p]
+_lib.orc_table_builder_add_tombstone.argtypes = [
+    C.c_void_p, C.c_char_p, C.c_size_t, C.c_char_p, C.c_size_t, C.c_uint64]
 _lib.orc_snap_dict_table.argtypes = [C.c_char_p, C.c_uint32,
                                      C.POINTER(C.c_uint32)]
 _lib.orc_snappy_compress_dict.restype = C.c_size_t
@@ -312,13 +314,17 @@ def default_table_opts(**kw) -> TableOpts:
     return t
 
 
-def build_sst(entries, opts: TableOpts = None) -> bytes:
-    """entries: iterable of (internal_key: bytes, value: bytes), sorted."""
+def build_sst(entries, opts: TableOpts = None, tombstones=()) -> bytes:
+    """entries: iterable of (internal_key: bytes, value: bytes), sorted.
+    tombstones: iterable of (start_ukey, end_ukey, seq) -> the
+    "rocksdb.range_del" meta block."""
     if opts is None:
         opts = default_table_opts()
     b = _lib.orc_table_builder_new(C.byref(opts))
     for k, v in entries:
         _lib.orc_table_builder_add(b, k, len(k), v, len(v))
+    for st, en, seq in tombstones:
+        _lib.orc_table_builder_add_tombstone(b, st, len(st), en, len(en), seq)
     out = Buf()
     _lib.orc_table_builder_finish(b, C.byref(out))
     data = C.string_at(out.data, out.size)
@@ -354,6 +360,32 @@ def read_sst(data: bytes):
     _lib.orc_table_close(r)
     if rc != 0:
         raise ValueError("iterate failed")
+    return out
+
+
+_TOMBCB = C.CFUNCTYPE(C.c_int, C.c_void_p, C.POINTER(C.c_uint8), C.c_size_t,
+                      C.POINTER(C.c_uint8), C.c_size_t, C.c_uint64)
+_lib.orc_table_tombstones.restype = C.c_int64
+_lib.orc_table_tombstones.argtypes = [C.c_void_p, _TOMBCB, C.c_void_p]
+
+
+def read_tombstones(data: bytes):
+    """[(start_ukey, end_ukey, seq)] from an SST's range-del meta block."""
+    err = C.create_string_buffer(160)
+    r = _lib.orc_table_open(data, len(data), err, 160)
+    if not r:
+        raise ValueError("open failed: %s" % err.value.decode())
+    out = []
+
+    @_TOMBCB
+    def cb(_arg, s_, sl, e_, el, seq):
+        out.append((C.string_at(s_, sl), C.string_at(e_, el), seq))
+        return 0
+
+    n = _lib.orc_table_tombstones(r, cb, None)
+    _lib.orc_table_close(r)
+    if n < 0:
+        raise ValueError("tombstone block corrupt")
     return out
 
 

@@ -237,10 +237,149 @@ static int km_next(kmerge* m) {
   return 0;
 }
 
+/* ---------- range-deletion aggregator (envelope subset) ----------
+ * Restates CompactionRangeDelAggregator::ShouldDelete
+ * (db/range_del_aggregator.cc:407-413) + the CompactionIterator drop site
+ * (compaction_iterator.cc:1056-1063) for the SUPPORTED ENVELOPE:
+ * bottommost output, no snapshots, no grandparents.  In that envelope all
+ * tombstones are obsolete at emission (<= earliest_snapshot = +inf at
+ * bottommost), so outputs carry no range-del block and the only effect is
+ * dropping covered point keys; with a single snapshot stripe the rule is
+ * simply: drop key iff some tombstone [start,end) contains its user key
+ * with tombstone seq > key seq.  Fragmentation
+ * (db/range_tombstone_fragmenter.cc) reduces to interval max-seq here. */
+typedef struct rd_frag {
+  uint8_t* start;
+  size_t start_len;
+  uint64_t max_seq; /* over tombstones covering [start, next_start) */
+} rd_frag;
+typedef struct rd_aggr {
+  /* raw tombstones */
+  uint8_t** s;
+  size_t* sl;
+  uint8_t** e;
+  size_t* el;
+  uint64_t* seq;
+  size_t n, cap;
+  /* fragments: boundaries sorted asc; frag i covers [b[i], b[i+1]) */
+  rd_frag* frags;
+  size_t nfrags;
+} rd_aggr;
+
+static int uk_cmp(const uint8_t* a, size_t al, const uint8_t* b, size_t bl) {
+  size_t m = al < bl ? al : bl;
+  int c = memcmp(a, b, m);
+  if (c) return c;
+  return al < bl ? -1 : (al > bl ? 1 : 0);
+}
+
+static int rd_tomb_cb(void* arg, const uint8_t* start_uk, size_t slen,
+                      const uint8_t* end_uk, size_t elen, uint64_t seq) {
+  rd_aggr* a = (rd_aggr*)arg;
+  if (a->n == a->cap) {
+    a->cap = a->cap ? a->cap * 2 : 16;
+    a->s = (uint8_t**)realloc(a->s, a->cap * sizeof(void*));
+    a->sl = (size_t*)realloc(a->sl, a->cap * sizeof(size_t));
+    a->e = (uint8_t**)realloc(a->e, a->cap * sizeof(void*));
+    a->el = (size_t*)realloc(a->el, a->cap * sizeof(size_t));
+    a->seq = (uint64_t*)realloc(a->seq, a->cap * sizeof(uint64_t));
+  }
+  a->s[a->n] = (uint8_t*)malloc(slen ? slen : 1);
+  memcpy(a->s[a->n], start_uk, slen);
+  a->sl[a->n] = slen;
+  a->e[a->n] = (uint8_t*)malloc(elen ? elen : 1);
+  memcpy(a->e[a->n], end_uk, elen);
+  a->el[a->n] = elen;
+  a->seq[a->n] = seq;
+  a->n++;
+  return 0;
+}
+
+static void rd_build_frags(rd_aggr* a) {
+  size_t nb = a->n * 2;
+  if (!nb) return;
+  /* boundary list = all starts + ends, sorted unique */
+  const uint8_t** bp = (const uint8_t**)malloc(nb * sizeof(void*));
+  size_t* bl = (size_t*)malloc(nb * sizeof(size_t));
+  for (size_t i = 0; i < a->n; i++) {
+    bp[2 * i] = a->s[i];
+    bl[2 * i] = a->sl[i];
+    bp[2 * i + 1] = a->e[i];
+    bl[2 * i + 1] = a->el[i];
+  }
+  /* insertion sort (tombstone counts are small) */
+  for (size_t i = 1; i < nb; i++)
+    for (size_t j = i; j > 0 && uk_cmp(bp[j], bl[j], bp[j - 1], bl[j - 1]) < 0;
+         j--) {
+      const uint8_t* tp = bp[j];
+      size_t tl = bl[j];
+      bp[j] = bp[j - 1];
+      bl[j] = bl[j - 1];
+      bp[j - 1] = tp;
+      bl[j - 1] = tl;
+    }
+  a->frags = (rd_frag*)calloc(nb, sizeof(rd_frag));
+  a->nfrags = 0;
+  /* every boundary starts a fragment [b_i, b_{i+1}); max_seq 0 marks a
+   * gap (incl. the terminator starting at the last boundary) */
+  for (size_t i = 0; i + 1 <= nb - 1 + 1; i++) {
+    if (i + 1 < nb && uk_cmp(bp[i], bl[i], bp[i + 1], bl[i + 1]) == 0)
+      continue;
+    uint64_t mx = 0;
+    if (i + 1 < nb)
+      for (size_t t = 0; t < a->n; t++)
+        if (uk_cmp(a->s[t], a->sl[t], bp[i], bl[i]) <= 0 &&
+            uk_cmp(bp[i], bl[i], a->e[t], a->el[t]) < 0 && a->seq[t] > mx)
+          mx = a->seq[t];
+    rd_frag* f = &a->frags[a->nfrags++];
+    f->start = (uint8_t*)malloc(bl[i] ? bl[i] : 1);
+    memcpy(f->start, bp[i], bl[i]);
+    f->start_len = bl[i];
+    f->max_seq = mx;
+    if (i + 1 == nb) break;
+  }
+  free(bp);
+  free(bl);
+}
+
+/* covered iff the fragment containing uk has max_seq > seq; keys past the
+ * last boundary are uncovered (the last interval ends at the max end
+ * boundary which is itself a fragment start with max_seq 0) */
+static int rd_covers(const rd_aggr* a, const uint8_t* uk, size_t ul,
+                     uint64_t seq) {
+  if (!a->nfrags) return 0;
+  size_t lo = 0, hi = a->nfrags;
+  while (lo < hi) { /* first frag with start > uk */
+    size_t mid = (lo + hi) / 2;
+    if (uk_cmp(a->frags[mid].start, a->frags[mid].start_len, uk, ul) <= 0)
+      lo = mid + 1;
+    else
+      hi = mid;
+  }
+  if (lo == 0) return 0; /* before the first tombstone */
+  return a->frags[lo - 1].max_seq > seq;
+}
+
+static void rd_free(rd_aggr* a) {
+  for (size_t i = 0; i < a->n; i++) {
+    free(a->s[i]);
+    free(a->e[i]);
+  }
+  free(a->s);
+  free(a->sl);
+  free(a->e);
+  free(a->el);
+  free(a->seq);
+  for (size_t f = 0; f < a->nfrags; f++) free(a->frags[f].start);
+  free(a->frags);
+  memset(a, 0, sizeof(*a));
+}
+
 /* ---------- compaction iterator (compaction_iterator.cc FSM) ---------- */
 typedef struct citer {
   kmerge* in;
   const dcw_job_desc* d;
+  const rd_aggr* rd; /* range-deletion aggregator (NULL when no tombstones) */
   /* config */
   int visible_at_tip;
   uint64_t earliest_snapshot;
@@ -472,6 +611,10 @@ static void ci_next_from_input(citer* c) {
           c->at_next = 1;
         }
       }
+    } else if (c->rd && rd_covers(c->rd, c->cur_ukey, c->cur_ukey_len, seq)) {
+      /* dropped by a range tombstone (compaction_iterator.cc:1056-1063:
+       * the ShouldDelete check sits in the final keep branch only) */
+      if (km_next(in) != 0) goto in_err;
     } else {
       c->valid = 1; /* kNewUserKey */
     }
@@ -816,10 +959,51 @@ int32_t orc_execute(const dcw_job_desc* d, dcw_job_result* res) {
       }
     }
 
+  /* range-deletion tombstones from every input's meta block
+   * (MakeInputIterator adds all inputs' tombstones up front,
+   * db/version_set.cc:7269-7352) */
+  rd_aggr rd;
+  memset(&rd, 0, sizeof(rd));
+  for (uint32_t r = 0; r < d->num_runs; r++)
+    for (uint32_t f = 0; f < d->runs[r].num_files; f++) {
+      FILE* fp = fopen(d->runs[r].files[f], "rb");
+      if (!fp) continue;
+      fseek(fp, 0, SEEK_END);
+      long fsz = ftell(fp);
+      fseek(fp, 0, SEEK_SET);
+      uint8_t* img = (uint8_t*)malloc((size_t)fsz);
+      size_t got = fread(img, 1, (size_t)fsz, fp);
+      fclose(fp);
+      if (got == (size_t)fsz) {
+        char terr[128];
+        orc_table_reader* tr = orc_table_open(img, got, terr, sizeof(terr));
+        if (tr) {
+          (void)orc_table_tombstones(tr, rd_tomb_cb, &rd);
+          orc_table_close(tr);
+        }
+      }
+      free(img);
+    }
+  if (rd.n) {
+    /* supported envelope (see rd_aggr header comment); outside it the
+     * worker refuses and the DB runs the job locally */
+    if (!d->bottommost_level || d->num_snapshots || d->num_grandparents) {
+      snprintf(res->error, sizeof(res->error),
+               "range deletions outside supported envelope "
+               "(bottommost, no snapshots, no grandparents)");
+      rd_free(&rd);
+      km_free(&km);
+      res->status = 6;
+      return 6;
+    }
+    rd_build_frags(&rd);
+  }
+
   citer ci;
   memset(&ci, 0, sizeof(ci));
   ci.in = &km;
   ci.d = d;
+  ci.rd = rd.n ? &rd : NULL;
   ci.visible_at_tip = d->num_snapshots == 0;
   ci.earliest_snapshot = d->num_snapshots ? d->snapshots[0] : KMAXSEQ;
 
@@ -835,6 +1019,7 @@ int32_t orc_execute(const dcw_job_desc* d, dcw_job_result* res) {
       snprintf(res->error, sizeof(res->error), "%s", out.err);
       res->status = 4;
       km_free(&km);
+      rd_free(&rd);
       free(out.files);
       return 4;
     }
@@ -846,6 +1031,7 @@ int32_t orc_execute(const dcw_job_desc* d, dcw_job_result* res) {
     res->status = 5;
     if (out.builder) orc_table_builder_delete(out.builder);
     km_free(&km);
+    rd_free(&rd);
     free(out.files);
     return 5;
   }
@@ -854,11 +1040,13 @@ int32_t orc_execute(const dcw_job_desc* d, dcw_job_result* res) {
       snprintf(res->error, sizeof(res->error), "%s", out.err);
       res->status = 4;
       km_free(&km);
+      rd_free(&rd);
       free(out.files);
       return 4;
     }
   }
   km_free(&km);
+  rd_free(&rd);
   res->status = 0;
   res->files = out.files;
   res->num_files = out.num_files;

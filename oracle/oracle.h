@@ -115,6 +115,18 @@ typedef int (*orc_kv_cb)(void* arg, const uint8_t* ikey, size_t klen,
                          const uint8_t* val, size_t vlen);
 int orc_table_iterate(orc_table_reader* r, orc_kv_cb cb, void* arg);
 
+/* Range-deletion tombstones from the "rocksdb.range_del" meta block;
+ * returns count (0 when absent), -1 on corruption. */
+typedef int (*orc_tomb_cb)(void* arg, const uint8_t* start_uk, size_t slen,
+                           const uint8_t* end_uk, size_t elen, uint64_t seq);
+int64_t orc_table_tombstones(orc_table_reader* r, orc_tomb_cb cb, void* arg);
+
+/* builder: add a kTypeRangeDeletion entry ([start_uk, end_uk) at seq) */
+void orc_table_builder_add_tombstone(orc_table_builder* b,
+                                     const uint8_t* start_uk, size_t slen,
+                                     const uint8_t* end_uk, size_t elen,
+                                     uint64_t seq);
+
 /* ---------- dictionary snappy ("DZT dict codec v1", dzt.c spec) ---------- */
 void orc_snap_dict_table(const uint8_t* dict, uint32_t D, uint32_t* tab /*2048*/);
 size_t orc_snappy_compress_dict(const uint8_t* dict, uint32_t D,

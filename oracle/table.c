@@ -185,6 +185,8 @@ struct orc_table_builder {
   orc_buf last_key;
   uint64_t pending_off, pending_size;
   int has_pending; /* data block flushed, index entry not yet added */
+  oblk rangedel; /* (ikey(start,seq,0xF), end_ukey) entries, interval 1 */
+  int has_rangedel;
   /* props */
   uint64_t num_entries, num_deletions, num_range_deletions, num_merge_operands;
   uint64_t raw_key_size, raw_value_size, num_data_blocks, data_size, index_size;
@@ -214,7 +216,28 @@ orc_table_builder* orc_table_builder_new(const orc_table_opts* o) {
   oblk_init(&b->data, o->block_restart_interval, 0);
   oblk_init(&b->index_seq, o->index_block_restart_interval, 1);
   oblk_init(&b->index_user, o->index_block_restart_interval, 1);
+  oblk_init(&b->rangedel, 1, 0); /* range_del_block(1), builder.cc rep ctor */
   return b;
+}
+
+/* kTypeRangeDeletion entry -> the range-del meta block
+ * (block_based_table_builder.cc:1045-1067: counted in num_entries,
+ * num_deletions, num_range_deletions and the raw sizes) */
+void orc_table_builder_add_tombstone(orc_table_builder* b,
+                                     const uint8_t* start_uk, size_t slen,
+                                     const uint8_t* end_uk, size_t elen,
+                                     uint64_t seq) {
+  uint8_t ikey[512];
+  memcpy(ikey, start_uk, slen);
+  uint64_t tag = (seq << 8) | DCW_TYPE_RANGE_DELETION;
+  memcpy(ikey + slen, &tag, 8);
+  oblk_add(&b->rangedel, ikey, slen + 8, end_uk, elen, NULL, 0);
+  b->has_rangedel = 1;
+  b->num_entries++;
+  b->num_deletions++;
+  b->num_range_deletions++;
+  b->raw_key_size += slen + 8;
+  b->raw_value_size += elen;
 }
 
 /* write block contents (+5B trailer) at current offset; handle out.
@@ -409,6 +432,17 @@ int orc_table_builder_finish(orc_table_builder* b, orc_buf* out) {
   tb_write_block(b, idx->data, idx->size, 1 /*enable_index_compression*/, &index_off,
                  &index_size_comp);
 
+  /* 4. range deletion tombstone meta block (uncompressed,
+   * block_based_table_builder.cc:1735-1743; written between index and
+   * properties per Finish()'s layout comment :1950-1955) */
+  uint64_t rd_off = 0, rd_size = 0;
+  if (b->has_rangedel) {
+    orc_buf rdo = {0};
+    oblk_finish(&b->rangedel, &rdo);
+    tb_write_block(b, rdo.data, rdo.size, 0, &rd_off, &rd_size);
+    orc_buf_free(&rdo);
+  }
+
   /* 5. properties block */
   prop props[40];
   int np = 0;
@@ -487,6 +521,12 @@ int orc_table_builder_finish(orc_table_builder* b, orc_buf* out) {
     int hn = orc_varint64_put(hv, props_off);
     hn += orc_varint64_put(hv + hn, props_size);
     oblk_add(&mi, (const uint8_t*)"rocksdb.properties", 18, hv, (size_t)hn, NULL, 0);
+    if (b->has_rangedel) { /* sorted after "rocksdb.properties" */
+      uint8_t rv[24];
+      int rn = orc_varint64_put(rv, rd_off);
+      rn += orc_varint64_put(rv + rn, rd_size);
+      oblk_add(&mi, (const uint8_t*)"rocksdb.range_del", 17, rv, (size_t)rn, NULL, 0);
+    }
     orc_buf mio = {0};
     oblk_finish(&mi, &mio);
     uint64_t mi_off, mi_size;
@@ -522,6 +562,7 @@ void orc_table_builder_delete(orc_table_builder* b) {
   oblk_free(&b->data);
   oblk_free(&b->index_seq);
   oblk_free(&b->index_user);
+  oblk_free(&b->rangedel);
   orc_buf_free(&b->last_key);
   orc_buf_free(&b->scratch);
   orc_buf_free(&b->scratch2);
@@ -539,6 +580,7 @@ struct orc_table_reader {
   uint64_t* blk_size;
   size_t nblocks;
   uint64_t num_entries;
+  uint64_t mi_off, mi_sz; /* metaindex handle (meta blocks, e.g. range_del) */
 };
 
 static int rd_handle(const uint8_t* p, const uint8_t* lim, uint64_t* off, uint64_t* sz,
@@ -614,6 +656,8 @@ orc_table_reader* orc_table_open(const uint8_t* data, size_t size, char* err,
     free(r);
     return NULL;
   }
+  r->mi_off = mi_off;
+  r->mi_sz = mi_sz;
   size_t idx_n;
   uint8_t* idx = read_block(r, idx_off, idx_sz, &idx_n, err, errcap);
   if (!idx) { free(r); return NULL; }
@@ -735,4 +779,99 @@ int orc_table_iterate(orc_table_reader* r, orc_kv_cb cb, void* arg) {
     free(blk);
   }
   return 0;
+}
+
+/* ---- meta blocks: range deletions ---- */
+/* generic iterator over a decoded block's prefix-compressed entries */
+static int blk_iter_kvs(const uint8_t* blk, size_t n,
+                        int (*cb)(void*, const uint8_t*, size_t,
+                                  const uint8_t*, size_t),
+                        void* arg) {
+  if (n < 8) return -1;
+  uint32_t footer_u32;
+  memcpy(&footer_u32, blk + n - 4, 4);
+  uint32_t nres = footer_u32 & 0x7fffffff;
+  size_t end = n - 4 - 4 * (size_t)nres;
+  const uint8_t* p = blk;
+  const uint8_t* lim = blk + end;
+  uint8_t key[512];
+  size_t klen = 0;
+  while (p < lim) {
+    uint32_t sh, ns, vl;
+    int a = orc_varint32_get(p, lim, &sh);
+    if (a < 0) return -1;
+    p += a;
+    a = orc_varint32_get(p, lim, &ns);
+    if (a < 0) return -1;
+    p += a;
+    a = orc_varint32_get(p, lim, &vl);
+    if (a < 0) return -1;
+    p += a;
+    if (sh > klen || sh + ns > sizeof(key) || p + ns + vl > lim) return -1;
+    memcpy(key + sh, p, ns);
+    klen = sh + ns;
+    p += ns;
+    int rc = cb(arg, key, klen, p, vl);
+    if (rc) return rc;
+    p += vl;
+  }
+  return 0;
+}
+
+struct mi_scan {
+  const char* want;
+  size_t want_len;
+  uint64_t off, sz;
+  int found;
+};
+static int mi_cb(void* arg, const uint8_t* k, size_t kl, const uint8_t* v,
+                 size_t vl) {
+  struct mi_scan* m = (struct mi_scan*)arg;
+  if (kl == m->want_len && memcmp(k, m->want, kl) == 0) {
+    int adv = 0;
+    if (rd_handle(v, v + vl, &m->off, &m->sz, &adv) == 0) m->found = 1;
+    return 1;
+  }
+  return 0;
+}
+
+struct rd_scan {
+  orc_tomb_cb cb;
+  void* arg;
+  int64_t count;
+};
+static int rd_cb(void* arg, const uint8_t* k, size_t kl, const uint8_t* v,
+                 size_t vl) {
+  struct rd_scan* sscan = (struct rd_scan*)arg;
+  if (kl < 9) return -1;
+  uint64_t tag;
+  memcpy(&tag, k + kl - 8, 8);
+  if ((uint8_t)tag != DCW_TYPE_RANGE_DELETION) return -1;
+  sscan->count++;
+  if (sscan->cb)
+    return sscan->cb(sscan->arg, k, kl - 8, v, vl, tag >> 8);
+  return 0;
+}
+
+/* Iterate the "rocksdb.range_del" meta block's tombstones
+ * (block_based_table_builder.cc:1735-1743 write side;
+ * block_based_table_reader.cc:984 read side).  Returns the tombstone
+ * count, 0 when the block is absent, -1 on corruption. */
+int64_t orc_table_tombstones(orc_table_reader* r, orc_tomb_cb cb, void* arg) {
+  char err[128];
+  size_t mi_n;
+  uint8_t* mi = read_block(r, r->mi_off, r->mi_sz, &mi_n, err, sizeof(err));
+  if (!mi) return -1;
+  struct mi_scan m = {"rocksdb.range_del", 17, 0, 0, 0};
+  int rc = blk_iter_kvs(mi, mi_n, mi_cb, &m);
+  free(mi);
+  if (rc < 0) return -1;
+  if (!m.found) return 0;
+  size_t rd_n;
+  uint8_t* rd = read_block(r, m.off, m.sz, &rd_n, err, sizeof(err));
+  if (!rd) return -1;
+  struct rd_scan sscan = {cb, arg, 0};
+  rc = blk_iter_kvs(rd, rd_n, rd_cb, &sscan);
+  free(rd);
+  return rc < 0 ? -1 : sscan.count;
 }

@@ -111,6 +111,19 @@ class GpuJob {
   int seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* mx,
                  uint64_t* n_tombstones, std::string* err);
 
+  // ---- range deletions (envelope subset; oracle/compact.c rd_aggr) ----
+  // Fragment list sorted by (normkey, len): frag i covers user keys in
+  // [start_i, start_{i+1}); max_seq 0 marks a gap/terminator.  Set BEFORE
+  // dedup(); the FSM drops covered point keys in its final keep branch.
+  struct RdFrag {
+    uint64_t k0, k1;   // zero-padded big-endian start-key words
+    uint32_t len;      // true byte length (zero-pad tie-break)
+    uint64_t max_seq;
+  };
+  void set_range_del_frags(const std::vector<RdFrag>& frags) {
+    rd_frags_ = frags;
+  }
+
   // ---- DcwZipTable ("DZT1") output path (BASELINE configs[3]);
   //      format spec: oracle/dzt.c header comment ----
   struct DztVBlock {
@@ -161,6 +174,7 @@ class GpuJob {
   std::vector<uint8_t> h_shared_, h_klen_;
   std::vector<uint32_t> h_vlen_;
   std::vector<uint32_t> run_blocks_;
+  std::vector<RdFrag> rd_frags_;
 };
 
 // device management

@@ -184,6 +184,77 @@ ParsedSst parse_sst(const uint8_t* data, size_t size) {
     prev_sz = sz;
     have_prev = true;
   }
+  // metaindex -> "rocksdb.range_del" meta block -> tombstones
+  // (block_based_table_builder.cc:1735-1743; the block is uncompressed)
+  if (mi_off + mi_sz + kTrailerSize <= size && mi_sz >= 8) {
+    const uint8_t* mb = data + mi_off;
+    if (r.checksum_type == 0 ||
+        load32(mb + mi_sz + 1) ==
+            block_checksum(r.checksum_type, &g_crc, mb, mi_sz, mb[mi_sz])) {
+      uint32_t mnr = load32(mb + mi_sz - 4) & 0x7fffffff;
+      size_t mend = mi_sz - 4 - 4 * (size_t)mnr;
+      const uint8_t* mp = mb;
+      const uint8_t* ml = mb + mend;
+      std::string mkey;
+      uint64_t rd_off = 0, rd_sz = 0;
+      bool rd_found = false;
+      while (mp < ml) {
+        uint32_t sh, ns, vl;
+        if ((a = varint32_get(mp, ml, &sh)) < 0) break;
+        mp += a;
+        if ((a = varint32_get(mp, ml, &ns)) < 0) break;
+        mp += a;
+        if ((a = varint32_get(mp, ml, &vl)) < 0) break;
+        mp += a;
+        if (sh > mkey.size()) break;
+        mkey.resize(sh);
+        mkey.append((const char*)mp, ns);
+        mp += ns;
+        if (mkey == "rocksdb.range_del") {
+          const uint8_t* vp = mp;
+          int b1 = varint64_get(vp, mp + vl, &rd_off);
+          if (b1 > 0 && varint64_get(vp + b1, mp + vl, &rd_sz) > 0)
+            rd_found = true;
+        }
+        mp += vl;
+      }
+      if (rd_found && rd_off + rd_sz + kTrailerSize <= size && rd_sz >= 8) {
+        const uint8_t* rb = data + rd_off;
+        if (r.checksum_type != 0 &&
+            load32(rb + rd_sz + 1) !=
+                block_checksum(r.checksum_type, &g_crc, rb, rd_sz, rb[rd_sz])) {
+          r.error = "range_del block checksum mismatch";
+          return r;
+        }
+        uint32_t rnr = load32(rb + rd_sz - 4) & 0x7fffffff;
+        size_t rend = rd_sz - 4 - 4 * (size_t)rnr;
+        const uint8_t* rp = rb;
+        const uint8_t* rl = rb + rend;
+        std::string rkey;
+        while (rp < rl) {
+          uint32_t sh, ns, vl;
+          if ((a = varint32_get(rp, rl, &sh)) < 0) break;
+          rp += a;
+          if ((a = varint32_get(rp, rl, &ns)) < 0) break;
+          rp += a;
+          if ((a = varint32_get(rp, rl, &vl)) < 0) break;
+          rp += a;
+          if (sh > rkey.size() || rp + ns + vl > rl) break;
+          rkey.resize(sh);
+          rkey.append((const char*)rp, ns);
+          rp += ns;
+          if (rkey.size() < 9) break;
+          uint64_t tag = load64((const uint8_t*)rkey.data() + rkey.size() - 8);
+          SstTombstone t;
+          t.start.assign(rkey, 0, rkey.size() - 8);
+          t.end.assign((const char*)rp, vl);
+          t.seq = tag >> 8;
+          r.tombstones.push_back(std::move(t));
+          rp += vl;
+        }
+      }
+    }
+  }
   r.ok = true;
   return r;
 }

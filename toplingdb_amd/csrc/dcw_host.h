@@ -79,9 +79,14 @@ class BlockBuilder {
 struct SstIndexEntry {
   uint64_t off, size; // BlockHandle of a data block
 };
+struct SstTombstone {
+  std::string start, end; // user keys, [start, end)
+  uint64_t seq = 0;
+};
 struct ParsedSst {
   uint32_t checksum_type = 4;
   std::vector<SstIndexEntry> data_blocks;
+  std::vector<SstTombstone> tombstones; // "rocksdb.range_del" meta block
   std::string error;
   bool ok = false;
 };

@@ -518,6 +518,130 @@ __global__ void k_decode_entries(
   }
 }
 
+// ---------------- general key shapes (mixed/long user keys) ----------------
+// The fast path packs the whole <=16 B uniform user key into the normkey
+// (k0,k1).  GENERAL mode (mixed lengths or 16 < ukey <= DCW_GKEY_MAX)
+// keeps (k0,k1) as the zero-padded 16-byte PREFIX and stores the FULL
+// user key in a fixed-stride side table `kext` indexed by the entry's
+// payload index w; comparisons gather the tail only on prefix ties
+// (db/dbformat.h:1057-1096 arbitrary-length bytewise contract).
+#define DCW_GKEY_MAX 48
+#define DCW_GKEY_STRIDE 48
+
+// full user-key bytewise compare from the side table (klen = ikey length)
+__device__ __forceinline__ int gkey_cmp(const uint8_t* __restrict__ kext,
+                                        const uint8_t* __restrict__ klen,
+                                        uint64_t wa, uint64_t wb) {
+  const uint8_t* A = kext + wa * DCW_GKEY_STRIDE;
+  const uint8_t* B = kext + wb * DCW_GKEY_STRIDE;
+  uint32_t la = klen[wa] - 8, lb = klen[wb] - 8;
+  uint32_t m = la < lb ? la : lb;
+  uint32_t t = 0;
+  for (; t + 8 <= m; t += 8) {
+    uint64_t x, y;
+    memcpy(&x, A + t, 8);
+    memcpy(&y, B + t, 8);
+    if (x != y) {
+      x = __builtin_bswap64(x);
+      y = __builtin_bswap64(y);
+      return x < y ? -1 : 1;
+    }
+  }
+  for (; t < m; t++)
+    if (A[t] != B[t]) return A[t] < B[t] ? -1 : 1;
+  if (la != lb) return la < lb ? -1 : 1;
+  return 0;
+}
+
+// general entry order: prefix words, then full-key gather, then ~tag
+__device__ __forceinline__ bool ent_le_g(const ulong4& a, const ulong4& b,
+                                         const uint8_t* __restrict__ kext,
+                                         const uint8_t* __restrict__ klen) {
+  if (a.x != b.x) return a.x < b.x;
+  if (a.y != b.y) return a.y < b.y;
+  if (kext) {
+    int c = gkey_cmp(kext, klen, a.w, b.w);
+    if (c) return c < 0;
+  }
+  return a.z <= b.z;
+}
+
+// general decode: running key kept in a local buffer; full ukey bytes go
+// to the side table, prefix normkey to the entry array
+__global__ void k_decode_entries_g(
+    const uint8_t* __restrict__ ublob, const uint64_t* __restrict__ uoff,
+    const uint32_t* __restrict__ usize, const uint32_t* __restrict__ nrestarts,
+    const uint32_t* __restrict__ iv_block, const uint32_t* __restrict__ iv_local,
+    const uint32_t* __restrict__ iv_base, uint32_t nintervals,
+    ulong4* __restrict__ ents, uint64_t* __restrict__ voff,
+    uint32_t* __restrict__ vlen_out, uint8_t* __restrict__ klen_out,
+    uint8_t* __restrict__ kext, uint32_t* err_flag) {
+  for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nintervals;
+       i += gridDim.x * blockDim.x) {
+    uint32_t b = iv_block[i];
+    const uint8_t* ublk = ublob + uoff[b];
+    uint32_t beg, end;
+    if (interval_bounds(ublk, usize[b], nrestarts[b], iv_local[i], &beg, &end) != 0)
+      return;
+    ByteWin W;
+    W.base = ublk;
+    if (beg < end) W.fill(beg);
+    uint32_t pos = beg;
+    uint8_t cur[DCW_GKEY_MAX + 8];
+    uint32_t klen = 0;
+    uint32_t out = iv_base[i];
+    while (pos < end) {
+      uint32_t shared, non_shared, vl;
+      if (!win_varint32(W, pos, end, &shared)) break;
+      if (!win_varint32(W, pos, end, &non_shared)) break;
+      if (!win_varint32(W, pos, end, &vl)) break;
+      if (shared > klen || shared + non_shared > DCW_GKEY_MAX + 8 ||
+          (uint64_t)pos + non_shared + vl > end) {
+        set_err(err_flag,
+                shared + non_shared > DCW_GKEY_MAX + 8 ? DE_UKEY_LEN
+                                                       : DE_BLOCK_PARSE);
+        return;
+      }
+      for (uint32_t t = 0; t < non_shared; t++) cur[shared + t] = W.at(pos + t);
+      klen = shared + non_shared;
+      pos += non_shared;
+      if (klen < 9) {
+        set_err(err_flag, DE_BLOCK_PARSE);
+        return;
+      }
+      uint32_t ulen = klen - 8;
+      uint64_t tag;
+      memcpy(&tag, cur + ulen, 8);
+      uint8_t vt = (uint8_t)tag;
+      if (!(vt == kTypeValue || vt == kTypeDeletion || vt == kTypeSingleDeletion)) {
+        set_err(err_flag, DE_TYPE);
+        return;
+      }
+      // prefix normkey: first 16 bytes zero-padded, big-endian words
+      uint64_t u0 = 0, u1 = 0;
+      uint32_t p0 = ulen < 8 ? ulen : 8;
+      memcpy(&u0, cur, 8);
+      if (ulen < 8) u0 &= p0 ? ((p0 == 8) ? ~0ull : ((1ull << (8 * p0)) - 1)) : 0;
+      if (ulen > 8) {
+        uint32_t p1 = ulen - 8 < 8 ? ulen - 8 : 8;
+        memcpy(&u1, cur + 8, 8);
+        u1 &= (p1 == 8) ? ~0ull : ((1ull << (8 * p1)) - 1);
+      }
+      ents[out] = make_ulong4(__builtin_bswap64(u0), __builtin_bswap64(u1),
+                              ~tag, out);
+      // full ukey into the side table (zero-pad the slot tail)
+      uint8_t* slot = kext + (uint64_t)out * DCW_GKEY_STRIDE;
+      for (uint32_t t = 0; t < ulen; t++) slot[t] = cur[t];
+      for (uint32_t t = ulen; t < DCW_GKEY_STRIDE; t++) slot[t] = 0;
+      voff[out] = uoff[b] + pos;
+      vlen_out[out] = vl;
+      klen_out[out] = (uint8_t)klen;
+      pos += vl;
+      out++;
+    }
+  }
+}
+
 // ------------------------------------------------------------------
 // merge
 // ------------------------------------------------------------------
@@ -648,7 +772,35 @@ struct FsmParams {
   const uint64_t* lb_lg_k1;
   const uint32_t* lb_level_beg; // num_levels+1
   uint32_t num_levels;
+  // range-deletion fragments (GpuJob::RdFrag SoA; envelope subset)
+  const uint64_t* rd_k0;
+  const uint64_t* rd_k1;
+  const uint32_t* rd_len;
+  const uint64_t* rd_seq;
+  uint32_t num_rd;
+  uint32_t rd_ukey_len; // the job's uniform user-key length
 };
+
+// covered iff the fragment containing the user key has max_seq > seq
+// (range_del_aggregator.cc:407-413 single-stripe rule; zero-padded
+// normkey compare with byte-length tie-break == raw bytewise order)
+__device__ __forceinline__ bool fsm_rd_covers(const FsmParams& P, uint64_t k0,
+                                              uint64_t k1, uint64_t seq) {
+  uint32_t lo = 0, hi = P.num_rd;
+  while (lo < hi) { // last frag with (k0,k1,len) <= (key, ukey_len)
+    uint32_t mid = (lo + hi) / 2;
+    bool le = P.rd_k0[mid] < k0 ||
+              (P.rd_k0[mid] == k0 &&
+               (P.rd_k1[mid] < k1 ||
+                (P.rd_k1[mid] == k1 && P.rd_len[mid] <= P.rd_ukey_len)));
+    if (le)
+      lo = mid + 1;
+    else
+      hi = mid;
+  }
+  if (lo == 0) return false;
+  return P.rd_seq[lo - 1] > seq;
+}
 
 __device__ uint64_t fsm_find_earliest(const FsmParams& P, uint64_t seq,
                                       uint64_t* prev) {
@@ -796,6 +948,9 @@ __global__ void k_group_fsm(const ulong4* __restrict__ e, uint64_t n,
         } else {
           consumed = g1 - pos; // drop delete and all covered versions
         }
+      } else if (P.num_rd && fsm_rd_covers(P, k0, k1, seq)) {
+        // dropped by a range tombstone (compaction_iterator.cc:1056-1063:
+        // the ShouldDelete site is the final keep branch only)
       } else {
         out_this = true; // kNewUserKey / plain keep
       }
@@ -1844,6 +1999,8 @@ struct GpuJob::Impl {
        *d_dzt_kbs = nullptr, *d_dzt_keyarea = nullptr, *d_dzt_kidx = nullptr,
        *d_dzt_img = nullptr;
   uint64_t dzt_ccap = 0;
+  void *d_rd_k0 = nullptr, *d_rd_k1 = nullptr, *d_rd_len = nullptr,
+       *d_rd_seq = nullptr;
   void* h_plan = nullptr; // pinned host landing for next+meta
   size_t h_plan_cap = 0;
   bool h_plan_pageable = false;
@@ -1934,6 +2091,7 @@ void GpuJob::reset() {
   h_klen_.clear();
   h_vlen_.clear();
   run_blocks_.clear();
+  rd_frags_.clear();
   ms_decode = ms_merge = ms_dedup = ms_emit = ms_h2d = ms_d2h = 0;
 }
 GpuJob::~GpuJob() {
@@ -1966,6 +2124,7 @@ GpuJob::~GpuJob() {
   F(p->d_dzt_vstage); F(p->d_dzt_dict); F(p->d_dzt_dict_tab);
   F(p->d_dzt_cblob); F(p->d_dzt_bsize); F(p->d_dzt_btype); F(p->d_dzt_csum);
   F(p->d_dzt_kbs); F(p->d_dzt_keyarea); F(p->d_dzt_kidx); F(p->d_dzt_img);
+  F(p->d_rd_k0); F(p->d_rd_k1); F(p->d_rd_len); F(p->d_rd_seq);
   if (p->h_plan) {
     if (p->h_plan_pageable)
       free(p->h_plan);
@@ -2385,6 +2544,31 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   P.lb_lg_k0 = p->d_lb_lg0;
   P.lb_lg_k1 = p->d_lb_lg1;
   P.lb_level_beg = p->d_lb_beg;
+  // range-deletion fragments
+  P.num_rd = (uint32_t)rd_frags_.size();
+  P.rd_ukey_len = ukey_len;
+  if (P.num_rd) {
+    std::vector<uint64_t> fk0(P.num_rd), fk1(P.num_rd), fsq(P.num_rd);
+    std::vector<uint32_t> fln(P.num_rd);
+    for (uint32_t i = 0; i < P.num_rd; i++) {
+      fk0[i] = rd_frags_[i].k0;
+      fk1[i] = rd_frags_[i].k1;
+      fln[i] = rd_frags_[i].len;
+      fsq[i] = rd_frags_[i].max_seq;
+    }
+    HIPCHK(p->ens(&p->d_rd_k0, P.num_rd * 8));
+    HIPCHK(p->ens(&p->d_rd_k1, P.num_rd * 8));
+    HIPCHK(p->ens(&p->d_rd_len, P.num_rd * 4));
+    HIPCHK(p->ens(&p->d_rd_seq, P.num_rd * 8));
+    HIPCHK(p->h2d_meta(p->d_rd_k0, fk0.data(), P.num_rd * 8));
+    HIPCHK(p->h2d_meta(p->d_rd_k1, fk1.data(), P.num_rd * 8));
+    HIPCHK(p->h2d_meta(p->d_rd_len, fln.data(), P.num_rd * 4));
+    HIPCHK(p->h2d_meta(p->d_rd_seq, fsq.data(), P.num_rd * 8));
+    P.rd_k0 = (const uint64_t*)p->d_rd_k0;
+    P.rd_k1 = (const uint64_t*)p->d_rd_k1;
+    P.rd_len = (const uint32_t*)p->d_rd_len;
+    P.rd_seq = (const uint64_t*)p->d_rd_seq;
+  }
 
   HIPCHK(p->ens((void**)&p->d_survive, n));
   HIPCHK(p->ens((void**)&p->d_newtag, sizeof(uint64_t) * n));

@@ -18,6 +18,7 @@
 #include <csignal>
 #include <unistd.h>
 #include <cstring>
+#include <algorithm>
 #include <condition_variable>
 #include <deque>
 #include <future>
@@ -119,6 +120,7 @@ uint64_t g_next_stage_handle = 1;
 struct StagedJob {
   StagedInput dev;
   uint64_t in_bytes = 0;
+  std::vector<SstTombstone> tombstones;
 };
 std::unordered_map<uint64_t, StagedJob*> g_staged;
 std::mutex g_cancel_mu;
@@ -248,6 +250,7 @@ struct LoadedInputs {
   RawBuf blob; // all files concatenated (pinned)
   GpuInputs gi;
   uint64_t in_bytes = 0;
+  std::vector<SstTombstone> tombstones; // all inputs' range deletions
 };
 
 int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err) {
@@ -338,6 +341,7 @@ int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err) {
       }
       for (auto& h : ps.data_blocks)
         L->gi.blocks.push_back({base + h.off, (uint32_t)h.size});
+      for (auto& t : ps.tombstones) L->tombstones.push_back(t);
       L->in_bytes += sz;
     }
     L->gi.run_block_begin.push_back((uint32_t)L->gi.blocks.size());
@@ -702,6 +706,7 @@ uint64_t dcw_stage_inputs(const dcw_job_desc* d) {
     return 0;
   }
   sj->in_bytes = L.in_bytes;
+  sj->tombstones = std::move(L.tombstones);
   uint64_t h = g_next_stage_handle++;
   g_staged[h] = sj;
   return h;
@@ -762,16 +767,55 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   uint64_t in_bytes = 0;
   LoadedInputs L;
   StagedJob* staged = nullptr;
+  const std::vector<SstTombstone>* tombstones = nullptr;
   if (d->staged_handle) {
     std::lock_guard<std::mutex> lk(g_mu);
     auto it = g_staged.find(d->staged_handle);
     if (it == g_staged.end()) return fail(res, 15, "bad staged handle");
     staged = it->second;
     in_bytes = staged->in_bytes;
+    tombstones = &staged->tombstones;
   } else {
     if (load_inputs(d, &L, &err) != 0) return fail(res, 16, err);
     wp.mark(0, now_usec());
     in_bytes = L.in_bytes;
+    tombstones = &L.tombstones;
+  }
+  // range deletions: supported envelope only (mirrors oracle/compact.c
+  // rd_aggr — bottommost, no snapshots, no grandparents; all tombstones
+  // are then obsolete at emission and only drop covered point keys)
+  std::vector<GpuJob::RdFrag> rd_frags;
+  if (tombstones && !tombstones->empty()) {
+    if (!d->bottommost_level || d->num_snapshots || d->num_grandparents)
+      return fail(res, 29,
+                  "range deletions outside supported envelope "
+                  "(bottommost, no snapshots, no grandparents)");
+    for (auto& t : *tombstones)
+      if (t.start.size() > 16 || t.end.size() > 16)
+        return fail(res, 29, "range tombstone bound exceeds 16 B envelope");
+    // fragment boundaries (same construction as the oracle, incl. the
+    // zero-seq gap/terminator fragments)
+    std::vector<std::string> bounds;
+    for (auto& t : *tombstones) {
+      bounds.push_back(t.start);
+      bounds.push_back(t.end);
+    }
+    std::sort(bounds.begin(), bounds.end());
+    bounds.erase(std::unique(bounds.begin(), bounds.end()), bounds.end());
+    for (size_t i = 0; i < bounds.size(); i++) {
+      uint64_t mx = 0;
+      if (i + 1 < bounds.size())
+        for (auto& t : *tombstones)
+          if (t.start <= bounds[i] && bounds[i] < t.end && t.seq > mx)
+            mx = t.seq;
+      GpuJob::RdFrag fr;
+      uint64_t c;
+      make_normkey((const uint8_t*)bounds[i].data(),
+                   (uint32_t)bounds[i].size(), 0, &fr.k0, &fr.k1, &c);
+      fr.len = (uint32_t)bounds[i].size();
+      fr.max_seq = mx;
+      rd_frags.push_back(fr);
+    }
   }
   res->t_read_usec = now_usec() - t0;
 
@@ -788,6 +832,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   wp.mark(2, now_usec());
   if (job.merge(&err) != 0) return fail(res, 19, err);
   wp.mark(3, now_usec());
+  job.set_range_del_frags(rd_frags);
   if (job.dedup(d, &err) != 0) return fail(res, 20, err);
   wp.mark(4, now_usec());
 
