@@ -646,14 +646,15 @@ __global__ void k_decode_entries_g(
 // merge
 // ------------------------------------------------------------------
 __device__ uint64_t merge_diag(const ulong4* A, uint64_t nA, const ulong4* B,
-                               uint64_t nB, uint64_t d) {
+                               uint64_t nB, uint64_t d,
+                               const uint8_t* kext, const uint8_t* klen) {
   // largest a in [max(0,d-nB), min(d,nA)] s.t. A[0..a) all <= B from b=d-a on
   uint64_t lo = d > nB ? d - nB : 0;
   uint64_t hi = d < nA ? d : nA;
   while (lo < hi) {
     uint64_t mid = (lo + hi + 1) / 2;
     // A[mid-1] vs B[d-mid]: A goes first on ties
-    if (ent_le(A[mid - 1], B[d - mid]))
+    if (ent_le_g(A[mid - 1], B[d - mid], kext, klen))
       lo = mid;
     else
       hi = mid - 1;
@@ -696,7 +697,8 @@ __device__ __forceinline__ bool ent_le_s(const ulong4& a, const ulong4& b) {
 }
 __global__ __launch_bounds__(MT_TPB) void k_merge_tiled(
     const ulong4* __restrict__ A, uint64_t nA, const ulong4* __restrict__ B,
-    uint64_t nB, ulong4* __restrict__ out) {
+    uint64_t nB, ulong4* __restrict__ out, const uint8_t* __restrict__ kext,
+    const uint8_t* __restrict__ klen) {
   __shared__ ulong4 S[MT_TILE]; // A-segment then B-segment
   __shared__ uint32_t seg[2];   // nA_t, a0 broadcast... [0]=nA_t
   __shared__ uint64_t base[2];  // a0, b0
@@ -706,8 +708,8 @@ __global__ __launch_bounds__(MT_TPB) void k_merge_tiled(
     uint64_t d0 = t * MT_TILE;
     uint64_t d1 = d0 + MT_TILE < total ? d0 + MT_TILE : total;
     if (threadIdx.x == 0) {
-      uint64_t a0 = merge_diag(A, nA, B, nB, d0);
-      uint64_t a1 = merge_diag(A, nA, B, nB, d1);
+      uint64_t a0 = merge_diag(A, nA, B, nB, d0, kext, klen);
+      uint64_t a1 = merge_diag(A, nA, B, nB, d1, kext, klen);
       base[0] = a0;
       base[1] = d0 - a0;
       seg[0] = (uint32_t)(a1 - a0);
@@ -729,7 +731,7 @@ __global__ __launch_bounds__(MT_TPB) void k_merge_tiled(
       uint32_t hi = td0 < na ? td0 : na;
       while (lo < hi) {
         uint32_t mid = (lo + hi + 1) / 2;
-        if (ent_le_s(S[mid - 1], S[na + (td0 - mid)]))
+        if (ent_le_g(S[mid - 1], S[na + (td0 - mid)], kext, klen))
           lo = mid;
         else
           hi = mid - 1;
@@ -737,7 +739,7 @@ __global__ __launch_bounds__(MT_TPB) void k_merge_tiled(
       uint32_t a = lo, b = td0 - lo;
       ulong4 regs[MT_ITEMS];
       for (uint32_t d = td0; d < td1; d++) {
-        bool takeA = a < na && (b >= nbt || ent_le_s(S[a], S[na + b]));
+        bool takeA = a < na && (b >= nbt || ent_le_g(S[a], S[na + b], kext, klen));
         regs[d - td0] = takeA ? S[a++] : S[na + b++];
       }
       for (uint32_t d = td0; d < td1; d++) out[d0 + d] = regs[d - td0];
@@ -750,10 +752,14 @@ __global__ __launch_bounds__(MT_TPB) void k_merge_tiled(
 // dedup / visibility
 // ------------------------------------------------------------------
 __global__ void k_mark_heads(const ulong4* __restrict__ e, uint64_t n,
+                             const uint8_t* __restrict__ kext,
+                             const uint8_t* __restrict__ klen,
                              uint8_t* __restrict__ head) {
   for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (uint64_t)gridDim.x * blockDim.x) {
-    head[i] = (i == 0) || e[i].x != e[i - 1].x || e[i].y != e[i - 1].y;
+    bool h = (i == 0) || e[i].x != e[i - 1].x || e[i].y != e[i - 1].y;
+    if (!h && kext) h = gkey_cmp(kext, klen, e[i].w, e[i - 1].w) != 0;
+    head[i] = h;
   }
 }
 
@@ -1012,7 +1018,8 @@ __global__ void k_gather_survivors(
     const uint32_t* __restrict__ vlen, const uint8_t* __restrict__ klen,
     uint64_t* __restrict__ s_k0, uint64_t* __restrict__ s_k1,
     uint64_t* __restrict__ s_tag, uint64_t* __restrict__ s_voff,
-    uint32_t* __restrict__ s_vlen, uint8_t* __restrict__ s_klen) {
+    uint32_t* __restrict__ s_vlen, uint8_t* __restrict__ s_klen,
+    uint32_t* __restrict__ s_w) {
   for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (uint64_t)gridDim.x * blockDim.x) {
     if (!survive[i]) continue;
@@ -1024,6 +1031,7 @@ __global__ void k_gather_survivors(
     s_voff[o] = voff[ref];
     s_vlen[o] = clearv[i] ? 0 : vlen[ref];
     s_klen[o] = klen[ref];
+    s_w[o] = (uint32_t)ref;
   }
 }
 
@@ -1033,6 +1041,21 @@ __device__ __forceinline__ void build_ikey(uint64_t k0, uint64_t k1, uint64_t ta
   memcpy(out, &b0, 8);
   memcpy(out + 8, &b1, 8);
   memcpy(out + (klen - 8), &tag, 8);
+}
+// internal key of survivor i: fast path from the normkey, general path
+// from the full-key side table (out must hold DCW_GKEY_MAX+8)
+__device__ __forceinline__ void load_ikey(
+    uint64_t k0, uint64_t k1, uint64_t tag, uint32_t klen,
+    const uint8_t* __restrict__ kext, const uint32_t* __restrict__ s_w,
+    uint64_t i, uint8_t* out) {
+  if (!kext) {
+    build_ikey(k0, k1, tag, klen, out);
+    return;
+  }
+  const uint8_t* slot = kext + (uint64_t)s_w[i] * 48u;
+  uint32_t ulen = klen - 8;
+  for (uint32_t t = 0; t < ulen; t++) out[t] = slot[t];
+  memcpy(out + ulen, &tag, 8);
 }
 
 __global__ void k_shared_prefix(const uint64_t* __restrict__ s_k0,
