@@ -102,10 +102,10 @@ class GpuJob {
   int gp_positions(const dcw_job_desc* d, std::vector<uint32_t>* pos,
                    std::vector<uint8_t>* nback, std::string* err);
   // min/max sequence + tombstone count over survivor range (post zeroing)
-  // per-block records prefetched by emit_blocks: 96 B each =
-  // [klen_first u8 | first ikey (<=32) | klen_last u8 at 32 | last ikey |
-  //  minseq u64 at 64 | maxseq u64 at 72 | n_tombstones u64 at 80]
-  static constexpr size_t kBlkStatStride = 96;
+  // per-block records prefetched by emit_blocks: 160 B each =
+  // [klen_first u8 | first ikey (<=56) at 1 | klen_last u8 at 64 | last
+  //  ikey at 65 | minseq u64 at 128 | maxseq at 136 | n_tombstones at 144]
+  static constexpr size_t kBlkStatStride = 160;
   const uint8_t* chunk_stats() const;
   void block_stats(size_t b, uint64_t* mn, uint64_t* mx, uint64_t* tomb);
   int seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* mx,
@@ -163,7 +163,11 @@ class GpuJob {
   double ms_decode = 0, ms_merge = 0, ms_dedup = 0, ms_emit = 0, ms_h2d = 0,
          ms_d2h = 0;
   // set by decode(): uniform user key length of the job's entries
+  // (0 in general-key mode)
   uint32_t ukey_len = 0;
+  // decode() retried in general mode: mixed/long user keys (<= 48 B) via
+  // the prefix normkey + full-key side table
+  bool general_keys = false;
 
   struct Impl; // implementation detail (dcw_kernels.hip)
 

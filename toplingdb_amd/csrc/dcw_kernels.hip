@@ -672,7 +672,7 @@ __global__ void k_merge_pair(const ulong4* __restrict__ A, uint64_t nA,
        c += (uint64_t)gridDim.x * blockDim.x) {
     uint64_t d0 = c * ITEMS;
     uint64_t d1 = d0 + ITEMS < total ? d0 + ITEMS : total;
-    uint64_t a = merge_diag(A, nA, B, nB, d0);
+    uint64_t a = merge_diag(A, nA, B, nB, d0, nullptr, nullptr);
     uint64_t b = d0 - a;
     for (uint64_t d = d0; d < d1; d++) {
       bool takeA = a < nA && (b >= nB || ent_le(A[a], B[b]));
@@ -1062,6 +1062,8 @@ __global__ void k_shared_prefix(const uint64_t* __restrict__ s_k0,
                                 const uint64_t* __restrict__ s_k1,
                                 const uint64_t* __restrict__ s_tag,
                                 const uint8_t* __restrict__ s_klen, uint64_t n,
+                                const uint8_t* __restrict__ kext,
+                                const uint32_t* __restrict__ s_w,
                                 uint8_t* __restrict__ s_shared) {
   for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (uint64_t)gridDim.x * blockDim.x) {
@@ -1069,10 +1071,10 @@ __global__ void k_shared_prefix(const uint64_t* __restrict__ s_k0,
       s_shared[0] = 0;
       continue;
     }
-    uint8_t a[24], b[24];
+    uint8_t a[DCW_GKEY_MAX + 8], b[DCW_GKEY_MAX + 8];
     uint32_t ka = s_klen[i - 1], kb = s_klen[i];
-    build_ikey(s_k0[i - 1], s_k1[i - 1], s_tag[i - 1], ka, a);
-    build_ikey(s_k0[i], s_k1[i], s_tag[i], kb, b);
+    load_ikey(s_k0[i - 1], s_k1[i - 1], s_tag[i - 1], ka, kext, s_w, i - 1, a);
+    load_ikey(s_k0[i], s_k1[i], s_tag[i], kb, kext, s_w, i, b);
     uint32_t m = ka < kb ? ka : kb;
     uint32_t s = 0;
     while (s < m && a[s] == b[s]) s++;
@@ -1106,7 +1108,8 @@ __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
                        const uint8_t* __restrict__ s_shared,
                        const uint8_t* __restrict__ ublob,
                        uint8_t* __restrict__ ucblob, uint32_t restart_interval,
-                       uint32_t* err_flag) {
+                       const uint8_t* __restrict__ kext,
+                       const uint32_t* __restrict__ s_w, uint32_t* err_flag) {
   __shared__ uint32_t offs[EMIT_MAX_ENTRIES];
   __shared__ uint32_t esz[EMIT_MAX_ENTRIES];
   for (uint32_t b = blockIdx.x; b < nblocks; b += gridDim.x) {
@@ -1140,8 +1143,8 @@ __global__ void k_emit(const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
     for (uint32_t li = threadIdx.x; li < d.count; li += blockDim.x) {
       uint32_t i = d.first + li;
       uint32_t klen = s_klen[i];
-      uint8_t key[24];
-      build_ikey(s_k0[i], s_k1[i], s_tag[i], klen, key);
+      uint8_t key[DCW_GKEY_MAX + 8];
+      load_ikey(s_k0[i], s_k1[i], s_tag[i], klen, kext, s_w, i, key);
       uint32_t shared = (li % restart_interval == 0) ? 0 : s_shared[i];
       uint32_t non_shared = klen - shared;
       uint32_t vl = s_vlen[i];
@@ -1604,22 +1607,27 @@ __global__ void k_pack(const EmitBlockDesc* __restrict__ bds, uint32_t b0,
 // per-block metadata record, fetched once per chunk together with the
 // compressed sizes: boundary keys (for index separators) + seq-range and
 // tombstone count (for file properties; replaces a per-file reduce pass)
-#define BLKSTAT_STRIDE 96
+#define BLKSTAT_STRIDE 160
+// record layout (BLKSTAT_STRIDE=160): [klen_f u8 | first ikey <= 56 B at 1]
+// [klen_l u8 at 64 | last ikey at 65] [minseq u64 at 128 | maxseq at 136 |
+// n_tombstones at 144]
 __global__ void k_block_stats(const EmitBlockDesc* __restrict__ bds, uint32_t b0,
                               uint32_t b1, const uint64_t* __restrict__ s_k0,
                               const uint64_t* __restrict__ s_k1,
                               const uint64_t* __restrict__ s_tag,
                               const uint8_t* __restrict__ s_klen,
+                              const uint8_t* __restrict__ kext,
+                              const uint32_t* __restrict__ s_w,
                               uint8_t* __restrict__ out) {
   for (uint32_t b = b0 + blockIdx.x * blockDim.x + threadIdx.x; b < b1;
        b += gridDim.x * blockDim.x) {
     uint8_t* o = out + (uint64_t)(b - b0) * BLKSTAT_STRIDE;
     uint32_t f = bds[b].first, l = bds[b].first + bds[b].count - 1;
-    memset(o, 0, 64);
+    memset(o, 0, 128);
     o[0] = s_klen[f];
-    build_ikey(s_k0[f], s_k1[f], s_tag[f], s_klen[f], o + 1);
-    o[32] = s_klen[l];
-    build_ikey(s_k0[l], s_k1[l], s_tag[l], s_klen[l], o + 33);
+    load_ikey(s_k0[f], s_k1[f], s_tag[f], s_klen[f], kext, s_w, f, o + 1);
+    o[64] = s_klen[l];
+    load_ikey(s_k0[l], s_k1[l], s_tag[l], s_klen[l], kext, s_w, l, o + 65);
     uint64_t mn = ~0ull, mx = 0, tomb = 0;
     for (uint32_t i = f; i <= l; i++) {
       uint64_t tag = s_tag[i];
@@ -1629,9 +1637,9 @@ __global__ void k_block_stats(const EmitBlockDesc* __restrict__ bds, uint32_t b0
       uint8_t vt = (uint8_t)tag;
       if (vt == kTypeDeletion || vt == kTypeSingleDeletion) tomb++;
     }
-    memcpy(o + 64, &mn, 8);
-    memcpy(o + 72, &mx, 8);
-    memcpy(o + 80, &tomb, 8);
+    memcpy(o + 128, &mn, 8);
+    memcpy(o + 136, &mx, 8);
+    memcpy(o + 144, &tomb, 8);
   }
 }
 
@@ -1644,6 +1652,8 @@ __global__ void k_gather_range(const uint64_t* __restrict__ s_k0,
                                const uint8_t* __restrict__ s_klen,
                                const uint8_t* __restrict__ ublob, uint64_t first,
                                uint32_t count, const uint64_t* __restrict__ recoff,
+                               const uint8_t* __restrict__ kext,
+                               const uint32_t* __restrict__ s_w,
                                uint8_t* __restrict__ out) {
   for (uint32_t i = blockIdx.x * blockDim.x + threadIdx.x; i < count;
        i += gridDim.x * blockDim.x) {
@@ -1651,7 +1661,7 @@ __global__ void k_gather_range(const uint64_t* __restrict__ s_k0,
     uint8_t* p = out + recoff[i];
     uint32_t klen = s_klen[idx];
     p[0] = (uint8_t)klen;
-    build_ikey(s_k0[idx], s_k1[idx], s_tag[idx], klen, p + 1);
+    load_ikey(s_k0[idx], s_k1[idx], s_tag[idx], klen, kext, s_w, idx, p + 1);
     uint32_t vl = s_vlen[idx];
     memcpy(p + 1 + klen, &vl, 4);
     const uint8_t* src = ublob + s_voff[idx];
@@ -2024,6 +2034,8 @@ struct GpuJob::Impl {
   uint64_t dzt_ccap = 0;
   void *d_rd_k0 = nullptr, *d_rd_k1 = nullptr, *d_rd_len = nullptr,
        *d_rd_seq = nullptr;
+  void *d_kext = nullptr; // general-key side table (48 B/entry full ukeys)
+  void *d_sw = nullptr;   // survivor -> original payload index (u32)
   void* h_plan = nullptr; // pinned host landing for next+meta
   size_t h_plan_cap = 0;
   bool h_plan_pageable = false;
@@ -2110,6 +2122,7 @@ void GpuJob::reset() {
   n_entries_ = 0;
   n_surv_ = 0;
   ukey_len = 0;
+  general_keys = false;
   h_shared_.clear();
   h_klen_.clear();
   h_vlen_.clear();
@@ -2148,6 +2161,7 @@ GpuJob::~GpuJob() {
   F(p->d_dzt_cblob); F(p->d_dzt_bsize); F(p->d_dzt_btype); F(p->d_dzt_csum);
   F(p->d_dzt_kbs); F(p->d_dzt_keyarea); F(p->d_dzt_kidx); F(p->d_dzt_img);
   F(p->d_rd_k0); F(p->d_rd_k1); F(p->d_rd_len); F(p->d_rd_seq);
+  F(p->d_kext); F(p->d_sw);
   if (p->h_plan) {
     if (p->h_plan_pageable)
       free(p->h_plan);
@@ -2410,11 +2424,44 @@ int GpuJob::decode(std::string* err) {
   p->kresolve();
   (void)hipEventDestroy(t0);
   (void)hipEventDestroy(t1);
+  if (err_host == DE_UKEY_LEN) {
+    // GENERAL-KEY retry: mixed lengths / 16 < ukey <= DCW_GKEY_MAX via the
+    // prefix normkey + full-key side table (arbitrary-length bytewise
+    // contract, db/dbformat.h:1057-1096)
+    hipEvent_t g0, g1;
+    (void)hipEventCreate(&g0);
+    (void)hipEventCreate(&g1);
+    (void)hipEventRecord(g0, p->stream);
+    HIPCHK(hipMemsetAsync(p->d_err, 0, 8, p->stream));
+    HIPCHK(p->ens(&p->d_kext, (uint64_t)total_entries * 48u + 64));
+    p->kbegin("decode_entries_g",
+              (double)p->ublob_size + 93.0 * total_entries);
+    hipLaunchKernelGGL(k_decode_entries_g, dim3(grid_for(niv)), dim3(256), 0,
+                       p->stream, p->d_ublob, p->d_uoff, p->d_usize,
+                       p->d_nrestarts, p->d_iv_block, p->d_iv_local,
+                       p->d_iv_base, (uint32_t)niv, p->d_ent[0], p->d_voff,
+                       p->d_vlen, p->d_klen, (uint8_t*)p->d_kext, p->d_err);
+    p->kend();
+    HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost,
+                          p->stream));
+    (void)hipEventRecord(g1, p->stream);
+    HIPCHK(hipStreamSynchronize(p->stream));
+    ms_decode += ms_between(g0, g1);
+    p->kresolve();
+    (void)hipEventDestroy(g0);
+    (void)hipEventDestroy(g1);
+    if (err_host) {
+      if (err) *err = "general-key decode failed, code " +
+                      std::to_string(err_host) +
+                      (err_host == DE_UKEY_LEN ? " (user key > 48 B)" : "");
+      return -1;
+    }
+    general_keys = true;
+    ukey_len = 0;
+    return 0;
+  }
   if (err_host) {
-    if (err) *err = "entry decode failed, code " + std::to_string(err_host) +
-                    (err_host == DE_UKEY_LEN
-                         ? " (non-uniform or >16B user keys: outside the round-1 worker envelope)"
-                         : "");
+    if (err) *err = "entry decode failed, code " + std::to_string(err_host);
     return -1;
   }
   ukey_len = uklen;
@@ -2441,7 +2488,9 @@ int GpuJob::merge(std::string* err) {
       hipLaunchKernelGGL(k_merge_tiled,
                          dim3((uint32_t)(ntiles < 8192 ? ntiles : 8192)),
                          dim3(MT_TPB), 0, p->stream, p->d_ent[cur] + a0, nA,
-                         p->d_ent[cur] + a1, nB, p->d_ent[cur ^ 1] + a0);
+                         p->d_ent[cur] + a1, nB, p->d_ent[cur ^ 1] + a0,
+                         general_keys ? (const uint8_t*)p->d_kext : nullptr,
+                         p->d_klen);
       p->kend();
       nbounds.push_back(b1);
     }
@@ -2504,7 +2553,9 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   HIPCHK(p->ens((void**)&p->d_head, n));
   p->kbegin("mark_heads", 17.0 * n);
   hipLaunchKernelGGL(k_mark_heads, dim3(grid_for(n)), dim3(256), 0, p->stream,
-                     ents, n, p->d_head);
+                     ents, n,
+                     general_keys ? (const uint8_t*)p->d_kext : nullptr,
+                     p->d_klen, p->d_head);
   p->kend();
   // head positions via scan, then gather head indices on host
   HIPCHK(p->ens((void**)&p->d_pos, sizeof(uint32_t) * n));
@@ -2653,16 +2704,19 @@ int GpuJob::dedup(const dcw_job_desc* d, std::string* err) {
   HIPCHK(p->ens((void**)&p->d_svlen, sizeof(uint32_t) * (nsurv + 1)));
   HIPCHK(p->ens((void**)&p->d_sklen, nsurv + 1));
   HIPCHK(p->ens((void**)&p->d_sshared, nsurv + 1));
+  HIPCHK(p->ens(&p->d_sw, sizeof(uint32_t) * (nsurv + 1)));
   p->kbegin("gather_survivors", 90.0 * n);
   hipLaunchKernelGGL(k_gather_survivors, dim3(grid_for(n)), dim3(256), 0,
                      p->stream, ents, n, p->d_survive, p->d_pos, p->d_newtag,
                      p->d_clearv, p->d_voff, p->d_vlen, p->d_klen, p->d_sk0,
-                     p->d_sk1, p->d_stag, p->d_svoff, p->d_svlen, p->d_sklen);
+                     p->d_sk1, p->d_stag, p->d_svoff, p->d_svlen, p->d_sklen,
+                     (uint32_t*)p->d_sw);
   p->kend();
   p->kbegin("shared_prefix", 50.0 * nsurv);
   hipLaunchKernelGGL(k_shared_prefix, dim3(grid_for(nsurv)), dim3(256), 0,
                      p->stream, p->d_sk0, p->d_sk1, p->d_stag, p->d_sklen, nsurv,
-                     p->d_sshared);
+                     general_keys ? (const uint8_t*)p->d_kext : nullptr,
+                     (const uint32_t*)p->d_sw, p->d_sshared);
   p->kend();
   // plan metadata D2H
   h_shared_.resize(nsurv);
@@ -2778,7 +2832,9 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   hipLaunchKernelGGL(k_emit, dim3(nb < 4096 ? nb : 4096), dim3(256), 0, p->stream,
                      p->d_bds, nb, p->d_sk0, p->d_sk1, p->d_stag, p->d_svoff,
                      p->d_svlen, p->d_sklen, p->d_sshared, p->d_ublob,
-                     p->d_ucblob, o.block_restart_interval, p->d_err);
+                     p->d_ucblob, o.block_restart_interval,
+                     general_keys ? (const uint8_t*)p->d_kext : nullptr,
+                     (const uint32_t*)p->d_sw, p->d_err);
   p->kend();
   if (o.compression == 1) {
     uint32_t max_unc = 0;
@@ -2842,7 +2898,9 @@ int GpuJob::emit_blocks(const std::vector<PlannedBlock>& blocks, const TableOpts
   }
   hipLaunchKernelGGL(k_block_stats, dim3(grid_for(nb)), dim3(256), 0, p->stream,
                      p->d_bds, 0u, nb, p->d_sk0, p->d_sk1, p->d_stag,
-                     p->d_sklen, (uint8_t*)p->d_scratch_keys);
+                     p->d_sklen,
+                     general_keys ? (const uint8_t*)p->d_kext : nullptr,
+                     (const uint32_t*)p->d_sw, (uint8_t*)p->d_scratch_keys);
   comp_sizes->resize(nb);
   HIPCHK(hipMemcpyAsync(comp_sizes->data(), p->d_ebsize, sizeof(uint32_t) * nb,
                         hipMemcpyDeviceToHost, p->stream));
@@ -2933,7 +2991,7 @@ int GpuJob::fetch_block_keys(size_t b0, size_t b1,
   for (size_t i = b0; i < b1; i++) {
     const uint8_t* o = (const uint8_t*)p->h_keys + i * BLKSTAT_STRIDE;
     first_keys->emplace_back((const char*)o + 1, o[0]);
-    last_keys->emplace_back((const char*)o + 33, o[32]);
+    last_keys->emplace_back((const char*)o + 65, o[64]);
   }
   return 0;
 }
@@ -2945,9 +3003,9 @@ const uint8_t* GpuJob::chunk_stats() const {
 
 void GpuJob::block_stats(size_t b, uint64_t* mn, uint64_t* mx, uint64_t* tomb) {
   const uint8_t* o = (const uint8_t*)p_->h_keys + b * BLKSTAT_STRIDE;
-  memcpy(mn, o + 64, 8);
-  memcpy(mx, o + 72, 8);
-  memcpy(tomb, o + 80, 8);
+  memcpy(mn, o + 128, 8);
+  memcpy(mx, o + 136, 8);
+  memcpy(tomb, o + 144, 8);
 }
 
 int GpuJob::gather_entries(uint64_t first, uint32_t count,
@@ -2971,7 +3029,8 @@ int GpuJob::gather_entries(uint64_t first, uint32_t count,
   hipLaunchKernelGGL(k_gather_range, dim3(grid_for(count)), dim3(256), 0,
                      p->stream, p->d_sk0, p->d_sk1, p->d_stag, p->d_svoff,
                      p->d_svlen, p->d_sklen, p->d_ublob, first, count, d_recoff,
-                     d_out);
+                     general_keys ? (const uint8_t*)p->d_kext : nullptr,
+                     (const uint32_t*)p->d_sw, d_out);
   std::vector<uint8_t> h(acc);
   HIPCHK(hipMemcpyAsync(h.data(), d_out, acc, hipMemcpyDeviceToHost, p->stream));
   HIPCHK(hipStreamSynchronize(p->stream));

@@ -356,7 +356,7 @@ int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err) {
 // avoids per-block heap strings on the main thread
 struct BKey {
   uint8_t len = 0;
-  char b[32];
+  char b[72]; // internal key <= 56 B (general-key envelope)
 };
 
 int fail(dcw_job_result* res, int code, const std::string& msg) {
@@ -830,6 +830,17 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     return fail(res, 30 /*DCW_CANCELLED*/, "job cancelled");
   if (job.decode(&err) != 0) return fail(res, 18, err);
   wp.mark(2, now_usec());
+  if (job.general_keys) {
+    // general-key mode envelope: these combinations still fall back local
+    if (!rd_frags.empty())
+      return fail(res, 29, "range deletions with general-shape keys "
+                           "outside envelope");
+    if (d->num_grandparents > 0)
+      return fail(res, 29, "grandparent cutting with general-shape keys "
+                           "outside envelope");
+    if (d->output_table_factory == 1)
+      return fail(res, 29, "DcwZipTable requires uniform user-key length");
+  }
   if (job.merge(&err) != 0) return fail(res, 19, err);
   wp.mark(3, now_usec());
   job.set_range_del_frags(rd_frags);
@@ -1098,15 +1109,15 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
         BKey fk, lk;
         fk.len = r[0];
         memcpy(fk.b, r + 1, fk.len);
-        lk.len = r[32];
-        memcpy(lk.b, r + 33, lk.len);
+        lk.len = r[64];
+        memcpy(lk.b, r + 65, lk.len);
         first_keys.push_back(fk);
         last_keys.push_back(lk);
         block_counts.push_back(blocks[b].count);
         uint64_t bmn, bmx, bt;
-        memcpy(&bmn, r + 64, 8);
-        memcpy(&bmx, r + 72, 8);
-        memcpy(&bt, r + 80, 8);
+        memcpy(&bmn, r + 128, 8);
+        memcpy(&bmx, r + 136, 8);
+        memcpy(&bt, r + 144, 8);
         if (bmn < mn_seq) mn_seq = bmn;
         if (bmx > mx_seq) mx_seq = bmx;
         n_tomb += bt;
