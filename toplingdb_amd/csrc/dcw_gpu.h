@@ -52,6 +52,11 @@ class GpuJob {
   // H2D of the input blob + block table; or adopt an existing staged input
   // (no H2D, buffers borrowed — not freed by this job).
   int stage(const GpuInputs& in, std::string* err);
+  // split staging for read/H2D overlap: begin -> chunk per input file as
+  // its read completes -> stage(in) uploads the block tables (skips the
+  // blob copy when begin/chunk ran)
+  int stage_begin(size_t blob_size, std::string* err);
+  int stage_chunk(uint64_t off, const void* src, size_t n, std::string* err);
   int stage_adopt(const StagedInput& s, std::string* err);
   // move this job's staged buffers out into `s` (for dcw_stage_inputs)
   int stage_release(StagedInput* s, std::string* err);

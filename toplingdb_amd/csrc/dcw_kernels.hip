@@ -455,8 +455,14 @@ __global__ void k_decode_entries(
       if (!win_varint32(W, pos, end, &shared)) break;
       if (!win_varint32(W, pos, end, &non_shared)) break;
       if (!win_varint32(W, pos, end, &vl)) break;
-      if (shared > klen || shared + non_shared > 24 ||
-          (uint64_t)pos + non_shared + vl > end) {
+      if (shared + non_shared > 24) {
+        // beyond the fast path's 24-byte ikey registers: the host retries
+        // in general-key mode (could still be corruption — the general
+        // decode re-validates against DCW_GKEY_MAX)
+        set_err(err_flag, DE_UKEY_LEN);
+        return;
+      }
+      if (shared > klen || (uint64_t)pos + non_shared + vl > end) {
         set_err(err_flag, DE_BLOCK_PARSE);
         return;
       }
@@ -2034,6 +2040,8 @@ struct GpuJob::Impl {
   uint64_t dzt_ccap = 0;
   void *d_rd_k0 = nullptr, *d_rd_k1 = nullptr, *d_rd_len = nullptr,
        *d_rd_seq = nullptr;
+  bool staged_split = false; // stage_begin/chunk in progress
+  hipEvent_t stage_t0 = nullptr;
   void *d_kext = nullptr; // general-key side table (48 B/entry full ukeys)
   void *d_sw = nullptr;   // survivor -> original payload index (u32)
   void* h_plan = nullptr; // pinned host landing for next+meta
@@ -2189,17 +2197,40 @@ static uint32_t grid_for(uint64_t work, uint32_t block = 256) {
   return (uint32_t)g;
 }
 
+// split staging: begin allocates, chunk streams each input file's bytes
+// as soon as its read completes (overlapping H2D with the remaining
+// reads), finish uploads the block tables
+int GpuJob::stage_begin(size_t blob_size, std::string* err) {
+  Impl* p = p_;
+  (void)hipEventCreate(&p->stage_t0);
+  (void)hipEventRecord(p->stage_t0, p->stream);
+  HIPCHK(p->ens((void**)&p->d_blob_own, blob_size ? blob_size : 1));
+  p->d_blob = p->d_blob_own;
+  p->borrowed_stage = false;
+  p->staged_split = true;
+  return 0;
+}
+int GpuJob::stage_chunk(uint64_t off, const void* src, size_t n,
+                        std::string* err) {
+  Impl* p = p_;
+  HIPCHK(hipMemcpyAsync(p->d_blob + off, src, n, hipMemcpyHostToDevice,
+                        p->stream));
+  return 0;
+}
+
 int GpuJob::stage(const GpuInputs& in, std::string* err) {
   Impl* p = p_;
   hipEvent_t t0, t1;
   (void)hipEventCreate(&t0);
   (void)hipEventCreate(&t1);
   (void)hipEventRecord(t0, p->stream);
-  HIPCHK(p->ens((void**)&p->d_blob_own, in.blob_size));
-  p->d_blob = p->d_blob_own;
-  p->borrowed_stage = false;
-  HIPCHK(hipMemcpyAsync(p->d_blob, in.blob, in.blob_size, hipMemcpyHostToDevice,
-                        p->stream));
+  if (!p->staged_split) {
+    HIPCHK(p->ens((void**)&p->d_blob_own, in.blob_size));
+    p->d_blob = p->d_blob_own;
+    p->borrowed_stage = false;
+    HIPCHK(hipMemcpyAsync(p->d_blob, in.blob, in.blob_size,
+                          hipMemcpyHostToDevice, p->stream));
+  }
   p->n_blocks = (uint32_t)in.blocks.size();
   p->checksum_type = in.checksum_type;
   std::vector<uint64_t> boff(p->n_blocks);
@@ -2223,7 +2254,11 @@ int GpuJob::stage(const GpuInputs& in, std::string* err) {
   HIPCHK(hipMemsetAsync(p->d_uklen_probe, 0xff, 4, p->stream));
   (void)hipEventRecord(t1, p->stream);
   HIPCHK(hipStreamSynchronize(p->stream));
-  ms_h2d += ms_between(t0, t1);
+  ms_h2d += ms_between(p->staged_split ? p->stage_t0 : t0, t1);
+  if (p->staged_split) {
+    (void)hipEventDestroy(p->stage_t0);
+    p->staged_split = false;
+  }
   (void)hipEventDestroy(t0);
   (void)hipEventDestroy(t1);
   // remember run boundaries (translated to entries later)

@@ -142,7 +142,8 @@ uint64_t now_usec() {
 }
 
 struct LoadedInputs; // fwd (RawBuf defined below the pin pool)
-int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err);
+int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err,
+                GpuJob* job = nullptr);
 
 TableOpts opts_from_desc(const dcw_job_desc* d) {
   TableOpts o;
@@ -253,7 +254,8 @@ struct LoadedInputs {
   std::vector<SstTombstone> tombstones; // all inputs' range deletions
 };
 
-int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err) {
+int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err,
+                GpuJob* job) {
   // pre-size the pinned blob so reads land in place (no regrow memcpy)
   uint64_t total = 0;
   for (uint32_t r = 0; r < d->num_runs; r++)
@@ -266,6 +268,9 @@ int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err) {
       total += (uint64_t)st.st_size;
     }
   L->blob.reserve(total);
+  // overlap H2D with the remaining file reads: each file's bytes start
+  // streaming to the device the moment its read completes
+  if (job && job->stage_begin(total, err) != 0) return -1;
   L->gi.run_block_begin.push_back(0);
   uint32_t cstype = 0xffffffff;
   for (uint32_t r = 0; r < d->num_runs; r++) {
@@ -329,6 +334,8 @@ int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err) {
         return -1;
       }
       L->blob.len = base + sz;
+      if (job && job->stage_chunk(base, L->blob.p + base, sz, err) != 0)
+        return -1;
       ParsedSst ps = parse_sst(L->blob.p + base, sz);
       if (!ps.ok) {
         *err = std::string(path) + ": " + ps.error;
@@ -776,7 +783,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     in_bytes = staged->in_bytes;
     tombstones = &staged->tombstones;
   } else {
-    if (load_inputs(d, &L, &err) != 0) return fail(res, 16, err);
+    if (load_inputs(d, &L, &err, &job) != 0) return fail(res, 16, err);
     wp.mark(0, now_usec());
     in_bytes = L.in_bytes;
     tombstones = &L.tombstones;
