@@ -149,6 +149,18 @@ typedef struct dcw_job_desc {
    *     in toplingdb_amd/csrc/dcw_dzt.h). */
   uint32_t output_table_factory;
 
+  /* ---- flush offload (BuildTable, db/builder.cc:56) ----
+   * When flush_kv != NULL the job's input is a SORTED raw KV stream (one
+   * memtable) instead of SST runs (num_runs must be 0): records are
+   * [klen u32 LE][internal key][vlen u32 LE][value], flush_offsets[i] =
+   * byte offset of record i (num_entries+1 entries, last = blob size).
+   * Semantically a flush is a single-run compaction at output_level 0
+   * (same CompactionIterator pass, never bottommost). */
+  const uint8_t* flush_kv;
+  uint64_t flush_kv_bytes;
+  const uint64_t* flush_offsets;
+  uint64_t flush_num_entries;
+
   /* bench hook: when nonzero the worker keeps input SST images and the
    * device staging for this handle alive across calls (see dcw_stage_*) */
   uint64_t staged_handle;

@@ -939,6 +939,56 @@ int32_t orc_execute(const dcw_job_desc* d, dcw_job_result* res) {
     res->status = 2;
     return 2;
   }
+  /* flush offload: reduce the raw sorted KV stream to a
+   * single-run compaction by materializing it as an uncompressed temp SST
+   * (BlockBasedTable encode/decode round-trips the stream exactly, so the
+   * reduction is semantics-preserving; BuildTable = same iterator pass,
+   * db/builder.cc:56) */
+  char flush_tmp[600];
+  flush_tmp[0] = 0;
+  const char* flush_files[1];
+  dcw_run flush_run;
+  dcw_job_desc d2;
+  if (d->flush_kv != NULL) {
+    if (d->num_runs) {
+      snprintf(res->error, sizeof(res->error), "flush job must carry no runs");
+      res->status = 2;
+      return 2;
+    }
+    orc_table_opts t;
+    orc_table_opts_default(&t);
+    orc_table_builder* tb = orc_table_builder_new(&t);
+    for (uint64_t i = 0; i < d->flush_num_entries; i++) {
+      const uint8_t* p = d->flush_kv + d->flush_offsets[i];
+      uint32_t klen, vl;
+      memcpy(&klen, p, 4);
+      memcpy(&vl, p + 4 + klen, 4);
+      orc_table_builder_add(tb, p + 4, klen, p + 8 + klen, vl);
+    }
+    orc_buf img = {0};
+    orc_table_builder_finish(tb, &img);
+    orc_table_builder_delete(tb);
+    snprintf(flush_tmp, sizeof(flush_tmp), "%s/.flush_input.tmp",
+             d->output_dir);
+    FILE* fp = fopen(flush_tmp, "wb");
+    if (!fp || fwrite(img.data, 1, img.size, fp) != img.size) {
+      if (fp) fclose(fp);
+      orc_buf_free(&img);
+      snprintf(res->error, sizeof(res->error), "flush temp write failed");
+      res->status = 3;
+      return 3;
+    }
+    fclose(fp);
+    orc_buf_free(&img);
+    d2 = *d;
+    flush_files[0] = flush_tmp;
+    flush_run.files = flush_files;
+    flush_run.num_files = 1;
+    d2.runs = &flush_run;
+    d2.num_runs = 1;
+    d = &d2;
+  }
+
   kmerge km;
   memset(&km, 0, sizeof(km));
   if (km_init(&km, d->runs, d->num_runs) != 0) {
@@ -1047,6 +1097,10 @@ int32_t orc_execute(const dcw_job_desc* d, dcw_job_result* res) {
   }
   km_free(&km);
   rd_free(&rd);
+  if (flush_tmp[0]) {
+    remove(flush_tmp);
+    in_bytes = d->flush_kv_bytes; /* the metric numerator is the raw stream */
+  }
   res->status = 0;
   res->files = out.files;
   res->num_files = out.num_files;

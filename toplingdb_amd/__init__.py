@@ -66,6 +66,10 @@ class JobDesc(C.Structure):
         ("block_size_deviation", C.c_uint64),
         ("comparator_name", C.c_char_p),
         ("output_table_factory", C.c_uint32),
+        ("flush_kv", C.POINTER(C.c_uint8)),
+        ("flush_kv_bytes", C.c_uint64),
+        ("flush_offsets", C.POINTER(C.c_uint64)),
+        ("flush_num_entries", C.c_uint64),
         ("staged_handle", C.c_uint64),
     ]
 
@@ -249,6 +253,23 @@ def make_job(runs, output_dir, **kw) -> JobDesc:
     d.block_size_deviation = kw.pop("block_size_deviation", 10)
     d.comparator_name = b"leveldb.BytewiseComparator"
     d.output_table_factory = kw.pop("output_table_factory", 0)
+    flush_entries = kw.pop("flush_entries", None)
+    if flush_entries is not None:
+        # sorted [(internal_key, value)] -> the raw flush record blob
+        blob = bytearray()
+        offs = []
+        for k, v in flush_entries:
+            offs.append(len(blob))
+            blob += len(k).to_bytes(4, "little") + k
+            blob += len(v).to_bytes(4, "little") + v
+        offs.append(len(blob))
+        bbuf = (C.c_uint8 * len(blob)).from_buffer_copy(bytes(blob))
+        obuf = (C.c_uint64 * len(offs))(*offs)
+        d._keep += [bbuf, obuf]
+        d.flush_kv = bbuf
+        d.flush_kv_bytes = len(blob)
+        d.flush_offsets = obuf
+        d.flush_num_entries = len(flush_entries)
     d.staged_handle = kw.pop("staged_handle", 0)
     for k, v in kw.items():
         setattr(d, k, v)

@@ -62,6 +62,9 @@ class GpuJob {
   int stage_release(StagedInput* s, std::string* err);
   // checksum-verify, decompress, parse blocks -> per-run entry arrays
   int decode(std::string* err);
+  // flush offload: raw sorted KV records -> entry arrays
+  // (single run, no merge needed)
+  int decode_flush(const dcw_job_desc* d, std::string* err);
   // k-way merge (pairwise merge-path rounds) -> one sorted entry array
   int merge(std::string* err);
   // visibility/dedup FSM + survivor compaction + plan metadata D2H

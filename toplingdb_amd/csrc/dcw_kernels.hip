@@ -648,6 +648,60 @@ __global__ void k_decode_entries_g(
   }
 }
 
+// flush offload input: one thread per raw KV record.
+// Record: [klen u32][ikey][vlen u32][value]; offs[i] = record offset,
+// offs[n] = blob size.  general: keys > 16 B / mixed via the side table.
+__global__ void k_decode_flush(const uint8_t* __restrict__ blob,
+                               const uint64_t* __restrict__ offs, uint64_t n,
+                               int general, uint32_t uniform_ulen,
+                               ulong4* __restrict__ ents,
+                               uint64_t* __restrict__ voff,
+                               uint32_t* __restrict__ vlen_out,
+                               uint8_t* __restrict__ klen_out,
+                               uint8_t* __restrict__ kext, uint32_t* err_flag) {
+  for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    const uint8_t* p = blob + offs[i];
+    const uint8_t* lim = blob + offs[i + 1];
+    uint32_t klen, vl;
+    memcpy(&klen, p, 4);
+    const uint8_t* key = p + 4;
+    memcpy(&vl, key + klen, 4);
+    if (p + 8 + klen + vl > lim || klen < 9 ||
+        klen > (general ? DCW_GKEY_MAX + 8u : 24u)) {
+      set_err(err_flag, DE_BLOCK_PARSE);
+      return;
+    }
+    uint32_t ulen = klen - 8;
+    if (!general && ulen != uniform_ulen) {
+      set_err(err_flag, DE_UKEY_LEN);
+      return;
+    }
+    uint64_t tag;
+    memcpy(&tag, key + ulen, 8);
+    uint8_t vt = (uint8_t)tag;
+    if (!(vt == kTypeValue || vt == kTypeDeletion || vt == kTypeSingleDeletion)) {
+      set_err(err_flag, DE_TYPE);
+      return;
+    }
+    uint64_t u0 = 0, u1 = 0;
+    for (uint32_t t = 0; t < ulen && t < 8; t++)
+      u0 |= (uint64_t)key[t] << (8 * t);
+    for (uint32_t t = 8; t < ulen && t < 16; t++)
+      u1 |= (uint64_t)key[t] << (8 * (t - 8));
+    ents[i] = make_ulong4(__builtin_bswap64(u0), __builtin_bswap64(u1), ~tag,
+                          i);
+    if (general) {
+      uint8_t* slot = kext + i * (uint64_t)DCW_GKEY_STRIDE;
+      for (uint32_t t = 0; t < ulen; t++) slot[t] = key[t];
+      for (uint32_t t = ulen; t < DCW_GKEY_STRIDE; t++) slot[t] = 0;
+    }
+    voff[i] = (uint64_t)(key + klen + 4 - blob);
+    vlen_out[i] = vl;
+    klen_out[i] = (uint8_t)klen;
+  }
+}
+
 // ------------------------------------------------------------------
 // merge
 // ------------------------------------------------------------------
@@ -2044,6 +2098,7 @@ struct GpuJob::Impl {
   hipEvent_t stage_t0 = nullptr;
   void *d_kext = nullptr; // general-key side table (48 B/entry full ukeys)
   void *d_sw = nullptr;   // survivor -> original payload index (u32)
+  void *d_flush_offs = nullptr; // flush-offload record offsets (u64)
   void* h_plan = nullptr; // pinned host landing for next+meta
   size_t h_plan_cap = 0;
   bool h_plan_pageable = false;
@@ -2169,7 +2224,7 @@ GpuJob::~GpuJob() {
   F(p->d_dzt_cblob); F(p->d_dzt_bsize); F(p->d_dzt_btype); F(p->d_dzt_csum);
   F(p->d_dzt_kbs); F(p->d_dzt_keyarea); F(p->d_dzt_kidx); F(p->d_dzt_img);
   F(p->d_rd_k0); F(p->d_rd_k1); F(p->d_rd_len); F(p->d_rd_seq);
-  F(p->d_kext); F(p->d_sw);
+  F(p->d_kext); F(p->d_sw); F(p->d_flush_offs);
   if (p->h_plan) {
     if (p->h_plan_pageable)
       free(p->h_plan);
@@ -2500,6 +2555,80 @@ int GpuJob::decode(std::string* err) {
     return -1;
   }
   ukey_len = uklen;
+  return 0;
+}
+
+int GpuJob::decode_flush(const dcw_job_desc* d, std::string* err) {
+  Impl* p = p_;
+  uint64_t n = d->flush_num_entries;
+  // host pre-scan of record key lengths: pick fast vs general mode and
+  // validate bounds (the blob is caller host memory)
+  bool general = false;
+  uint32_t ulen0 = 0;
+  for (uint64_t i = 0; i < n; i++) {
+    uint32_t klen;
+    if (d->flush_offsets[i] + 8 > d->flush_kv_bytes) {
+      if (err) *err = "flush record out of bounds";
+      return -1;
+    }
+    memcpy(&klen, d->flush_kv + d->flush_offsets[i], 4);
+    if (klen < 9 || klen > DCW_GKEY_MAX + 8u) {
+      if (err) *err = "flush key length outside envelope";
+      return -1;
+    }
+    uint32_t ul = klen - 8;
+    if (i == 0) ulen0 = ul;
+    if (ul != ulen0 || ul > 16) general = true;
+  }
+  hipEvent_t t0, t1;
+  (void)hipEventCreate(&t0);
+  (void)hipEventCreate(&t1);
+  (void)hipEventRecord(t0, p->stream);
+  HIPCHK(p->ens((void**)&p->d_ublob, d->flush_kv_bytes + 64));
+  p->ublob_size = d->flush_kv_bytes;
+  HIPCHK(hipMemcpyAsync(p->d_ublob, d->flush_kv, d->flush_kv_bytes,
+                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(p->ens(&p->d_flush_offs, (n + 1) * 8));
+  HIPCHK(hipMemcpyAsync(p->d_flush_offs, d->flush_offsets, (n + 1) * 8,
+                        hipMemcpyHostToDevice, p->stream));
+  if (!p->d_err) HIPCHK(hipMalloc(&p->d_err, 8));
+  HIPCHK(hipMemsetAsync(p->d_err, 0, 8, p->stream));
+  if (!p->d_crc) HIPCHK(hipMalloc(&p->d_crc, sizeof(Crc32cTables)));
+  HIPCHK(hipMemcpyAsync(p->d_crc, &g_crc, sizeof(Crc32cTables),
+                        hipMemcpyHostToDevice, p->stream));
+  HIPCHK(p->ens((void**)&p->d_ent[0], sizeof(ulong4) * (n + 1)));
+  HIPCHK(p->ens((void**)&p->d_ent[1], sizeof(ulong4) * (n + 1)));
+  HIPCHK(p->ens((void**)&p->d_voff, sizeof(uint64_t) * (n + 1)));
+  HIPCHK(p->ens((void**)&p->d_vlen, sizeof(uint32_t) * (n + 1)));
+  HIPCHK(p->ens((void**)&p->d_klen, n + 1));
+  if (general) HIPCHK(p->ens(&p->d_kext, n * 48ull + 64));
+  p->kbegin("decode_flush", (double)d->flush_kv_bytes + 45.0 * n);
+  hipLaunchKernelGGL(k_decode_flush, dim3(grid_for(n)), dim3(256), 0,
+                     p->stream, p->d_ublob, (const uint64_t*)p->d_flush_offs,
+                     n, general ? 1 : 0, ulen0, p->d_ent[0], p->d_voff,
+                     p->d_vlen, p->d_klen, (uint8_t*)p->d_kext, p->d_err);
+  p->kend();
+  uint32_t err_host = 0;
+  HIPCHK(hipMemcpyAsync(&err_host, p->d_err, 4, hipMemcpyDeviceToHost,
+                        p->stream));
+  (void)hipEventRecord(t1, p->stream);
+  HIPCHK(hipStreamSynchronize(p->stream));
+  ms_decode += ms_between(t0, t1);
+  p->kresolve();
+  (void)hipEventDestroy(t0);
+  (void)hipEventDestroy(t1);
+  if (err_host) {
+    if (err) *err = "flush decode failed, code " + std::to_string(err_host);
+    return -1;
+  }
+  p->n_entries = n;
+  n_entries_ = n;
+  p->run_entry_begin.clear();
+  p->run_entry_begin.push_back(0);
+  p->run_entry_begin.push_back(n);
+  p->final_buf = 0;
+  general_keys = general;
+  ukey_len = general ? 0 : ulen0;
   return 0;
 }
 

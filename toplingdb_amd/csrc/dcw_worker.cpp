@@ -59,6 +59,10 @@ struct TailPool {
   void start_locked() {
     unsigned n = std::thread::hardware_concurrency();
     unsigned workers = n ? (n < 32 ? n : 32) : 8;
+    if (const char* e = getenv("DCW_TAIL_WORKERS")) {
+      int w = atoi(e);
+      if (w > 0 && w <= 256) workers = (unsigned)w;
+    }
     for (unsigned i = 0; i < workers; i++)
       threads.emplace_back([this]() {
         for (;;) {
@@ -294,9 +298,14 @@ int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err,
       // ~10 GB/s and sits on the job's critical path (the write side is
       // already segmented)
       uint8_t* dstp = L->blob.p + base;
+      static const uint64_t max_seg = [] {
+        const char* e = getenv("DCW_READ_SEGS");
+        int v = e ? atoi(e) : 4;
+        return (uint64_t)(v > 0 && v <= 16 ? v : 4);
+      }();
       const uint64_t kSeg = 16u << 20;
       uint64_t nseg = (sz + kSeg - 1) / kSeg;
-      if (nseg > 4) nseg = 4;
+      if (nseg > max_seg) nseg = max_seg;
       if (nseg == 0) nseg = 1;
       uint64_t seg = (sz + nseg - 1) / nseg;
       std::vector<std::future<bool>> segr;
@@ -775,7 +784,16 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   LoadedInputs L;
   StagedJob* staged = nullptr;
   const std::vector<SstTombstone>* tombstones = nullptr;
-  if (d->staged_handle) {
+  const bool flush_mode = d->flush_kv != nullptr;
+  if (flush_mode) {
+    // flush offload: the input is one sorted raw-KV
+    // memtable stream; semantically a single-run compaction at L0
+    if (d->num_runs || d->staged_handle)
+      return fail(res, 14, "flush job must carry no SST runs");
+    if (d->flush_num_entries == 0 || !d->flush_offsets)
+      return fail(res, 14, "empty flush job");
+    in_bytes = d->flush_kv_bytes;
+  } else if (d->staged_handle) {
     std::lock_guard<std::mutex> lk(g_mu);
     auto it = g_staged.find(d->staged_handle);
     if (it == g_staged.end()) return fail(res, 15, "bad staged handle");
@@ -827,16 +845,23 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
   res->t_read_usec = now_usec() - t0;
 
   // ---- GPU pipeline ----
-  if (staged) {
+  if (flush_mode) {
+    if (consume_cancel(d->job_id))
+      return fail(res, 30 /*DCW_CANCELLED*/, "job cancelled");
+    if (job.decode_flush(d, &err) != 0) return fail(res, 18, err);
+    wp.mark(2, now_usec());
+  } else if (staged) {
     if (job.stage_adopt(staged->dev, &err) != 0) return fail(res, 17, err);
     wp.mark(1, now_usec());
   } else {
     if (job.stage(L.gi, &err) != 0) return fail(res, 17, err);
   }
-  if (consume_cancel(d->job_id))
-    return fail(res, 30 /*DCW_CANCELLED*/, "job cancelled");
-  if (job.decode(&err) != 0) return fail(res, 18, err);
-  wp.mark(2, now_usec());
+  if (!flush_mode) {
+    if (consume_cancel(d->job_id))
+      return fail(res, 30 /*DCW_CANCELLED*/, "job cancelled");
+    if (job.decode(&err) != 0) return fail(res, 18, err);
+    wp.mark(2, now_usec());
+  }
   if (job.general_keys) {
     // general-key mode envelope: these combinations still fall back local
     if (!rd_frags.empty())
