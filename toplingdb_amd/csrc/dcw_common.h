@@ -637,6 +637,133 @@ DCW_HD size_t snappy_uncompress_dict(const uint8_t* dict, size_t D,
   return op == oend ? (size_t)ulen : 0;
 }
 
+// ---------- XXPH3 (filter hash) ----------
+// RocksDB's FROZEN xxh3-preview fork (util/xxph3.h; differs from final
+// XXH3) — the hash behind GetSliceHash64 (util/hash.h:97) that the bloom
+// filter path consumes.  Restated for inputs <= 128 B (user keys are
+// <= 48 B in this worker); pinned by golden vectors generated from the
+// reference's own header (oracle/_ref).
+static const uint8_t kXxph3Secret[128] = {
+    0xb8, 0xfe, 0x6c, 0x39, 0x23, 0xa4, 0x4b, 0xbe, 0x7c, 0x01, 0x81, 0x2c,
+    0xf7, 0x21, 0xad, 0x1c, 0xde, 0xd4, 0x6d, 0xe9, 0x83, 0x90, 0x97, 0xdb,
+    0x72, 0x40, 0xa4, 0xa4, 0xb7, 0xb3, 0x67, 0x1f, 0xcb, 0x79, 0xe6, 0x4e,
+    0xcc, 0xc0, 0xe5, 0x78, 0x82, 0x5a, 0xd0, 0x7d, 0xcc, 0xff, 0x72, 0x21,
+    0xb8, 0x08, 0x46, 0x74, 0xf7, 0x43, 0x24, 0x8e, 0xe0, 0x35, 0x90, 0xe6,
+    0x81, 0x3a, 0x26, 0x4c, 0x3c, 0x28, 0x52, 0xbb, 0x91, 0xc3, 0x00, 0xcb,
+    0x88, 0xd0, 0x65, 0x8b, 0x1b, 0x53, 0x2e, 0xa3, 0x71, 0x64, 0x48, 0x97,
+    0xa2, 0x0d, 0xf9, 0x4e, 0x38, 0x19, 0xef, 0x46, 0xa9, 0xde, 0xac, 0xd8,
+    0xa8, 0xfa, 0x76, 0x3f, 0xe3, 0x9c, 0x34, 0x3f, 0xf9, 0xdc, 0xbb, 0xc7,
+    0xc7, 0x0b, 0x4f, 0x1d, 0x8a, 0x51, 0xe0, 0x4b, 0xcd, 0xb4, 0x59, 0x31,
+    0xc8, 0x9f, 0x7e, 0xc9, 0xd9, 0x78, 0x73, 0x64};
+
+DCW_HD uint64_t xxph3_load64(const uint8_t* p) {
+  uint64_t v;
+  memcpy(&v, p, 8);
+  return v;
+}
+DCW_HD uint64_t xxph3_mul128_fold64(uint64_t a, uint64_t b) {
+#if defined(__SIZEOF_INT128__)
+  __uint128_t prod = (__uint128_t)a * b;
+  return (uint64_t)prod ^ (uint64_t)(prod >> 64);
+#else
+  uint64_t lo_lo = (a & 0xffffffffu) * (b & 0xffffffffu);
+  uint64_t hi_lo = (a >> 32) * (b & 0xffffffffu);
+  uint64_t lo_hi = (a & 0xffffffffu) * (b >> 32);
+  uint64_t hi_hi = (a >> 32) * (b >> 32);
+  uint64_t cross = (lo_lo >> 32) + (hi_lo & 0xffffffffu) + lo_hi;
+  uint64_t upper = (hi_lo >> 32) + (cross >> 32) + hi_hi;
+  uint64_t lower = (cross << 32) | (lo_lo & 0xffffffffu);
+  return lower ^ upper;
+#endif
+}
+DCW_HD uint64_t xxph3_avalanche(uint64_t h) {
+  h ^= h >> 37;
+  h *= 1609587929392839161ull; // PRIME64_3
+  h ^= h >> 32;
+  return h;
+}
+DCW_HD uint64_t xxph3_mix16(const uint8_t* in, const uint8_t* sec) {
+  return xxph3_mul128_fold64(xxph3_load64(in) ^ xxph3_load64(sec),
+                             xxph3_load64(in + 8) ^ xxph3_load64(sec + 8));
+}
+// XXPH3_64bits for len <= 128 (seed 0); out of range is a caller bug
+DCW_HD uint64_t xxph3_64(const uint8_t* in, size_t len) {
+  const uint64_t P64_1 = 11400714785074694791ull;
+  const uint64_t P64_2 = 14029467366897019727ull;
+  const uint32_t P32_1 = 2654435761u;
+  const uint8_t* sec = kXxph3Secret;
+  if (len == 0)
+    return xxph3_mul128_fold64(xxph3_load64(sec), P64_2);
+  if (len <= 3) {
+    uint32_t c1 = in[0], c2 = in[len >> 1], c3 = in[len - 1];
+    uint32_t comb = c1 | (c2 << 8) | (c3 << 16) | ((uint32_t)len << 24);
+    uint32_t s32;
+    memcpy(&s32, sec, 4);
+    uint64_t keyed = (uint64_t)comb ^ (uint64_t)s32;
+    return xxph3_avalanche(keyed * P64_1);
+  }
+  if (len <= 8) {
+    uint32_t lo, hi;
+    memcpy(&lo, in, 4);
+    memcpy(&hi, in + len - 4, 4);
+    uint64_t in64 = (uint64_t)lo | ((uint64_t)hi << 32);
+    uint64_t keyed = in64 ^ xxph3_load64(sec);
+    uint64_t mix64 = len + (keyed ^ (keyed >> 51)) * P32_1;
+    return xxph3_avalanche((mix64 ^ (mix64 >> 47)) * P64_2);
+  }
+  if (len <= 16) {
+    uint64_t lo = xxph3_load64(in) ^ xxph3_load64(sec);
+    uint64_t hi = xxph3_load64(in + len - 8) ^ xxph3_load64(sec + 8);
+    uint64_t acc = len + lo + hi + xxph3_mul128_fold64(lo, hi);
+    return xxph3_avalanche(acc);
+  }
+  uint64_t acc = len * P64_1;
+  if (len > 32) {
+    if (len > 64) {
+      if (len > 96) {
+        acc += xxph3_mix16(in + 48, sec + 96);
+        acc += xxph3_mix16(in + len - 64, sec + 112);
+      }
+      acc += xxph3_mix16(in + 32, sec + 64);
+      acc += xxph3_mix16(in + len - 48, sec + 80);
+    }
+    acc += xxph3_mix16(in + 16, sec + 32);
+    acc += xxph3_mix16(in + len - 32, sec + 48);
+  }
+  acc += xxph3_mix16(in, sec);
+  acc += xxph3_mix16(in + len - 16, sec + 16);
+  return xxph3_avalanche(acc);
+}
+
+// ---------- FastLocalBloom (util/bloom_impl.h:144-223) ----------
+DCW_HD uint32_t bloom_fastrange32(uint32_t a, uint32_t b) {
+  return (uint32_t)(((uint64_t)a * b) >> 32);
+}
+// ChooseNumProbes (bloom_impl.h:156-199)
+DCW_HD int bloom_num_probes(int millibits_per_key) {
+  if (millibits_per_key <= 2080) return 1;
+  if (millibits_per_key <= 3580) return 2;
+  if (millibits_per_key <= 5100) return 3;
+  if (millibits_per_key <= 6640) return 4;
+  if (millibits_per_key <= 8300) return 5;
+  if (millibits_per_key <= 10070) return 6;
+  if (millibits_per_key <= 11720) return 7;
+  if (millibits_per_key <= 14001) return 8;
+  if (millibits_per_key <= 16050) return 9;
+  if (millibits_per_key <= 18300) return 10;
+  if (millibits_per_key <= 22001) return 11;
+  if (millibits_per_key <= 25501) return 12;
+  if (millibits_per_key > 50000) return 24;
+  return (millibits_per_key - 1) / 2000 - 1;
+}
+// CalculateSpace (filter_policy.cc:409-424; no malloc rounding — the
+// default optimize_filters_for_memory=false path)
+DCW_HD uint64_t bloom_len_with_metadata(uint64_t num_keys, uint32_t millibits) {
+  uint64_t raw = (num_keys * millibits + 7999) / 8000;
+  if (raw >= 0xffffffc0ull) raw = 0xffffffc0ull;
+  return ((raw + 63) & ~63ull) + 5;
+}
+
 // ---------- internal keys ----------
 // Normalized 24-byte sort key for uniform-length user keys (<=16 B):
 // k0,k1 = big-endian words of the zero-padded user key, k2 = ~tag.
