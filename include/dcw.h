@@ -149,6 +149,13 @@ typedef struct dcw_job_desc {
    *     in toplingdb_amd/csrc/dcw_dzt.h). */
   uint32_t output_table_factory;
 
+  /* Bloom filter build: BlockBasedTableOptions::
+   * filter_policy = BloomFilterPolicy(bits_per_key) with whole-key
+   * filtering (full_filter_block.cc; FastLocalBloom, util/bloom_impl.h:
+   * 144-223; hash = XXPH3 of the user key, util/hash.h:97).  Millibits
+   * per key; 0 = no filter (the default).  10000 == 10.0 bits/key. */
+  uint32_t bloom_millibits_per_key;
+
   /* ---- flush offload (BuildTable, db/builder.cc:56) ----
    * When flush_kv != NULL the job's input is a SORTED raw KV stream (one
    * memtable) instead of SST runs (num_runs must be 0): records are

@@ -73,6 +73,7 @@ class JobDesc(C.Structure):
         ("block_size_deviation", C.c_uint64),
         ("comparator_name", C.c_char_p),
         ("output_table_factory", C.c_uint32),
+        ("bloom_millibits_per_key", C.c_uint32),
         ("flush_kv", C.POINTER(C.c_uint8)),
         ("flush_kv_bytes", C.c_uint64),
         ("flush_offsets", C.POINTER(C.c_uint64)),
@@ -122,6 +123,8 @@ _lib.orc_crc32c_masked.restype = C.c_uint32
 _lib.orc_crc32c_masked.argtypes = [C.c_void_p, C.c_size_t]
 _lib.orc_xxh3_64.restype = C.c_uint64
 _lib.orc_xxh3_64.argtypes = [C.c_void_p, C.c_size_t]
+_lib.orc_xxph3_64.restype = C.c_uint64
+_lib.orc_xxph3_64.argtypes = [C.c_void_p, C.c_size_t]
 _lib.orc_block_checksum.restype = C.c_uint32
 _lib.orc_block_checksum.argtypes = [C.c_uint32, C.c_void_p, C.c_size_t, C.c_uint8]
 _lib.orc_snappy_max_compressed.restype = C.c_size_t
@@ -149,6 +152,10 @@ def crc32c_masked(data: bytes) -> int:
 
 def xxh3_64(data: bytes) -> int:
     return _lib.orc_xxh3_64(data, len(data))
+
+
+def xxph3_64(data: bytes) -> int:
+    return _lib.orc_xxph3_64(data, len(data))
 
 
 def block_checksum(cstype: int, data: bytes, last_byte: int) -> int:
@@ -487,6 +494,7 @@ def make_job(runs, output_dir, **kw) -> JobDesc:
     d.block_size_deviation = kw.pop("block_size_deviation", 10)
     d.comparator_name = b"leveldb.BytewiseComparator"
     d.output_table_factory = kw.pop("output_table_factory", 0)
+    d.bloom_millibits_per_key = kw.pop("bloom_millibits_per_key", 0)
     flush_entries = kw.pop("flush_entries", None)
     if flush_entries is not None:
         # sorted [(internal_key, value)] -> the raw flush record blob

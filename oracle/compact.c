@@ -799,6 +799,7 @@ static void fill_topts(outputs* o, orc_table_opts* t) {
   t->file_creation_time = d->current_time;
   t->oldest_key_time = 0;
   t->level_at_creation = d->output_level;
+  t->bloom_millibits_per_key = d->bloom_millibits_per_key;
 }
 
 static void out_open(outputs* o) {

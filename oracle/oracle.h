@@ -40,6 +40,7 @@ extern "C" {
 uint32_t orc_crc32c(const void* data, size_t n);           /* unmasked */
 uint32_t orc_crc32c_masked(const void* data, size_t n);    /* crc32c::Mask(Value()) */
 uint64_t orc_xxh3_64(const void* data, size_t n);          /* XXH3_64bits, seed 0 */
+uint64_t orc_xxph3_64(const void* data, size_t n);         /* XXPH3 (filter hash), n <= 128 */
 uint32_t orc_block_checksum(uint32_t checksum_type, const void* data, size_t n,
                             uint8_t last_byte); /* ComputeBuiltinChecksumWithLastByte */
 
@@ -88,6 +89,9 @@ typedef struct orc_table_opts {
   uint64_t file_creation_time;
   uint64_t oldest_key_time;     /* 0 for compaction outputs */
   int32_t level_at_creation;
+  /* bloom filter (FastLocalBloom, util/bloom_impl.h:144): millibits per
+   * key, 0 = no filter.  10000 == BloomFilterPolicy(10.0). */
+  uint32_t bloom_millibits_per_key;
 } orc_table_opts;
 
 void orc_table_opts_default(orc_table_opts* o);

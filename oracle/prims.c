@@ -557,6 +557,78 @@ size_t orc_snappy_uncompress_dict(const uint8_t* dict, size_t D,
   return op == oend ? ulen : 0;
 }
 
+/* ---------------- XXPH3 (filter hash) ----------------
+ * RocksDB's FROZEN xxh3-preview fork (util/xxph3.h; NOT final XXH3) —
+ * the hash behind GetSliceHash64 (util/hash.h:97) used by bloom filters.
+ * Restated for inputs <= 128 B; pinned by tests/golden/xxph3_vectors.json
+ * (generated from the reference's own header via oracle/_ref/xxph3_ref). */
+static const uint8_t xxph3_secret[128] = {
+    0xb8, 0xfe, 0x6c, 0x39, 0x23, 0xa4, 0x4b, 0xbe, 0x7c, 0x01, 0x81, 0x2c,
+    0xf7, 0x21, 0xad, 0x1c, 0xde, 0xd4, 0x6d, 0xe9, 0x83, 0x90, 0x97, 0xdb,
+    0x72, 0x40, 0xa4, 0xa4, 0xb7, 0xb3, 0x67, 0x1f, 0xcb, 0x79, 0xe6, 0x4e,
+    0xcc, 0xc0, 0xe5, 0x78, 0x82, 0x5a, 0xd0, 0x7d, 0xcc, 0xff, 0x72, 0x21,
+    0xb8, 0x08, 0x46, 0x74, 0xf7, 0x43, 0x24, 0x8e, 0xe0, 0x35, 0x90, 0xe6,
+    0x81, 0x3a, 0x26, 0x4c, 0x3c, 0x28, 0x52, 0xbb, 0x91, 0xc3, 0x00, 0xcb,
+    0x88, 0xd0, 0x65, 0x8b, 0x1b, 0x53, 0x2e, 0xa3, 0x71, 0x64, 0x48, 0x97,
+    0xa2, 0x0d, 0xf9, 0x4e, 0x38, 0x19, 0xef, 0x46, 0xa9, 0xde, 0xac, 0xd8,
+    0xa8, 0xfa, 0x76, 0x3f, 0xe3, 0x9c, 0x34, 0x3f, 0xf9, 0xdc, 0xbb, 0xc7,
+    0xc7, 0x0b, 0x4f, 0x1d, 0x8a, 0x51, 0xe0, 0x4b, 0xcd, 0xb4, 0x59, 0x31,
+    0xc8, 0x9f, 0x7e, 0xc9, 0xd9, 0x78, 0x73, 0x64};
+static uint64_t xp_fold64(uint64_t a, uint64_t b) {
+  __uint128_t p = (__uint128_t)a * b;
+  return (uint64_t)p ^ (uint64_t)(p >> 64);
+}
+static uint64_t xp_avalanche(uint64_t h) {
+  h ^= h >> 37;
+  h *= 1609587929392839161ull;
+  h ^= h >> 32;
+  return h;
+}
+static uint64_t xp_mix16(const uint8_t* in, const uint8_t* sec) {
+  return xp_fold64(rd64(in) ^ rd64(sec), rd64(in + 8) ^ rd64(sec + 8));
+}
+uint64_t orc_xxph3_64(const void* data, size_t len) {
+  /* XXPH3 primes == the xxhash primes already defined above (P64_*, P32_1) */
+  const uint8_t* in = (const uint8_t*)data;
+  const uint8_t* sec = xxph3_secret;
+  if (len == 0) return xp_fold64(rd64(sec), P64_2);
+  if (len <= 3) {
+    uint32_t comb = (uint32_t)in[0] | ((uint32_t)in[len >> 1] << 8) |
+                    ((uint32_t)in[len - 1] << 16) | ((uint32_t)len << 24);
+    return xp_avalanche(((uint64_t)comb ^ (uint64_t)rd32(sec)) * P64_1);
+  }
+  if (len <= 8) {
+    uint64_t in64 = (uint64_t)rd32(in) | ((uint64_t)rd32(in + len - 4) << 32);
+    uint64_t keyed = in64 ^ rd64(sec);
+    uint64_t mix64 = len + (keyed ^ (keyed >> 51)) * P32_1;
+    return xp_avalanche((mix64 ^ (mix64 >> 47)) * P64_2);
+  }
+  if (len <= 16) {
+    uint64_t lo = rd64(in) ^ rd64(sec);
+    uint64_t hi = rd64(in + len - 8) ^ rd64(sec + 8);
+    return xp_avalanche(len + lo + hi + xp_fold64(lo, hi));
+  }
+  if (len <= 128) {
+    uint64_t acc = len * P64_1;
+    if (len > 32) {
+      if (len > 64) {
+        if (len > 96) {
+          acc += xp_mix16(in + 48, sec + 96);
+          acc += xp_mix16(in + len - 64, sec + 112);
+        }
+        acc += xp_mix16(in + 32, sec + 64);
+        acc += xp_mix16(in + len - 48, sec + 80);
+      }
+      acc += xp_mix16(in + 16, sec + 32);
+      acc += xp_mix16(in + len - 32, sec + 48);
+    }
+    acc += xp_mix16(in, sec);
+    acc += xp_mix16(in + len - 16, sec + 16);
+    return xp_avalanche(acc);
+  }
+  return 0; /* out of restated range (keys are <= 48 B) */
+}
+
 /* internal key compare (dbformat.h:1057-1096): ukey bytewise asc;
  * shorter-prefix first; tie -> 8-byte LE tag numeric DESC. */
 int orc_ikey_compare(const uint8_t* a, size_t alen, const uint8_t* b, size_t blen) {

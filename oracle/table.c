@@ -187,6 +187,11 @@ struct orc_table_builder {
   int has_pending; /* data block flushed, index entry not yet added */
   oblk rangedel; /* (ikey(start,seq,0xF), end_ukey) entries, interval 1 */
   int has_rangedel;
+  /* bloom filter hash entries (XXPH3 of user keys; consecutive equal
+   * hashes deduped — XXPH3FilterBitsBuilder::AddKey) */
+  uint64_t* fhash;
+  size_t nfh, fh_cap;
+  uint64_t filter_size; /* props.filter_size (content, no trailer) */
   /* props */
   uint64_t num_entries, num_deletions, num_range_deletions, num_merge_operands;
   uint64_t raw_key_size, raw_value_size, num_data_blocks, data_size, index_size;
@@ -368,8 +373,21 @@ static void tb_flush_data(orc_table_builder* b) {
   b->has_pending = 1;
 }
 
+static void tb_filter_add(orc_table_builder* b, const uint8_t* ukey,
+                          size_t ulen) {
+  /* XXPH3FilterBitsBuilder::AddKey: consecutive-equal-hash dedup */
+  uint64_t h = orc_xxph3_64(ukey, ulen);
+  if (b->nfh && b->fhash[b->nfh - 1] == h) return;
+  if (b->nfh == b->fh_cap) {
+    b->fh_cap = b->fh_cap ? b->fh_cap * 2 : 1024;
+    b->fhash = (uint64_t*)realloc(b->fhash, b->fh_cap * 8);
+  }
+  b->fhash[b->nfh++] = h;
+}
+
 void orc_table_builder_add(orc_table_builder* b, const uint8_t* ikey, size_t klen,
                            const uint8_t* value, size_t vlen) {
+  if (b->o.bloom_millibits_per_key) tb_filter_add(b, ikey, klen - 8);
   uint8_t type = ikey[klen - 8];
   /* flush policy (flush_block_policy.cc:37-52) */
   int should_flush = 0;
@@ -424,7 +442,44 @@ int orc_table_builder_finish(orc_table_builder* b, orc_buf* out) {
   if (b->has_pending) tb_add_index_entry(b, NULL, 0);
   b->tail_start_offset = b->file.size;
 
-  /* 1. (no filter block)  2. index block */
+  /* 1. full filter block (FastLocalBloom; filter_policy.cc Finish +
+   * bloom_impl.h AddHash; uncompressed, builder.cc:1488-1526) */
+  uint64_t filter_off = 0, filter_sz = 0;
+  int has_filter = b->o.bloom_millibits_per_key && b->nfh > 0;
+  if (has_filter) {
+    uint64_t lwm = ((uint64_t)b->nfh * b->o.bloom_millibits_per_key + 7999) / 8000;
+    if (lwm >= 0xffffffc0ull) lwm = 0xffffffc0ull;
+    lwm = ((lwm + 63) & ~63ull) + 5; /* CalculateSpace */
+    uint32_t len = (uint32_t)(lwm - 5);
+    int probes; /* ChooseNumProbes(millibits_per_key), bloom_impl.h:156 */
+    {
+      int m = (int)b->o.bloom_millibits_per_key;
+      probes = m <= 2080 ? 1 : m <= 3580 ? 2 : m <= 5100 ? 3 : m <= 6640 ? 4
+              : m <= 8300 ? 5 : m <= 10070 ? 6 : m <= 11720 ? 7
+              : m <= 14001 ? 8 : m <= 16050 ? 9 : m <= 18300 ? 10
+              : m <= 22001 ? 11 : m <= 25501 ? 12
+              : m > 50000 ? 24 : (m - 1) / 2000 - 1;
+    }
+    uint8_t* fdata = (uint8_t*)calloc(1, lwm);
+    for (size_t i = 0; i < b->nfh; i++) {
+      uint32_t h1 = (uint32_t)b->fhash[i];
+      uint32_t h2 = (uint32_t)(b->fhash[i] >> 32);
+      uint32_t line = (uint32_t)(((uint64_t)h1 * (len >> 6)) >> 32) << 6;
+      uint32_t h = h2;
+      for (int p = 0; p < probes; p++, h *= 0x9e3779b9u) {
+        int bitpos = h >> (32 - 9);
+        fdata[line + (bitpos >> 3)] |= (uint8_t)(1u << (bitpos & 7));
+      }
+    }
+    fdata[len] = (uint8_t)(int8_t)-1; /* marker: newer Bloom impls */
+    fdata[len + 1] = 0;               /* sub-implementation */
+    fdata[len + 2] = (uint8_t)probes;
+    tb_write_block(b, fdata, lwm, 0, &filter_off, &filter_sz);
+    free(fdata);
+    b->filter_size = lwm;
+  }
+
+  /* 2. index block */
   uint64_t index_off, index_size_comp;
   orc_buf* idx = &b->scratch;
   oblk_finish(b->sep_is_key_plus_seq ? &b->index_seq : &b->index_user, idx);
@@ -456,12 +511,12 @@ int orc_table_builder_finish(orc_table_builder* b, orc_buf* out) {
                b->sep_is_key_plus_seq ? 0 : 1);
   prop_add_int(props, &np, "rocksdb.index.value.is.delta.encoded", 1);
   prop_add_int(props, &np, "rocksdb.num.entries", b->num_entries);
-  prop_add_int(props, &np, "rocksdb.num.filter_entries", 0);
+  prop_add_int(props, &np, "rocksdb.num.filter_entries", (uint64_t)b->nfh);
   prop_add_int(props, &np, "rocksdb.deleted.keys", b->num_deletions);
   prop_add_int(props, &np, "rocksdb.merge.operands", b->num_merge_operands);
   prop_add_int(props, &np, "rocksdb.num.range-deletions", b->num_range_deletions);
   prop_add_int(props, &np, "rocksdb.num.data.blocks", b->num_data_blocks);
-  prop_add_int(props, &np, "rocksdb.filter.size", 0);
+  prop_add_int(props, &np, "rocksdb.filter.size", b->filter_size);
   prop_add_int(props, &np, "rocksdb.format.version", b->o.format_version);
   prop_add_int(props, &np, "rocksdb.fixed.key.length", 0);
   prop_add_int(props, &np, "rocksdb.column.family.id", b->o.cf_id);
@@ -477,6 +532,8 @@ int orc_table_builder_finish(orc_table_builder* b, orc_buf* out) {
   if (b->o.db_host_id && b->o.db_host_id[0])
     prop_add_str(props, &np, "rocksdb.creating.host.identity", b->o.db_host_id);
   prop_add_str(props, &np, "rocksdb.comparator", "leveldb.BytewiseComparator");
+  if (b->o.bloom_millibits_per_key)
+    prop_add_str(props, &np, "rocksdb.filter.policy", "bloomfilter");
   prop_add_str(props, &np, "rocksdb.merge.operator", "nullptr");
   prop_add_str(props, &np, "rocksdb.prefix.extractor.name", "nullptr");
   prop_add_str(props, &np, "rocksdb.property.collectors", "[]");
@@ -517,6 +574,14 @@ int orc_table_builder_finish(orc_table_builder* b, orc_buf* out) {
   {
     oblk mi;
     oblk_init(&mi, 1, 0);
+    if (has_filter) {
+      uint8_t fv[24];
+      int fn2 = orc_varint64_put(fv, filter_off);
+      fn2 += orc_varint64_put(fv + fn2, filter_sz);
+      oblk_add(&mi,
+               (const uint8_t*)"fullfilter.rocksdb.BuiltinBloomFilter", 37,
+               fv, (size_t)fn2, NULL, 0);
+    }
     uint8_t hv[24];
     int hn = orc_varint64_put(hv, props_off);
     hn += orc_varint64_put(hv + hn, props_size);
@@ -563,6 +628,7 @@ void orc_table_builder_delete(orc_table_builder* b) {
   oblk_free(&b->index_seq);
   oblk_free(&b->index_user);
   oblk_free(&b->rangedel);
+  free(b->fhash);
   orc_buf_free(&b->last_key);
   orc_buf_free(&b->scratch);
   orc_buf_free(&b->scratch2);

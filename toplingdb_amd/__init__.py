@@ -66,6 +66,7 @@ class JobDesc(C.Structure):
         ("block_size_deviation", C.c_uint64),
         ("comparator_name", C.c_char_p),
         ("output_table_factory", C.c_uint32),
+        ("bloom_millibits_per_key", C.c_uint32),
         ("flush_kv", C.POINTER(C.c_uint8)),
         ("flush_kv_bytes", C.c_uint64),
         ("flush_offsets", C.POINTER(C.c_uint64)),
@@ -253,6 +254,7 @@ def make_job(runs, output_dir, **kw) -> JobDesc:
     d.block_size_deviation = kw.pop("block_size_deviation", 10)
     d.comparator_name = b"leveldb.BytewiseComparator"
     d.output_table_factory = kw.pop("output_table_factory", 0)
+    d.bloom_millibits_per_key = kw.pop("bloom_millibits_per_key", 0)
     flush_entries = kw.pop("flush_entries", None)
     if flush_entries is not None:
         # sorted [(internal_key, value)] -> the raw flush record blob

@@ -118,6 +118,10 @@ class GpuJob {
   void block_stats(size_t b, uint64_t* mn, uint64_t* mx, uint64_t* tomb);
   int seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* mx,
                  uint64_t* n_tombstones, std::string* err);
+  // per-file bloom filter: FastLocalBloom content bytes
+  // (incl. 5-byte metadata) for survivors [first, first+count)
+  int filter_build(uint64_t first, uint64_t count, uint32_t millibits,
+                   std::string* content, uint64_t* n_added, std::string* err);
 
   // ---- range deletions (envelope subset; oracle/compact.c rd_aggr) ----
   // Fragment list sorted by (normkey, len): frag i covers user keys in

@@ -288,8 +288,18 @@ SstIndexEntry append_block(std::string& out, const TableOpts& o,
 std::string build_tail(const TableOpts& o, const TailStats& st,
                        const std::vector<SstIndexEntry>& handles,
                        const std::vector<std::string>& separators,
-                       bool sep_is_user_key, uint64_t tail_start) {
+                       bool sep_is_user_key, uint64_t tail_start,
+                       const std::string& filter_content,
+                       uint64_t num_filter_entries) {
   std::string out;
+  // full filter block FIRST in the tail (uncompressed;
+  // block_based_table_builder.cc:1488-1526, layout comment :1950-1955)
+  SstIndexEntry filter_handle{0, 0};
+  if (!filter_content.empty()) {
+    filter_handle = append_block(out, o, (const uint8_t*)filter_content.data(),
+                                 filter_content.size(), false);
+    filter_handle.off += tail_start;
+  }
   // index block (restart_interval=1, value delta encoding for fmt>=4)
   BlockBuilder ib(o.index_block_restart_interval, true);
   for (size_t i = 0; i < handles.size(); i++) {
@@ -325,12 +335,12 @@ std::string build_tail(const TableOpts& o, const TailStats& st,
   addi("rocksdb.index.key.is.user.key", sep_is_user_key ? 1 : 0);
   addi("rocksdb.index.value.is.delta.encoded", 1);
   addi("rocksdb.num.entries", st.num_entries);
-  addi("rocksdb.num.filter_entries", 0);
+  addi("rocksdb.num.filter_entries", num_filter_entries);
   addi("rocksdb.deleted.keys", st.num_deletions);
   addi("rocksdb.merge.operands", st.num_merge_operands);
   addi("rocksdb.num.range-deletions", st.num_range_deletions);
   addi("rocksdb.num.data.blocks", st.num_data_blocks);
-  addi("rocksdb.filter.size", 0);
+  addi("rocksdb.filter.size", filter_content.size());
   addi("rocksdb.format.version", o.format_version);
   addi("rocksdb.fixed.key.length", 0);
   addi("rocksdb.column.family.id", o.cf_id);
@@ -344,6 +354,8 @@ std::string build_tail(const TableOpts& o, const TailStats& st,
     adds("rocksdb.creating.session.identity", o.db_session_id);
   if (!o.db_host_id.empty()) adds("rocksdb.creating.host.identity", o.db_host_id);
   adds("rocksdb.comparator", "leveldb.BytewiseComparator");
+  if (o.bloom_millibits_per_key)
+    adds("rocksdb.filter.policy", "bloomfilter");
   adds("rocksdb.merge.operator", "nullptr");
   adds("rocksdb.prefix.extractor.name", "nullptr");
   adds("rocksdb.property.collectors", "[]");
@@ -367,8 +379,14 @@ std::string build_tail(const TableOpts& o, const TailStats& st,
                                             pcontents.size(), false);
   props_handle.off += tail_start;
 
-  // metaindex (restart_interval=1)
+  // metaindex (restart_interval=1; keys sorted: "fullfilter." < "rocksdb.")
   BlockBuilder mi(1, false);
+  if (!filter_content.empty()) {
+    std::string fv;
+    put_varint64(fv, filter_handle.off);
+    put_varint64(fv, filter_handle.size);
+    mi.Add("fullfilter.rocksdb.BuiltinBloomFilter", fv);
+  }
   std::string hv;
   put_varint64(hv, props_handle.off);
   put_varint64(hv, props_handle.size);

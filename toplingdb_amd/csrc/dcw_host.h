@@ -40,6 +40,8 @@ struct TableOpts {
   uint64_t file_creation_time = 0;
   uint64_t oldest_key_time = 0;
   int level_at_creation = 0;
+  // bloom filter (FastLocalBloom): millibits/key, 0 = none
+  uint32_t bloom_millibits_per_key = 0;
 };
 
 // ---- generic delta-encoded block builder (host-side blocks) ----
@@ -110,7 +112,9 @@ struct TailStats {
 std::string build_tail(const TableOpts& o, const TailStats& st,
                        const std::vector<SstIndexEntry>& handles,
                        const std::vector<std::string>& separators,
-                       bool sep_is_user_key, uint64_t tail_start_offset);
+                       bool sep_is_user_key, uint64_t tail_start_offset,
+                       const std::string& filter_content = std::string(),
+                       uint64_t num_filter_entries = 0);
 
 // write one block + 5-byte trailer to out; returns handle
 SstIndexEntry append_block(std::string& out, const TableOpts& o,

@@ -2099,6 +2099,8 @@ struct GpuJob::Impl {
   void *d_kext = nullptr; // general-key side table (48 B/entry full ukeys)
   void *d_sw = nullptr;   // survivor -> original payload index (u32)
   void *d_flush_offs = nullptr; // flush-offload record offsets (u64)
+  void *d_fhash = nullptr;      // per-file filter hashes (u64)
+  void *d_filter = nullptr;     // filter bit array
   void* h_plan = nullptr; // pinned host landing for next+meta
   size_t h_plan_cap = 0;
   bool h_plan_pageable = false;
@@ -2224,7 +2226,7 @@ GpuJob::~GpuJob() {
   F(p->d_dzt_cblob); F(p->d_dzt_bsize); F(p->d_dzt_btype); F(p->d_dzt_csum);
   F(p->d_dzt_kbs); F(p->d_dzt_keyarea); F(p->d_dzt_kidx); F(p->d_dzt_img);
   F(p->d_rd_k0); F(p->d_rd_k1); F(p->d_rd_len); F(p->d_rd_seq);
-  F(p->d_kext); F(p->d_sw); F(p->d_flush_offs);
+  F(p->d_kext); F(p->d_sw); F(p->d_flush_offs); F(p->d_fhash); F(p->d_filter);
   if (p->h_plan) {
     if (p->h_plan_pageable)
       free(p->h_plan);
@@ -3271,6 +3273,96 @@ int GpuJob::seq_minmax(uint64_t first, uint64_t count, uint64_t* mn, uint64_t* m
   *mn = out[0];
   *mx = out[1];
   *n_tombstones = out[2];
+  return 0;
+}
+
+// ------------------------------------------------------------------
+// bloom filter build: XXPH3 over user keys + FastLocalBloom
+// (util/bloom_impl.h:144-223).  Consecutive-equal-hash dedup matches
+// XXPH3FilterBitsBuilder::AddKey exactly.
+// ------------------------------------------------------------------
+__global__ void k_filter_hashes(const uint64_t* __restrict__ s_k0,
+                                const uint64_t* __restrict__ s_k1,
+                                const uint8_t* __restrict__ s_klen,
+                                const uint8_t* __restrict__ kext,
+                                const uint32_t* __restrict__ s_w,
+                                uint64_t first, uint64_t count,
+                                uint64_t* __restrict__ fhash) {
+  for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < count;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    uint64_t e = first + i;
+    uint8_t key[DCW_GKEY_MAX + 8];
+    uint32_t klen = s_klen[e];
+    load_ikey(s_k0[e], s_k1[e], 0 /*tag unused*/, klen, kext, s_w, e, key);
+    fhash[i] = xxph3_64(key, klen - 8);
+  }
+}
+
+__global__ void k_filter_bits(const uint64_t* __restrict__ fhash,
+                              uint64_t count, uint32_t len, int num_probes,
+                              uint32_t* __restrict__ filter_words) {
+  for (uint64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < count;
+       i += (uint64_t)gridDim.x * blockDim.x) {
+    if (i > 0 && fhash[i] == fhash[i - 1]) continue; // AddKey dedup
+    uint32_t h1 = (uint32_t)fhash[i];
+    uint32_t h2 = (uint32_t)(fhash[i] >> 32);
+    uint32_t line = bloom_fastrange32(h1, len >> 6) << 6; // byte offset
+    uint32_t h = h2;
+    for (int p = 0; p < num_probes; p++, h *= 0x9e3779b9u) {
+      int bitpos = h >> (32 - 9); // within the 512-bit cache line
+      // LE u32 word bit (8*byte + bit) == bitpos & 31
+      atomicOr(&filter_words[(line << 3 >> 5) + (bitpos >> 5)],
+               1u << (bitpos & 31));
+    }
+  }
+}
+
+int GpuJob::filter_build(uint64_t first, uint64_t count, uint32_t millibits,
+                         std::string* content, uint64_t* n_added,
+                         std::string* err) {
+  Impl* p = p_;
+  HIPCHK(p->ens(&p->d_fhash, count * 8));
+  p->kbegin("filter_hashes", 32.0 * count);
+  hipLaunchKernelGGL(k_filter_hashes, dim3(grid_for(count)), dim3(256), 0,
+                     p->stream, p->d_sk0, p->d_sk1, p->d_sklen,
+                     general_keys ? (const uint8_t*)p->d_kext : nullptr,
+                     (const uint32_t*)p->d_sw, first, count,
+                     (uint64_t*)p->d_fhash);
+  p->kend();
+  // host counts the deduped entries (the filter length depends on it)
+  std::vector<uint64_t> h(count);
+  HIPCHK(hipMemcpyAsync(h.data(), p->d_fhash, count * 8,
+                        hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  uint64_t n = 0;
+  for (uint64_t i = 0; i < count; i++)
+    if (i == 0 || h[i] != h[i - 1]) n++;
+  *n_added = n;
+  if (n == 0) {
+    content->clear();
+    return 0;
+  }
+  uint64_t lwm = bloom_len_with_metadata(n, millibits);
+  uint32_t len = (uint32_t)(lwm - 5);
+  int probes = bloom_num_probes((int)millibits);
+  HIPCHK(p->ens(&p->d_filter, lwm + 8));
+  HIPCHK(hipMemsetAsync(p->d_filter, 0, lwm + 8, p->stream));
+  p->kbegin("filter_bits", 8.0 * count + len);
+  hipLaunchKernelGGL(k_filter_bits, dim3(grid_for(count)), dim3(256), 0,
+                     p->stream, (const uint64_t*)p->d_fhash, count, len,
+                     probes, (uint32_t*)p->d_filter);
+  p->kend();
+  content->resize(lwm);
+  HIPCHK(hipMemcpyAsync(&(*content)[0], p->d_filter, lwm,
+                        hipMemcpyDeviceToHost, p->stream));
+  HIPCHK(hipStreamSynchronize(p->stream));
+  p->kresolve();
+  // metadata bytes (filter_policy.cc:378-385)
+  (*content)[len] = (char)(int8_t)-1;
+  (*content)[len + 1] = 0;
+  (*content)[len + 2] = (char)probes;
+  (*content)[len + 3] = 0;
+  (*content)[len + 4] = 0;
   return 0;
 }
 

@@ -170,6 +170,7 @@ TableOpts opts_from_desc(const dcw_job_desc* d) {
   o.file_creation_time = d->current_time;
   o.oldest_key_time = 0;
   o.level_at_creation = d->output_level;
+  o.bloom_millibits_per_key = d->bloom_millibits_per_key;
   return o;
 }
 
@@ -1209,6 +1210,14 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     if (cut) cur = cut_entry;
     uint64_t file_count = cur - s;
     if (handles.empty()) break; // nothing left
+    // per-file bloom filter (built on the GPU from the file's survivors)
+    std::string filter_content;
+    uint64_t n_filter = 0;
+    if (o.bloom_millibits_per_key && file_count > 0) {
+      if (job.filter_build(file_first, file_count, o.bloom_millibits_per_key,
+                           &filter_content, &n_filter, &err) != 0)
+        return fail(res, 37, err);
+    }
     uint64_t tw1 = now_usec();
     // hand the whole per-file tail (separators, stats, meta tail, file
     // write) to a background thread — the GPU starts the next file now.
@@ -1234,6 +1243,8 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       std::vector<BKey> first_keys, last_keys;
       uint64_t file_first, file_count, mn_seq, mx_seq, n_tomb;
       std::string path;
+      std::string filter_content;
+      uint64_t n_filter = 0;
     };
     auto tj = std::make_shared<TailJob>();
     tj->o = o;
@@ -1251,6 +1262,8 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     tj->mx_seq = mx_seq;
     tj->n_tomb = n_tomb;
     tj->path = path;
+    tj->filter_content = std::move(filter_content);
+    tj->n_filter = n_filter;
     const uint8_t* klen_p = klen.data();
     const uint32_t* vlen_p = vlen.data();
     uint64_t* out_bytes_p = &total_out_bytes;
@@ -1285,7 +1298,8 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
       st.num_deletions = tj->n_tomb;
       uint64_t tail_start = tj->image.len;
       std::string tail =
-          build_tail(tj->o, st, tj->handles, seps, !sep_key_plus_seq, tail_start);
+          build_tail(tj->o, st, tj->handles, seps, !sep_key_plus_seq,
+                     tail_start, tj->filter_content, tj->n_filter);
       tj->image.append(tail.data(), tail.size());
       // parallel segmented write: tmpfs writes are page-clear + memcpy
       // bound per thread, and the last file's write sits on the job's
