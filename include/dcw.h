@@ -42,6 +42,10 @@ enum {
 enum {
   DCW_COMPRESSION_NONE = 0x0,
   DCW_COMPRESSION_SNAPPY = 0x1,
+  DCW_COMPRESSION_ZSTD = 0x7, /* INPUT blocks only: decoded host-side at
+                                 load (util/compression.h:1332 framing:
+                                 varint32 decompressed size + zstd frame);
+                                 output compression stays none/snappy */
 };
 
 /* ChecksumType (include/rocksdb/table.h; default kXXH3, table.h:257) */

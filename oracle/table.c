@@ -247,12 +247,32 @@ void orc_table_builder_add_tombstone(orc_table_builder* b,
 
 /* write block contents (+5B trailer) at current offset; handle out.
  * try_compress: data blocks and index blocks (enable_index_compression=true) */
+/* minimal libzstd ABI (no dev header in this image; runtime lib present;
+ * the worker only DECODES zstd — these encode calls exist so tests can
+ * build zstd inputs) */
+extern size_t ZSTD_compressBound(size_t);
+extern size_t ZSTD_compress(void*, size_t, const void*, size_t, int);
+extern size_t ZSTD_decompress(void*, size_t, const void*, size_t);
+extern unsigned ZSTD_isError(size_t);
+
 static void tb_write_block(orc_table_builder* b, const uint8_t* data, size_t n,
                            int try_compress, uint64_t* hoff, uint64_t* hsize) {
   uint8_t type = DCW_COMPRESSION_NONE;
   const uint8_t* out = data;
   size_t outn = n;
-  if (try_compress && b->o.compression == DCW_COMPRESSION_SNAPPY) {
+  if (try_compress && b->o.compression == DCW_COMPRESSION_ZSTD) {
+    /* ZSTD_Compress framing (util/compression.h:1332-1377): varint32
+     * decompressed size + zstd frame, level 3 (kDefaultCompressionLevel) */
+    buf_reserve(&b->comp, ZSTD_compressBound(n) + 8);
+    int hn = orc_varint32_put(b->comp.data, (uint32_t)n);
+    size_t cn = ZSTD_compress(b->comp.data + hn, ZSTD_compressBound(n), data,
+                              n, 3);
+    if (!ZSTD_isError(cn) && hn + cn <= ((uint64_t)896 * n) >> 10) {
+      type = DCW_COMPRESSION_ZSTD;
+      out = b->comp.data;
+      outn = hn + cn;
+    }
+  } else if (try_compress && b->o.compression == DCW_COMPRESSION_SNAPPY) {
     buf_reserve(&b->comp, orc_snappy_max_compressed(n));
     size_t cn = orc_snappy_compress(data, n, b->comp.data);
     if (cn <= ((uint64_t)896 * n) >> 10) { /* GoodCompressionRatio, default 896/KiB */
@@ -692,6 +712,24 @@ static uint8_t* read_block(const orc_table_reader* r, uint64_t off, uint64_t sz,
     if (orc_snappy_uncompress(p, sz, out, un) != un) {
       free(out);
       snprintf(err, errcap, "snappy corruption");
+      return NULL;
+    }
+    *out_n = un;
+    return out;
+  }
+  if (type == DCW_COMPRESSION_ZSTD) {
+    /* ZSTD_Uncompress framing (util/compression.h): varint32 size + frame */
+    uint32_t un;
+    int hn = orc_varint32_get(p, p + (sz < 5 ? sz : 5), &un);
+    if (hn < 0) {
+      snprintf(err, errcap, "bad zstd preamble");
+      return NULL;
+    }
+    uint8_t* out = (uint8_t*)malloc(un ? un : 1);
+    size_t got = ZSTD_decompress(out, un, p + hn, sz - hn);
+    if (ZSTD_isError(got) || got != un) {
+      free(out);
+      snprintf(err, errcap, "zstd corruption");
       return NULL;
     }
     *out_n = un;

@@ -57,6 +57,9 @@ class GpuJob {
   // blob copy when begin/chunk ran)
   int stage_begin(size_t blob_size, std::string* err);
   int stage_chunk(uint64_t off, const void* src, size_t n, std::string* err);
+  // abandon split staging (e.g. the blob was rewritten for zstd inputs);
+  // the following stage() performs the full copy
+  void stage_cancel();
   int stage_adopt(const StagedInput& s, std::string* err);
   // move this job's staged buffers out into `s` (for dcw_stage_inputs)
   int stage_release(StagedInput* s, std::string* err);

@@ -2267,6 +2267,15 @@ int GpuJob::stage_begin(size_t blob_size, std::string* err) {
   p->staged_split = true;
   return 0;
 }
+void GpuJob::stage_cancel() {
+  Impl* p = p_;
+  if (p->staged_split) {
+    (void)hipStreamSynchronize(p->stream); // drain issued chunk copies
+    (void)hipEventDestroy(p->stage_t0);
+    p->staged_split = false;
+  }
+}
+
 int GpuJob::stage_chunk(uint64_t off, const void* src, size_t n,
                         std::string* err) {
   Impl* p = p_;

@@ -244,6 +244,11 @@ struct RawBuf {
     len += n;
   }
   void release() { p = nullptr; len = cap = 0; }
+  void swap(RawBuf& o) {
+    std::swap(p, o.p);
+    std::swap(len, o.len);
+    std::swap(cap, o.cap);
+  }
   ~RawBuf() {
     if (p) g_pin_pool.release(p, cap);
   }
@@ -259,7 +264,15 @@ struct LoadedInputs {
   GpuInputs gi;
   uint64_t in_bytes = 0;
   std::vector<SstTombstone> tombstones; // all inputs' range deletions
+  bool host_decoded = false; // zstd inputs were rewritten (see below)
 };
+
+// minimal libzstd ABI (runtime lib only in this image); the worker only
+// DECODES zstd inputs — SURVEY §8f-1's input side.  The GPU codec stays
+// snappy; zstd blocks are decoded host-side at load and handed to the
+// device as raw blocks with recomputed trailers.
+extern "C" size_t ZSTD_decompress(void*, size_t, const void*, size_t);
+extern "C" unsigned ZSTD_isError(size_t);
 
 int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err,
                 GpuJob* job) {
@@ -365,9 +378,69 @@ int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err,
     }
     L->gi.run_block_begin.push_back((uint32_t)L->gi.blocks.size());
   }
+  // zstd input blocks (type byte 0x7 in the trailer): verify + decode on
+  // the host, rebuild the staged blob with raw blocks + recomputed
+  // trailers (decompression output is unique, so parity is unaffected;
+  // the reference reads its inputs through the same libzstd,
+  // util/compression.h ZSTD_Uncompress)
+  bool any_zstd = false;
+  uint32_t ct = cstype == 0xffffffff ? 4 : cstype;
+  for (auto& blk : L->gi.blocks)
+    if (L->blob.p[blk.off + blk.size] == 7) {
+      any_zstd = true;
+      break;
+    }
+  if (any_zstd) {
+    if (job) { // already-streamed chunks are superseded by a full restage
+      job->stage_cancel();
+    }
+    RawBuf nb;
+    uint64_t est = L->blob.len * 2 + (16u << 20);
+    nb.reserve(est);
+    for (auto& blk : L->gi.blocks) {
+      const uint8_t* body = L->blob.p + blk.off;
+      uint8_t type = body[blk.size];
+      uint64_t noff = nb.len;
+      if (type == 7) {
+        uint32_t stored;
+        memcpy(&stored, body + blk.size + 1, 4);
+        if (ct != 0 &&
+            stored != block_checksum(ct, &g_crc, body, blk.size, type)) {
+          *err = "zstd input block checksum mismatch";
+          return -1;
+        }
+        uint32_t un;
+        int hn = varint32_get(body, body + (blk.size < 5 ? blk.size : 5), &un);
+        if (hn < 0) {
+          *err = "bad zstd preamble in input block";
+          return -1;
+        }
+        nb.reserve(nb.len + un + 8);
+        size_t got = ZSTD_decompress(nb.p + nb.len, un, body + hn,
+                                     blk.size - hn);
+        if (ZSTD_isError(got) || got != un) {
+          *err = "zstd input block corrupt";
+          return -1;
+        }
+        nb.len += un;
+        uint8_t tr[5];
+        tr[0] = 0;
+        uint32_t cs = block_checksum(ct, &g_crc, nb.p + noff, un, 0);
+        memcpy(tr + 1, &cs, 4);
+        nb.append(tr, 5);
+        blk.off = noff;
+        blk.size = un;
+      } else {
+        nb.append(body, blk.size + 5);
+        blk.off = noff;
+      }
+    }
+    L->blob.swap(nb); // nb's dtor returns the old blob to the pool
+    L->host_decoded = true;
+  }
   L->gi.blob = L->blob.p;
   L->gi.blob_size = L->blob.len;
-  L->gi.checksum_type = cstype == 0xffffffff ? 4 : cstype;
+  L->gi.checksum_type = ct;
   return 0;
 }
 
