@@ -99,6 +99,9 @@ std::string BlockBuilder::Finish() {
 }
 
 // ---------------- input SST parsing ----------------
+extern "C" size_t ZSTD_decompress(void*, size_t, const void*, size_t);
+extern "C" unsigned ZSTD_isError(size_t);
+
 ParsedSst parse_sst(const uint8_t* data, size_t size) {
   ParsedSst r;
   if (size < 53) {
@@ -142,6 +145,18 @@ ParsedSst parse_sst(const uint8_t* data, size_t size) {
     dec.resize(ul);
     if (snappy_uncompress(ib, idx_sz, (uint8_t*)dec.data(), ul) != ul) {
       r.error = "index snappy corrupt";
+      return r;
+    }
+    ib = (const uint8_t*)dec.data();
+    idx_sz = ul;
+  } else if (type == 7 /*zstd: host decode, same framing as data blocks*/) {
+    uint32_t ul;
+    int hn = varint32_get(ib, ib + (idx_sz < 5 ? idx_sz : 5), &ul);
+    if (hn < 0) { r.error = "index zstd preamble"; return r; }
+    dec.resize(ul);
+    size_t got = ZSTD_decompress((void*)dec.data(), ul, ib + hn, idx_sz - hn);
+    if (ZSTD_isError(got) || got != ul) {
+      r.error = "index zstd corrupt";
       return r;
     }
     ib = (const uint8_t*)dec.data();

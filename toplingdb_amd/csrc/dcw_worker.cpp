@@ -766,7 +766,7 @@ using namespace dcw;
 
 extern "C" {
 
-const char* dcw_version(void) { return "toplingdb_amd dcompact worker r1 (gfx950)"; }
+const char* dcw_version(void) { return "toplingdb_amd dcompact worker r2 (gfx950)"; }
 
 static void dcw_segv_handler(int sig, siginfo_t* si, void*) {
   char buf[128];
