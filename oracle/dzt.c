@@ -188,8 +188,9 @@ int orc_dzt_builder_finish(orc_dzt_builder* b, orc_buf* out) {
 
   /* ---- value blocks ---- */
   uint8_t* ublock = (uint8_t*)malloc(DZT_VBLK_ULEN_MAX + 65536);
-  uint8_t* cblock = (uint8_t*)malloc(orc_snappy_max_compressed(
-                                         DZT_VBLK_ULEN_MAX + 65536) + 16);
+  /* dict copies may use the 5-byte form: worst ~1.4x block size */
+  uint8_t* cblock = (uint8_t*)malloc(DZT_VBLK_ULEN_MAX +
+                                     DZT_VBLK_ULEN_MAX / 2 + 256);
   uint32_t* voff_entry = (uint32_t*)malloc(n ? n * 4 : 4);
   size_t i = 0;
   while (i < n) {

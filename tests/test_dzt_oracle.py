@@ -33,6 +33,23 @@ def test_dict_codec_roundtrip():
     assert len(with_dict) < len(without)
 
 
+def test_dict_codec_expansion_worst_case():
+    # adversarial: alternating 4-byte dict matches (5-byte copy form at
+    # >64K offsets) + 1-byte literals can EXPAND the block to ~1.4x its
+    # size — the regression that overflowed the old 32+n+n/6 slot bound
+    rnd = random.Random(99)
+    dictionary = bytes(rnd.getrandbits(8) for _ in range(48 * 1024))
+    payload = bytearray()
+    i = 0
+    while len(payload) < 12000:
+        payload += dictionary[i:i + 4] + bytes([rnd.getrandbits(8)])
+        i = (i + 97 * 4) % (len(dictionary) - 4)
+    payload = bytes(payload)
+    enc = oracle.snappy_compress_dict(dictionary, payload)
+    assert oracle.snappy_uncompress_dict(dictionary, enc) == payload
+    assert len(enc) > len(payload) * 1.15  # would overflow the old bound
+
+
 def test_dict_codec_large_offsets():
     # dictionary matches reach past 64 KiB -> 4-byte-offset copy form
     rnd = random.Random(11)

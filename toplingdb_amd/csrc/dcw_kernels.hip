@@ -3436,6 +3436,7 @@ __global__ void k_dzt_gather(const GpuJob::DztVBlock* __restrict__ vbs,
 
 // dict-snappy per value block, wave-parallel (spec v4 segmentation; the
 // segment encoder is the SAME DCW_HD function the oracle restates)
+#define DZT_FRAG_MAX 400 // >= ceil(1.4 * 256-byte segment) + headers
 __global__ __launch_bounds__(256) void k_dzt_compress(
     const GpuJob::DztVBlock* __restrict__ vbs, uint32_t nvb,
     const uint8_t* __restrict__ vstage, const uint8_t* __restrict__ dict,
@@ -3469,14 +3470,16 @@ __global__ __launch_bounds__(256) void k_dzt_compress(
     wave_lds_sync();
     uint32_t seg = (uint32_t)snap_segment_size(n);
     uint32_t s0 = lane * seg;
-    uint8_t frag[SNAP_FRAG_MAX];
+    // dict copies may use the 5-byte (4-byte-offset) form: worst case is
+    // ~1.4 bytes out per segment byte (alternating 1-literal + 4-byte
+    // match), so the fragment buffer must exceed SNAP_FRAG_MAX
+    uint8_t frag[DZT_FRAG_MAX];
     uint32_t fl = 0;
     if (s0 < n) {
       uint32_t s1 = s0 + seg < n ? s0 + seg : n;
       uint8_t* e = snap_encode_segment_dict(dict, D, gin, D + s0, D + s1, tab,
                                             frag);
       fl = (uint32_t)(e - frag);
-      if (fl > SNAP_FRAG_MAX) set_err(err_flag, DE_SNAPPY);
     }
     uint32_t inc = fl;
     for (int sh = 1; sh < WAVE; sh <<= 1) {
@@ -3670,7 +3673,10 @@ int GpuJob::dzt_values(const std::vector<DztVBlock>& vbs,
                      (const uint32_t*)p->d_dzt_voff, ent_base, p->d_svoff,
                      p->d_svlen, p->d_ublob, (uint8_t*)p->d_dzt_vstage);
   p->kend();
-  p->dzt_ccap = (snappy_max_compressed(SNAP_MAX_UNC) + 63) & ~(uint64_t)63;
+  // worst-case dict encode is ~1.4x the block (5-byte copy form), larger
+  // than snappy_max_compressed's 32+n+n/6 bound for plain snappy
+  p->dzt_ccap =
+      ((uint64_t)SNAP_MAX_UNC + SNAP_MAX_UNC / 2 + 128 + 63) & ~(uint64_t)63;
   HIPCHK(p->ens(&p->d_dzt_bsize, nvb * 4));
   HIPCHK(p->ens(&p->d_dzt_btype, nvb));
   HIPCHK(p->ens(&p->d_dzt_csum, nvb * 4));
