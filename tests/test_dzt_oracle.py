@@ -33,10 +33,13 @@ def test_dict_codec_roundtrip():
     assert len(with_dict) < len(without)
 
 
-def test_dict_codec_expansion_worst_case():
-    # adversarial: alternating 4-byte dict matches (5-byte copy form at
-    # >64K offsets) + 1-byte literals can EXPAND the block to ~1.4x its
-    # size — the regression that overflowed the old 32+n+n/6 slot bound
+def test_dict_codec_adversarial_match_mix():
+    # alternating 4-byte dict matches + 1-byte literals: the densest
+    # copy/literal interleaving the encoder can emit.  (With the DZT
+    # constants — dict <= 48 KiB, blocks <= 16 KiB — the max virtual
+    # offset is 65536, so the 5-byte copy form is reachable only at one
+    # corner; the device/oracle output slots are nevertheless sized for
+    # the 1.4x worst case defensively.)
     rnd = random.Random(99)
     dictionary = bytes(rnd.getrandbits(8) for _ in range(48 * 1024))
     payload = bytearray()
@@ -47,7 +50,6 @@ def test_dict_codec_expansion_worst_case():
     payload = bytes(payload)
     enc = oracle.snappy_compress_dict(dictionary, payload)
     assert oracle.snappy_uncompress_dict(dictionary, enc) == payload
-    assert len(enc) > len(payload) * 1.15  # would overflow the old bound
 
 
 def test_dict_codec_large_offsets():
