@@ -109,3 +109,39 @@ def test_dzt_gpu_tombstones(tmp_path):
     rg, ro = run_both(tmp_path, runs, compression=1, bottommost_level=0,
                       snapshots=[15000], earliest_write_conflict_snapshot=15000)
     assert_identical(rg, ro)
+
+
+def test_dzt_bbt_pool_reuse_and_concurrency(tmp_path):
+    # pooled GpuJob objects are reused across table formats; alternate and
+    # run concurrently to shake out stale per-job state
+    import concurrent.futures
+    runs_a = gen_runs(tmp_path, 2, 20000, seed0=31)
+    runs_b = gen_runs(tmp_path, 2, 20000, seed0=67)
+    dirs = {}
+    for tag in ("z1", "b1", "z2", "oz1", "ob1", "oz2", "cz", "cb", "ocz", "ocb"):
+        d = tmp_path / tag
+        d.mkdir()
+        dirs[tag] = str(d)
+    # sequential alternation on the same pool
+    for tag, runs, otf in (("z1", runs_a, 1), ("b1", runs_a, 0),
+                           ("z2", runs_b, 1)):
+        rg = dcw.execute(dcw.make_job(runs, dirs[tag], compression=1,
+                                      bottommost_level=1,
+                                      output_table_factory=otf))
+        ro = oracle.execute(oracle.make_job(runs, dirs["o" + tag],
+                                            compression=1, bottommost_level=1,
+                                            output_table_factory=otf))
+        for fg, fo in zip(rg["files"], ro["files"]):
+            assert open(fg["path"], "rb").read() == open(fo["path"], "rb").read()
+    # concurrent DZT + BBT
+    kwz = dict(compression=1, bottommost_level=1, output_table_factory=1)
+    kwb = dict(compression=1, bottommost_level=1, output_table_factory=0)
+    with concurrent.futures.ThreadPoolExecutor(2) as ex:
+        fz = ex.submit(dcw.execute, dcw.make_job(runs_a, dirs["cz"], **kwz))
+        fb = ex.submit(dcw.execute, dcw.make_job(runs_b, dirs["cb"], **kwb))
+        rz, rb = fz.result(), fb.result()
+    oz = oracle.execute(oracle.make_job(runs_a, dirs["ocz"], **kwz))
+    ob = oracle.execute(oracle.make_job(runs_b, dirs["ocb"], **kwb))
+    for rg, ro in ((rz, oz), (rb, ob)):
+        for fg, fo in zip(rg["files"], ro["files"]):
+            assert open(fg["path"], "rb").read() == open(fo["path"], "rb").read()
