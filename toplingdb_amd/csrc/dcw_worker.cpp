@@ -544,7 +544,14 @@ int dzt_finish(const dcw_job_desc* d, dcw_job_result* res, GpuJob& job,
   auto put32 = [](std::string& s, uint32_t v) { s.append((const char*)&v, 4); };
   auto put64 = [](std::string& s, uint64_t v) { s.append((const char*)&v, 8); };
 
+  uint64_t dzt_us[8] = {0}; // sample,values,keyarea,pack,kindex,props,write,meta
   for (auto& fp : files) {
+    uint64_t tph = now_usec();
+    auto mark = [&](int slot) {
+      uint64_t now = now_usec();
+      dzt_us[slot] += now - tph;
+      tph = now;
+    };
     TableOpts o = base;
     o.orig_file_number = next_file_number++;
     // ---- dict (sampling rule identical to oracle/dzt.c) ----
@@ -566,12 +573,14 @@ int dzt_finish(const dcw_job_desc* d, dcw_job_result* res, GpuJob& job,
       dict.append((const char*)samp.data() + si * 256, take);
     }
     // ---- value blocks on the GPU ----
+    mark(0);
     std::vector<uint32_t> csize, csum;
     std::vector<uint8_t> btype;
     if (job.dzt_values(fp.vbs, fp.voff, fp.first, (const uint8_t*)dict.data(),
                        (uint32_t)dict.size(), o, &csize, &btype, &csum,
                        &err) != 0)
       return fail(res, 32, err);
+    mark(1);
     std::vector<uint64_t> outoff(fp.vbs.size());
     uint64_t vtotal = 0;
     for (size_t b = 0; b < fp.vbs.size(); b++) {
@@ -594,10 +603,12 @@ int dzt_finish(const dcw_job_desc* d, dcw_job_result* res, GpuJob& job,
     if (job.dzt_keyarea(fp.kbs, fp.voff, fp.first, fp.key_area_size, image.p,
                         first_ikeys.data(), &err) != 0)
       return fail(res, 33, err);
+    mark(2);
     memcpy(image.p + dict_off, dict.data(), dict.size());
     if (job.dzt_pack_values(fp.vbs, outoff, vtotal, image.p + value_off,
                             &err) != 0)
       return fail(res, 34, err);
+    mark(3);
     // ---- key index ----
     std::string kindex;
     kindex.reserve(kindex_size);
@@ -608,6 +619,7 @@ int dzt_finish(const dcw_job_desc* d, dcw_job_result* res, GpuJob& job,
       put32(kindex, (uint32_t)(kend - fp.kbs[kb].koff));
       put64(kindex, fp.kbs[kb].first - fp.first);
     }
+    mark(4);
     memcpy(image.p + kindex_off, kindex.data(), kindex.size());
     // ---- value index ----
     std::string vindex;
@@ -666,6 +678,7 @@ int dzt_finish(const dcw_job_desc* d, dcw_job_result* res, GpuJob& job,
     put64(footer, 0x313050495A574344ull); // "DCWZIP01"
     image.append(props.data(), props.size());
     image.append(footer.data(), footer.size());
+    mark(5);
     // ---- write + meta ----
     char path[600];
     snprintf(path, sizeof(path), "%s/%06" PRIu64 ".sst", d->output_dir,
@@ -719,6 +732,7 @@ int dzt_finish(const dcw_job_desc* d, dcw_job_result* res, GpuJob& job,
       close(fd);
       if (!ok) return fail(res, 35, std::string("short write ") + path);
     }
+    mark(6);
     dcw_output_file of;
     memset(&of, 0, sizeof(of));
     snprintf(of.path, sizeof(of.path), "%s", path);
@@ -740,6 +754,15 @@ int dzt_finish(const dcw_job_desc* d, dcw_job_result* res, GpuJob& job,
     out_files.push_back(of);
     total_out_bytes += image.len;
     total_out_entries += fp.count;
+    mark(7);
+  }
+  if (getenv("DCW_PHASE_DEBUG")) {
+    static const char* nm[8] = {"plan+dict", "values", "keyarea", "pack",
+                                "kindex",    "props",  "write",   "meta"};
+    fprintf(stderr, "[dzt]");
+    for (int i = 0; i < 8; i++)
+      fprintf(stderr, " %s=%.1fms", nm[i], dzt_us[i] / 1000.0);
+    fprintf(stderr, "\n");
   }
 
   res->num_files = (uint32_t)out_files.size();
