@@ -544,6 +544,7 @@ int dzt_finish(const dcw_job_desc* d, dcw_job_result* res, GpuJob& job,
   auto put32 = [](std::string& s, uint32_t v) { s.append((const char*)&v, 4); };
   auto put64 = [](std::string& s, uint64_t v) { s.append((const char*)&v, 8); };
 
+  std::vector<std::future<bool>> dzt_writes; // per-file background writes
   uint64_t dzt_us[8] = {0}; // sample,values,keyarea,pack,kindex,props,write,meta
   for (auto& fp : files) {
     uint64_t tph = now_usec();
@@ -599,7 +600,8 @@ int dzt_finish(const dcw_job_desc* d, dcw_job_result* res, GpuJob& job,
     // ---- GPU emits straight into the host image ----
     RawBuf image;
     std::vector<uint8_t> first_ikeys((uint64_t)nkb * IK);
-    image.resize_uninit(props_off); // props+footer appended after
+    image.reserve(props_off + (64u << 10)); // props+footer appended after
+    image.resize_uninit(props_off);
     if (job.dzt_keyarea(fp.kbs, fp.voff, fp.first, fp.key_area_size, image.p,
                         first_ikeys.data(), &err) != 0)
       return fail(res, 33, err);
@@ -683,61 +685,67 @@ int dzt_finish(const dcw_job_desc* d, dcw_job_result* res, GpuJob& job,
     char path[600];
     snprintf(path, sizeof(path), "%s/%06" PRIu64 ".sst", d->output_dir,
              o.orig_file_number);
-    // segmented parallel write (tmpfs page-clear+memcpy bound per thread;
-    // DZT files are large — up to target_file_size — so this was the
-    // dominant cost of the configs[3] job)
+    // segmented parallel write on a background task, overlapping the next
+    // file's GPU phases (the write was the dominant configs[3] cost)
+    const uint64_t image_len = image.len; // meta reads it after the swap
     {
-      int fd = open(path, O_WRONLY | O_CREAT | O_TRUNC, 0644);
-      if (fd < 0) return fail(res, 35, std::string("cannot write ") + path);
-      size_t len = image.len;
-      if (ftruncate(fd, (off_t)len) != 0) {
-        close(fd);
-        return fail(res, 35, std::string("ftruncate failed ") + path);
-      }
-      const size_t kSeg = 32u << 20;
-      size_t nseg = (len + kSeg - 1) / kSeg;
-      if (nseg > 8) nseg = 8;
-      if (nseg == 0) nseg = 1;
-      size_t seg = (len + nseg - 1) / nseg;
-      std::vector<std::future<bool>> segw;
-      const uint8_t* src = image.p;
-      for (size_t si = 1; si < nseg; si++) {
-        size_t off = si * seg;
-        size_t cnt = off < len ? std::min(seg, len - off) : 0;
-        segw.emplace_back(std::async(std::launch::async, [fd, src, off, cnt]() {
+      auto img = std::make_shared<RawBuf>();
+      image.swap(*img);
+      std::string pth(path);
+      dzt_writes.emplace_back(std::async(std::launch::async, [img, pth]() {
+        int fd = open(pth.c_str(), O_WRONLY | O_CREAT | O_TRUNC, 0644);
+        if (fd < 0) return false;
+        size_t len = img->len;
+        if (ftruncate(fd, (off_t)len) != 0) {
+          close(fd);
+          return false;
+        }
+        const size_t kSeg = 32u << 20;
+        size_t nseg = (len + kSeg - 1) / kSeg;
+        if (nseg > 8) nseg = 8;
+        if (nseg == 0) nseg = 1;
+        size_t seg = (len + nseg - 1) / nseg;
+        std::vector<std::future<bool>> segw;
+        const uint8_t* src = img->p;
+        for (size_t si = 1; si < nseg; si++) {
+          size_t off = si * seg;
+          size_t cnt = off < len ? std::min(seg, len - off) : 0;
+          segw.emplace_back(
+              std::async(std::launch::async, [fd, src, off, cnt]() {
+                size_t done = 0;
+                while (done < cnt) {
+                  ssize_t w = pwrite(fd, src + off + done, cnt - done,
+                                     (off_t)(off + done));
+                  if (w <= 0) return false;
+                  done += (size_t)w;
+                }
+                return true;
+              }));
+        }
+        bool ok = true;
+        {
+          size_t cnt = std::min(seg, len);
           size_t done = 0;
           while (done < cnt) {
-            ssize_t w = pwrite(fd, src + off + done, cnt - done,
-                               (off_t)(off + done));
-            if (w <= 0) return false;
+            ssize_t w = pwrite(fd, src + done, cnt - done, (off_t)done);
+            if (w <= 0) {
+              ok = false;
+              break;
+            }
             done += (size_t)w;
           }
-          return true;
-        }));
-      }
-      bool ok = true;
-      {
-        size_t cnt = std::min(seg, len);
-        size_t done = 0;
-        while (done < cnt) {
-          ssize_t w = pwrite(fd, src + done, cnt - done, (off_t)done);
-          if (w <= 0) {
-            ok = false;
-            break;
-          }
-          done += (size_t)w;
         }
-      }
-      for (auto& fw : segw) ok = fw.get() && ok;
-      close(fd);
-      if (!ok) return fail(res, 35, std::string("short write ") + path);
+        for (auto& fw : segw) ok = fw.get() && ok;
+        close(fd);
+        return ok;
+      }));
     }
     mark(6);
     dcw_output_file of;
     memset(&of, 0, sizeof(of));
     snprintf(of.path, sizeof(of.path), "%s", path);
     of.file_number = o.orig_file_number;
-    of.file_size = image.len;
+    of.file_size = image_len;
     memcpy(of.smallest_ikey, first_ikeys.data(), IK);
     of.smallest_len = IK;
     std::vector<std::pair<std::string, std::string>> lastkv;
@@ -752,10 +760,12 @@ int dzt_finish(const dcw_job_desc* d, dcw_job_result* res, GpuJob& job,
     of.largest_seqno = mx;
     of.num_entries = fp.count;
     out_files.push_back(of);
-    total_out_bytes += image.len;
+    total_out_bytes += image_len;
     total_out_entries += fp.count;
     mark(7);
   }
+  for (auto& w : dzt_writes)
+    if (!w.get()) return fail(res, 35, "DZT output file write failed");
   if (getenv("DCW_PHASE_DEBUG")) {
     static const char* nm[8] = {"plan+dict", "values", "keyarea", "pack",
                                 "kindex",    "props",  "write",   "meta"};
