@@ -689,6 +689,14 @@ int dzt_finish(const dcw_job_desc* d, dcw_job_result* res, GpuJob& job,
     // file's GPU phases (the write was the dominant configs[3] cost)
     const uint64_t image_len = image.len; // meta reads it after the swap
     {
+      // bound in-flight images: each holds a pinned buffer; unbounded
+      // overlap forces fresh hipHostMalloc pins that cost more than the
+      // write overlap saves
+      while (dzt_writes.size() >= 2) {
+        if (!dzt_writes.front().get())
+          return fail(res, 35, "DZT output file write failed");
+        dzt_writes.erase(dzt_writes.begin());
+      }
       auto img = std::make_shared<RawBuf>();
       image.swap(*img);
       std::string pth(path);
