@@ -136,5 +136,95 @@ def main():
     dcw.shutdown()
 
 
+def run_r2_feature_fuzz(seed, n_cases=20):
+    """Round-2 feature-mix fuzz: random jobs drawing from {BBT, DZT} x
+    {no filter, bloom} x {uniform16, mixed<=48 keys} x {snappy, zstd
+    inputs} x {plain, tombstoned (bottommost envelope)} x {SST runs,
+    flush stream}; every output file bit-compared GPU vs oracle."""
+    import random
+    rnd = random.Random(seed)
+    for c in range(n_cases):
+        tmp = tempfile.mkdtemp(prefix="sweepf_", dir="/dev/shm")
+        mixed_keys = rnd.random() < 0.35
+        flush = rnd.random() < 0.25
+        dzt = (not mixed_keys) and rnd.random() < 0.35
+        bloom = rnd.choice([0, 10000, 15500]) if not dzt else 0
+        zstd_in = (not flush) and rnd.random() < 0.3
+        tomb = ((not mixed_keys) and (not flush) and rnd.random() < 0.3)
+        bottom = 1 if tomb else rnd.choice([0, 1])
+        snaps = [] if tomb else (sorted(rnd.sample(range(1, 50000), 2))
+                                 if rnd.random() < 0.4 else [])
+        n_runs = rnd.choice([1, 2, 3])
+        nent = rnd.choice([3000, 12000, 30000])
+        tag = "fuzz%d dzt=%d bloom=%d mk=%d z=%d t=%d fl=%d b=%d" % (
+            c, dzt, bloom, mixed_keys, zstd_in, tomb, flush, bottom)
+        kw = dict(compression=rnd.choice([0, 1]),
+                  target_file_size=rnd.choice([1 << 20, 64 << 20]),
+                  snapshots=snaps, bottommost_level=bottom,
+                  output_table_factory=1 if dzt else 0,
+                  bloom_millibits_per_key=bloom)
+        seq = 1
+        def gen_kvs(n):
+            nonlocal seq
+            out = []
+            for _ in range(n):
+                if mixed_keys:
+                    ln = rnd.choice([6, 16, 24, 44])
+                    uk = bytes(rnd.getrandbits(8) for _ in range(ln))
+                else:
+                    uk = b"k%015d" % rnd.randrange(n * 4)
+                t = 0 if rnd.random() < 0.15 else 1
+                out.append((uk, seq, t, b"" if t == 0 else b"v%d" % seq))
+                seq += 1
+            out.sort(key=lambda e: (e[0], -e[1]))
+            dedup = []
+            for e in out:
+                if dedup and dedup[-1][0] == e[0] and dedup[-1][1] == e[1]:
+                    continue
+                dedup.append(e)
+            return [(oracle.make_ikey(k, s2, t2), v) for k, s2, t2, v in dedup]
+        og = os.path.join(tmp, "g")
+        oo = os.path.join(tmp, "o")
+        os.makedirs(og)
+        os.makedirs(oo)
+        if flush:
+            es = gen_kvs(nent)
+            rg = dcw.execute(dcw.make_job([], og, flush_entries=es, **kw))
+            ro = oracle.execute(oracle.make_job([], oo, flush_entries=es, **kw))
+        else:
+            runs = []
+            for r in range(n_runs):
+                es = gen_kvs(nent)
+                ts = []
+                if tomb:
+                    for _ in range(rnd.randrange(1, 4)):
+                        a = rnd.randrange(nent * 4)
+                        b2 = a + rnd.randrange(1, nent)
+                        ts.append((b"k%015d" % a, b"k%015d" % b2,
+                                   seq + rnd.randrange(100)))
+                opts = oracle.default_table_opts(
+                    compression=7 if zstd_in else rnd.choice([0, 1]))
+                p = os.path.join(tmp, "in%d.sst" % r)
+                with open(p, "wb") as f:
+                    f.write(oracle.build_sst(es, opts, tombstones=ts))
+                runs.append([p])
+            rg = dcw.execute(dcw.make_job(runs, og, **kw))
+            ro = oracle.execute(oracle.make_job(runs, oo, **kw))
+        assert rg["out_entries"] == ro["out_entries"], tag
+        assert len(rg["files"]) == len(ro["files"]), tag
+        for fg, fo in zip(rg["files"], ro["files"]):
+            with open(fg["path"], "rb") as a, open(fo["path"], "rb") as b:
+                assert a.read() == b.read(), (tag, fg["path"])
+        print("OK %-48s files=%d entries=%d" %
+              (tag, len(rg["files"]), rg["out_entries"]), flush=True)
+
+
 if __name__ == "__main__":
-    main()
+    if len(sys.argv) > 1 and sys.argv[1] == "r2fuzz":
+        dcw.init(0)
+        run_r2_feature_fuzz(int(sys.argv[2]) if len(sys.argv) > 2 else 1,
+                            int(sys.argv[3]) if len(sys.argv) > 3 else 20)
+        dcw.shutdown()
+        print("R2 FUZZ DONE")
+    else:
+        main()
