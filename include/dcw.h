@@ -149,8 +149,8 @@ typedef struct dcw_job_desc {
    * 0 = BlockBasedTable format_version=5 (default).
    * 1 = DcwZipTable "DZT1": the searchable-compressed SST of BASELINE.json
    *     configs[3] (the reference's ToplingZipTable is absent/private,
-   *     README.md:53 — own design, parity self-pinned; layout documented
-   *     in toplingdb_amd/csrc/dcw_dzt.h). */
+   *     README.md:53 — own design, parity self-pinned; format spec in the
+   *     oracle/dzt.c header comment). */
   uint32_t output_table_factory;
 
   /* Bloom filter build: BlockBasedTableOptions::
