@@ -111,6 +111,27 @@ def test_dzt_gpu_tombstones(tmp_path):
     assert_identical(rg, ro)
 
 
+def test_bbt_dzt_bbt_outoff_realloc(tmp_path):
+    # regression (r2 feature fuzz case 9): d_outoff is shared between the
+    # BBT pack and the DZT value pack; when the DZT path reallocated it
+    # through a different capacity tracker, a big-BBT -> tiny-DZT ->
+    # big-BBT sequence on one pooled GpuJob left the BBT capacity stale
+    # and the offset upload overran the (smaller) live allocation
+    # ("HIP error: invalid argument at h2d_meta(d_outoff, ...)")
+    big = gen_runs(tmp_path, 2, 50000, seed0=91)
+    tiny = gen_runs(tmp_path, 1, 800, seed0=95)
+    for tag, runs, otf in (("p1", big, 0), ("p2", tiny, 1), ("p3", big, 0)):
+        dg = tmp_path / ("g" + tag)
+        do = tmp_path / ("o" + tag)
+        dg.mkdir()
+        do.mkdir()
+        kw = dict(compression=1, bottommost_level=1, output_table_factory=otf)
+        rg = dcw.execute(dcw.make_job(runs, str(dg), **kw))
+        ro = oracle.execute(oracle.make_job(runs, str(do), **kw))
+        for fg, fo in zip(rg["files"], ro["files"]):
+            assert open(fg["path"], "rb").read() == open(fo["path"], "rb").read()
+
+
 def test_dzt_bbt_pool_reuse_and_concurrency(tmp_path):
     # pooled GpuJob objects are reused across table formats; alternate and
     # run concurrently to shake out stale per-job state

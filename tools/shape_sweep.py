@@ -136,7 +136,7 @@ def main():
     dcw.shutdown()
 
 
-def run_r2_feature_fuzz(seed, n_cases=20):
+def run_r2_feature_fuzz(seed, n_cases=20, start_exec=0):
     """Round-2 feature-mix fuzz: random jobs drawing from {BBT, DZT} x
     {no filter, bloom} x {uniform16, mixed<=48 keys} x {snappy, zstd
     inputs} x {plain, tombstoned (bottommost envelope)} x {SST runs,
@@ -158,6 +158,7 @@ def run_r2_feature_fuzz(seed, n_cases=20):
         nent = rnd.choice([3000, 12000, 30000])
         tag = "fuzz%d dzt=%d bloom=%d mk=%d z=%d t=%d fl=%d b=%d" % (
             c, dzt, bloom, mixed_keys, zstd_in, tomb, flush, bottom)
+        print("RUN", tag, flush=True)
         kw = dict(compression=rnd.choice([0, 1]),
                   target_file_size=rnd.choice([1 << 20, 64 << 20]),
                   snapshots=snaps, bottommost_level=bottom,
@@ -187,6 +188,20 @@ def run_r2_feature_fuzz(seed, n_cases=20):
         oo = os.path.join(tmp, "o")
         os.makedirs(og)
         os.makedirs(oo)
+        if c < start_exec:
+            # advance the rng identically without executing
+            if flush:
+                gen_kvs(nent)
+            else:
+                for r in range(n_runs):
+                    gen_kvs(nent)
+                    if tomb:
+                        for _ in range(rnd.randrange(1, 4)):
+                            rnd.randrange(nent * 4)
+                            rnd.randrange(1, nent)
+                            rnd.randrange(100)
+                    rnd.choice([0, 1]) if not zstd_in else None
+            continue
         if flush:
             es = gen_kvs(nent)
             rg = dcw.execute(dcw.make_job([], og, flush_entries=es, **kw))
@@ -223,7 +238,8 @@ if __name__ == "__main__":
     if len(sys.argv) > 1 and sys.argv[1] == "r2fuzz":
         dcw.init(0)
         run_r2_feature_fuzz(int(sys.argv[2]) if len(sys.argv) > 2 else 1,
-                            int(sys.argv[3]) if len(sys.argv) > 3 else 20)
+                            int(sys.argv[3]) if len(sys.argv) > 3 else 20,
+                            int(sys.argv[4]) if len(sys.argv) > 4 else 0)
         dcw.shutdown()
         print("R2 FUZZ DONE")
     else:

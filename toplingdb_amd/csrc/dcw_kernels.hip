@@ -3748,7 +3748,10 @@ int GpuJob::dzt_pack_values(const std::vector<DztVBlock>& vbs,
   Impl* p = p_;
   uint32_t nvb = (uint32_t)vbs.size();
   if (!nvb) return 0;
-  HIPCHK(p->ens(&p->d_outoff, sizeof(uint64_t) * nvb));
+  // d_outoff is shared with pack_into (BBT); it is tracked by outoff_cap,
+  // NOT the ens() caps map — mixing the two leaves outoff_cap stale after
+  // an ens() realloc and a later pack_into copies past the real allocation
+  ENSURE(p->d_outoff, p->outoff_cap, sizeof(uint64_t) * nvb);
   HIPCHK(p->h2d_meta(p->d_outoff, outoff.data(), sizeof(uint64_t) * nvb));
   HIPCHK(p->ens(&p->d_dzt_img, total_bytes));
   p->kbegin("dzt_pack", 2.0 * (double)total_bytes);
