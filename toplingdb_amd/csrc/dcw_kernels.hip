@@ -1276,15 +1276,19 @@ __device__ __forceinline__ uint8_t* snap_emit_literal4v(uint8_t* op,
   for (; t < len; t++) op[t] = lit[t];
   return op + len;
 }
+template <typename TAB>
 __device__ __forceinline__ uint8_t* snap_encode_segment_dev(
     const uint8_t* __restrict__ in, uint32_t s0, uint32_t s1,
-    const uint32_t* __restrict__ tab, uint8_t* op) {
+    const TAB* __restrict__ tab, uint8_t* op) {
+  // TAB = uint32_t or uint16_t: the table stores the same first-occurrence
+  // (minimum) positions either way, so the emitted stream is identical
+  const uint32_t kNone = (uint32_t)(TAB)~(TAB)0;
   uint32_t lit = s0, p = s0;
   while (p + 4 <= s1) {
     uint32_t w = load32(in + p);
     uint32_t h = (w * kSnapHashMul) >> (32 - kSnapHashBits);
     uint32_t c = tab[h];
-    if (c != 0xffffffffu && c < p && load32(in + c) == w) {
+    if (c != kNone && c < p && load32(in + c) == w) {
       uint32_t l = 4;
       while (p + l + 4 <= s1) {
         uint32_t a = load32(in + c + l);
@@ -1309,16 +1313,37 @@ __device__ __forceinline__ uint8_t* snap_encode_segment_dev(
   return snap_emit_literal4v(op, in + lit, s1 - lit);
 }
 
-__global__ __launch_bounds__(256) void k_compress(
+// min-position insert into a u16 table half-word: CAS on the containing
+// aligned u32 keeps atomicMin's exact semantics (valid positions are
+// <= SNAP_MAX_UNC-4 < 0xffff, so the sentinel never collides)
+__device__ __forceinline__ void lds_min_u16(uint16_t* tab, uint32_t h,
+                                            uint32_t p) {
+  uint32_t* w = (uint32_t*)tab + (h >> 1);
+  uint32_t sh = (h & 1u) * 16;
+  uint32_t old = *(volatile uint32_t*)w;
+  for (;;) {
+    if (p >= ((old >> sh) & 0xffffu)) return;
+    uint32_t neu = (old & ~(0xffffu << sh)) | (p << sh);
+    uint32_t got = atomicCAS(w, old, neu);
+    if (got == old) return;
+    old = got;
+  }
+}
+
+__global__ __launch_bounds__(256, 8) void k_compress(
     const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
     const uint8_t* __restrict__ ucblob, uint8_t* __restrict__ cblob,
     uint64_t ccap_per_block, uint32_t* __restrict__ bsize,
     uint8_t* __restrict__ btype, uint32_t* err_flag) {
-  __shared__ uint32_t tabs[4][1u << kSnapHashBits]; // 8 KiB per wave
+  // u16 positions halve the table to 4 KiB per wave: 16 KiB per
+  // workgroup lifts LDS-limited occupancy 20 -> 28 waves/CU (SGPR-bound),
+  // which is what a latency-chain-bound kernel wants; emitted bytes are
+  // unchanged (same min-position table content)
+  __shared__ uint16_t tabs[4][1u << kSnapHashBits]; // 4 KiB per wave
   uint32_t wid = threadIdx.x / WAVE;  // wave within workgroup
   uint32_t lane = threadIdx.x % WAVE;
   uint32_t waves = blockDim.x / WAVE;
-  uint32_t* tab = tabs[wid];
+  uint16_t* tab = tabs[wid];
   for (uint32_t b = blockIdx.x * waves + wid; b < nblocks;
        b += gridDim.x * waves) {
     EmitBlockDesc d = bds[b];
@@ -1332,12 +1357,15 @@ __global__ __launch_bounds__(256) void k_compress(
     }
     const uint8_t* gin = ucblob + d.uout;
     uint32_t n = d.unc_size;
-    for (uint32_t t = lane; t < (1u << kSnapHashBits); t += WAVE)
-      tab[t] = 0xffffffffu;
+    {
+      uint32_t* tw = (uint32_t*)tab;
+      for (uint32_t t = lane; t < (1u << kSnapHashBits) / 2; t += WAVE)
+        tw[t] = 0xffffffffu;
+    }
     wave_lds_sync();
     for (uint32_t p = lane; p + 4 <= n; p += WAVE) {
       uint32_t h = (load32(gin + p) * kSnapHashMul) >> (32 - kSnapHashBits);
-      atomicMin(&tab[h], p);
+      lds_min_u16(tab, h, p);
     }
     wave_lds_sync();
     uint32_t seg = (uint32_t)snap_segment_size(n);
