@@ -536,17 +536,22 @@ DCW_HD void snap_dict_table(const uint8_t* dict, uint32_t D,
     if (tab[h] == 0xffffffffu) tab[h] = p;
   }
 }
-// encode one block segment; s0/s1 are VIRTUAL positions (>= D)
+// encode one block segment; s0/s1 are VIRTUAL positions (>= D).
+// TAB = uint32_t (host) or uint16_t (device LDS): virtual positions are
+// bounded by dict(<=48 KiB)+block(<=16 KiB)-4 = 65532 < 0xffff, so the
+// narrow table stores identical min positions and the stream is unchanged
+template <typename TAB>
 DCW_HD uint8_t* snap_encode_segment_dict(const uint8_t* dict, uint32_t D,
                                          const uint8_t* in, uint32_t s0,
-                                         uint32_t s1, const uint32_t* tab,
+                                         uint32_t s1, const TAB* tab,
                                          uint8_t* op) {
+  const uint32_t kNone = (uint32_t)(TAB)~(TAB)0;
   uint32_t lit = s0, p = s0;
   while (p + 4 <= s1) {
     uint32_t w = load32(in + (p - D));
     uint32_t h = (w * kSnapHashMul) >> (32 - kSnapHashBits);
     uint32_t c = tab[h];
-    if (c != 0xffffffffu && c < p && snap_vload32(dict, D, in, c) == w) {
+    if (c != kNone && c < p && snap_vload32(dict, D, in, c) == w) {
       uint32_t l = 4;
       while (p + l < s1 && snap_vbyte(dict, D, in, c + l) == in[p + l - D]) l++;
       op = snap_emit_literal(op, in + (lit - D), p - lit);

@@ -141,11 +141,17 @@ __global__ void k_verify_usize(const uint8_t* __restrict__ blob,
 // one wave per block: raw -> cooperative copy; snappy -> staged through LDS
 // (compressed in + decoded out both in LDS, lane 0 runs the serial decoder,
 // all lanes copy in/out); oversized blocks fall back to the direct path.
-#define DEC_MAX 4992 // 2*4992*4 waves = 39.9 KB LDS/WG -> 4 WGs (16 decoders)/CU
+#define DEC_MAX 4992 // decoded-side LDS bound (typical block_size + slack)
+// compressed-side LDS bound: asymmetric (snappy output ~2x the input on
+// these blocks), so in+out = 8 KiB/wave -> 5 WGs (20 decoders)/CU instead
+// of the symmetric layout's 4.  Inputs in (DEC_IN_MAX, DEC_MAX] keep the
+// decoded side in LDS and read input from global (the v1 path: ~10%
+// slower per block, but it is the minority tail).
+#define DEC_IN_MAX 3200
 struct DecLds {
-  uint8_t in[DEC_MAX]; // compressed input + decoded output both staged in
-  uint8_t out[DEC_MAX]; // LDS: the serial byte decoder is LDS-latency bound
-                        // (a global-input variant measured ~10% slower)
+  uint8_t in[DEC_IN_MAX]; // compressed input staged in LDS: the serial
+  uint8_t out[DEC_MAX];   // byte decoder is LDS-latency bound (the
+                          // global-input variant measured ~10% slower)
 };
 __device__ __forceinline__ void wave_lds_sync2() {
   __builtin_amdgcn_s_waitcnt(0);
@@ -176,14 +182,18 @@ __global__ __launch_bounds__(256) void k_decompress(
       }
       if (lane == 0)
         for (uint32_t t = n & ~15u; t < n; t++) dst[t] = src[t];
-    } else if (n <= DEC_MAX && usize[i] <= DEC_MAX) {
-      for (uint32_t t = lane * 4; t < n; t += WAVE * 4) {
-        uint32_t chunk = n - t < 4 ? n - t : 4;
-        for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = src[t + x];
+    } else if (usize[i] <= DEC_MAX) {
+      bool in_lds = n <= DEC_IN_MAX;
+      if (in_lds) {
+        for (uint32_t t = lane * 4; t < n; t += WAVE * 4) {
+          uint32_t chunk = n - t < 4 ? n - t : 4;
+          for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = src[t + x];
+        }
+        wave_lds_sync2();
       }
-      wave_lds_sync2();
       if (lane == 0) {
-        if (snappy_uncompress(L.in, n, L.out, usize[i]) != usize[i])
+        if (snappy_uncompress(in_lds ? L.in : src, n, L.out, usize[i]) !=
+            usize[i])
           set_err(err_flag, DE_SNAPPY);
       }
       wave_lds_sync2();
@@ -3465,18 +3475,20 @@ __global__ void k_dzt_gather(const GpuJob::DztVBlock* __restrict__ vbs,
 // dict-snappy per value block, wave-parallel (spec v4 segmentation; the
 // segment encoder is the SAME DCW_HD function the oracle restates)
 #define DZT_FRAG_MAX 400 // >= ceil(1.4 * 256-byte segment) + headers
-__global__ __launch_bounds__(256) void k_dzt_compress(
+__global__ __launch_bounds__(256, 8) void k_dzt_compress(
     const GpuJob::DztVBlock* __restrict__ vbs, uint32_t nvb,
     const uint8_t* __restrict__ vstage, const uint8_t* __restrict__ dict,
     uint32_t D, const uint32_t* __restrict__ dict_tab,
     uint8_t* __restrict__ cblob, uint64_t ccap,
     uint32_t* __restrict__ bsize, uint8_t* __restrict__ btype,
     uint32_t* err_flag) {
-  __shared__ uint32_t tabs[4][1u << kSnapHashBits];
+  // u16 virtual positions (max 65532 < 0xffff sentinel), same occupancy
+  // lift as k_compress: 4 KiB table per wave
+  __shared__ uint16_t tabs[4][1u << kSnapHashBits];
   uint32_t wid = threadIdx.x / WAVE;
   uint32_t lane = threadIdx.x % WAVE;
   uint32_t waves = blockDim.x / WAVE;
-  uint32_t* tab = tabs[wid];
+  uint16_t* tab = tabs[wid];
   for (uint32_t b = blockIdx.x * waves + wid; b < nvb; b += gridDim.x * waves) {
     GpuJob::DztVBlock v = vbs[b];
     uint32_t n = v.ulen;
@@ -3488,12 +3500,14 @@ __global__ __launch_bounds__(256) void k_dzt_compress(
       continue;
     }
     const uint8_t* gin = vstage + v.stage_off;
-    for (uint32_t t = lane; t < (1u << kSnapHashBits); t += WAVE)
-      tab[t] = dict_tab[t];
+    for (uint32_t t = lane; t < (1u << kSnapHashBits); t += WAVE) {
+      uint32_t dv = dict_tab[t]; // dict positions < D <= 48 KiB fit u16
+      tab[t] = (uint16_t)(dv > 0xffffu ? 0xffffu : dv);
+    }
     wave_lds_sync();
     for (uint32_t p = lane; p + 4 <= n; p += WAVE) {
       uint32_t h = (load32(gin + p) * kSnapHashMul) >> (32 - kSnapHashBits);
-      atomicMin(&tab[h], D + p); // dict positions (< D) always win
+      lds_min_u16(tab, h, D + p); // dict positions (< D) always win
     }
     wave_lds_sync();
     uint32_t seg = (uint32_t)snap_segment_size(n);
