@@ -141,17 +141,16 @@ __global__ void k_verify_usize(const uint8_t* __restrict__ blob,
 // one wave per block: raw -> cooperative copy; snappy -> staged through LDS
 // (compressed in + decoded out both in LDS, lane 0 runs the serial decoder,
 // all lanes copy in/out); oversized blocks fall back to the direct path.
-#define DEC_MAX 4992 // decoded-side LDS bound (typical block_size + slack)
-// compressed-side LDS bound: asymmetric (snappy output ~2x the input on
-// these blocks), so in+out = 8 KiB/wave -> 5 WGs (20 decoders)/CU instead
-// of the symmetric layout's 4.  Inputs in (DEC_IN_MAX, DEC_MAX] keep the
-// decoded side in LDS and read input from global (the v1 path: ~10%
-// slower per block, but it is the minority tail).
-#define DEC_IN_MAX 3200
+#define DEC_MAX 4992 // 2*4992*4 waves = 39.9 KB LDS/WG -> 4 WGs (16 decoders)/CU
+// (an asymmetric in[3200]+out[4992] layout reaching 20 decoders/CU was
+// measured 24% SLOWER end-to-end: compressible blocks sit right under the
+// 896/1024 ratio bound, so the csize in (3200, 4992] tail that must read
+// input from global is large, and that path costs more than the
+// occupancy buys)
 struct DecLds {
-  uint8_t in[DEC_IN_MAX]; // compressed input staged in LDS: the serial
-  uint8_t out[DEC_MAX];   // byte decoder is LDS-latency bound (the
-                          // global-input variant measured ~10% slower)
+  uint8_t in[DEC_MAX]; // compressed input + decoded output both staged in
+  uint8_t out[DEC_MAX]; // LDS: the serial byte decoder is LDS-latency bound
+                        // (a global-input variant measured ~10% slower)
 };
 __device__ __forceinline__ void wave_lds_sync2() {
   __builtin_amdgcn_s_waitcnt(0);
@@ -182,18 +181,14 @@ __global__ __launch_bounds__(256) void k_decompress(
       }
       if (lane == 0)
         for (uint32_t t = n & ~15u; t < n; t++) dst[t] = src[t];
-    } else if (usize[i] <= DEC_MAX) {
-      bool in_lds = n <= DEC_IN_MAX;
-      if (in_lds) {
-        for (uint32_t t = lane * 4; t < n; t += WAVE * 4) {
-          uint32_t chunk = n - t < 4 ? n - t : 4;
-          for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = src[t + x];
-        }
-        wave_lds_sync2();
+    } else if (n <= DEC_MAX && usize[i] <= DEC_MAX) {
+      for (uint32_t t = lane * 4; t < n; t += WAVE * 4) {
+        uint32_t chunk = n - t < 4 ? n - t : 4;
+        for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = src[t + x];
       }
+      wave_lds_sync2();
       if (lane == 0) {
-        if (snappy_uncompress(in_lds ? L.in : src, n, L.out, usize[i]) !=
-            usize[i])
+        if (snappy_uncompress(L.in, n, L.out, usize[i]) != usize[i])
           set_err(err_flag, DE_SNAPPY);
       }
       wave_lds_sync2();
@@ -765,10 +760,28 @@ __device__ __forceinline__ bool ent_le_s(const ulong4& a, const ulong4& b) {
   if (a.y != b.y) return a.y < b.y;
   return a.z <= b.z;
 }
+// tile boundaries for k_merge_tiled, one thread per boundary: with ~1
+// tile per workgroup the two ~21-step dependent-load binary searches
+// would otherwise run serially on thread 0 of every workgroup (255
+// threads idle) and dominate the pair-merge latency; here they all
+// overlap in one small launch
+__global__ void k_merge_diags(const ulong4* __restrict__ A, uint64_t nA,
+                              const ulong4* __restrict__ B, uint64_t nB,
+                              const uint8_t* __restrict__ kext,
+                              const uint8_t* __restrict__ klen,
+                              uint64_t ntiles, uint64_t* __restrict__ diags) {
+  uint64_t total = nA + nB;
+  for (uint64_t t = blockIdx.x * blockDim.x + threadIdx.x; t <= ntiles;
+       t += (uint64_t)gridDim.x * blockDim.x) {
+    uint64_t d = t * MT_TILE;
+    if (d > total) d = total;
+    diags[t] = merge_diag(A, nA, B, nB, d, kext, klen);
+  }
+}
 __global__ __launch_bounds__(MT_TPB) void k_merge_tiled(
     const ulong4* __restrict__ A, uint64_t nA, const ulong4* __restrict__ B,
     uint64_t nB, ulong4* __restrict__ out, const uint8_t* __restrict__ kext,
-    const uint8_t* __restrict__ klen) {
+    const uint8_t* __restrict__ klen, const uint64_t* __restrict__ diags) {
   __shared__ ulong4 S[MT_TILE]; // A-segment then B-segment
   __shared__ uint32_t seg[2];   // nA_t, a0 broadcast... [0]=nA_t
   __shared__ uint64_t base[2];  // a0, b0
@@ -778,8 +791,9 @@ __global__ __launch_bounds__(MT_TPB) void k_merge_tiled(
     uint64_t d0 = t * MT_TILE;
     uint64_t d1 = d0 + MT_TILE < total ? d0 + MT_TILE : total;
     if (threadIdx.x == 0) {
-      uint64_t a0 = merge_diag(A, nA, B, nB, d0, kext, klen);
-      uint64_t a1 = merge_diag(A, nA, B, nB, d1, kext, klen);
+      uint64_t a0 = diags ? diags[t] : merge_diag(A, nA, B, nB, d0, kext, klen);
+      uint64_t a1 =
+          diags ? diags[t + 1] : merge_diag(A, nA, B, nB, d1, kext, klen);
       base[0] = a0;
       base[1] = d0 - a0;
       seg[0] = (uint32_t)(a1 - a0);
@@ -2118,6 +2132,7 @@ struct GpuJob::Impl {
   void* d_scratch_recoff = nullptr;
   void* d_scratch_gather = nullptr;
   void* d_scratch_mm = nullptr;
+  void* d_merge_diag = nullptr;
   void *d_gp_sm0 = nullptr, *d_gp_sm1 = nullptr, *d_gp_lg0 = nullptr,
        *d_gp_lg1 = nullptr, *d_gp_tie = nullptr, *d_gp_pos = nullptr,
        *d_gp_nback = nullptr;
@@ -2256,6 +2271,7 @@ GpuJob::~GpuJob() {
   F(p->d_scratch64); F(p->d_out_img); F(p->d_outoff);
   F(p->d_scratch_keys); F(p->d_scratch_recoff); F(p->d_scratch_gather);
   F(p->d_scratch_mm);
+  F(p->d_merge_diag);
   F(p->d_gp_sm0); F(p->d_gp_sm1); F(p->d_gp_lg0); F(p->d_gp_lg1);
   F(p->d_gp_tie); F(p->d_gp_pos); F(p->d_gp_nback);
   F(p->d_plan_next); F(p->d_plan_meta); F(p->d_plan_nr);
@@ -2689,6 +2705,16 @@ int GpuJob::merge(std::string* err) {
   (void)hipEventRecord(t0, p->stream);
   std::vector<uint64_t> bounds = p->run_entry_begin; // size k+1
   int cur = 0;
+  // tile-boundary scratch shared by every pair (launches on one stream
+  // serialize, so no pair's diags outlive its own merge launch); env
+  // DCW_MERGE_V=0 falls back to in-kernel thread-0 searches
+  static const bool diag_precompute = [] {
+    const char* v = getenv("DCW_MERGE_V");
+    return !v || atoi(v) != 0;
+  }();
+  if (diag_precompute && bounds.size() > 2)
+    HIPCHK(p->ens(&p->d_merge_diag,
+                  sizeof(uint64_t) * (bounds.back() / MT_TILE + 2)));
   while (bounds.size() > 2) {
     std::vector<uint64_t> nbounds;
     nbounds.push_back(0);
@@ -2696,14 +2722,24 @@ int GpuJob::merge(std::string* err) {
     for (size_t i = 0; i + 1 < k; i += 2) {
       uint64_t a0 = bounds[i], a1 = bounds[i + 1], b1 = bounds[i + 2];
       uint64_t nA = a1 - a0, nB = b1 - a1;
-      p->kbegin("merge_pair", 64.0 * (double)(nA + nB));
       uint64_t ntiles = (nA + nB + MT_TILE - 1) / MT_TILE;
+      const uint64_t* diags = nullptr;
+      if (diag_precompute) {
+        hipLaunchKernelGGL(k_merge_diags,
+                           dim3((uint32_t)((ntiles + 256) / 256)), dim3(256), 0,
+                           p->stream, p->d_ent[cur] + a0, nA,
+                           p->d_ent[cur] + a1, nB,
+                           general_keys ? (const uint8_t*)p->d_kext : nullptr,
+                           p->d_klen, ntiles, (uint64_t*)p->d_merge_diag);
+        diags = (const uint64_t*)p->d_merge_diag;
+      }
+      p->kbegin("merge_pair", 64.0 * (double)(nA + nB));
       hipLaunchKernelGGL(k_merge_tiled,
                          dim3((uint32_t)(ntiles < 8192 ? ntiles : 8192)),
                          dim3(MT_TPB), 0, p->stream, p->d_ent[cur] + a0, nA,
                          p->d_ent[cur] + a1, nB, p->d_ent[cur ^ 1] + a0,
                          general_keys ? (const uint8_t*)p->d_kext : nullptr,
-                         p->d_klen);
+                         p->d_klen, diags);
       p->kend();
       nbounds.push_back(b1);
     }
