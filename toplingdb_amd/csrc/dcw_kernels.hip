@@ -141,17 +141,125 @@ __global__ void k_verify_usize(const uint8_t* __restrict__ blob,
 // one wave per block: raw -> cooperative copy; snappy -> staged through LDS
 // (compressed in + decoded out both in LDS, lane 0 runs the serial decoder,
 // all lanes copy in/out); oversized blocks fall back to the direct path.
-#define DEC_MAX 4992 // 2*4992*4 waves = 39.9 KB LDS/WG -> 4 WGs (16 decoders)/CU
+#define DEC_MAX 4992 // 2*(4992+16)*4 waves = 40.1 KB LDS/WG -> 4 WGs (16 decoders)/CU
 // (an asymmetric in[3200]+out[4992] layout reaching 20 decoders/CU was
-// measured 24% SLOWER end-to-end: compressible blocks sit right under the
-// 896/1024 ratio bound, so the csize in (3200, 4992] tail that must read
-// input from global is large, and that path costs more than the
-// occupancy buys)
+// measured slower end-to-end; LDS staging of BOTH sides is load-bearing)
+#define DEC_PAD 16 // slack for the 8/16-byte moves of the fast decoder
 struct DecLds {
-  uint8_t in[DEC_MAX]; // compressed input + decoded output both staged in
-  uint8_t out[DEC_MAX]; // LDS: the serial byte decoder is LDS-latency bound
-                        // (a global-input variant measured ~10% slower)
+  uint8_t in[DEC_MAX + DEC_PAD]; // compressed input + decoded output both
+  uint8_t out[DEC_MAX + DEC_PAD]; // staged in LDS: the serial byte decoder
+                                  // is LDS-latency bound (a global-input
+                                  // variant measured ~10% slower)
 };
+
+__device__ __forceinline__ uint64_t lds_ld64(const uint8_t* p) {
+  uint64_t v;
+  memcpy(&v, p, 8);
+  return v;
+}
+__device__ __forceinline__ void lds_st64(uint8_t* p, uint64_t v) {
+  memcpy(p, &v, 8);
+}
+// Serial snappy decoder specialized for LDS-staged blocks.  The generic
+// DCW_HD decoder compiles to a dependent ds_read -> waitcnt -> ds_write
+// chain PER WORD (and 2-3 chained byte reads per tag), which is what made
+// k_decompress ~30% of all kernel time.  Here each op does ONE unaligned
+// 8-byte header fetch (tag + length/offset bytes decoded from registers),
+// literals move 16 B per round trip, and short-offset overlapped copies
+// use pattern doubling.  Stores may run up to 15 B past the logical
+// output end and header fetches up to 7 B past the input end: both stay
+// inside DEC_PAD, and every ACCEPT decision is bounds-checked against the
+// true n/ulen first, so accepted output bytes are identical to the
+// generic decoder's and corrupt blocks are rejected the same way.
+__device__ uint32_t snap_dec_lds(const uint8_t* __restrict__ in, uint32_t n,
+                                 uint8_t* __restrict__ out, uint32_t cap) {
+  uint32_t ulen = 0, ip = 0;
+  {
+    uint64_t h = lds_ld64(in);
+    uint32_t s = 0;
+    for (;;) {
+      if (ip >= n || ip >= 5) return 0;
+      uint8_t b = (uint8_t)(h >> (8 * ip));
+      ulen |= (uint32_t)(b & 0x7f) << s;
+      ip++;
+      if (!(b & 0x80)) break;
+      s += 7;
+    }
+  }
+  if (ulen > cap) return 0;
+  uint32_t op = 0;
+  while (ip < n) {
+    uint64_t h = lds_ld64(in + ip);
+    uint8_t tag = (uint8_t)h;
+    if ((tag & 3) == 0) { // literal
+      uint32_t len = (uint32_t)(tag >> 2) + 1;
+      uint32_t hb = 1;
+      if (len > 60) {
+        uint32_t nb = len - 60; // 1..4 length bytes, little-endian
+        if (ip + 1 + nb > n) return 0;
+        len = (uint32_t)((h >> 8) & (0xffffffffull >> (8 * (4 - nb)))) + 1;
+        hb = 1 + nb;
+      }
+      ip += hb;
+      if (ip + len > n || op + len > ulen) return 0;
+      for (uint32_t i = 0; i < len; i += 16) {
+        uint64_t a = lds_ld64(in + ip + i);
+        uint64_t b = lds_ld64(in + ip + i + 8);
+        lds_st64(out + op + i, a);
+        lds_st64(out + op + i + 8, b);
+      }
+      ip += len;
+      op += len;
+    } else { // copy
+      uint32_t len, offset, hb;
+      if ((tag & 3) == 1) {
+        len = ((uint32_t)(tag >> 2) & 7) + 4;
+        offset = ((uint32_t)(tag >> 5) << 8) | (uint8_t)(h >> 8);
+        hb = 2;
+      } else if ((tag & 3) == 2) {
+        len = (uint32_t)(tag >> 2) + 1;
+        offset = (uint32_t)(h >> 8) & 0xffffu;
+        hb = 3;
+      } else {
+        len = (uint32_t)(tag >> 2) + 1;
+        offset = (uint32_t)(h >> 8);
+        hb = 5;
+      }
+      if (ip + hb > n) return 0;
+      ip += hb;
+      if (offset == 0 || offset > op || op + len > ulen) return 0;
+      uint32_t src = op - offset;
+      uint32_t end = op + len;
+      if (offset >= 16) {
+        for (uint32_t i = 0; i < len; i += 16) {
+          uint64_t a = lds_ld64(out + src + i);
+          uint64_t b = lds_ld64(out + src + i + 8);
+          lds_st64(out + op + i, a);
+          lds_st64(out + op + i + 8, b);
+        }
+      } else if (offset >= 8) {
+        for (uint32_t i = 0; i < len; i += 8)
+          lds_st64(out + op + i, lds_ld64(out + src + i));
+      } else {
+        // pattern doubling: each 8-B store validates (op-src) more bytes
+        // and the usable distance doubles; reads of not-yet-valid bytes
+        // land beyond the advancing point and are overwritten next round
+        while (op < end) {
+          lds_st64(out + op, lds_ld64(out + src));
+          uint32_t d = op - src;
+          if (d >= 8) {
+            for (uint32_t i = 8; op + i < end; i += 8)
+              lds_st64(out + op + i, lds_ld64(out + src + i));
+            break;
+          }
+          op += d < end - op ? d : end - op;
+        }
+      }
+      op = end;
+    }
+  }
+  return op == ulen ? ulen : 0;
+}
 __device__ __forceinline__ void wave_lds_sync2() {
   __builtin_amdgcn_s_waitcnt(0);
   __builtin_amdgcn_wave_barrier();
@@ -188,7 +296,7 @@ __global__ __launch_bounds__(256) void k_decompress(
       }
       wave_lds_sync2();
       if (lane == 0) {
-        if (snappy_uncompress(L.in, n, L.out, usize[i]) != usize[i])
+        if (snap_dec_lds(L.in, n, L.out, usize[i]) != usize[i])
           set_err(err_flag, DE_SNAPPY);
       }
       wave_lds_sync2();
