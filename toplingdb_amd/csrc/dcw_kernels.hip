@@ -1570,15 +1570,17 @@ __device__ __forceinline__ uint32_t snap_copy_len(uint32_t offset, uint32_t len)
   }
   return c;
 }
+template <typename TAB>
 __device__ __forceinline__ uint32_t snap_measure_segment_dev(
     const uint8_t* __restrict__ in, uint32_t s0, uint32_t s1,
-    const uint32_t* __restrict__ tab) {
+    const TAB* __restrict__ tab) {
+  const uint32_t kNone = (uint32_t)(TAB)~(TAB)0;
   uint32_t lit = s0, p = s0, out = 0;
   while (p + 4 <= s1) {
     uint32_t w = load32(in + p);
     uint32_t h = (w * kSnapHashMul) >> (32 - kSnapHashBits);
     uint32_t c = tab[h];
-    if (c != 0xffffffffu && c < p && load32(in + c) == w) {
+    if (c != kNone && c < p && load32(in + c) == w) {
       uint32_t l = 4;
       while (p + l + 4 <= s1) {
         uint32_t a = load32(in + c + l);
@@ -1701,17 +1703,17 @@ __global__ __launch_bounds__(256) void k_compress_ldsin(
 
 #define SNAP_LDSIN_MAX 5376 // LDS-staged input bound (default 4 KiB blocks)
 template <int LDSIN>
-__global__ __launch_bounds__(256) void k_compress_2p(
+__global__ __launch_bounds__(256, 8) void k_compress_2p(
     const EmitBlockDesc* __restrict__ bds, uint32_t nblocks,
     const uint8_t* __restrict__ ucblob, uint8_t* __restrict__ cblob,
     uint64_t ccap_per_block, uint32_t* __restrict__ bsize,
     uint8_t* __restrict__ btype, uint32_t* err_flag) {
-  __shared__ uint32_t tabs[4][1u << kSnapHashBits]; // 8 KiB per wave
+  __shared__ uint16_t tabs[4][1u << kSnapHashBits]; // 4 KiB per wave
   __shared__ uint8_t ins[LDSIN ? 4 : 1][LDSIN ? SNAP_LDSIN_MAX : 4];
   uint32_t wid = threadIdx.x / WAVE;
   uint32_t lane = threadIdx.x % WAVE;
   uint32_t waves = blockDim.x / WAVE;
-  uint32_t* tab = tabs[wid];
+  uint16_t* tab = tabs[wid];
   for (uint32_t b = blockIdx.x * waves + wid; b < nblocks;
        b += gridDim.x * waves) {
     EmitBlockDesc d = bds[b];
@@ -1740,12 +1742,15 @@ __global__ __launch_bounds__(256) void k_compress_2p(
       }
       in = li;
     }
-    for (uint32_t t = lane; t < (1u << kSnapHashBits); t += WAVE)
-      tab[t] = 0xffffffffu;
+    {
+      uint32_t* tw = (uint32_t*)tab;
+      for (uint32_t t = lane; t < (1u << kSnapHashBits) / 2; t += WAVE)
+        tw[t] = 0xffffffffu;
+    }
     wave_lds_sync();
     for (uint32_t p = lane; p + 4 <= n; p += WAVE) {
       uint32_t h = (load32(in + p) * kSnapHashMul) >> (32 - kSnapHashBits);
-      atomicMin(&tab[h], p);
+      lds_min_u16(tab, h, p);
     }
     wave_lds_sync();
     uint32_t seg = (uint32_t)snap_segment_size(n);
