@@ -24,6 +24,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdio>
+#include <sys/time.h>
 #include <cstring>
 #include <map>
 #include <mutex>
@@ -2109,15 +2110,25 @@ struct GpuJob::Impl {
   size_t meta_off = 0;
 
   void arena_reset() {
-    // keep only the largest block to stay grow-only without hoarding
-    if (meta_blocks.size() > 1) {
-      size_t best = 0;
+    // grow-only: a job stages ~20-40 MB of metadata through 10-30 blocks,
+    // and the old keep-one-block policy re-freed and re-pinned them every
+    // job — hipHostFree is device-synchronizing and the cycle cost
+    // ~200 ms/job of pipeline wall.  Keep everything up to a cap and only
+    // trim (smallest first) beyond it.
+    const size_t kArenaCap = 512u << 20;
+    size_t total = 0;
+    for (auto& b : meta_blocks) total += b.cap;
+    static const bool dbg = getenv("DCW_PHASE_DEBUG") != nullptr;
+    if (dbg)
+      fprintf(stderr, "[arena] blocks=%zu total=%.1fMB\n", meta_blocks.size(),
+              total / 1048576.0);
+    while (total > kArenaCap && meta_blocks.size() > 1) {
+      size_t small = 0;
       for (size_t i = 1; i < meta_blocks.size(); i++)
-        if (meta_blocks[i].cap > meta_blocks[best].cap) best = i;
-      for (size_t i = 0; i < meta_blocks.size(); i++)
-        if (i != best && meta_blocks[i].p) (void)hipHostFree(meta_blocks[i].p);
-      meta_blocks[0] = meta_blocks[best];
-      meta_blocks.resize(1);
+        if (meta_blocks[i].cap < meta_blocks[small].cap) small = i;
+      total -= meta_blocks[small].cap;
+      if (meta_blocks[small].p) (void)hipHostFree(meta_blocks[small].p);
+      meta_blocks.erase(meta_blocks.begin() + small);
     }
     meta_cur_block = 0;
     meta_off = 0;
@@ -2334,9 +2345,20 @@ void GpuJob::drain_d2h() {
 // buffers belong to their StagedInput; owned ones are freed).
 void GpuJob::reset() {
   Impl* p = p_;
+  static const bool dbg = getenv("DCW_PHASE_DEBUG") != nullptr;
+  auto us = [] {
+    struct timeval tv;
+    gettimeofday(&tv, nullptr);
+    return (uint64_t)tv.tv_sec * 1000000 + tv.tv_usec;
+  };
+  uint64_t t0 = dbg ? us() : 0;
   (void)hipStreamSynchronize(p->stream);
   (void)hipStreamSynchronize(p->d2h_stream);
+  uint64_t t1 = dbg ? us() : 0;
   p->arena_reset();
+  if (dbg)
+    fprintf(stderr, "[reset] sync=%.1fms arena=%.1fms\n", (t1 - t0) / 1000.0,
+            (us() - t1) / 1000.0);
   // own-slot staged buffers are grow-only (kept for the next job);
   // borrowed ones belong to their StagedInput — either way just unalias
   p->d_blob = nullptr;

@@ -113,6 +113,8 @@ struct JobPool {
         return j;
       }
     }
+    if (getenv("DCW_PHASE_DEBUG"))
+      fprintf(stderr, "[pool] miss -> new GpuJob\n");
     return new GpuJob();
   }
   void put(GpuJob* j) {
@@ -265,6 +267,8 @@ struct LoadedInputs {
   uint64_t in_bytes = 0;
   std::vector<SstTombstone> tombstones; // all inputs' range deletions
   bool host_decoded = false; // zstd inputs were rewritten (see below)
+  // sub-phase wall attribution (printed under DCW_PHASE_DEBUG)
+  uint64_t us_reserve = 0, us_pread = 0, us_parse = 0, us_stage = 0;
 };
 
 // minimal libzstd ABI (runtime lib only in this image); the worker only
@@ -287,10 +291,14 @@ int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err,
       }
       total += (uint64_t)st.st_size;
     }
+  uint64_t tphase = now_usec();
   L->blob.reserve(total);
+  L->us_reserve += now_usec() - tphase;
+  tphase = now_usec();
   // overlap H2D with the remaining file reads: each file's bytes start
   // streaming to the device the moment its read completes
   if (job && job->stage_begin(total, err) != 0) return -1;
+  L->us_stage += now_usec() - tphase;
   L->gi.run_block_begin.push_back(0);
   uint32_t cstype = 0xffffffff;
   for (uint32_t r = 0; r < d->num_runs; r++) {
@@ -309,7 +317,10 @@ int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err,
       }
       uint64_t sz = (uint64_t)st.st_size;
       uint64_t base = L->blob.len;
+      uint64_t tf = now_usec();
       L->blob.reserve(base + sz);
+      L->us_reserve += now_usec() - tf;
+      tf = now_usec();
       // segmented parallel read: a single-thread fread of a ~64 MiB SST is
       // ~10 GB/s and sits on the job's critical path (the write side is
       // already segmented)
@@ -359,9 +370,14 @@ int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err,
         return -1;
       }
       L->blob.len = base + sz;
+      L->us_pread += now_usec() - tf;
+      tf = now_usec();
       if (job && job->stage_chunk(base, L->blob.p + base, sz, err) != 0)
         return -1;
+      L->us_stage += now_usec() - tf;
+      tf = now_usec();
       ParsedSst ps = parse_sst(L->blob.p + base, sz);
+      L->us_parse += now_usec() - tf;
       if (!ps.ok) {
         *err = std::string(path) + ": " + ps.error;
         return -1;
@@ -385,11 +401,13 @@ int load_inputs(const dcw_job_desc* d, LoadedInputs* L, std::string* err,
   // util/compression.h ZSTD_Uncompress)
   bool any_zstd = false;
   uint32_t ct = cstype == 0xffffffff ? 4 : cstype;
+  tphase = now_usec();
   for (auto& blk : L->gi.blocks)
     if (L->blob.p[blk.off + blk.size] == 7) {
       any_zstd = true;
       break;
     }
+  L->us_parse += now_usec() - tphase;
   if (any_zstd) {
     if (job) { // already-streamed chunks are superseded by a full restage
       job->stage_cancel();
@@ -938,6 +956,7 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     ~PooledJob() { g_jobs.put(j); }
   } pj;
   pj.j->reset();
+  uint64_t us_reset = now_usec() - t0;
   GpuJob& job = *pj.j;
   uint64_t in_bytes = 0;
   LoadedInputs L;
@@ -1541,6 +1560,11 @@ int32_t dcw_execute(const dcw_job_desc* d, dcw_job_result* res) {
     for (int i = 0; i < 13; i++)
       fprintf(stderr, " %s=%.1fms", wp.names[i], wp.us[i] / 1000.0);
     fprintf(stderr, " total=%.1fms\n", (now_usec() - t_start) / 1000.0);
+    fprintf(stderr,
+            "[load] reset=%.1fms reserve=%.1fms pread=%.1fms stage=%.1fms "
+            "parse=%.1fms\n",
+            us_reset / 1000.0, L.us_reserve / 1000.0, L.us_pread / 1000.0,
+            L.us_stage / 1000.0, L.us_parse / 1000.0);
   }
 
   res->num_files = (uint32_t)out_files.size();
