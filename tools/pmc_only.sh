@@ -12,7 +12,7 @@ rocprofv3 --pmc WRITE_SIZE -d /tmp/pmcw -o w -- \
 FDB=$(find /tmp/pmcf -name "*.db" | head -1)
 WDB=$(find /tmp/pmcw -name "*.db" | head -1)
 python tools/pmc_per_launch.py "$FDB" "$WDB" gpurun_out/pmc_per_launch.json \
-  "r2 capture: 2 jobs (steps1+warmup1, jif1), decode word-window build" \
+  "r2-final capture: 2 jobs (steps1+warmup1, jif1), u16 tables + fast decoder" \
   > gpurun_out/pmc_summary.txt 2>&1 || cat gpurun_out/pmc_summary.txt
 cp /tmp/pmcf.json gpurun_out/pmc_bench_f.json 2>/dev/null
 # keep only small artifacts for the merge

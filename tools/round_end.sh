@@ -11,6 +11,13 @@ python -c "import __graft_entry__ as g; g.smoke()" > gpurun_out/re_smoke.log 2>&
 echo "smoke: $(tail -1 gpurun_out/re_smoke.log)"
 
 python bench.py > gpurun_out/re_bench_default.json 2> gpurun_out/re_bench_default.log
+# kernel-trace cross-check of the same command (contract: rocprof average
+# duration for the dominant kernel must agree with the HIP-event timing)
+export TMPDIR=/tmp
+timeout 420 rocprofv3 --kernel-trace --stats -f csv -d /tmp/kt -o kt -- \
+  python bench.py --steps 3 --warmup 1 --skip-cpu-baseline --skip-hbm-resident \
+  > gpurun_out/re_kt_bench.json 2> /tmp/kt_run.log
+for f in $(find /tmp/kt -name "*stats*.csv" 2>/dev/null); do cp "$f" gpurun_out/re_kt_$(basename $f); done
 python - <<'EOF'
 import json
 d = json.loads([l for l in open('gpurun_out/re_bench_default.json') if l.strip().startswith('{')][-1])
