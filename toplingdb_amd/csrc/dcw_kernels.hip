@@ -1615,12 +1615,12 @@ __global__ __launch_bounds__(256) void k_compress_ldsin(
     const uint8_t* __restrict__ ucblob, uint8_t* __restrict__ cblob,
     uint64_t ccap_per_block, uint32_t* __restrict__ bsize,
     uint8_t* __restrict__ btype, uint32_t* err_flag) {
-  __shared__ uint32_t tabs[4][1u << kSnapHashBits]; // 8 KiB per wave
+  __shared__ uint16_t tabs[4][1u << kSnapHashBits]; // 4 KiB per wave
   __shared__ uint8_t ins[4][5376];
   uint32_t wid = threadIdx.x / WAVE;
   uint32_t lane = threadIdx.x % WAVE;
   uint32_t waves = blockDim.x / WAVE;
-  uint32_t* tab = tabs[wid];
+  uint16_t* tab = tabs[wid];
   for (uint32_t b = blockIdx.x * waves + wid; b < nblocks;
        b += gridDim.x * waves) {
     EmitBlockDesc d = bds[b];
@@ -1649,12 +1649,15 @@ __global__ __launch_bounds__(256) void k_compress_ldsin(
       }
       in = li;
     }
-    for (uint32_t t = lane; t < (1u << kSnapHashBits); t += WAVE)
-      tab[t] = 0xffffffffu;
+    {
+      uint32_t* tw = (uint32_t*)tab;
+      for (uint32_t t = lane; t < (1u << kSnapHashBits) / 2; t += WAVE)
+        tw[t] = 0xffffffffu;
+    }
     wave_lds_sync();
     for (uint32_t p = lane; p + 4 <= n; p += WAVE) {
       uint32_t h = (load32(in + p) * kSnapHashMul) >> (32 - kSnapHashBits);
-      atomicMin(&tab[h], p);
+      lds_min_u16(tab, h, p);
     }
     wave_lds_sync();
     uint32_t seg = (uint32_t)snap_segment_size(n);
