@@ -172,6 +172,7 @@ __device__ __forceinline__ void lds_st64(uint8_t* p, uint64_t v) {
 // inside DEC_PAD, and every ACCEPT decision is bounds-checked against the
 // true n/ulen first, so accepted output bytes are identical to the
 // generic decoder's and corrupt blocks are rejected the same way.
+template <int PIPE>
 __device__ uint32_t snap_dec_lds(const uint8_t* __restrict__ in, uint32_t n,
                                  uint8_t* __restrict__ out, uint32_t cap) {
   uint32_t ulen = 0, ip = 0;
@@ -189,8 +190,12 @@ __device__ uint32_t snap_dec_lds(const uint8_t* __restrict__ in, uint32_t n,
   }
   if (ulen > cap) return 0;
   uint32_t op = 0;
+  // PIPE=1 software-pipelines the header: the next op's position is known
+  // before the current op's copy runs, so its 8-byte fetch is issued ahead
+  // of the copy and the LDS round trip overlaps the data movement
+  uint64_t h = lds_ld64(in + ip);
   while (ip < n) {
-    uint64_t h = lds_ld64(in + ip);
+    if (!PIPE) h = lds_ld64(in + ip);
     uint8_t tag = (uint8_t)h;
     if ((tag & 3) == 0) { // literal
       uint32_t len = (uint32_t)(tag >> 2) + 1;
@@ -203,6 +208,7 @@ __device__ uint32_t snap_dec_lds(const uint8_t* __restrict__ in, uint32_t n,
       }
       ip += hb;
       if (ip + len > n || op + len > ulen) return 0;
+      uint64_t hn = PIPE ? lds_ld64(in + ip + len) : 0;
       for (uint32_t i = 0; i < len; i += 16) {
         uint64_t a = lds_ld64(in + ip + i);
         uint64_t b = lds_ld64(in + ip + i + 8);
@@ -211,6 +217,7 @@ __device__ uint32_t snap_dec_lds(const uint8_t* __restrict__ in, uint32_t n,
       }
       ip += len;
       op += len;
+      if (PIPE) h = hn;
     } else { // copy
       uint32_t len, offset, hb;
       if ((tag & 3) == 1) {
@@ -229,6 +236,7 @@ __device__ uint32_t snap_dec_lds(const uint8_t* __restrict__ in, uint32_t n,
       if (ip + hb > n) return 0;
       ip += hb;
       if (offset == 0 || offset > op || op + len > ulen) return 0;
+      uint64_t hn = PIPE ? lds_ld64(in + ip) : 0;
       uint32_t src = op - offset;
       uint32_t end = op + len;
       if (offset >= 16) {
@@ -257,6 +265,7 @@ __device__ uint32_t snap_dec_lds(const uint8_t* __restrict__ in, uint32_t n,
         }
       }
       op = end;
+      if (PIPE) h = hn;
     }
   }
   return op == ulen ? ulen : 0;
@@ -269,7 +278,8 @@ __global__ __launch_bounds__(256) void k_decompress(
     const uint8_t* __restrict__ blob, const uint64_t* __restrict__ boff,
     const uint32_t* __restrict__ bsize, const uint8_t* __restrict__ btype,
     const uint64_t* __restrict__ uoff, const uint32_t* __restrict__ usize,
-    uint32_t nblocks, uint8_t* __restrict__ ublob, uint32_t* err_flag) {
+    uint32_t nblocks, uint8_t* __restrict__ ublob, uint32_t* err_flag,
+    uint32_t dec_pipe) {
   __shared__ DecLds lds[4];
   uint32_t waves_per_wg = blockDim.x / WAVE;
   uint32_t wid = threadIdx.x / WAVE;
@@ -297,8 +307,9 @@ __global__ __launch_bounds__(256) void k_decompress(
       }
       wave_lds_sync2();
       if (lane == 0) {
-        if (snap_dec_lds(L.in, n, L.out, usize[i]) != usize[i])
-          set_err(err_flag, DE_SNAPPY);
+        uint32_t r = dec_pipe ? snap_dec_lds<1>(L.in, n, L.out, usize[i])
+                              : snap_dec_lds<0>(L.in, n, L.out, usize[i]);
+        if (r != usize[i]) set_err(err_flag, DE_SNAPPY);
       }
       wave_lds_sync2();
       uint32_t un = usize[i];
@@ -2626,10 +2637,14 @@ int GpuJob::decode(std::string* err) {
                        p->d_btype_in, p->d_uoff, p->d_usize, nb, p->d_ublob,
                        p->d_err);
   else
+    static const uint32_t dec_pipe = [] {
+      const char* v = getenv("DCW_DEC_PIPE");
+      return (uint32_t)(v ? atoi(v) : 0);
+    }();
     hipLaunchKernelGGL(k_decompress, dim3(grid_for(nb * 4ull)), dim3(256), 0,
                        p->stream, p->d_blob, p->d_boff, p->d_bsize,
                        p->d_btype_in, p->d_uoff, p->d_usize, nb, p->d_ublob,
-                       p->d_err);
+                       p->d_err, dec_pipe);
   p->kend();
   HIPCHK(p->ens((void**)&p->d_nrestarts, sizeof(uint32_t) * nb));
   hipLaunchKernelGGL(k_num_restarts, dim3(grid_for(nb)), dim3(256), 0, p->stream,
