@@ -2636,7 +2636,7 @@ int GpuJob::decode(std::string* err) {
                        p->stream, p->d_blob, p->d_boff, p->d_bsize,
                        p->d_btype_in, p->d_uoff, p->d_usize, nb, p->d_ublob,
                        p->d_err);
-  else
+  else {
     static const uint32_t dec_pipe = [] {
       const char* v = getenv("DCW_DEC_PIPE");
       return (uint32_t)(v ? atoi(v) : 0);
@@ -2645,6 +2645,7 @@ int GpuJob::decode(std::string* err) {
                        p->stream, p->d_blob, p->d_boff, p->d_bsize,
                        p->d_btype_in, p->d_uoff, p->d_usize, nb, p->d_ublob,
                        p->d_err, dec_pipe);
+  }
   p->kend();
   HIPCHK(p->ens((void**)&p->d_nrestarts, sizeof(uint32_t) * nb));
   hipLaunchKernelGGL(k_num_restarts, dim3(grid_for(nb)), dim3(256), 0, p->stream,
