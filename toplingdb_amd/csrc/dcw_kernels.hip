@@ -153,13 +153,20 @@ struct DecLds {
                                   // variant measured ~10% slower)
 };
 
+// two unaligned 32-bit DS ops: unaligned ds_read/write_b32 is penalty-free
+// on CDNA4, while an align(1) 8-byte memcpy is lowered to EIGHT ds_*_u8
+// (and an unaligned _b64 would replay at 64 cycles) — this is the decoder's
+// dominant instruction, so the width matters more than anything else in it
 __device__ __forceinline__ uint64_t lds_ld64(const uint8_t* p) {
-  uint64_t v;
-  memcpy(&v, p, 8);
-  return v;
+  uint32_t lo, hi;
+  memcpy(&lo, p, 4);
+  memcpy(&hi, p + 4, 4);
+  return (uint64_t)lo | ((uint64_t)hi << 32);
 }
 __device__ __forceinline__ void lds_st64(uint8_t* p, uint64_t v) {
-  memcpy(p, &v, 8);
+  uint32_t lo = (uint32_t)v, hi = (uint32_t)(v >> 32);
+  memcpy(p, &lo, 4);
+  memcpy(p + 4, &hi, 4);
 }
 // Serial snappy decoder specialized for LDS-staged blocks.  The generic
 // DCW_HD decoder compiles to a dependent ds_read -> waitcnt -> ds_write
@@ -301,10 +308,16 @@ __global__ __launch_bounds__(256) void k_decompress(
       if (lane == 0)
         for (uint32_t t = n & ~15u; t < n; t++) dst[t] = src[t];
     } else if (n <= DEC_MAX && usize[i] <= DEC_MAX) {
-      for (uint32_t t = lane * 4; t < n; t += WAVE * 4) {
-        uint32_t chunk = n - t < 4 ? n - t : 4;
-        for (uint32_t x = 0; x < chunk; x++) L.in[t + x] = src[t + x];
+      // 16-byte staging: byte-granular copies issued ~5000 DS ops per
+      // block per wave and saturated the CU's DS issue pipe that the 16
+      // resident serial decoders depend on
+      for (uint32_t t = lane * 16; t + 16 <= n; t += WAVE * 16) {
+        ulong2 v;
+        memcpy(&v, src + t, 16);
+        memcpy(L.in + t, &v, 16);
       }
+      if (lane == 0)
+        for (uint32_t t = n & ~15u; t < n; t++) L.in[t] = src[t];
       wave_lds_sync2();
       if (lane == 0) {
         uint32_t r = dec_pipe ? snap_dec_lds<1>(L.in, n, L.out, usize[i])
@@ -313,10 +326,13 @@ __global__ __launch_bounds__(256) void k_decompress(
       }
       wave_lds_sync2();
       uint32_t un = usize[i];
-      for (uint32_t t = lane * 4; t < un; t += WAVE * 4) {
-        uint32_t chunk = un - t < 4 ? un - t : 4;
-        for (uint32_t x = 0; x < chunk; x++) dst[t + x] = L.out[t + x];
+      for (uint32_t t = lane * 16; t + 16 <= un; t += WAVE * 16) {
+        ulong2 v;
+        memcpy(&v, L.out + t, 16);
+        memcpy(dst + t, &v, 16);
       }
+      if (lane == 0)
+        for (uint32_t t = un & ~15u; t < un; t++) dst[t] = L.out[t];
       wave_lds_sync2();
     } else {
       if (lane == 0) {
