@@ -179,6 +179,146 @@ __device__ __forceinline__ void lds_st64(uint8_t* p, uint64_t v) {
 // inside DEC_PAD, and every ACCEPT decision is bounds-checked against the
 // true n/ulen first, so accepted output bytes are identical to the
 // generic decoder's and corrupt blocks are rejected the same way.
+__device__ __forceinline__ void wave_lds_sync2() {
+  __builtin_amdgcn_s_waitcnt(0);
+  __builtin_amdgcn_wave_barrier();
+}
+
+// Wave-cooperative decoder (DCW_DEC_PIPE=2): all 64 lanes run the tag
+// parse in LOCKSTEP (same data, same branches - zero divergence, so the
+// serial parse chain is paid once per wave, not once per op), each lane
+// keeps every 64th op in registers, and the batch then executes in
+// dependency rounds: literals and copies whose source lies wholly below
+// the done-prefix frontier run concurrently across lanes; the frontier
+// op itself is always executable (pattern doubling reads only its own
+// writes and the done region), so progress is guaranteed.  Bounds
+// checks are identical to the serial decoder and lane-uniform.
+__device__ uint32_t snap_dec_wave(const uint8_t* __restrict__ in, uint32_t n,
+                                  uint8_t* __restrict__ out, uint32_t cap,
+                                  uint32_t lane) {
+  uint32_t ulen = 0, ip = 0;
+  {
+    uint64_t h = lds_ld64(in);
+    uint32_t s = 0;
+    for (;;) {
+      if (ip >= n || ip >= 5) return 0;
+      uint8_t b = (uint8_t)(h >> (8 * ip));
+      ulen |= (uint32_t)(b & 0x7f) << s;
+      ip++;
+      if (!(b & 0x80)) break;
+      s += 7;
+    }
+  }
+  if (ulen > cap) return 0;
+  uint32_t opos = 0;
+  while (ip < n) {
+    uint32_t nops = 0;
+    uint32_t my_dst = 0, my_src = 0, my_len = 0, my_kind = 0;
+    while (ip < n && nops < WAVE) {
+      uint64_t h = lds_ld64(in + ip);
+      uint8_t tag = (uint8_t)h;
+      uint32_t len, srcp, kind;
+      if ((tag & 3) == 0) { // literal
+        len = (uint32_t)(tag >> 2) + 1;
+        uint32_t hb = 1;
+        if (len > 60) {
+          uint32_t nb = len - 60;
+          if (ip + 1 + nb > n) return 0;
+          len = (uint32_t)((h >> 8) & (0xffffffffull >> (8 * (4 - nb)))) + 1;
+          hb = 1 + nb;
+        }
+        ip += hb;
+        if (ip + len > n || opos + len > ulen) return 0;
+        srcp = ip;
+        kind = 1;
+        ip += len;
+      } else { // copy
+        uint32_t offset, hb;
+        if ((tag & 3) == 1) {
+          len = ((uint32_t)(tag >> 2) & 7) + 4;
+          offset = ((uint32_t)(tag >> 5) << 8) | (uint8_t)(h >> 8);
+          hb = 2;
+        } else if ((tag & 3) == 2) {
+          len = (uint32_t)(tag >> 2) + 1;
+          offset = (uint32_t)(h >> 8) & 0xffffu;
+          hb = 3;
+        } else {
+          len = (uint32_t)(tag >> 2) + 1;
+          offset = (uint32_t)(h >> 8);
+          hb = 5;
+        }
+        if (ip + hb > n) return 0;
+        ip += hb;
+        if (offset == 0 || offset > opos || opos + len > ulen) return 0;
+        srcp = opos - offset;
+        kind = 2;
+      }
+      if (nops == lane) {
+        my_dst = opos;
+        my_src = srcp;
+        my_len = len;
+        my_kind = kind;
+      }
+      opos += len;
+      nops++;
+    }
+    uint64_t done_mask = 0;
+    uint32_t frontier = 0;
+    while (frontier < nops) {
+      uint32_t fdst = __shfl(my_dst, (int)frontier);
+      bool pending = lane < nops && !((done_mask >> lane) & 1);
+      bool ready = pending && (my_kind == 1 ||
+                               my_src + my_len <= fdst || lane == frontier);
+      if (ready) {
+        // EXACT-length stores: rounds execute ops out of order, so the
+        // serial decoder's 15-byte overshoot would clobber a neighbour
+        // op's already-written bytes
+        const uint8_t* sbase = my_kind == 1 ? in : out;
+        uint32_t offset = my_kind == 1 ? 0xffffffffu : my_dst - my_src;
+        uint32_t t = 0;
+        if (offset >= 16) {
+          for (; t + 16 <= my_len; t += 16) {
+            uint64_t a = lds_ld64(sbase + my_src + t);
+            uint64_t b = lds_ld64(sbase + my_src + t + 8);
+            lds_st64(out + my_dst + t, a);
+            lds_st64(out + my_dst + t + 8, b);
+          }
+          for (; t < my_len; t++) out[my_dst + t] = sbase[my_src + t];
+        } else if (offset >= 8) {
+          for (; t + 8 <= my_len; t += 8)
+            lds_st64(out + my_dst + t, lds_ld64(out + my_src + t));
+          for (; t < my_len; t++) out[my_dst + t] = out[my_src + t];
+        } else {
+          uint32_t o = my_dst, src = my_src, end = my_dst + my_len;
+          while (o < end) {
+            uint32_t d = o - src;
+            if (end - o < 8 || d < 8) {
+              if (end - o < 8) { // byte-exact tail
+                for (; o < end; o++) out[o] = out[o - offset];
+                break;
+              }
+              lds_st64(out + o, lds_ld64(out + src));
+              o += d < end - o ? d : end - o;
+            } else {
+              // pattern established through o at distance d (a power-of-two
+              // multiple of offset): chunk-copy at that distance, byte tail
+              uint32_t dist = d;
+              for (; o + 8 <= end; o += 8)
+                lds_st64(out + o, lds_ld64(out + o - dist));
+              for (; o < end; o++) out[o] = out[o - offset];
+              break;
+            }
+          }
+        }
+      }
+      wave_lds_sync2();
+      done_mask |= __ballot(ready);
+      while (frontier < nops && ((done_mask >> frontier) & 1)) frontier++;
+    }
+  }
+  return opos == ulen ? ulen : 0;
+}
+
 template <int PIPE>
 __device__ uint32_t snap_dec_lds(const uint8_t* __restrict__ in, uint32_t n,
                                  uint8_t* __restrict__ out, uint32_t cap) {
@@ -277,10 +417,6 @@ __device__ uint32_t snap_dec_lds(const uint8_t* __restrict__ in, uint32_t n,
   }
   return op == ulen ? ulen : 0;
 }
-__device__ __forceinline__ void wave_lds_sync2() {
-  __builtin_amdgcn_s_waitcnt(0);
-  __builtin_amdgcn_wave_barrier();
-}
 __global__ __launch_bounds__(256) void k_decompress(
     const uint8_t* __restrict__ blob, const uint64_t* __restrict__ boff,
     const uint32_t* __restrict__ bsize, const uint8_t* __restrict__ btype,
@@ -319,7 +455,10 @@ __global__ __launch_bounds__(256) void k_decompress(
       if (lane == 0)
         for (uint32_t t = n & ~15u; t < n; t++) L.in[t] = src[t];
       wave_lds_sync2();
-      if (lane == 0) {
+      if (dec_pipe == 2) {
+        uint32_t r = snap_dec_wave(L.in, n, L.out, usize[i], lane);
+        if (lane == 0 && r != usize[i]) set_err(err_flag, DE_SNAPPY);
+      } else if (lane == 0) {
         uint32_t r = dec_pipe ? snap_dec_lds<1>(L.in, n, L.out, usize[i])
                               : snap_dec_lds<0>(L.in, n, L.out, usize[i]);
         if (r != usize[i]) set_err(err_flag, DE_SNAPPY);
