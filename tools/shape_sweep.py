@@ -136,7 +136,7 @@ def main():
     dcw.shutdown()
 
 
-def run_r2_feature_fuzz(seed, n_cases=20, start_exec=0):
+def run_r2_feature_fuzz(seed, n_cases=20, start_exec=0, scale=1):
     """Round-2 feature-mix fuzz: random jobs drawing from {BBT, DZT} x
     {no filter, bloom} x {uniform16, mixed<=48 keys} x {snappy, zstd
     inputs} x {plain, tombstoned (bottommost envelope)} x {SST runs,
@@ -155,7 +155,7 @@ def run_r2_feature_fuzz(seed, n_cases=20, start_exec=0):
         snaps = [] if tomb else (sorted(rnd.sample(range(1, 50000), 2))
                                  if rnd.random() < 0.4 else [])
         n_runs = rnd.choice([1, 2, 3])
-        nent = rnd.choice([3000, 12000, 30000])
+        nent = rnd.choice([3000, 12000, 30000]) * scale
         tag = "fuzz%d dzt=%d bloom=%d mk=%d z=%d t=%d fl=%d b=%d" % (
             c, dzt, bloom, mixed_keys, zstd_in, tomb, flush, bottom)
         print("RUN", tag, flush=True)
@@ -239,7 +239,8 @@ if __name__ == "__main__":
         dcw.init(0)
         run_r2_feature_fuzz(int(sys.argv[2]) if len(sys.argv) > 2 else 1,
                             int(sys.argv[3]) if len(sys.argv) > 3 else 20,
-                            int(sys.argv[4]) if len(sys.argv) > 4 else 0)
+                            int(sys.argv[4]) if len(sys.argv) > 4 else 0,
+                            int(sys.argv[5]) if len(sys.argv) > 5 else 1)
         dcw.shutdown()
         print("R2 FUZZ DONE")
     else:
