@@ -2305,7 +2305,15 @@ struct GpuJob::Impl {
   hipError_t h2d_meta(void* dst, const void* src, size_t n) {
     if (n == 0) return hipSuccess;
     static const bool sync_meta = getenv("DCW_SYNC_META") != nullptr;
-    if (sync_meta) return hipMemcpy(dst, src, n, hipMemcpyHostToDevice);
+    if (sync_meta) {
+      // a NULL-stream hipMemcpy does NOT order against these NON-BLOCKING
+      // job streams (caught by the suite: a grandparent-cut SST differed
+      // by one byte under this env) — stage on the job stream and drain
+      hipError_t e = hipMemcpyAsync(dst, src, n, hipMemcpyHostToDevice,
+                                    stream);
+      if (e != hipSuccess) return e;
+      return hipStreamSynchronize(stream);
+    }
     size_t need = (n + 63) & ~(size_t)63;
     while (meta_cur_block < meta_blocks.size() &&
            meta_off + need > meta_blocks[meta_cur_block].cap) {
